@@ -1,0 +1,276 @@
+"""User-facing functional API: hetu_amd.<fn>(...) builds graph ops.
+
+Plays the role of the reference's ops.yml-generated `hetu.*` functions
+(/root/reference/python/hetu/_binding/codegen/ops.yml): each function makes
+an op in the current graph and returns its output Tensor(s).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+import torch
+
+from ...parallel.dstates import DistributedStates
+from ..graph import current_graph
+from ..tensor import Tensor
+from . import basics as B
+from . import nnops as N
+from . import comm as C
+
+
+def _cg():
+    return current_graph()
+
+
+def _wrap_const(x, like: Tensor) -> Tensor:
+    if isinstance(x, Tensor):
+        return x
+    raise TypeError("use *_scalar ops or pass a Tensor")
+
+
+# ---- sources --------------------------------------------------------------
+
+def placeholder(shape, dtype=torch.float32, name="placeholder",
+                ds: Optional[DistributedStates] = None, device_group=None
+                ) -> Tensor:
+    op = _cg().make_op(B.PlaceholderOp(), [], {"shape": tuple(shape),
+                                               "dtype": dtype},
+                       name=name, ds_list=[ds] if ds else None,
+                       device_group=device_group)
+    return op.output()
+
+
+# alias matching the reference's parallel_placeholder
+def parallel_placeholder(shape, dtype=torch.float32, ds=None,
+                         device_group=None, name="placeholder") -> Tensor:
+    return placeholder(shape, dtype, name, ds, device_group)
+
+
+def variable(data: torch.Tensor, name="variable", requires_grad=True,
+             ds: Optional[DistributedStates] = None, device_group=None
+             ) -> Tensor:
+    g = _cg()
+    op = g.make_op(B.VariableOp(), [], {"shape": tuple(data.shape),
+                                        "dtype": data.dtype},
+                   name=name, ds_list=[ds] if ds else None,
+                   device_group=device_group)
+    t = op.output()
+    t.set_data(data)
+    t.is_parameter = requires_grad
+    t.requires_grad = requires_grad
+    if requires_grad:
+        g.parameters.append(t)
+    return t
+
+
+def constant(shape, value, dtype=torch.float32, name="const") -> Tensor:
+    return _cg().make_op(B.ConstantOp(), [], {"shape": tuple(shape),
+                                              "value": value,
+                                              "dtype": dtype},
+                         name=name).output()
+
+
+# ---- arithmetic ------------------------------------------------------------
+
+def add(a: Tensor, b) -> Tensor:
+    if not isinstance(b, Tensor):
+        return _cg().make_op(B.AddScalarOp(), [a], {"value": float(b)}).output()
+    return _cg().make_op(B.AddOp(), [a, b], {}).output()
+
+
+def sub(a: Tensor, b) -> Tensor:
+    if not isinstance(b, Tensor):
+        return _cg().make_op(B.AddScalarOp(), [a], {"value": -float(b)}).output()
+    return _cg().make_op(B.SubOp(), [a, b], {}).output()
+
+
+def mul(a: Tensor, b) -> Tensor:
+    if not isinstance(b, Tensor):
+        return _cg().make_op(B.MulScalarOp(), [a], {"value": float(b)}).output()
+    return _cg().make_op(B.MulOp(), [a, b], {}).output()
+
+
+def div(a: Tensor, b) -> Tensor:
+    if not isinstance(b, Tensor):
+        return _cg().make_op(B.MulScalarOp(), [a],
+                             {"value": 1.0 / float(b)}).output()
+    return _cg().make_op(B.DivOp(), [a, b], {}).output()
+
+
+def neg(a: Tensor) -> Tensor:
+    return _cg().make_op(B.NegOp(), [a], {}).output()
+
+
+def pow(a: Tensor, p: float) -> Tensor:  # noqa: A001
+    return _cg().make_op(B.PowScalarOp(), [a], {"value": float(p)}).output()
+
+
+def exp(a):
+    return _cg().make_op(B.ExpOp(), [a], {}).output()
+
+
+def log(a):
+    return _cg().make_op(B.LogOp(), [a], {}).output()
+
+
+def sqrt(a):
+    return _cg().make_op(B.SqrtOp(), [a], {}).output()
+
+
+def rsqrt(a):
+    return _cg().make_op(B.RsqrtOp(), [a], {}).output()
+
+
+def add_n(ts: Sequence[Tensor]) -> Tensor:
+    return B.make_add_n(_cg(), list(ts))
+
+
+# ---- shape -----------------------------------------------------------------
+
+def reshape(a: Tensor, shape) -> Tensor:
+    return _cg().make_op(B.ReshapeOp(), [a], {"shape": tuple(shape)}).output()
+
+
+def transpose(a: Tensor, dim0: int, dim1: int) -> Tensor:
+    return _cg().make_op(B.TransposeOp(), [a], {"dim0": dim0,
+                                                "dim1": dim1}).output()
+
+
+def slice_(a: Tensor, dim: int, start, length) -> Tensor:
+    return _cg().make_op(B.SliceOp(), [a], {"dim": dim, "start": start,
+                                            "length": length}).output()
+
+
+def concat(ts: Sequence[Tensor], dim: int = 0) -> Tensor:
+    return _cg().make_op(B.ConcatOp(), list(ts), {"dim": dim}).output()
+
+
+def contiguous(a: Tensor) -> Tensor:
+    return _cg().make_op(B.ContiguousOp(), [a], {}).output()
+
+
+def cast(a: Tensor, dtype) -> Tensor:
+    if a.dtype == dtype:
+        return a
+    return _cg().make_op(B.CastOp(), [a], {"dtype": dtype}).output()
+
+
+# ---- reductions ------------------------------------------------------------
+
+def reduce_sum(a: Tensor, dim=None, keepdim=False) -> Tensor:
+    return _cg().make_op(B.ReduceOp_(), [a], {"mode": "sum", "dim": dim,
+                                              "keepdim": keepdim}).output()
+
+
+def reduce_mean(a: Tensor, dim=None, keepdim=False) -> Tensor:
+    return _cg().make_op(B.ReduceOp_(), [a], {"mode": "mean", "dim": dim,
+                                              "keepdim": keepdim}).output()
+
+
+def reduce_max(a: Tensor, dim=None, keepdim=False) -> Tensor:
+    return _cg().make_op(B.ReduceOp_(), [a], {"mode": "max", "dim": dim,
+                                              "keepdim": keepdim}).output()
+
+
+# ---- GEMM ------------------------------------------------------------------
+
+def matmul(a: Tensor, b: Tensor, trans_a=False, trans_b=False) -> Tensor:
+    return _cg().make_op(B.MatMul2DOp(), [a, b],
+                         {"trans_a": trans_a, "trans_b": trans_b}).output()
+
+
+def linear(x: Tensor, w: Tensor, bias: Optional[Tensor] = None) -> Tensor:
+    ins = [x, w] + ([bias] if bias is not None else [])
+    return _cg().make_op(B.LinearOp(), ins, {}).output()
+
+
+def bmm(a: Tensor, b: Tensor) -> Tensor:
+    return _cg().make_op(B.BatchMatMulOp(), [a, b], {}).output()
+
+
+# ---- nn --------------------------------------------------------------------
+
+def relu(a):
+    return _cg().make_op(N.ReluOp(), [a], {}).output()
+
+
+def gelu(a):
+    return _cg().make_op(N.GeluOp(), [a], {}).output()
+
+
+def silu(a):
+    return _cg().make_op(N.SiluOp(), [a], {}).output()
+
+
+def tanh(a):
+    return _cg().make_op(N.TanhOp(), [a], {}).output()
+
+
+def sigmoid(a):
+    return _cg().make_op(N.SigmoidOp(), [a], {}).output()
+
+
+def swiglu(a):
+    return _cg().make_op(N.SwiGLUOp(), [a], {}).output()
+
+
+def softmax(a, dim=-1):
+    return _cg().make_op(N.SoftmaxOp(), [a], {"dim": dim}).output()
+
+
+_DROPOUT_SEED = [12345]
+_DROPOUT_OFFSET = [0]
+
+
+def dropout(a, p: float):
+    _DROPOUT_OFFSET[0] += 1 << 20
+    return _cg().make_op(N.DropoutOp(), [a],
+                         {"p": p, "seed": _DROPOUT_SEED[0],
+                          "offset": _DROPOUT_OFFSET[0]}).output()
+
+
+def layer_norm(x, w, b, eps=1e-5):
+    return _cg().make_op(N.LayerNormOp(), [x, w, b], {"eps": eps}).output(0)
+
+
+def rms_norm(x, w, eps=1e-6):
+    return _cg().make_op(N.RMSNormOp(), [x, w], {"eps": eps}).output(0)
+
+
+def embedding(table, ids):
+    return _cg().make_op(N.EmbeddingOp(), [table, ids], {}).output()
+
+
+def rotary(x, cos, sin):
+    return _cg().make_op(N.RotaryOp(), [x, cos, sin], {}).output()
+
+
+def attention(q, k, v, causal=True, scale=None):
+    return _cg().make_op(N.AttentionOp(), [q, k, v],
+                         {"causal": causal, "scale": scale}).output(0)
+
+
+def softmax_cross_entropy_sparse(logits, labels, ignore_index=-100):
+    return _cg().make_op(N.SoftmaxCrossEntropySparseOp(), [logits, labels],
+                         {"ignore_index": ignore_index}).output(0)
+
+
+def mse_loss(x, y):
+    return _cg().make_op(N.MSELossOp(), [x, y], {}).output()
+
+
+# ---- comm ------------------------------------------------------------------
+
+def comm(x: Tensor, dst_ds: DistributedStates, name="comm") -> Tensor:
+    return C.make_comm(_cg(), x, dst_ds, name=name)
+
+
+# ---- autodiff --------------------------------------------------------------
+
+def gradients(ys, xs, grad_ys=None):
+    g = _cg()
+    single = isinstance(ys, Tensor)
+    ys_l = [ys] if single else list(ys)
+    gy_l = None if grad_ys is None else (
+        [grad_ys] if isinstance(grad_ys, Tensor) else list(grad_ys))
+    return g.gradients(ys_l, list(xs), gy_l)

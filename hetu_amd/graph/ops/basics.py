@@ -1,0 +1,905 @@
+"""Core op set: sources, arithmetic, shape ops, reductions, GEMM family.
+
+Re-designs the reference op families (/root/reference/hetu/graph/ops/ —
+Arithmetics.cc, matmul.cc, Linear.cc, BatchMatMul.cc, Reshape/Transpose/
+Slice/Concat, Reduce.cc, variable.cc, placeholder.cc, sum.cc) for torch-ROCm
+execution. Each op implements infer_meta / compute / gradient and, where the
+layout is nontrivial, deduce_states for SPMD propagation.
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional
+
+import torch
+
+from ...core.symbol import IntSymbol, resolve_dim, resolve_shape
+from ...parallel.dstates import DistributedStates, ds_from_index_table
+from ..op import Op, OpInterface
+from ..tensor import Tensor, TensorMeta
+
+
+def _g(t: Tensor):
+    return t.graph
+
+
+def _make(graph, iface, inputs, attrs=None, name="", **kw):
+    return graph.make_op(iface, inputs, attrs or {}, name=name, **kw)
+
+
+# ---------------------------------------------------------------------------
+# Sources
+# ---------------------------------------------------------------------------
+
+class PlaceholderOp(OpInterface):
+    type = "Placeholder"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(attrs["shape"], attrs["dtype"])]
+
+    def compute(self, op, inputs, ctx):
+        raise RuntimeError("placeholder must be fed")
+
+
+class VariableOp(OpInterface):
+    type = "Variable"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(attrs["shape"], attrs["dtype"])]
+
+    def compute(self, op, inputs, ctx):
+        data = op.outputs[0].get_data()
+        if data is None:
+            raise RuntimeError(f"variable {op.name} not initialized")
+        return [data]
+
+    def gradient(self, op, grad_outputs):
+        return []
+
+
+class ConstantOp(OpInterface):
+    type = "Constant"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(attrs["shape"], attrs["dtype"])]
+
+    def compute(self, op, inputs, ctx):
+        a = op.attrs
+        return [torch.full(resolve_shape(a["shape"]), a["value"],
+                           dtype=a["dtype"], device=ctx.device)]
+
+
+class OnesLikeOp(OpInterface):
+    type = "OnesLike"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, inputs[0].dtype)]
+
+    def deduce_states(self, op):
+        # gradient-seed semantics: d(sum)/d(partial contribution) == 1
+        # everywhere, so a partial input yields a *duplicate* seed.
+        src = op.inputs[0].ds
+        if src is not None:
+            states = dict(src.states)
+            if -2 in states:
+                states[-1] = states.get(-1, 1) * states.pop(-2)
+            order = [-1 if d == -2 else d for d in src.order]
+            dedup = []
+            for d in order:
+                if d not in dedup:
+                    dedup.append(d)
+            op.outputs[0].ds = DistributedStates(src.device_num, states, dedup)
+        op.outputs[0].device_group = op.inputs[0].device_group
+
+    def compute(self, op, inputs, ctx):
+        return [torch.ones_like(inputs[0])]
+
+
+class ZerosLikeOp(OpInterface):
+    type = "ZerosLike"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        return [torch.zeros_like(inputs[0])]
+
+
+def make_ones_like(graph, t: Tensor) -> Tensor:
+    return _make(graph, OnesLikeOp(), [t], name=f"ones_like({t.name})").output()
+
+
+# ---------------------------------------------------------------------------
+# Elementwise arithmetic (broadcasting; grads reduce back to input shape)
+# ---------------------------------------------------------------------------
+
+def _bcast_shape(s1, s2):
+    out = []
+    l1, l2 = len(s1), len(s2)
+    for i in range(max(l1, l2)):
+        d1 = s1[l1 - 1 - i] if i < l1 else 1
+        d2 = s2[l2 - 1 - i] if i < l2 else 1
+        if isinstance(d1, IntSymbol) or isinstance(d2, IntSymbol):
+            out.append(d1 if not (isinstance(d1, int) and d1 == 1) else d2)
+        else:
+            out.append(max(d1, d2))
+    return tuple(reversed(out))
+
+
+def _reduce_to_shape(graph, g: Tensor, target: Tensor) -> Tensor:
+    if tuple(g.shape) == tuple(target.shape):
+        return g
+    return _make(graph, ReduceToShapeOp(), [g, target],
+                 name=f"reduce_to({target.name})").output()
+
+
+class ReduceToShapeOp(OpInterface):
+    """Sum-reduce a broadcasted gradient back to the shape of inputs[1]."""
+    type = "ReduceToShape"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[1].shape, inputs[0].dtype)]
+
+    def deduce_states(self, op):
+        op.outputs[0].ds = op.inputs[1].ds
+        op.outputs[0].device_group = op.inputs[1].device_group
+
+    def compute(self, op, inputs, ctx):
+        g, ref = inputs
+        tgt = list(ref.shape)
+        while g.ndim > len(tgt):
+            g = g.sum(0)
+        for i, d in enumerate(tgt):
+            if g.shape[i] != d:
+                g = g.sum(i, keepdim=True)
+        return [g]
+
+
+class _BinaryOp(OpInterface):
+    def infer_meta(self, attrs, inputs):
+        dtype = inputs[0].dtype
+        return [TensorMeta(_bcast_shape(inputs[0].shape, inputs[1].shape),
+                           dtype)]
+
+
+class AddOp(_BinaryOp):
+    type = "Add"
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0] + inputs[1]]
+
+    def gradient(self, op, g):
+        gy = g[0]
+        gr = _g(op.outputs[0])
+        return [_reduce_to_shape(gr, gy, op.inputs[0]),
+                _reduce_to_shape(gr, gy, op.inputs[1])]
+
+
+class SubOp(_BinaryOp):
+    type = "Sub"
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0] - inputs[1]]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_reduce_to_shape(gr, g[0], op.inputs[0]),
+                _reduce_to_shape(gr, make_neg(gr, g[0]), op.inputs[1])]
+
+
+class MulOp(_BinaryOp):
+    type = "Mul"
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0] * inputs[1]]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        a, b = op.inputs
+        ga = _make(gr, MulOp(), [g[0], b]).output()
+        gb = _make(gr, MulOp(), [g[0], a]).output()
+        return [_reduce_to_shape(gr, ga, a), _reduce_to_shape(gr, gb, b)]
+
+
+class DivOp(_BinaryOp):
+    type = "Div"
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0] / inputs[1]]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        a, b = op.inputs
+        ga = _make(gr, DivOp(), [g[0], b]).output()
+        # gb = -g * a / b^2
+        gb_num = _make(gr, MulOp(), [g[0], a]).output()
+        b2 = _make(gr, MulOp(), [b, b]).output()
+        gb = make_neg(gr, _make(gr, DivOp(), [gb_num, b2]).output())
+        return [_reduce_to_shape(gr, ga, a), _reduce_to_shape(gr, gb, b)]
+
+
+class _ScalarOp(OpInterface):
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, inputs[0].dtype)]
+
+
+class AddScalarOp(_ScalarOp):
+    type = "AddScalar"
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0] + op.attrs["value"]]
+
+    def gradient(self, op, g):
+        return [g[0]]
+
+
+class MulScalarOp(_ScalarOp):
+    type = "MulScalar"
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0] * op.attrs["value"]]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, MulScalarOp(), [g[0]],
+                      {"value": op.attrs["value"]}).output()]
+
+
+class PowScalarOp(_ScalarOp):
+    type = "PowScalar"
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0] ** op.attrs["value"]]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        p = op.attrs["value"]
+        xp = _make(gr, PowScalarOp(), [op.inputs[0]], {"value": p - 1}).output()
+        gx = _make(gr, MulOp(), [g[0], xp]).output()
+        return [_make(gr, MulScalarOp(), [gx], {"value": p}).output()]
+
+
+class NegOp(_ScalarOp):
+    type = "Neg"
+
+    def compute(self, op, inputs, ctx):
+        return [-inputs[0]]
+
+    def gradient(self, op, g):
+        return [make_neg(_g(op.outputs[0]), g[0])]
+
+
+def make_neg(graph, t):
+    return _make(graph, NegOp(), [t]).output()
+
+
+class _UnaryTorch(_ScalarOp):
+    fn = None
+
+    def compute(self, op, inputs, ctx):
+        return [self.__class__.fn(inputs[0])]
+
+
+class ExpOp(_UnaryTorch):
+    type = "Exp"
+    fn = torch.exp
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, MulOp(), [g[0], op.outputs[0]]).output()]
+
+
+class LogOp(_UnaryTorch):
+    type = "Log"
+    fn = torch.log
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, DivOp(), [g[0], op.inputs[0]]).output()]
+
+
+class SqrtOp(_UnaryTorch):
+    type = "Sqrt"
+    fn = torch.sqrt
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        half = _make(gr, MulScalarOp(), [op.outputs[0]], {"value": 2.0}).output()
+        return [_make(gr, DivOp(), [g[0], half]).output()]
+
+
+class RsqrtOp(_UnaryTorch):
+    type = "Rsqrt"
+    fn = torch.rsqrt
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        y3 = _make(gr, PowScalarOp(), [op.outputs[0]], {"value": 3.0}).output()
+        gx = _make(gr, MulOp(), [g[0], y3]).output()
+        return [_make(gr, MulScalarOp(), [gx], {"value": -0.5}).output()]
+
+
+class AddNOp(OpInterface):
+    """n-ary sum (reference sum.cc) — grad accumulation node."""
+    type = "AddN"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        out = inputs[0].clone()
+        for t in inputs[1:]:
+            out += t
+        return [out]
+
+    def gradient(self, op, g):
+        return [g[0]] * len(op.inputs)
+
+
+def make_add_n(graph, ts: List[Tensor]) -> Tensor:
+    return _make(graph, AddNOp(), ts, name="grad_sum").output()
+
+
+# ---------------------------------------------------------------------------
+# Shape ops
+# ---------------------------------------------------------------------------
+
+class ReshapeOp(OpInterface):
+    type = "Reshape"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(attrs["shape"], inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        shape = resolve_shape(op.attrs["shape"])
+        # allow a single -1
+        return [inputs[0].reshape(shape)]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, ReshapeOp(), [g[0]],
+                      {"shape": op.inputs[0].shape}).output()]
+
+
+class TransposeOp(OpInterface):
+    type = "Transpose"
+
+    def infer_meta(self, attrs, inputs):
+        d0, d1 = attrs["dim0"], attrs["dim1"]
+        shape = list(inputs[0].shape)
+        shape[d0], shape[d1] = shape[d1], shape[d0]
+        return [TensorMeta(shape, inputs[0].dtype)]
+
+    def deduce_states(self, op):
+        src = op.inputs[0].ds
+        if src is not None:
+            d0, d1 = op.attrs["dim0"], op.attrs["dim1"]
+            nd = op.inputs[0].ndim
+            d0 = d0 % nd
+            d1 = d1 % nd
+            states = {}
+            for d, n in src.states.items():
+                nd_ = d
+                if d == d0:
+                    nd_ = d1
+                elif d == d1:
+                    nd_ = d0
+                states[nd_] = n
+            order = [d1 if d == d0 else d0 if d == d1 else d
+                     for d in src.order]
+            op.outputs[0].ds = DistributedStates(src.device_num, states, order)
+        op.outputs[0].device_group = op.inputs[0].device_group
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0].transpose(op.attrs["dim0"], op.attrs["dim1"])
+                .contiguous()]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, TransposeOp(), [g[0]],
+                      {"dim0": op.attrs["dim0"], "dim1": op.attrs["dim1"]}
+                      ).output()]
+
+
+class SliceOp(OpInterface):
+    """Slice along one dim: [start, start+length)."""
+    type = "Slice"
+
+    def infer_meta(self, attrs, inputs):
+        shape = list(inputs[0].shape)
+        shape[attrs["dim"]] = attrs["length"]
+        return [TensorMeta(shape, inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        a = op.attrs
+        return [inputs[0].narrow(a["dim"], resolve_dim(a["start"]),
+                                 resolve_dim(a["length"])).contiguous()]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, SliceGradOp(), [g[0], op.inputs[0]],
+                      dict(op.attrs)).output()]
+
+
+class SliceGradOp(OpInterface):
+    type = "SliceGrad"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[1].shape, inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        g, ref = inputs
+        out = torch.zeros_like(ref)
+        a = op.attrs
+        out.narrow(a["dim"], resolve_dim(a["start"]),
+                   resolve_dim(a["length"])).copy_(g)
+        return [out]
+
+
+class ConcatOp(OpInterface):
+    type = "Concat"
+
+    def infer_meta(self, attrs, inputs):
+        dim = attrs["dim"]
+        shape = list(inputs[0].shape)
+        total = 0
+        symbolic = False
+        for t in inputs:
+            d = t.shape[dim]
+            if isinstance(d, IntSymbol):
+                symbolic = True
+                break
+            total += d
+        if symbolic:
+            shape[dim] = inputs[0].shape[dim]
+        else:
+            shape[dim] = total
+        return [TensorMeta(shape, inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        return [torch.cat(inputs, dim=op.attrs["dim"])]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        dim = op.attrs["dim"]
+        grads = []
+        start = 0
+        for t in op.inputs:
+            grads.append(_make(gr, SliceOp(), [g[0]],
+                               {"dim": dim, "start": start,
+                                "length": t.shape[dim]}).output())
+            start += resolve_dim(t.shape[dim]) if not isinstance(
+                t.shape[dim], IntSymbol) else 0
+        return grads
+
+
+class ContiguousOp(_ScalarOp):
+    type = "Contiguous"
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0].contiguous()]
+
+    def gradient(self, op, g):
+        return [g[0]]
+
+
+class CastOp(OpInterface):
+    """dtype transfer (reference data_transfer.cc)."""
+    type = "Cast"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, attrs["dtype"])]
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0].to(op.attrs["dtype"])]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, CastOp(), [g[0]],
+                      {"dtype": op.inputs[0].dtype}).output()]
+
+
+# ---------------------------------------------------------------------------
+# Reductions
+# ---------------------------------------------------------------------------
+
+class ReduceOp_(OpInterface):
+    type = "Reduce"
+
+    def infer_meta(self, attrs, inputs):
+        dim = attrs["dim"]
+        keepdim = attrs["keepdim"]
+        shape = list(inputs[0].shape)
+        if dim is None:
+            shape = [1] * len(shape) if keepdim else []
+        else:
+            dims = [dim] if isinstance(dim, int) else list(dim)
+            dims = [d % len(shape) for d in dims]
+            if keepdim:
+                for d in dims:
+                    shape[d] = 1
+            else:
+                shape = [s for i, s in enumerate(shape) if i not in dims]
+        dtype = inputs[0].dtype
+        return [TensorMeta(shape, dtype)]
+
+    def deduce_states(self, op):
+        src = op.inputs[0].ds
+        if src is not None:
+            dim = op.attrs["dim"]
+            keepdim = op.attrs["keepdim"]
+            nd = op.inputs[0].ndim
+            dims = (list(range(nd)) if dim is None
+                    else [dim % nd] if isinstance(dim, int)
+                    else [d % nd for d in dim])
+            states = {}
+            part = src.partial
+            for d, n in src.states.items():
+                if d >= 0 and d in dims:
+                    if op.attrs["mode"] in ("sum", "mean"):
+                        part *= n     # reduced over a split dim -> partial
+                    else:
+                        raise ValueError("max/min reduce over split dim")
+                elif d >= 0:
+                    nd_ = d - sum(1 for r in dims if r < d) if not keepdim else d
+                    states[nd_] = n
+                elif d == -1:
+                    states[-1] = n
+            if part > 1:
+                states[-2] = part
+            order = []
+            for d in src.order:
+                if d >= 0 and d in dims:
+                    order.append(-2)
+                elif d >= 0 and not keepdim:
+                    order.append(d - sum(1 for r in dims if r < d))
+                else:
+                    order.append(d)
+            op.outputs[0].ds = DistributedStates(src.device_num, states,
+                                                 [d for d in order])
+        op.outputs[0].device_group = op.inputs[0].device_group
+
+    def _split_factor(self, op):
+        """Product of input splits over the reduced dims: a mean over a
+        split dim must divide by the GLOBAL count so that the partial-sum
+        representation (sum over ranks == global mean) holds."""
+        src = op.inputs[0].ds
+        if src is None:
+            return 1
+        dim = op.attrs["dim"]
+        nd = op.inputs[0].ndim
+        dims = (list(range(nd)) if dim is None
+                else [dim % nd] if isinstance(dim, int)
+                else [d % nd for d in dim])
+        f = 1
+        for d in dims:
+            f *= src.get_dim(d)
+        return f
+
+    def compute(self, op, inputs, ctx):
+        x = inputs[0]
+        mode = op.attrs["mode"]
+        dim = op.attrs["dim"]
+        keepdim = op.attrs["keepdim"]
+        if mode == "sum":
+            return [x.sum() if dim is None else x.sum(dim, keepdim=keepdim)]
+        if mode == "mean":
+            y = x.mean() if dim is None else x.mean(dim, keepdim=keepdim)
+            f = self._split_factor(op)
+            if f > 1:
+                y = y / f
+            return [y]
+        if mode == "max":
+            if dim is None:
+                return [x.max()]
+            return [x.max(dim, keepdim=keepdim).values]
+        raise ValueError(mode)
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        mode = op.attrs["mode"]
+        if mode not in ("sum", "mean"):
+            raise NotImplementedError("grad for max/min reduce")
+        return [_make(gr, ReduceGradOp(), [g[0], op.inputs[0]],
+                      dict(op.attrs)).output()]
+
+
+class ReduceGradOp(OpInterface):
+    type = "ReduceGrad"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[1].shape, inputs[0].dtype)]
+
+    def deduce_states(self, op):
+        op.outputs[0].ds = op.inputs[1].ds
+        op.outputs[0].device_group = op.inputs[1].device_group
+
+    def compute(self, op, inputs, ctx):
+        g, ref = inputs
+        mode = op.attrs["mode"]
+        dim = op.attrs["dim"]
+        keepdim = op.attrs["keepdim"]
+        nd = ref.ndim
+        if dim is None:
+            dims = list(range(nd))
+        else:
+            dims = [dim % nd] if isinstance(dim, int) else [d % nd for d in dim]
+        if not keepdim:
+            for d in sorted(dims):
+                g = g.unsqueeze(d)
+        out = g.expand_as(ref).contiguous()
+        if mode == "mean":
+            n = 1
+            for d in dims:
+                n *= ref.shape[d]
+            src = op.inputs[1].ds
+            if src is not None:
+                for d in dims:
+                    n *= src.get_dim(d)   # global count over split dims
+            out = out / n
+        return [out]
+
+
+# ---------------------------------------------------------------------------
+# GEMM family — matmul / linear / batched matmul
+# hand-written MFMA path via ops.functional.linear
+# ---------------------------------------------------------------------------
+
+def _deduce_matmul_ds(op, x: Tensor, w: Tensor, out: Tensor,
+                      x_row_dim: int, x_k_dim: int, w_k_dim: int,
+                      w_col_dim: int, out_row_dim: int, out_col_dim: int):
+    """Shared DS deduction for matmul-like ops via per-device index tables."""
+    dsx, dsw = x.ds, w.ds
+    if dsx is None and dsw is None:
+        return
+    n = (dsx or dsw).device_num
+    if dsx is None:
+        dsx = DistributedStates(n, {-1: n} if n > 1 else {})
+    if dsw is None:
+        dsw = DistributedStates(n, {-1: n} if n > 1 else {})
+    if dsx.device_num != dsw.device_num:
+        raise ValueError("matmul inputs on different-size device groups")
+    kx, kw = dsx.get_dim(x_k_dim), dsw.get_dim(w_k_dim)
+    if kx != kw:
+        raise ValueError(
+            f"contraction-dim splits differ: x {kx} vs w {kw}")
+    nrow = dsx.get_dim(x_row_dim)
+    ncol = dsw.get_dim(w_col_dim)
+    npart = dsx.partial * dsw.partial * kx
+    table = []
+    for i in range(n):
+        sx = dsx.map_device_to_state_index(i)
+        sw = dsw.map_device_to_state_index(i)
+        ipart = (sx.get(-2, 0) * dsw.partial + sw.get(-2, 0)) * kx \
+            + sx.get(x_k_dim, 0)
+        ent = {}
+        if nrow > 1:
+            ent[out_row_dim] = sx.get(x_row_dim, 0)
+        if ncol > 1:
+            ent[out_col_dim] = sw.get(w_col_dim, 0)
+        if npart > 1:
+            ent[-2] = ipart
+        table.append(ent)
+    counts = {out_row_dim: nrow, out_col_dim: ncol, -2: npart}
+    out.ds = ds_from_index_table(n, table, counts)
+    out.device_group = x.device_group or w.device_group
+
+
+class LinearOp(OpInterface):
+    """y = x @ W^T (+ b); W stored [out_features, in_features] (torch
+    convention). inputs: x [..., K], w [N, K], optional bias [N]."""
+    type = "Linear"
+
+    def infer_meta(self, attrs, inputs):
+        x, w = inputs[0], inputs[1]
+        shape = list(x.shape[:-1]) + [w.shape[0]]
+        return [TensorMeta(shape, x.dtype)]
+
+    def deduce_states(self, op):
+        x, w = op.inputs[0], op.inputs[1]
+        nd = x.ndim
+        _deduce_matmul_ds(op, x, w, op.outputs[0],
+                          x_row_dim=0, x_k_dim=nd - 1,
+                          w_k_dim=1, w_col_dim=0,
+                          out_row_dim=0, out_col_dim=nd - 1)
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        x, w = inputs[0], inputs[1]
+        b = inputs[2] if len(inputs) > 2 else None
+        return [F.linear(x, w, b, trans_w=True)]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        gy = g[0]
+        x, w = op.inputs[0], op.inputs[1]
+        # dx = gy @ W          : Linear with w not transposed
+        dx = _make(gr, MatMul2DOp(), [gy, w], {"trans_a": False,
+                                               "trans_b": False},
+                   name="linear_dx").output()
+        # dw = gy^T @ x  (flattened over leading dims)
+        dw = _make(gr, MatMulGradWOp(), [gy, x], name="linear_dw").output()
+        grads = [dx, dw]
+        if len(op.inputs) > 2:
+            db = _make(gr, ReduceLeadingOp(), [gy, op.inputs[2]],
+                       name="linear_db").output()
+            grads.append(db)
+        return grads
+
+
+class MatMul2DOp(OpInterface):
+    """General matmul on last two dims with optional transposes.
+    a [..., M, K] @ b [K, N] -> [..., M, N]."""
+    type = "MatMul"
+
+    def infer_meta(self, attrs, inputs):
+        a, b = inputs
+        ta, tb = attrs.get("trans_a", False), attrs.get("trans_b", False)
+        ash = list(a.shape)
+        bsh = list(b.shape)
+        m = ash[-1] if ta else ash[-2] if len(ash) >= 2 else 1
+        n = bsh[-2] if tb else bsh[-1]
+        if len(ash) == 2 and len(bsh) == 2:
+            shape = [m, n]
+        else:
+            shape = list(ash[:-1]) + [n]
+        return [TensorMeta(shape, a.dtype)]
+
+    def deduce_states(self, op):
+        a, b = op.inputs
+        ta = op.attrs.get("trans_a", False)
+        tb = op.attrs.get("trans_b", False)
+        nda, ndb = a.ndim, b.ndim
+        _deduce_matmul_ds(
+            op, a, b, op.outputs[0],
+            x_row_dim=(nda - 1 if ta else 0),
+            x_k_dim=(nda - 2 if ta else nda - 1),
+            w_k_dim=(ndb - 1 if tb else ndb - 2),
+            w_col_dim=(ndb - 2 if tb else ndb - 1),
+            out_row_dim=0, out_col_dim=nda - 1)
+
+    def compute(self, op, inputs, ctx):
+        a, b = inputs
+        if op.attrs.get("trans_a", False):
+            a = a.transpose(-1, -2)
+        if op.attrs.get("trans_b", False):
+            b = b.transpose(-1, -2)
+        return [torch.matmul(a, b)]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        a, b = op.inputs
+        ta = op.attrs.get("trans_a", False)
+        tb = op.attrs.get("trans_b", False)
+        if ta or tb:
+            raise NotImplementedError("grad for transposed matmul")
+        gy = g[0]
+        da = _make(gr, MatMul2DOp(), [gy, b],
+                   {"trans_a": False, "trans_b": True}).output()
+        db = _make(gr, MatMulGradBOp(), [a, gy]).output()
+        return [da, db]
+
+
+class MatMulGradWOp(OpInterface):
+    """dw = gy^T @ x with gy [..., N], x [..., K] flattened -> [N, K]."""
+    type = "MatMulGradW"
+
+    def infer_meta(self, attrs, inputs):
+        gy, x = inputs
+        return [TensorMeta([gy.shape[-1], x.shape[-1]], x.dtype)]
+
+    def deduce_states(self, op):
+        gy, x = op.inputs
+        nd = gy.ndim
+        # gy rows (tokens) are the contraction; gy cols -> out dim 0;
+        # x cols -> out dim 1
+        _deduce_matmul_ds(op, gy, x, op.outputs[0],
+                          x_row_dim=nd - 1, x_k_dim=0,
+                          w_k_dim=0, w_col_dim=x.ndim - 1,
+                          out_row_dim=0, out_col_dim=1)
+
+    def compute(self, op, inputs, ctx):
+        gy, x = inputs
+        gy2 = gy.reshape(-1, gy.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        return [torch.matmul(gy2.t(), x2)]
+
+
+class MatMulGradBOp(OpInterface):
+    """db = a^T @ gy flattened: a [..., K] gy [..., N] -> [K, N]."""
+    type = "MatMulGradB"
+
+    def infer_meta(self, attrs, inputs):
+        a, gy = inputs
+        return [TensorMeta([a.shape[-1], gy.shape[-1]], a.dtype)]
+
+    def deduce_states(self, op):
+        a, gy = op.inputs
+        _deduce_matmul_ds(op, a, gy, op.outputs[0],
+                          x_row_dim=a.ndim - 1, x_k_dim=0,
+                          w_k_dim=0, w_col_dim=gy.ndim - 1,
+                          out_row_dim=0, out_col_dim=1)
+
+    def compute(self, op, inputs, ctx):
+        a, gy = inputs
+        a2 = a.reshape(-1, a.shape[-1])
+        g2 = gy.reshape(-1, gy.shape[-1])
+        return [torch.matmul(a2.t(), g2)]
+
+
+class ReduceLeadingOp(OpInterface):
+    """Sum over all leading dims to match inputs[1]'s (1-D) shape — bias
+    gradient."""
+    type = "ReduceLeading"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[1].shape, inputs[0].dtype)]
+
+    def deduce_states(self, op):
+        gy, ref = op.inputs
+        out = op.outputs[0]
+        if gy.ds is not None:
+            n = gy.ds.device_num
+            table = []
+            ncol = gy.ds.get_dim(gy.ndim - 1)
+            # token splits become partial
+            npart = gy.ds.partial
+            for d in gy.ds.split_dims():
+                if d != gy.ndim - 1:
+                    npart *= gy.ds.get_dim(d)
+            for i in range(n):
+                st = gy.ds.map_device_to_state_index(i)
+                ipart = st.get(-2, 0)
+                for d in gy.ds.split_dims():
+                    if d != gy.ndim - 1:
+                        ipart = ipart * gy.ds.get_dim(d) + st.get(d, 0)
+                ent = {}
+                if ncol > 1:
+                    ent[0] = st.get(gy.ndim - 1, 0)
+                if npart > 1:
+                    ent[-2] = ipart
+                table.append(ent)
+            out.ds = ds_from_index_table(n, table, {0: ncol, -2: npart})
+        out.device_group = ref.device_group
+
+    def compute(self, op, inputs, ctx):
+        gy = inputs[0]
+        return [gy.reshape(-1, gy.shape[-1]).sum(0)]
+
+
+class BatchMatMulOp(OpInterface):
+    type = "BatchMatMul"
+
+    def infer_meta(self, attrs, inputs):
+        a, b = inputs
+        shape = list(a.shape[:-1]) + [b.shape[-1]]
+        return [TensorMeta(shape, a.dtype)]
+
+    def compute(self, op, inputs, ctx):
+        return [torch.matmul(inputs[0], inputs[1])]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        a, b = op.inputs
+        gy = g[0]
+        da = _make(gr, BatchMatMulTNOp(), [gy, b], {"mode": "nt"}).output()
+        db = _make(gr, BatchMatMulTNOp(), [a, gy], {"mode": "tn"}).output()
+        return [da, db]
+
+
+class BatchMatMulTNOp(OpInterface):
+    type = "BatchMatMulTN"
+
+    def infer_meta(self, attrs, inputs):
+        a, b = inputs
+        if attrs["mode"] == "nt":   # a @ b^T
+            shape = list(a.shape[:-1]) + [b.shape[-2]]
+        else:                        # a^T @ b
+            shape = list(a.shape[:-2]) + [a.shape[-1], b.shape[-1]]
+        return [TensorMeta(shape, a.dtype)]
+
+    def compute(self, op, inputs, ctx):
+        a, b = inputs
+        if op.attrs["mode"] == "nt":
+            return [torch.matmul(a, b.transpose(-1, -2))]
+        return [torch.matmul(a.transpose(-1, -2), b)]

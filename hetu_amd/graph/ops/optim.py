@@ -1,0 +1,151 @@
+"""Optimizer update ops + Optimizer.minimize.
+
+Reference parity: hetu/graph/optim/optimizer.h:13-118 (SGD, Adam ->
+MakeAdamOp) and ops/optimizer_update.h:9-130; the fused Adam kernel is
+ops/hip/optimizers.hip (reference Optimizers.cu:145 AdamCuda). Parameter
+gradients arrive partial over the data-parallel dim and are reduced here via
+a CommOp to the parameter's layout (the engine's fast path replaces this
+with bucketed flat-buffer allreduce overlapped with backward).
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+
+from ..op import OpInterface
+from ..tensor import Tensor, TensorMeta
+from .basics import _make
+from .comm import make_comm
+
+
+class OptimizerUpdateOp(OpInterface):
+    """Base: inputs [param, grad]; output: dummy scalar (dependency token)."""
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta([], torch.float32)]
+
+    def deduce_states(self, op):
+        op.outputs[0].ds = None
+        op.outputs[0].device_group = op.inputs[0].device_group
+
+
+class SGDStepOp(OptimizerUpdateOp):
+    type = "SGDStep"
+
+    def __init__(self):
+        self.state: Dict = {}
+
+    def compute(self, op, inputs, ctx):
+        param, grad = inputs
+        lr = op.attrs["lr"]
+        momentum = op.attrs.get("momentum", 0.0)
+        if momentum > 0.0:
+            buf = self.state.get("momentum_buffer")
+            if buf is None:
+                buf = torch.zeros_like(param, dtype=torch.float32)
+                self.state["momentum_buffer"] = buf
+            buf.mul_(momentum).add_(grad.float())
+            upd = buf
+        else:
+            upd = grad.float()
+        param -= (lr * upd).to(param.dtype)
+        return [torch.zeros((), device=param.device)]
+
+
+class AdamStepOp(OptimizerUpdateOp):
+    """Fused Adam with fp32 master weights + m/v states; updates the
+    variable's storage in place (bf16/fp16 params re-materialized from the
+    fp32 master every step, as the reference's transfer params do)."""
+    type = "AdamStep"
+
+    def __init__(self):
+        self.state: Dict = {}
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        param, grad = inputs
+        a = op.attrs
+        st = self.state
+        if "m" not in st:
+            st["master"] = param.detach().float().clone()
+            st["m"] = torch.zeros_like(st["master"])
+            st["v"] = torch.zeros_like(st["master"])
+            st["step"] = 0
+        st["step"] += 1
+        out16 = param if param.dtype != torch.float32 else None
+        F.adam_step(st["master"], grad, st["m"], st["v"],
+                    a["lr"], a["beta1"], a["beta2"], a["eps"],
+                    a.get("weight_decay", 0.0), st["step"], out16)
+        if out16 is None:
+            param.copy_(st["master"])
+        return [torch.zeros((), device=param.device)]
+
+
+class GroupOp(OpInterface):
+    """Join node over update ops (reference ops/group.cc)."""
+    type = "Group"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta([], torch.float32)]
+
+    def deduce_states(self, op):
+        op.outputs[0].ds = None
+
+    def compute(self, op, inputs, ctx):
+        dev = inputs[0].device if inputs else "cpu"
+        return [torch.zeros((), device=dev)]
+
+
+class Optimizer:
+    def __init__(self, lr: float):
+        self.lr = lr
+        self.update_ops: List = []
+
+    def _make_update(self, graph, param: Tensor, grad: Tensor) -> Tensor:
+        raise NotImplementedError
+
+    def minimize(self, loss: Tensor, params: Optional[List[Tensor]] = None
+                 ) -> Tensor:
+        graph = loss.graph
+        params = params if params is not None else list(graph.parameters)
+        grads = graph.gradients([loss], params)
+        updates = []
+        for p, g in zip(params, grads):
+            if g is None:
+                continue
+            # parameter-grad reduction: partial (over dp) -> param layout
+            if (g.ds is not None and p.ds is not None
+                    and not g.ds.check_equal(p.ds)):
+                g = make_comm(graph, g, p.ds, name=f"grad_allreduce_{p.name}")
+            updates.append(self._make_update(graph, p, g))
+        self.update_ops = updates
+        return _make(graph, GroupOp(), updates, name="train_op").output()
+
+
+class SGD(Optimizer):
+    def __init__(self, lr: float = 0.01, momentum: float = 0.0):
+        super().__init__(lr)
+        self.momentum = momentum
+
+    def _make_update(self, graph, param, grad):
+        return _make(graph, SGDStepOp(), [param, grad],
+                     {"lr": self.lr, "momentum": self.momentum},
+                     name=f"sgd_{param.name}").output()
+
+
+class Adam(Optimizer):
+    def __init__(self, lr: float = 1e-3, beta1: float = 0.9,
+                 beta2: float = 0.999, eps: float = 1e-8,
+                 weight_decay: float = 0.0):
+        super().__init__(lr)
+        self.beta1, self.beta2 = beta1, beta2
+        self.eps = eps
+        self.weight_decay = weight_decay
+
+    def _make_update(self, graph, param, grad):
+        return _make(graph, AdamStepOp(), [param, grad],
+                     {"lr": self.lr, "beta1": self.beta1,
+                      "beta2": self.beta2, "eps": self.eps,
+                      "weight_decay": self.weight_decay},
+                     name=f"adam_{param.name}").output()

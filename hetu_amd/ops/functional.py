@@ -1,0 +1,430 @@
+"""Kernel dispatch: hand-written HIP/CDNA4 extension on GPU, torch on CPU.
+
+Every hot op of the reference kernel set
+(/root/reference/hetu/impl/kernel/ — RMSNorm.cu, FusedLayerNorm.cu,
+SwiGLU.cu, rotary.cu, Softmax.cu, SoftmaxCrossEntropySparse.cu,
+VocabParallelCrossEntropyLoss.cu, Dropout.cu, EmbeddingLookup.cu,
+Optimizers.cu AdamCuda, FlashAttention.cu, MatMul.cu) has an MI355X-native
+equivalent here. On a ROCm GPU the call MUST go through the in-tree
+_hetu_hip extension (hand-written gfx950 kernels); if the extension is
+missing on a CUDA/HIP device we raise — no silent eager fallback. On CPU the
+plain torch implementations below serve as the numerics reference used by
+the unit tests.
+"""
+from __future__ import annotations
+
+import math
+import os
+from typing import Optional, Tuple
+
+import torch
+
+_EXT = None
+_EXT_ERR = None
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        import importlib.util
+        here = os.path.dirname(__file__)
+        cands = [os.path.join(here, "hip", f) for f in
+                 os.listdir(os.path.join(here, "hip"))
+                 if f.startswith("_hetu_hip") and f.endswith(".so")] \
+            if os.path.isdir(os.path.join(here, "hip")) else []
+        if not cands:
+            raise ImportError("_hetu_hip extension not built "
+                              "(run python setup.py build_ext --inplace)")
+        spec = importlib.util.spec_from_file_location("_hetu_hip", cands[0])
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _EXT = mod
+    except Exception as e:  # noqa: BLE001
+        _EXT_ERR = e
+        _EXT = None
+    return _EXT
+
+
+def ext():
+    e = _load_ext()
+    if e is None:
+        raise RuntimeError(
+            f"hetu_amd HIP extension required on GPU but not available: "
+            f"{_EXT_ERR}")
+    return e
+
+
+def has_ext() -> bool:
+    return _load_ext() is not None
+
+
+def _gpu(*ts) -> bool:
+    return any(isinstance(t, torch.Tensor) and t.is_cuda for t in ts)
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm (fused fwd/bwd; reference RMSNorm.cu — block per row)
+# ---------------------------------------------------------------------------
+
+def rmsnorm_fwd(x: torch.Tensor, w: torch.Tensor, eps: float
+                ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """returns (y, rstd[rows] fp32)"""
+    if _gpu(x):
+        return ext().rmsnorm_fwd(x.contiguous(), w.contiguous(), eps)
+    xf = x.float()
+    rstd = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    y = (xf * rstd) * w.float()
+    return y.to(x.dtype), rstd.squeeze(-1)
+
+
+def rmsnorm_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
+                rstd: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    if _gpu(x):
+        return ext().rmsnorm_bwd(dy.contiguous(), x.contiguous(),
+                                 w.contiguous(), rstd.contiguous())
+    xf, dyf, wf = x.float(), dy.float(), w.float()
+    r = rstd.unsqueeze(-1)
+    xhat = xf * r
+    wdy = dyf * wf
+    c = (wdy * xhat).mean(-1, keepdim=True)
+    dx = (wdy - xhat * c) * r
+    dw = (dyf * xhat).reshape(-1, x.shape[-1]).sum(0)
+    return dx.to(x.dtype), dw.to(w.dtype)
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm (fused, Welford; reference FusedLayerNorm.cu:455-760)
+# ---------------------------------------------------------------------------
+
+def layernorm_fwd(x, w, b, eps):
+    if _gpu(x):
+        return ext().layernorm_fwd(x.contiguous(), w.contiguous(),
+                                   b.contiguous(), eps)
+    xf = x.float()
+    mean = xf.mean(-1, keepdim=True)
+    var = xf.var(-1, unbiased=False, keepdim=True)
+    rstd = torch.rsqrt(var + eps)
+    y = (xf - mean) * rstd * w.float() + b.float()
+    return y.to(x.dtype), mean.squeeze(-1), rstd.squeeze(-1)
+
+
+def layernorm_bwd(dy, x, w, mean, rstd):
+    if _gpu(x):
+        return ext().layernorm_bwd(dy.contiguous(), x.contiguous(),
+                                   w.contiguous(), mean.contiguous(),
+                                   rstd.contiguous())
+    xf, dyf, wf = x.float(), dy.float(), w.float()
+    mu = mean.unsqueeze(-1)
+    r = rstd.unsqueeze(-1)
+    xhat = (xf - mu) * r
+    wdy = dyf * wf
+    c1 = wdy.mean(-1, keepdim=True)
+    c2 = (wdy * xhat).mean(-1, keepdim=True)
+    dx = (wdy - c1 - xhat * c2) * r
+    D = x.shape[-1]
+    dw = (dyf * xhat).reshape(-1, D).sum(0)
+    db = dyf.reshape(-1, D).sum(0)
+    return dx.to(x.dtype), dw.to(w.dtype), db.to(w.dtype)
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU: y = silu(x1) * x2 over last-dim halves (reference SwiGLU.cu:14,30)
+# ---------------------------------------------------------------------------
+
+def swiglu_fwd(x):
+    if _gpu(x):
+        return ext().swiglu_fwd(x.contiguous())
+    x1, x2 = x.float().chunk(2, dim=-1)
+    return (torch.nn.functional.silu(x1) * x2).to(x.dtype)
+
+
+def swiglu_bwd(dy, x):
+    if _gpu(x):
+        return ext().swiglu_bwd(dy.contiguous(), x.contiguous())
+    x1, x2 = x.float().chunk(2, dim=-1)
+    dyf = dy.float()
+    sig = torch.sigmoid(x1)
+    silu = x1 * sig
+    dsilu = sig * (1 + x1 * (1 - sig))
+    dx1 = dyf * x2 * dsilu
+    dx2 = dyf * silu
+    return torch.cat([dx1, dx2], dim=-1).to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
+# RoPE (reference rotary.cu:97-185; NeoX-style half rotation)
+# ---------------------------------------------------------------------------
+
+def rope_fwd(x, cos, sin, interleaved: bool = False):
+    """x: [B, S, H, D] (or [S, H, D] packed); cos/sin: [S, D/2] fp32."""
+    if _gpu(x):
+        return ext().rope_fwd(x.contiguous(), cos.contiguous(),
+                              sin.contiguous())
+    return _rope_ref(x, cos, sin, False)
+
+
+def rope_bwd(dy, cos, sin, interleaved: bool = False):
+    if _gpu(dy):
+        return ext().rope_bwd(dy.contiguous(), cos.contiguous(),
+                              sin.contiguous())
+    return _rope_ref(dy, cos, sin, True)
+
+
+def _rope_ref(x, cos, sin, backward: bool):
+    xf = x.float()
+    D = x.shape[-1]
+    x1, x2 = xf[..., :D // 2], xf[..., D // 2:]
+    shape = [1] * x.ndim
+    shape[-3] = cos.shape[0]   # seq dim of [B, S, H, D] or packed [S, H, D]
+    shape[-1] = D // 2
+    c = cos.reshape(shape)
+    s = sin.reshape(shape)
+    if backward:
+        s = -s
+    y1 = x1 * c - x2 * s
+    y2 = x2 * c + x1 * s
+    return torch.cat([y1, y2], dim=-1).to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
+# Softmax (row; reference Softmax.cu:151,202,315)
+# ---------------------------------------------------------------------------
+
+def softmax_fwd(x, dim=-1):
+    if _gpu(x) and dim in (-1, x.ndim - 1):
+        return ext().softmax_fwd(x.contiguous())
+    return torch.softmax(x.float(), dim=dim).to(x.dtype)
+
+
+def softmax_bwd(dy, y, dim=-1):
+    if _gpu(dy) and dim in (-1, dy.ndim - 1):
+        return ext().softmax_bwd(dy.contiguous(), y.contiguous())
+    dyf, yf = dy.float(), y.float()
+    dx = (dyf - (dyf * yf).sum(dim, keepdim=True)) * yf
+    return dx.to(dy.dtype)
+
+
+# ---------------------------------------------------------------------------
+# Sparse softmax cross-entropy (reference SoftmaxCrossEntropySparse.cu)
+# and TP-sharded vocab-parallel CE (VocabParallelCrossEntropyLoss.cu:15,70)
+# ---------------------------------------------------------------------------
+
+def softmax_ce_fwd(logits, labels, ignore_index: int = -100):
+    """logits [N, V], labels [N] -> (loss[N] fp32, logsumexp[N] fp32)."""
+    if _gpu(logits):
+        return ext().softmax_ce_fwd(logits.contiguous(),
+                                    labels.contiguous(), ignore_index)
+    lf = logits.float()
+    lse = torch.logsumexp(lf, dim=-1)
+    mask = labels != ignore_index
+    safe = labels.clamp(min=0)
+    picked = lf.gather(-1, safe.unsqueeze(-1)).squeeze(-1)
+    loss = torch.where(mask, lse - picked, torch.zeros_like(lse))
+    return loss, lse
+
+
+def softmax_ce_bwd(dloss, logits, labels, lse, ignore_index: int = -100):
+    if _gpu(logits):
+        return ext().softmax_ce_bwd(dloss.contiguous(), logits.contiguous(),
+                                    labels.contiguous(), lse.contiguous(),
+                                    ignore_index)
+    lf = logits.float()
+    p = torch.exp(lf - lse.unsqueeze(-1))
+    mask = (labels != ignore_index)
+    safe = labels.clamp(min=0)
+    onehot = torch.zeros_like(lf).scatter_(-1, safe.unsqueeze(-1), 1.0)
+    g = (p - onehot) * (dloss * mask.to(dloss.dtype)).unsqueeze(-1)
+    return g.to(logits.dtype)
+
+
+def vocab_parallel_ce_local_stats(logits, labels, vocab_start, vocab_end,
+                                  ignore_index: int = -100):
+    """Per-rank stage of the vocab-parallel CE: local max, local sum-exp (at
+    given max), and predicted-logit for labels owned by this shard.
+    Cross-rank max/sum allreduce happens at op level (see graph/ops/loss.py).
+    """
+    if _gpu(logits):
+        return ext().vp_ce_local(logits.contiguous(), labels.contiguous(),
+                                 vocab_start, vocab_end, ignore_index)
+    lf = logits.float()
+    lmax = lf.max(-1).values
+    in_shard = (labels >= vocab_start) & (labels < vocab_end) & \
+               (labels != ignore_index)
+    local_idx = (labels - vocab_start).clamp(min=0, max=lf.shape[-1] - 1)
+    picked = lf.gather(-1, local_idx.unsqueeze(-1)).squeeze(-1)
+    picked = torch.where(in_shard, picked, torch.zeros_like(picked))
+    return lmax, picked
+
+
+# ---------------------------------------------------------------------------
+# Dropout (Philox-seeded, stateless; reference Dropout.cu)
+# ---------------------------------------------------------------------------
+
+def dropout_fwd(x, p: float, seed: int, offset: int):
+    if p <= 0.0:
+        return x, None
+    if _gpu(x):
+        return ext().dropout_fwd(x.contiguous(), p, seed, offset)
+    g = torch.Generator(device="cpu").manual_seed(seed + offset)
+    mask = (torch.rand(x.shape, generator=g, device=x.device) >= p)
+    y = x * mask.to(x.dtype) / (1.0 - p)
+    return y, mask
+
+
+def dropout_bwd(dy, mask, p: float, seed: int, offset: int):
+    if p <= 0.0:
+        return dy
+    if _gpu(dy):
+        return ext().dropout_bwd(dy.contiguous(), mask, p, seed, offset)
+    return dy * mask.to(dy.dtype) / (1.0 - p)
+
+
+# ---------------------------------------------------------------------------
+# Embedding (reference EmbeddingLookup.cu)
+# ---------------------------------------------------------------------------
+
+def embedding_fwd(table, ids):
+    if _gpu(table):
+        return ext().embedding_fwd(table.contiguous(), ids.contiguous())
+    return table[ids]
+
+
+def embedding_bwd(dy, ids, num_rows: int):
+    if _gpu(dy):
+        return ext().embedding_bwd(dy.contiguous(), ids.contiguous(),
+                                   num_rows)
+    D = dy.shape[-1]
+    g = torch.zeros(num_rows, D, dtype=torch.float32, device=dy.device)
+    g.index_add_(0, ids.reshape(-1), dy.reshape(-1, D).float())
+    return g.to(dy.dtype)
+
+
+# ---------------------------------------------------------------------------
+# Fused Adam (reference Optimizers.cu:145 AdamCuda — m/v update + bias corr
+# + weight decay in one pass; fp32 master weights)
+# ---------------------------------------------------------------------------
+
+def adam_step(param32, grad, m, v, lr, beta1, beta2, eps, weight_decay,
+              step, param_out16: Optional[torch.Tensor] = None):
+    if _gpu(param32):
+        ext().adam_step(param32, grad, m, v, lr, beta1, beta2, eps,
+                        weight_decay, step,
+                        param_out16 if param_out16 is not None else grad.new_empty(0))
+        return
+    gf = grad.float()
+    if weight_decay != 0.0:
+        gf = gf + weight_decay * param32
+    m.mul_(beta1).add_(gf, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
+    bc1 = 1 - beta1 ** step
+    bc2 = 1 - beta2 ** step
+    update = (m / bc1) / (torch.sqrt(v / bc2) + eps)
+    param32.add_(update, alpha=-lr)
+    if param_out16 is not None:
+        param_out16.copy_(param32.to(param_out16.dtype))
+
+
+# ---------------------------------------------------------------------------
+# Flash attention (reference FlashAttention.cu wraps flash_attn 2;
+# here: hand-written CDNA4 MFMA kernel, see ops/hip/attention.hip)
+# ---------------------------------------------------------------------------
+
+def flash_attn_fwd(q, k, v, causal: bool, scale: Optional[float] = None):
+    """q,k,v: [B, H, S, D] (kv may have fewer heads - GQA).
+    Returns (out, lse[B,H,S] fp32)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _gpu(q):
+        return ext().flash_attn_fwd(q.contiguous(), k.contiguous(),
+                                    v.contiguous(), causal, scale)
+    return _attn_ref_fwd(q, k, v, causal, scale)
+
+
+def flash_attn_bwd(dout, q, k, v, out, lse, causal: bool,
+                   scale: Optional[float] = None):
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _gpu(q):
+        return ext().flash_attn_bwd(dout.contiguous(), q.contiguous(),
+                                    k.contiguous(), v.contiguous(),
+                                    out.contiguous(), lse.contiguous(),
+                                    causal, scale)
+    return _attn_ref_bwd(dout, q, k, v, lse, causal, scale)
+
+
+def _repeat_kv(k, n_head):
+    if k.shape[1] == n_head:
+        return k
+    rep = n_head // k.shape[1]
+    return k.repeat_interleave(rep, dim=1)
+
+
+def _attn_ref_fwd(q, k, v, causal, scale):
+    B, H, S, D = q.shape
+    kf = _repeat_kv(k, H).float()
+    vf = _repeat_kv(v, H).float()
+    qf = q.float()
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        Skv = kf.shape[2]
+        mask = torch.ones(S, Skv, dtype=torch.bool, device=q.device).tril(
+            diagonal=Skv - S)
+        scores = scores.masked_fill(~mask, float("-inf"))
+    lse = torch.logsumexp(scores, dim=-1)
+    p = torch.exp(scores - lse.unsqueeze(-1))
+    out = torch.matmul(p, vf)
+    return out.to(q.dtype), lse
+
+
+def _attn_ref_bwd(dout, q, k, v, lse, causal, scale):
+    B, H, S, D = q.shape
+    Hkv = k.shape[1]
+    kf = _repeat_kv(k, H).float()
+    vf = _repeat_kv(v, H).float()
+    qf, dof = q.float(), dout.float()
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        Skv = kf.shape[2]
+        mask = torch.ones(S, Skv, dtype=torch.bool, device=q.device).tril(
+            diagonal=Skv - S)
+        scores = scores.masked_fill(~mask, float("-inf"))
+    p = torch.exp(scores - lse.unsqueeze(-1).float())
+    dv = torch.matmul(p.transpose(-1, -2), dof)
+    dp = torch.matmul(dof, vf.transpose(-1, -2))
+    delta = (dp * p).sum(-1, keepdim=True)
+    ds = p * (dp - delta) * scale
+    dq = torch.matmul(ds, kf)
+    dk = torch.matmul(ds.transpose(-1, -2), qf)
+    if Hkv != H:
+        rep = H // Hkv
+        dk = dk.reshape(B, Hkv, rep, -1, D).sum(2)
+        dv = dv.reshape(B, Hkv, rep, -1, D).sum(2)
+    return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype)
+
+
+# ---------------------------------------------------------------------------
+# GEMM — hand-written MFMA bf16 kernel for the transformer hot path;
+# library (hipBLASLt via torch.matmul) for general shapes.
+# ---------------------------------------------------------------------------
+
+_HETU_GEMM = os.environ.get("HETU_AMD_GEMM", "hip")  # hip | blas
+
+
+def linear(x, w, bias=None, trans_w: bool = True):
+    """x [..., K] @ (w [N, K] if trans_w else w [K, N]) + bias."""
+    if _gpu(x) and _HETU_GEMM == "hip" and x.dtype == torch.bfloat16:
+        xs = x.reshape(-1, x.shape[-1]).contiguous()
+        M, K = xs.shape
+        N = w.shape[0] if trans_w else w.shape[1]
+        if M % 16 == 0 and N % 64 == 0 and K % 64 == 0:
+            y = ext().gemm_bf16(xs, w.contiguous(), trans_w)
+            if bias is not None:
+                y = y + bias
+            return y.reshape(*x.shape[:-1], N)
+    wm = w.t() if trans_w else w
+    y = torch.matmul(x, wm.to(x.dtype))
+    if bias is not None:
+        y = y + bias
+    return y
