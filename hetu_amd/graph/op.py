@@ -28,13 +28,16 @@ class OpInterface:
         raise NotImplementedError
 
     def deduce_states(self, op: "Op") -> None:
-        """Default SPMD propagation: outputs inherit input[0]'s ds if all
-        input ds are equal; ops with nontrivial layouts override."""
+        """Default SPMD propagation: outputs inherit the common ds of the
+        non-pure-duplicate inputs (weights that are replicated don't change
+        an activation's layout); ops with nontrivial layouts override."""
         dss = [t.ds for t in op.inputs if t.ds is not None]
+        nontrivial = [d for d in dss if not d.is_pure_dup()]
         ds = None
-        if dss:
-            ds = dss[0]
-            for other in dss[1:]:
+        pick = nontrivial or dss
+        if pick:
+            ds = pick[0]
+            for other in pick[1:]:
                 if not other.check_equal(ds):
                     ds = None
                     break

@@ -141,8 +141,48 @@ class ReduceToShapeOp(OpInterface):
         return [TensorMeta(inputs[1].shape, inputs[0].dtype)]
 
     def deduce_states(self, op):
-        op.outputs[0].ds = op.inputs[1].ds
-        op.outputs[0].device_group = op.inputs[1].device_group
+        g, ref = op.inputs
+        out = op.outputs[0]
+        out.device_group = ref.device_group or g.device_group
+        if g.ds is None:
+            out.ds = ref.ds
+            return
+        n = g.ds.device_num
+        # dims of g that get summed away: leading extras + broadcast dims
+        gnd, rnd = g.ndim, ref.ndim
+        reduced = set(range(gnd - rnd))
+        for i in range(rnd):
+            gd = gnd - rnd + i
+            try:
+                if (not isinstance(ref.shape[i], IntSymbol)
+                        and not isinstance(g.shape[gd], IntSymbol)
+                        and ref.shape[i] == 1 and g.shape[gd] != 1):
+                    reduced.add(gd)
+            except Exception:  # noqa: BLE001
+                pass
+        table = []
+        counts: Dict[int, int] = {}
+        npart = g.ds.partial
+        for d in g.ds.split_dims():
+            if d in reduced:
+                npart *= g.ds.get_dim(d)
+        for i in range(n):
+            st = g.ds.map_device_to_state_index(i)
+            ent: Dict[int, int] = {}
+            ip = st.get(-2, 0)
+            for d in g.ds.split_dims():
+                if d in reduced:
+                    ip = ip * g.ds.get_dim(d) + st.get(d, 0)
+                else:
+                    rd = d - (gnd - rnd)
+                    if g.ds.get_dim(d) > 1:
+                        ent[rd] = st.get(d, 0)
+                        counts[rd] = g.ds.get_dim(d)
+            if npart > 1:
+                ent[-2] = ip
+            table.append(ent)
+        counts[-2] = npart
+        out.ds = ds_from_index_table(n, table, counts)
 
     def compute(self, op, inputs, ctx):
         g, ref = inputs
@@ -160,6 +200,63 @@ class _BinaryOp(OpInterface):
         dtype = inputs[0].dtype
         return [TensorMeta(_bcast_shape(inputs[0].shape, inputs[1].shape),
                            dtype)]
+
+    def deduce_states(self, op):
+        """Broadcast-aware elementwise SPMD rule: a dim split on one side
+        must be matched by an equal split (same device->shard mapping) or a
+        broadcast (absent / size-1) dim on the other; partial inputs are
+        rejected (they must be comm'ed first)."""
+        a, b = op.inputs[0], op.inputs[1]
+        out = op.outputs[0]
+        out.device_group = a.device_group or b.device_group
+        if a.ds is None and b.ds is None:
+            out.ds = None
+            return
+        if a.ds is None or b.ds is None:
+            src_t = a if a.ds is not None else b
+            # the un-annotated side is replicated; output inherits the
+            # annotated layout (split dims right-aligned to output rank).
+            out.ds = _bcast_ds(src_t.ds, src_t.ndim, out.meta.ndim)
+            return
+        if a.ds.partial > 1 or b.ds.partial > 1:
+            raise ValueError(
+                f"elementwise op {op.name} on partial input (a={a.ds}, "
+                f"b={b.ds}): insert a comm op first")
+        n = a.ds.device_num
+        ond = out.meta.ndim
+        table = []
+        counts: Dict[int, int] = {}
+        for i in range(n):
+            sa = a.ds.map_device_to_state_index(i)
+            sb = b.ds.map_device_to_state_index(i)
+            ent: Dict[int, int] = {}
+            for od in range(ond):
+                ad = od - (ond - a.ndim)
+                bd = od - (ond - b.ndim)
+                na = a.ds.get_dim(ad) if ad >= 0 else 1
+                nb = b.ds.get_dim(bd) if bd >= 0 else 1
+                # broadcast dims (size 1) cannot be split
+                if na > 1 and nb > 1:
+                    if na != nb or sa.get(ad, 0) != sb.get(bd, 0):
+                        raise ValueError(
+                            f"misaligned splits on dim {od} of {op.name}")
+                nn = max(na, nb)
+                if nn > 1:
+                    ent[od] = sa.get(ad, 0) if na > 1 else sb.get(bd, 0)
+                    counts[od] = nn
+            table.append(ent)
+        out.ds = ds_from_index_table(n, table, counts)
+
+
+def _bcast_ds(src: DistributedStates, src_tensor_ndim: int, out_ndim: int
+              ) -> DistributedStates:
+    """Shift src's split dims to the output rank (right-aligned)."""
+    shift = out_ndim - src_tensor_ndim
+    if shift == 0:
+        return src
+    states = {(d + shift if d >= 0 else d): c for d, c in src.states.items()}
+    order = [(d + shift if d >= 0 else d) for d in src.order]
+    return DistributedStates(src.device_num, states, order)
 
 
 class AddOp(_BinaryOp):

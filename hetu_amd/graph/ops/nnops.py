@@ -226,6 +226,25 @@ class DropoutGradOp(OpInterface):
 # LayerNorm / RMSNorm (fused kernels)
 # ---------------------------------------------------------------------------
 
+def _norm_deduce(op):
+    """Norm ops: y follows x; row-stat outputs follow x minus last dim.
+    Weight/bias inputs are duplicated and do not affect the layout."""
+    x = op.inputs[0]
+    if x.ds is not None:
+        if x.ds.get_dim(x.ndim - 1) > 1:
+            raise ValueError(
+                f"{op.name}: input split on the normalized dim; use a "
+                f"sequence-parallel layer (comm first)")
+        if x.ds.partial > 1:
+            raise ValueError(f"{op.name}: partial input needs comm first")
+        op.outputs[0].ds = x.ds
+        for stat in op.outputs[1:]:
+            if stat.meta.ndim == x.ndim - 1:
+                stat.ds = x.ds
+    for out in op.outputs:
+        out.device_group = x.device_group
+
+
 class LayerNormOp(OpInterface):
     type = "LayerNorm"
 
@@ -234,6 +253,9 @@ class LayerNormOp(OpInterface):
         return [TensorMeta(inputs[0].shape, inputs[0].dtype),
                 TensorMeta(rows, torch.float32),
                 TensorMeta(rows, torch.float32)]
+
+    def deduce_states(self, op):
+        _norm_deduce(op)
 
     def compute(self, op, inputs, ctx):
         from ...ops import functional as F
@@ -291,6 +313,9 @@ class RMSNormOp(OpInterface):
         rows = list(inputs[0].shape[:-1])
         return [TensorMeta(inputs[0].shape, inputs[0].dtype),
                 TensorMeta(rows, torch.float32)]
+
+    def deduce_states(self, op):
+        _norm_deduce(op)
 
     def compute(self, op, inputs, ctx):
         from ...ops import functional as F

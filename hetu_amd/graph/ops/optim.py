@@ -56,11 +56,30 @@ class SGDStepOp(OptimizerUpdateOp):
 class AdamStepOp(OptimizerUpdateOp):
     """Fused Adam with fp32 master weights + m/v states; updates the
     variable's storage in place (bf16/fp16 params re-materialized from the
-    fp32 master every step, as the reference's transfer params do)."""
+    fp32 master every step, as the reference's transfer params do).
+
+    hipGraph capture support: bias corrections flow host-pinned -> device
+    tensor -> kernel pointer; a captured step re-reads the pinned buffer,
+    which `set_replay_step` updates between replays."""
     type = "AdamStep"
+
+    _instances: List["AdamStepOp"] = []
 
     def __init__(self):
         self.state: Dict = {}
+        AdamStepOp._instances.append(self)
+
+    @classmethod
+    def set_replay_step(cls, step: int):
+        """Update every instance's pinned bias-correction buffer for a
+        graph replay at optimizer step `step` (1-based)."""
+        for inst in cls._instances:
+            st = inst.state
+            if "bc_host" in st:
+                b1, b2 = st["betas"]
+                st["bc_host"][0] = 1.0 - b1 ** step
+                st["bc_host"][1] = 1.0 - b2 ** step
+                st["step"] = step
 
     def compute(self, op, inputs, ctx):
         from ...ops import functional as F
@@ -72,11 +91,23 @@ class AdamStepOp(OptimizerUpdateOp):
             st["m"] = torch.zeros_like(st["master"])
             st["v"] = torch.zeros_like(st["master"])
             st["step"] = 0
+            st["betas"] = (a["beta1"], a["beta2"])
+            if param.is_cuda:
+                st["bc_host"] = torch.empty(2, dtype=torch.float32,
+                                            pin_memory=True)
+                st["bc_dev"] = torch.empty(2, dtype=torch.float32,
+                                           device=param.device)
         st["step"] += 1
+        bc_dev = None
+        if "bc_host" in st:
+            st["bc_host"][0] = 1.0 - a["beta1"] ** st["step"]
+            st["bc_host"][1] = 1.0 - a["beta2"] ** st["step"]
+            st["bc_dev"].copy_(st["bc_host"], non_blocking=True)
+            bc_dev = st["bc_dev"]
         out16 = param if param.dtype != torch.float32 else None
         F.adam_step(st["master"], grad, st["m"], st["v"],
                     a["lr"], a["beta1"], a["beta2"], a["eps"],
-                    a.get("weight_decay", 0.0), st["step"], out16)
+                    a.get("weight_decay", 0.0), st["step"], out16, bc_dev)
         if out16 is None:
             param.copy_(st["master"])
         return [torch.zeros((), device=param.device)]

@@ -307,11 +307,19 @@ def embedding_bwd(dy, ids, num_rows: int):
 # ---------------------------------------------------------------------------
 
 def adam_step(param32, grad, m, v, lr, beta1, beta2, eps, weight_decay,
-              step, param_out16: Optional[torch.Tensor] = None):
+              step, param_out16: Optional[torch.Tensor] = None,
+              bc_dev: Optional[torch.Tensor] = None):
+    """bc_dev: optional fp32 device tensor [2] = (1-b1^t, 1-b2^t); when
+    given, the kernel reads bias corrections from it — this keeps a
+    hipGraph-captured train step correct across replays (the host updates
+    the pinned source of bc_dev between replays)."""
     if _gpu(param32):
         ext().adam_step(param32, grad, m, v, lr, beta1, beta2, eps,
                         weight_decay, step,
-                        param_out16 if param_out16 is not None else grad.new_empty(0))
+                        param_out16 if param_out16 is not None
+                        else grad.new_empty(0),
+                        bc_dev if bc_dev is not None
+                        else param32.new_empty(0))
         return
     gf = grad.float()
     if weight_decay != 0.0:
