@@ -417,16 +417,23 @@ def _attn_ref_bwd(dout, q, k, v, lse, causal, scale):
 # library (hipBLASLt via torch.matmul) for general shapes.
 # ---------------------------------------------------------------------------
 
-_HETU_GEMM = os.environ.get("HETU_AMD_GEMM", "hip")  # hip | blas
+# GEMM routing: the hand-written MFMA kernel (ops/hip/gemm.hip, measured
+# ~910 TF bf16 on MI355X) vs hipBLASLt via torch.matmul (~1380-1550 TF on
+# the transformer shapes). Plain projection GEMMs default to the library
+# per the "library for plain GEMMs" rule; HETU_AMD_GEMM=hip forces the
+# hand kernel (used by tests/microbenchmarks, and the target of the
+# 256^2 8-phase upgrade).
+_HETU_GEMM = os.environ.get("HETU_AMD_GEMM", "blas")  # hip | blas
 
 
 def linear(x, w, bias=None, trans_w: bool = True):
     """x [..., K] @ (w [N, K] if trans_w else w [K, N]) + bias."""
-    if _gpu(x) and _HETU_GEMM == "hip" and x.dtype == torch.bfloat16:
+    if _gpu(x) and _HETU_GEMM == "hip" and x.dtype == torch.bfloat16 \
+            and trans_w:
         xs = x.reshape(-1, x.shape[-1]).contiguous()
         M, K = xs.shape
-        N = w.shape[0] if trans_w else w.shape[1]
-        if M % 16 == 0 and N % 64 == 0 and K % 64 == 0:
+        N = w.shape[0]
+        if M % 128 == 0 and N % 128 == 0 and K % 64 == 0:
             y = ext().gemm_bf16(xs, w.contiguous(), trans_w)
             if bias is not None:
                 y = y + bias

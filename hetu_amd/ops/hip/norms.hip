@@ -47,7 +47,10 @@ __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x,
   }
 }
 
-template <typename T>
+// dw accumulated per-thread in registers across this block's rows; ONE
+// atomicAdd per element per block at the end (grid is capped so total
+// atomic traffic is ~grid*D, not rows*D).
+template <typename T, int NCHUNK>
 __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
                                    const T* __restrict__ x,
                                    const T* __restrict__ w,
@@ -57,6 +60,12 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
                                    int64_t rows, int D) {
   constexpr int V = VecIO<T>::VEC;
   __shared__ float smem[16];
+  float dwacc[NCHUNK][V];
+#pragma unroll
+  for (int c = 0; c < NCHUNK; ++c)
+#pragma unroll
+    for (int j = 0; j < V; ++j) dwacc[c][j] = 0.f;
+
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* dyr = dy + row * D;
     const T* xr = x + row * D;
@@ -72,7 +81,8 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
       for (int j = 0; j < V; ++j) dot += dv[j] * wv[j] * xv[j] * r;
     }
     dot = block_sum(dot, smem) / D;
-    for (int i = threadIdx.x * V; i < D; i += BLOCK * V) {
+    int c = 0;
+    for (int i = threadIdx.x * V; i < D; i += BLOCK * V, ++c) {
       float dv[VecIO<T>::VEC], xv[VecIO<T>::VEC], wv[VecIO<T>::VEC];
       VecIO<T>::load(dyr + i, dv);
       VecIO<T>::load(xr + i, xv);
@@ -82,10 +92,17 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
       for (int j = 0; j < V; ++j) {
         float xhat = xv[j] * r;
         o[j] = (dv[j] * wv[j] - xhat * dot) * r;
-        atomicAdd(dw_accum + i + j, dv[j] * xhat);
+        dwacc[c][j] += dv[j] * xhat;
       }
       VecIO<T>::store(dxr + i, o);
     }
+  }
+  {
+    int c = 0;
+    for (int i = threadIdx.x * V; i < D; i += BLOCK * V, ++c)
+#pragma unroll
+      for (int j = 0; j < V; ++j)
+        atomicAdd(dw_accum + i + j, dwacc[c][j]);
   }
 }
 
@@ -130,7 +147,7 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
   }
 }
 
-template <typename T>
+template <typename T, int NCHUNK>
 __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
                                      const T* __restrict__ w,
@@ -142,6 +159,11 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
                                      int64_t rows, int D) {
   constexpr int V = VecIO<T>::VEC;
   __shared__ float smem[16];
+  float dwacc[NCHUNK][V], dbacc[NCHUNK][V];
+#pragma unroll
+  for (int c = 0; c < NCHUNK; ++c)
+#pragma unroll
+    for (int j = 0; j < V; ++j) { dwacc[c][j] = 0.f; dbacc[c][j] = 0.f; }
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* dyr = dy + row * D;
     const T* xr = x + row * D;
@@ -163,7 +185,8 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
     }
     c1 = block_sum(c1, smem) / D;
     c2 = block_sum(c2, smem) / D;
-    for (int i = threadIdx.x * V; i < D; i += BLOCK * V) {
+    int c = 0;
+    for (int i = threadIdx.x * V; i < D; i += BLOCK * V, ++c) {
       float dv[VecIO<T>::VEC], xv[VecIO<T>::VEC], wv[VecIO<T>::VEC];
       VecIO<T>::load(dyr + i, dv);
       VecIO<T>::load(xr + i, xv);
@@ -174,11 +197,20 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ dy,
         float xhat = (xv[j] - mu) * r;
         float wdy = dv[j] * wv[j];
         o[j] = (wdy - c1 - xhat * c2) * r;
-        atomicAdd(dw_accum + i + j, dv[j] * xhat);
-        atomicAdd(db_accum + i + j, dv[j]);
+        dwacc[c][j] += dv[j] * xhat;
+        dbacc[c][j] += dv[j];
       }
       VecIO<T>::store(dxr + i, o);
     }
+  }
+  {
+    int c = 0;
+    for (int i = threadIdx.x * V; i < D; i += BLOCK * V, ++c)
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        atomicAdd(dw_accum + i + j, dwacc[c][j]);
+        atomicAdd(db_accum + i + j, dbacc[c][j]);
+      }
   }
 }
 
@@ -216,15 +248,26 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dx = torch::empty_like(x);
   auto dw32 = torch::zeros({D}, x.options().dtype(at::kFloat));
   auto stream = hetu_current_stream();
+  int grid = (int)std::min<int64_t>(rows, 1024);
   DISPATCH_FLOAT(x, "rmsnorm_bwd", [&] {
-    hipLaunchKernelGGL(rmsnorm_bwd_kernel<scalar_t>, dim3(row_grid(rows)),
-                       dim3(BLOCK), 0, stream,
-                       (const scalar_t*)dy.data_ptr(),
-                       (const scalar_t*)x.data_ptr(),
-                       (const scalar_t*)w.data_ptr(),
-                       rstd.data_ptr<float>(),
-                       (scalar_t*)dx.data_ptr(), dw32.data_ptr<float>(),
-                       rows, D);
+    constexpr int V = VecIO<scalar_t>::VEC;
+    const int nchunk = (D + BLOCK * V - 1) / (BLOCK * V);
+    auto launch = [&](auto tag) {
+      constexpr int NC = decltype(tag)::value;
+      hipLaunchKernelGGL((rmsnorm_bwd_kernel<scalar_t, NC>), dim3(grid),
+                         dim3(BLOCK), 0, stream,
+                         (const scalar_t*)dy.data_ptr(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)w.data_ptr(),
+                         rstd.data_ptr<float>(),
+                         (scalar_t*)dx.data_ptr(), dw32.data_ptr<float>(),
+                         rows, D);
+    };
+    if (nchunk <= 1) launch(std::integral_constant<int, 1>{});
+    else if (nchunk <= 2) launch(std::integral_constant<int, 2>{});
+    else if (nchunk <= 4) launch(std::integral_constant<int, 4>{});
+    else if (nchunk <= 8) launch(std::integral_constant<int, 8>{});
+    else TORCH_CHECK(false, "rmsnorm_bwd: D too large");
   });
   return {dx, dw32.to(w.scalar_type())};
 }
@@ -260,15 +303,26 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dw32 = torch::zeros({D}, x.options().dtype(at::kFloat));
   auto db32 = torch::zeros({D}, x.options().dtype(at::kFloat));
   auto stream = hetu_current_stream();
+  int grid = (int)std::min<int64_t>(rows, 1024);
   DISPATCH_FLOAT(x, "layernorm_bwd", [&] {
-    hipLaunchKernelGGL(layernorm_bwd_kernel<scalar_t>, dim3(row_grid(rows)),
-                       dim3(BLOCK), 0, stream,
-                       (const scalar_t*)dy.data_ptr(),
-                       (const scalar_t*)x.data_ptr(),
-                       (const scalar_t*)w.data_ptr(),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       (scalar_t*)dx.data_ptr(), dw32.data_ptr<float>(),
-                       db32.data_ptr<float>(), rows, D);
+    constexpr int V = VecIO<scalar_t>::VEC;
+    const int nchunk = (D + BLOCK * V - 1) / (BLOCK * V);
+    auto launch = [&](auto tag) {
+      constexpr int NC = decltype(tag)::value;
+      hipLaunchKernelGGL((layernorm_bwd_kernel<scalar_t, NC>), dim3(grid),
+                         dim3(BLOCK), 0, stream,
+                         (const scalar_t*)dy.data_ptr(),
+                         (const scalar_t*)x.data_ptr(),
+                         (const scalar_t*)w.data_ptr(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         (scalar_t*)dx.data_ptr(), dw32.data_ptr<float>(),
+                         db32.data_ptr<float>(), rows, D);
+    };
+    if (nchunk <= 1) launch(std::integral_constant<int, 1>{});
+    else if (nchunk <= 2) launch(std::integral_constant<int, 2>{});
+    else if (nchunk <= 4) launch(std::integral_constant<int, 4>{});
+    else if (nchunk <= 8) launch(std::integral_constant<int, 8>{});
+    else TORCH_CHECK(false, "layernorm_bwd: D too large");
   });
   return {dx, dw32.to(w.scalar_type()), db32.to(w.scalar_type())};
 }
