@@ -127,8 +127,12 @@ def add_n(ts: Sequence[Tensor]) -> Tensor:
 
 # ---- shape -----------------------------------------------------------------
 
-def reshape(a: Tensor, shape) -> Tensor:
-    return _cg().make_op(B.ReshapeOp(), [a], {"shape": tuple(shape)}).output()
+def reshape(a: Tensor, shape, ds=None) -> Tensor:
+    """`ds` overrides the output layout (reshape cannot always map split
+    dims mechanically, e.g. [B,S,3h] -> [B,S,3,H,Dh] moves a tp split from
+    dim 2 to dim 3 — the caller knows the intent)."""
+    return _cg().make_op(B.ReshapeOp(), [a], {"shape": tuple(shape)},
+                         ds_list=[ds] if ds is not None else None).output()
 
 
 def transpose(a: Tensor, dim0: int, dim1: int) -> Tensor:
@@ -257,6 +261,20 @@ def softmax_cross_entropy_sparse(logits, labels, ignore_index=-100):
 
 def mse_loss(x, y):
     return _cg().make_op(N.MSELossOp(), [x, y], {}).output()
+
+
+def vocab_parallel_embedding(table, ids, vocab: int):
+    from . import parallel_ops as P
+    return _cg().make_op(P.VocabParallelEmbeddingOp(), [table, ids],
+                         {"vocab": vocab}).output()
+
+
+def vocab_parallel_cross_entropy(logits, labels, vocab: int,
+                                 ignore_index: int = -100):
+    from . import parallel_ops as P
+    return _cg().make_op(P.VocabParallelCrossEntropyOp(), [logits, labels],
+                         {"vocab": vocab,
+                          "ignore_index": ignore_index}).output(0)
 
 
 # ---- comm ------------------------------------------------------------------

@@ -48,7 +48,8 @@ class GeluOp(_ScalarOp):
     type = "Gelu"
 
     def compute(self, op, inputs, ctx):
-        return [torch.nn.functional.gelu(inputs[0], approximate="tanh")]
+        from ...ops import functional as F
+        return [F.gelu_fwd(inputs[0])]
 
     def gradient(self, op, g):
         gr = _g(op.outputs[0])
@@ -59,20 +60,16 @@ class GeluGradOp(_ScalarOp):
     type = "GeluGrad"
 
     def compute(self, op, inputs, ctx):
-        g, x = inputs
-        xf = x.float()
-        c = 0.7978845608028654  # sqrt(2/pi)
-        a = 0.044715
-        t = torch.tanh(c * (xf + a * xf ** 3))
-        dt = (1 - t * t) * c * (1 + 3 * a * xf * xf)
-        return [(g.float() * (0.5 * (1 + t) + 0.5 * xf * dt)).to(g.dtype)]
+        from ...ops import functional as F
+        return [F.gelu_bwd(inputs[0], inputs[1])]
 
 
 class SiluOp(_ScalarOp):
     type = "Silu"
 
     def compute(self, op, inputs, ctx):
-        return [torch.nn.functional.silu(inputs[0])]
+        from ...ops import functional as F
+        return [F.silu_fwd(inputs[0])]
 
     def gradient(self, op, g):
         gr = _g(op.outputs[0])
@@ -83,9 +80,8 @@ class SiluGradOp(_ScalarOp):
     type = "SiluGrad"
 
     def compute(self, op, inputs, ctx):
-        g, x = inputs
-        s = torch.sigmoid(x.float())
-        return [(g.float() * s * (1 + x.float() * (1 - s))).to(g.dtype)]
+        from ...ops import functional as F
+        return [F.silu_bwd(inputs[0], inputs[1])]
 
 
 class TanhOp(_ScalarOp):

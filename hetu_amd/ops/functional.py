@@ -154,6 +154,40 @@ def swiglu_bwd(dy, x):
 
 
 # ---------------------------------------------------------------------------
+# Activations (fused fwd/bwd; reference Gelu.cu / Activation.cu)
+# ---------------------------------------------------------------------------
+
+def gelu_fwd(x):
+    if _gpu(x):
+        return ext().gelu_fwd(x.contiguous())
+    return torch.nn.functional.gelu(x, approximate="tanh")
+
+
+def gelu_bwd(dy, x):
+    if _gpu(x):
+        return ext().gelu_bwd(dy.contiguous(), x.contiguous())
+    xf = x.float()
+    c = 0.7978845608028654  # sqrt(2/pi)
+    a = 0.044715
+    t = torch.tanh(c * (xf + a * xf ** 3))
+    dt = (1 - t * t) * c * (1 + 3 * a * xf * xf)
+    return (dy.float() * (0.5 * (1 + t) + 0.5 * xf * dt)).to(x.dtype)
+
+
+def silu_fwd(x):
+    if _gpu(x):
+        return ext().silu_fwd(x.contiguous())
+    return torch.nn.functional.silu(x)
+
+
+def silu_bwd(dy, x):
+    if _gpu(x):
+        return ext().silu_bwd(dy.contiguous(), x.contiguous())
+    s = torch.sigmoid(x.float())
+    return (dy.float() * s * (1 + x.float() * (1 - s))).to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
 # RoPE (reference rotary.cu:97-185; NeoX-style half rotation)
 # ---------------------------------------------------------------------------
 

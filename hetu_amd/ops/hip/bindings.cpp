@@ -16,6 +16,11 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
 // swiglu.hip
 torch::Tensor swiglu_fwd(torch::Tensor x);
 torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor x);
+// activations.hip
+torch::Tensor gelu_fwd(torch::Tensor x);
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x);
+torch::Tensor silu_fwd(torch::Tensor x);
+torch::Tensor silu_bwd(torch::Tensor dy, torch::Tensor x);
 // rope.hip
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cos, torch::Tensor sin);
 torch::Tensor rope_bwd(torch::Tensor dy, torch::Tensor cos, torch::Tensor sin);
@@ -67,6 +72,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("silu_fwd", &silu_fwd);
+  m.def("silu_bwd", &silu_bwd);
   m.def("rope_fwd", &rope_fwd);
   m.def("rope_bwd", &rope_bwd);
   m.def("softmax_fwd", &softmax_fwd);
