@@ -37,7 +37,7 @@ DEV int swz(int row, int byte_in_row, int mask) {
 // global_load_lds, inverse-swizzling the per-lane source (rule 21).
 template <int ROWS, int ROWB, int MASK>
 DEV void stage_rm(const bf16* __restrict__ gsrc, int64_t row_stride_elts,
-                  char* lds, int tid) {
+                  char* lds, int tid, int max_row = 1 << 30) {
   constexpr int BYTES = ROWS * ROWB;
   constexpr int NINST = BYTES / 4096;
 #pragma unroll
@@ -46,7 +46,10 @@ DEV void stage_rm(const bf16* __restrict__ gsrc, int64_t row_stride_elts,
     int row = lin / ROWB;
     int cin = lin % ROWB;
     int sc = swz(row, cin, MASK);
-    const bf16* src = gsrc + (int64_t)row * row_stride_elts + sc / 2;
+    // clamp OOB rows to the last valid one: garbage bits can be NaN and
+    // 0 * NaN = NaN inside the MFMAs even for masked entries
+    const bf16* src = gsrc + (int64_t)min(row, max_row) * row_stride_elts
+                    + sc / 2;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)src,
         (__attribute__((address_space(3))) void*)(lds + lin), 16, 0, 0);
@@ -117,7 +120,7 @@ __global__ __launch_bounds__(THREADS) void fa_fwd_kernel(
   const bf16* Vb = V + ((int64_t)(b * Hkv + hkv)) * Skv * D;
 
   // stage Q tile once
-  stage_rm<BM, ROWB, MASK>(Qb, D, q_lds, tid);
+  stage_rm<BM, ROWB, MASK>(Qb, D, q_lds, tid, S - 1 - q0);
 
   f32x4 o_acc[D / 16];
 #pragma unroll
@@ -137,7 +140,8 @@ __global__ __launch_bounds__(THREADS) void fa_fwd_kernel(
     // ---- stage K tile; V transposed (scalar transpose) ----
     asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();            // prior P-reads/V-reads done
-    stage_rm<BN, ROWB, MASK>(Kb + (int64_t)k0 * D, D, k_lds, tid);
+    stage_rm<BN, ROWB, MASK>(Kb + (int64_t)k0 * D, D, k_lds, tid,
+                             Skv - 1 - k0);
     {
       // Vt[d][key]: each thread copies 8 contiguous d of one key row.
       constexpr int CHUNKS = BN * D / (THREADS * 8);
@@ -146,7 +150,7 @@ __global__ __launch_bounds__(THREADS) void fa_fwd_kernel(
         int idx = (c * THREADS + tid) * 8;
         int key = idx / D;
         int d0 = idx % D;
-        const bf16* src = Vb + ((int64_t)(k0 + key)) * D + d0;
+        const bf16* src = Vb + (int64_t)min(k0 + key, Skv - 1) * D + d0;
         ushort8 u = *reinterpret_cast<const ushort8*>(src);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -336,8 +340,8 @@ __global__ __launch_bounds__(THREADS) void fa_bwd_kernel(
   const float* del_h = DELTA + (int64_t)bh * S;
 
   // stage K, V, Kt once per workgroup
-  stage_rm<BN, ROWB, MASK>(Kb, D, k_lds, tid);
-  stage_rm<BN, ROWB, MASK>(Vb, D, v_lds, tid);
+  stage_rm<BN, ROWB, MASK>(Kb, D, k_lds, tid, Skv - 1 - k0);
+  stage_rm<BN, ROWB, MASK>(Vb, D, v_lds, tid, Skv - 1 - k0);
   {
     constexpr int CHUNKS = BN * D / (THREADS * 8);
 #pragma unroll
@@ -346,7 +350,7 @@ __global__ __launch_bounds__(THREADS) void fa_bwd_kernel(
       int key = idx / D;
       int d0 = idx % D;
       ushort8 u = *reinterpret_cast<const ushort8*>(
-          Kb + (int64_t)key * D + d0);
+          Kb + (int64_t)min(key, Skv - 1 - k0) * D + d0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int row = d0 + j;
@@ -371,8 +375,10 @@ __global__ __launch_bounds__(THREADS) void fa_bwd_kernel(
     const int q0 = qt * BM;
     asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    stage_rm<BM, ROWB, MASK>(Qh + (int64_t)q0 * D, D, q_lds, tid);
-    stage_rm<BM, ROWB, MASK>(dOh + (int64_t)q0 * D, D, do_lds, tid);
+    stage_rm<BM, ROWB, MASK>(Qh + (int64_t)q0 * D, D, q_lds, tid,
+                             S - 1 - q0);
+    stage_rm<BM, ROWB, MASK>(dOh + (int64_t)q0 * D, D, do_lds, tid,
+                             S - 1 - q0);
     {
       constexpr int CHUNKS = BM * D / (THREADS * 8);
 #pragma unroll
@@ -381,9 +387,9 @@ __global__ __launch_bounds__(THREADS) void fa_bwd_kernel(
         int qr = idx / D;
         int d0 = idx % D;
         ushort8 uq = *reinterpret_cast<const ushort8*>(
-            Qh + (int64_t)(q0 + qr) * D + d0);
+            Qh + (int64_t)min(q0 + qr, S - 1) * D + d0);
         ushort8 ud = *reinterpret_cast<const ushort8*>(
-            dOh + (int64_t)(q0 + qr) * D + d0);
+            dOh + (int64_t)min(q0 + qr, S - 1) * D + d0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int row = d0 + j;
@@ -445,7 +451,8 @@ __global__ __launch_bounds__(THREADS) void fa_bwd_kernel(
         bool masked = (qcol >= S) || (krow >= Skv) ||
                       (causal && krow > qcol + diag_off);
         float p = masked ? 0.f : __expf(sv - lse_q);
-        float ds = p * (dpt_acc[n][j] - del_q) * scale;
+        // guard: OOB-staged rows can hold NaN bit patterns; 0*NaN = NaN
+        float ds = masked ? 0.f : p * (dpt_acc[n][j] - del_q) * scale;
         st_acc[n][j] = p;      // now holds P^T
         dpt_acc[n][j] = ds;    // now holds dS^T
         int row = wrow + kg * 4 + j;
@@ -536,6 +543,11 @@ __global__ __launch_bounds__(THREADS) void fa_bwd_kernel(
 
 }  // namespace
 
+bool fa2_fwd_supported(int D, int S);
+void fa2_fwd_launch(const void* q, const void* k, const void* v, void* o,
+                    float* lse, int B, int H, int Hkv, int S, int Skv,
+                    float scale, bool causal, hipStream_t stream);
+
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
                                           double scale) {
@@ -548,6 +560,16 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, S}, q.options().dtype(at::kFloat));
   auto stream = hetu_current_stream();
+  static const bool force_v1 = [] {
+    const char* e = getenv("HETU_AMD_FA1");
+    return e && e[0] == '1';
+  }();
+  if (!force_v1 && fa2_fwd_supported(D, S)) {
+    fa2_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                   lse.data_ptr<float>(), B, H, Hkv, S, Skv, (float)scale,
+                   causal, stream);
+    return {o, lse};
+  }
   dim3 grid((S + BM - 1) / BM, B * H);
   size_t lds = (size_t)BM * D * 2 + BN * D * 2 + D * BN * 2 + BM * BN * 2;
   if (D == 128) {

@@ -25,6 +25,7 @@ SOURCES = [
     "adam.hip",
     "gemm.hip",
     "attention.hip",
+    "attention_v2.hip",
     "bindings.cpp",
 ]
 

@@ -194,6 +194,9 @@ class TestAttention:
         dict(B=2, H=4, Hkv=4, S=256, D=128, causal=False),
         dict(B=1, H=8, Hkv=2, S=512, D=128, causal=True),   # GQA
         dict(B=2, H=4, Hkv=4, S=192, D=64, causal=True),    # ragged S
+        dict(B=1, H=2, Hkv=2, S=300, D=128, causal=True),   # ragged S, v2
+        dict(B=1, H=2, Hkv=2, S=1024, D=128, causal=True),  # multi q-block
+        dict(B=1, H=2, Hkv=2, S=1024, D=128, causal=False),
     ])
     def test_fwd_bwd(self, cfg):
         B, H, Hkv, S, D = cfg["B"], cfg["H"], cfg["Hkv"], cfg["S"], cfg["D"]
