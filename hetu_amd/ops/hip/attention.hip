@@ -588,6 +588,11 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   return {o, lse};
 }
 
+bool fa2_bwd_supported(int D, int S);
+std::vector<torch::Tensor> fa2_bwd_launch(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor out, torch::Tensor lse, bool causal, double scale);
+
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout,
                                           torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, torch::Tensor out,
@@ -595,6 +600,13 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout,
                                           double scale) {
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   const int Hkv = k.size(1), Skv = k.size(2);
+  static const bool bwd_force_v1 = [] {
+    const char* e = getenv("HETU_AMD_FA1");
+    return e && e[0] == '1';
+  }();
+  if (!bwd_force_v1 && fa2_bwd_supported(D, S)) {
+    return fa2_bwd_launch(dout, q, k, v, out, lse, causal, scale);
+  }
   auto stream = hetu_current_stream();
   auto delta = torch::empty({B, H, S}, q.options().dtype(at::kFloat));
   {

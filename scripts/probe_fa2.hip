@@ -135,6 +135,75 @@ __global__ __launch_bounds__(512) void probe_kernel(
     perm[lane] = (float)x;
     perm[64 + lane] = (float)y;
   }
+
+  // ---- probe 5: tr-read from the swizzled ROW-MAJOR K image --------------
+  // want frag own=d (l&31), k = row = 16s + 8hi + j (j=0..7, two reads)
+  float e5 = 0.f;
+  if (wid == 0) {
+    const int g = (lane >> 4) & 3;
+    for (int s = 0; s < 2; ++s)
+      for (int dt = 0; dt < 4; ++dt) {
+        int row1 = 16 * s + 8 * hi + ((lane >> 2) & 3);
+        int row2 = row1 + 4;
+        int dby = (32 * dt + 16 * (g & 1) + 4 * (lane & 3)) * 2;
+        unsigned a1 = (unsigned)(uintptr_t)(
+            (__attribute__((address_space(3))) char*)(
+                k_lds + row1 * 256 + (dby ^ ((row1 & 15) << 4))));
+        unsigned a2 = (unsigned)(uintptr_t)(
+            (__attribute__((address_space(3))) char*)(
+                k_lds + row2 * 256 + (dby ^ ((row2 & 15) << 4))));
+        u32x2 r1, r2;
+        asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                     "ds_read_b64_tr_b16 %1, %3\n\t"
+                     "s_waitcnt lgkmcnt(0)"
+                     : "=&v"(r1), "=&v"(r2) : "v"(a1), "v"(a2));
+        __builtin_amdgcn_sched_barrier(0);
+        union { u32x2 u[2]; __attribute__((ext_vector_type(8))) __bf16 v; } f;
+        f.u[0] = r1; f.u[1] = r2;
+        for (int j = 0; j < 8; ++j) {
+          float got = __bfloat162float((bf16)f.v[j]);
+          float want = __bfloat162float(
+              KV[(16 * s + 8 * hi + j) * 128 + 32 * dt + iq]);
+          e5 += fabsf(got - want);
+        }
+      }
+  }
+  if (wid == 0) smat[2048 + lane] = e5;   // stash after the 32x32 S area
+
+  // ---- probe 6: tr-read from a 72B-row P' image (rows=q, cols=key) -------
+  // image P'[32 q][32 key], row stride 72 B; frag own=key, k=q
+  float e6 = 0.f;
+  if (wid == 0) {
+    char* p_lds = v_lds;   // reuse V region as scratch
+    // fill: P'[q][key] = Q[q*32+key... use KV values: P'[q][key]=KV[q*128+key]
+    for (int idx = lane; idx < 32 * 32; idx += 64) {
+      int q = idx >> 5, key = idx & 31;
+      *(bf16*)(p_lds + q * 72 + key * 2) = KV[q * 128 + key];
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const int g = (lane >> 4) & 3;
+    for (int s = 0; s < 2; ++s) {
+      int row1 = 16 * s + 8 * hi + ((lane >> 2) & 3);
+      int kby = (16 * (g & 1) + 4 * (lane & 3)) * 2;
+      unsigned a1 = (unsigned)(uintptr_t)(
+          (__attribute__((address_space(3))) char*)(p_lds + row1 * 72 + kby));
+      u32x2 r1, r2;
+      asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                   "ds_read_b64_tr_b16 %1, %2 offset:288\n\t"
+                   "s_waitcnt lgkmcnt(0)"
+                   : "=&v"(r1), "=&v"(r2) : "v"(a1));
+      __builtin_amdgcn_sched_barrier(0);
+      union { u32x2 u[2]; __attribute__((ext_vector_type(8))) __bf16 v; } f;
+      f.u[0] = r1; f.u[1] = r2;
+      for (int j = 0; j < 8; ++j) {
+        float got = __bfloat162float((bf16)f.v[j]);
+        float want = __bfloat162float(
+            KV[(16 * s + 8 * hi + j) * 128 + iq]);   // P'[q][key=iq]
+        e6 += fabsf(got - want);
+      }
+    }
+    smat[2048 + 64 + lane] = e6;
+  }
 }
 
 int main() {
@@ -158,14 +227,14 @@ int main() {
   float *dE1, *dE2, *dS, *dP;
   hipMalloc(&dKV, 64 * 128 * 2); hipMalloc(&dQ, 32 * 128 * 2);
   hipMalloc(&dE1, 64 * 4); hipMalloc(&dE2, 64 * 4);
-  hipMalloc(&dS, 64 * 32 * 4); hipMalloc(&dP, 128 * 4);
+  hipMalloc(&dS, (2048 + 256) * 4); hipMalloc(&dP, 128 * 4);
   hipMemcpy(dKV, hKV, 64 * 128 * 2, hipMemcpyHostToDevice);
   hipMemcpy(dQ, hQ, 32 * 128 * 2, hipMemcpyHostToDevice);
   hipLaunchKernelGGL(probe_kernel, dim3(1), dim3(512), 32768, 0,
                      dKV, dQ, dE1, dE2, dS, dP);
   hipError_t e = hipDeviceSynchronize();
   printf("kernel: %s\n", hipGetErrorString(e));
-  float hE1[64], hE2[64], hS[64 * 32], hP[128];
+  float hE1[64], hE2[64], hS[2048 + 256], hP[128];
   hipMemcpy(hE1, dE1, sizeof hE1, hipMemcpyDeviceToHost);
   hipMemcpy(hE2, dE2, sizeof hE2, hipMemcpyDeviceToHost);
   hipMemcpy(hS, dS, sizeof hS, hipMemcpyDeviceToHost);
@@ -184,8 +253,11 @@ int main() {
       }
     }
   printf("probe3 (mfma S^T):     %s (err %.1f)\n", serr < 1 ? "PASS" : "FAIL", serr);
-  printf("probe4 perm: x[0]=%.0f x[32]=%.0f y[0]=%.0f y[32]=%.0f "
-         "(expect 1000, 2000, 2032? / see T21 semantics)\n",
+  printf("probe4 perm: x[0]=%.0f x[32]=%.0f y[0]=%.0f y[32]=%.0f\n",
          hP[0], hP[32], hP[64], hP[96]);
+  float s5 = 0, s6 = 0;
+  for (int i = 0; i < 64; ++i) { s5 += hS[2048 + i]; s6 += hS[2048 + 64 + i]; }
+  printf("probe5 (Krm tr-read):  %s (err %.1f)\n", s5 == 0 ? "PASS" : "FAIL", s5);
+  printf("probe6 (P' tr-read):   %s (err %.1f)\n", s6 == 0 ? "PASS" : "FAIL", s6);
   return 0;
 }
