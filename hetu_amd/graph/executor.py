@@ -79,10 +79,18 @@ class Executor:
         return plan
 
     def run(self, fetches: Sequence[Tensor], feed_dict: Dict,
-            ctx: Optional[ExecContext] = None) -> List[torch.Tensor]:
+            ctx: Optional[ExecContext] = None,
+            seed_values: Optional[Dict[int, torch.Tensor]] = None,
+            keep_values: Optional[Dict[int, torch.Tensor]] = None
+            ) -> List[torch.Tensor]:
+        """seed_values: {tensor_id: value} of already-computed tensors (their
+        producing ops are skipped) — the pipeline engine seeds the backward
+        pass with the forward pass's cached activations.  keep_values: a
+        dict the caller provides to capture every computed value (disables
+        the degree-based free)."""
         ctx = ctx or self.ctx or ExecContext()
         plan = self._get_plan(fetches)
-        values: Dict[int, torch.Tensor] = {}
+        values: Dict[int, torch.Tensor] = dict(seed_values or {})
 
         # feed_dict keys may be Tensors or names
         feeds: Dict[int, torch.Tensor] = {}
@@ -95,6 +103,8 @@ class Executor:
             feeds[t.id] = v
 
         for i, op in enumerate(plan.topo):
+            if op.outputs and all(t.id in values for t in op.outputs):
+                continue                     # seeded (already computed)
             ins = []
             for t in op.inputs:
                 if t.id in values:
@@ -115,6 +125,10 @@ class Executor:
             outs = op.interface.compute(op, ins, ctx)
             for t, v in zip(op.outputs, outs):
                 values[t.id] = v
+            if keep_values is not None:
+                for t, v in zip(op.outputs, outs):
+                    keep_values[t.id] = v
+                continue                     # caller owns lifetimes
             # free dead intermediates (degree-based free, as in ComputeFunc)
             for t in op.inputs:
                 if plan.last_use.get(t.id) == i and t.id in values:

@@ -195,6 +195,116 @@ class LlamaLMHeadModel(Module):
         return loss, logits
 
 
+def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
+                               seq_len: int, dtype=torch.bfloat16,
+                               lr: float = 1e-4,
+                               stage_layers=None):
+    """Build THIS rank's pipeline-stage subgraph (see parallel.pipeline).
+
+    Returns a StageModule whose graph exposes fwd (act_out|loss), bwd
+    (dx + param grads via grad_in) and update (train_op fed by grad
+    placeholders; dp-allreduce of accumulated grads happens here, once per
+    step) fetch sets."""
+    from ..parallel.pipeline import StageModule
+    B, S = micro_batch, seq_len
+    sid = pspec.my_stage()
+    spec = pspec.stage_spec(sid)
+    parts = stage_layers or pspec.partition_layers(cfg.n_layer)
+    my_layers = parts[sid]
+    is_first, is_last = sid == 0, sid == pspec.pp - 1
+
+    g = DefineAndRunGraph(f"llama_stage{sid}")
+    push_graph(g)
+    try:
+        h: Dict = {"act_shape": (B, S, cfg.hidden), "act_dtype": dtype}
+        ds_in = spec.ds_activation(0)
+        cos_d, sin_d = rope_tables(cfg, seq_len, torch.float32)
+        cos = ht.variable(cos_d, name="rope.cos", requires_grad=False,
+                          ds=spec.ds_weight_dup(),
+                          device_group=spec.device_group)
+        sin = ht.variable(sin_d, name="rope.sin", requires_grad=False,
+                          ds=spec.ds_weight_dup(),
+                          device_group=spec.device_group)
+        if is_first:
+            input_ids = ht.placeholder((B, S), dtype=torch.int64,
+                                       name="input_ids", ds=ds_in,
+                                       device_group=spec.device_group)
+            wte = VocabParallelEmbedding(cfg.vocab, cfg.hidden, spec,
+                                         dtype=dtype, name="wte",
+                                         init_std=cfg.init_std)
+            x = wte(input_ids)
+            h["input_ids"] = input_ids
+        else:
+            act_in = ht.placeholder((B, S, cfg.hidden), dtype=dtype,
+                                    name="act_in", ds=ds_in,
+                                    device_group=spec.device_group)
+            x = act_in
+            h["act_in"] = act_in
+        blocks = [LlamaBlock(cfg, spec, cos, sin, li, dtype)
+                  for li in my_layers]
+        for blk in blocks:
+            x = blk(x, B, S)
+        if is_last:
+            labels = ht.placeholder((B * S,), dtype=torch.int64,
+                                    name="labels", ds=ds_in,
+                                    device_group=spec.device_group)
+            lnf = ParallelRMSNorm(cfg.hidden, spec, cfg.rms_eps, dtype,
+                                  name="lnf")
+            lm_head = ColumnParallelLinear(
+                cfg.hidden, cfg.vocab, spec, bias=False, dtype=dtype,
+                name="lm_head", init_std=cfg.init_std)
+            xo = lnf(x)
+            logits = lm_head(
+                ht.reshape(xo, (B * S, cfg.hidden),
+                           ds=spec._ds({0: spec.dp, -1: spec.tp}, [0, -1])))
+            per_tok = vocab_parallel_cross_entropy(logits, labels, cfg.vocab)
+            loss = ht.reduce_mean(per_tok)
+            h["labels"] = labels
+            h["loss"] = loss
+        else:
+            h["act_out"] = x
+
+        params = list(g.parameters)
+        h["params"] = params
+        # ---- backward fetch set ----
+        xs = params + ([] if is_first else [h["act_in"]])
+        if is_last:
+            grads = g.gradients([loss], xs)
+        else:
+            grad_in = ht.placeholder((B, S, cfg.hidden), dtype=dtype,
+                                     name="grad_in", ds=ds_in,
+                                     device_group=spec.device_group)
+            h["grad_in"] = grad_in
+            grads = g.gradients([x], xs, grad_ys=[grad_in])
+        h["param_grads"] = grads[:len(params)]
+        if not is_first:
+            h["dx"] = grads[len(params)]
+        # ---- update graph: grad placeholders -> (dp allreduce) -> Adam ---
+        from ..graph.ops.optim import AdamStepOp, GroupOp
+        from ..graph.ops.basics import _make
+        from ..graph.ops.comm import make_comm
+        grad_phs, updates = [], []
+        opt_attrs = {"lr": lr, "beta1": 0.9, "beta2": 0.999, "eps": 1e-8,
+                     "weight_decay": 0.0}
+        for p, pg in zip(params, h["param_grads"]):
+            gds = pg.ds if pg is not None else None
+            ph = ht.placeholder(tuple(p.shape), dtype=torch.float32,
+                                name=f"gbuf_{p.name}", ds=gds,
+                                device_group=spec.device_group)
+            grad_phs.append(ph)
+            gt = ph
+            if gds is not None and p.ds is not None \
+                    and not gds.check_equal(p.ds):
+                gt = make_comm(g, ph, p.ds, name=f"gred_{p.name}")
+            updates.append(_make(g, AdamStepOp(), [p, gt], dict(opt_attrs),
+                                 name=f"adam_{p.name}").output())
+        h["grad_phs"] = grad_phs
+        h["train_op"] = _make(g, GroupOp(), updates, name="train_op").output()
+    finally:
+        pop_graph()
+    return StageModule(g, h)
+
+
 def build_llama_train_graph(cfg: LlamaConfig, micro_batch: int, seq_len: int,
                             dtype=torch.bfloat16, lr: float = 1e-4,
                             spec: Optional[ParallelSpec] = None,
