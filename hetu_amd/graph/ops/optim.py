@@ -113,6 +113,73 @@ class AdamStepOp(OptimizerUpdateOp):
         return [torch.zeros((), device=param.device)]
 
 
+class ZeroAdamStepOp(OptimizerUpdateOp):
+    """ZeRO-1/2 Adam: optimizer states sharded over the data-parallel group
+    (reference: `zero` flag in ds configs + SplitReduceScatter /
+    SplitAllGather bridge ops, hetu/graph/ops/Communication.h:660-786,
+    subgraph.h:19-24 OPTIMIZE_COMPUTE_BRIDGE).
+
+    compute(): grad -> reduce-scatter over the dp group -> fused Adam on the
+    LOCAL shard (fp32 master/m/v only for 1/dp of the param) -> all-gather
+    the updated bf16 shard back into the param storage.  Collectives ride
+    RCCL; at world_size 1 it degrades to plain Adam."""
+    type = "ZeroAdamStep"
+
+    def __init__(self):
+        self.state: Dict = {}
+
+    def _dp_ranks(self, op, ctx):
+        p = op.inputs[0]
+        if ctx.comm is None or p.ds is None or p.ds.dup <= 1:
+            return [ctx.comm.rank if ctx.comm else 0]
+        from .comm import _my_index, _ranks
+        my = _my_index(ctx, p.device_group)
+        return _ranks(p.device_group, p.ds.group_devices_along(-1), my)
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        param, grad = inputs
+        a = op.attrs
+        st = self.state
+        ranks = self._dp_ranks(op, ctx)
+        n = len(ranks)
+        numel = param.numel()
+        pad = (-numel) % n
+        shard_elems = (numel + pad) // n
+        my_idx = sorted(ranks).index(ctx.comm.rank) if ctx.comm and n > 1 \
+            else 0
+        if "m" not in st:
+            flat = param.detach().float().reshape(-1)
+            if pad:
+                flat = torch.cat([flat, flat.new_zeros(pad)])
+            st["master"] = flat[my_idx * shard_elems:(my_idx + 1)
+                                * shard_elems].clone()
+            st["m"] = torch.zeros_like(st["master"])
+            st["v"] = torch.zeros_like(st["master"])
+            st["step"] = 0
+            st["pad"] = pad
+        st["step"] += 1
+        gflat = grad.reshape(-1)
+        if pad:
+            gflat = torch.cat([gflat, gflat.new_zeros(pad)])
+        if n > 1:
+            gshard = ctx.comm.reducescatter(gflat, ranks, dim=0,
+                                            my_index=my_idx)
+        else:
+            gshard = gflat
+        out16 = torch.empty(shard_elems, dtype=param.dtype,
+                            device=param.device)
+        F.adam_step(st["master"], gshard, st["m"], st["v"],
+                    a["lr"], a["beta1"], a["beta2"], a["eps"],
+                    a.get("weight_decay", 0.0), st["step"], out16, None)
+        if n > 1:
+            full = ctx.comm.allgather(out16, ranks, dim=0)
+        else:
+            full = out16
+        param.reshape(-1).copy_(full[:numel])
+        return [torch.zeros((), device=param.device)]
+
+
 class GroupOp(OpInterface):
     """Join node over update ops (reference ops/group.cc)."""
     type = "Group"
@@ -129,8 +196,9 @@ class GroupOp(OpInterface):
 
 
 class Optimizer:
-    def __init__(self, lr: float):
+    def __init__(self, lr: float, zero: bool = False):
         self.lr = lr
+        self.zero = zero     # ZeRO: shard optimizer states over dp
         self.update_ops: List = []
 
     def _make_update(self, graph, param: Tensor, grad: Tensor) -> Tensor:
@@ -145,8 +213,10 @@ class Optimizer:
         for p, g in zip(params, grads):
             if g is None:
                 continue
-            # parameter-grad reduction: partial (over dp) -> param layout
-            if (g.ds is not None and p.ds is not None
+            # parameter-grad reduction: partial (over dp) -> param layout.
+            # Under ZeRO the update op itself reduce-scatters the partial
+            # grad (COMPUTE_OPTIMIZE_BRIDGE semantics), so no comm here.
+            if (not self.zero and g.ds is not None and p.ds is not None
                     and not g.ds.check_equal(p.ds)):
                 g = make_comm(graph, g, p.ds, name=f"grad_allreduce_{p.name}")
             updates.append(self._make_update(graph, p, g))
@@ -168,14 +238,15 @@ class SGD(Optimizer):
 class Adam(Optimizer):
     def __init__(self, lr: float = 1e-3, beta1: float = 0.9,
                  beta2: float = 0.999, eps: float = 1e-8,
-                 weight_decay: float = 0.0):
-        super().__init__(lr)
+                 weight_decay: float = 0.0, zero: bool = False):
+        super().__init__(lr, zero=zero)
         self.beta1, self.beta2 = beta1, beta2
         self.eps = eps
         self.weight_decay = weight_decay
 
     def _make_update(self, graph, param, grad):
-        return _make(graph, AdamStepOp(), [param, grad],
+        cls = ZeroAdamStepOp if self.zero else AdamStepOp
+        return _make(graph, cls(), [param, grad],
                      {"lr": self.lr, "beta1": self.beta1,
                       "beta2": self.beta2, "eps": self.eps,
                       "weight_decay": self.weight_decay},

@@ -134,7 +134,8 @@ def build_gpt_forward(cfg: GPTConfig, input_ids, micro_batch: int,
 def build_gpt_train_graph(cfg: GPTConfig, micro_batch: int, seq_len: int,
                           dtype=torch.bfloat16, lr: float = 1e-4,
                           dp: int = 1, device_group=None,
-                          graph: Optional[DefineAndRunGraph] = None
+                          graph: Optional[DefineAndRunGraph] = None,
+                          zero: bool = False
                           ) -> (DefineAndRunGraph, Dict):
     g = graph or DefineAndRunGraph("gpt_train")
     if dp > 1 and device_group is None:
@@ -157,7 +158,7 @@ def build_gpt_train_graph(cfg: GPTConfig, micro_batch: int, seq_len: int,
             loss_report = ht.comm(
                 loss, DistributedStates(dp, {-1: dp}, order=[-1]),
                 name="loss_allreduce")
-        opt = Adam(lr=lr)
+        opt = Adam(lr=lr, zero=zero)
         train_op = opt.minimize(loss)
     finally:
         pop_graph()

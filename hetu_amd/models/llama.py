@@ -308,7 +308,8 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
 def build_llama_train_graph(cfg: LlamaConfig, micro_batch: int, seq_len: int,
                             dtype=torch.bfloat16, lr: float = 1e-4,
                             spec: Optional[ParallelSpec] = None,
-                            graph: Optional[DefineAndRunGraph] = None
+                            graph: Optional[DefineAndRunGraph] = None,
+                            zero: bool = False
                             ) -> (DefineAndRunGraph, Dict):
     g = graph or DefineAndRunGraph("llama_train")
     spec = spec or ParallelSpec()
@@ -328,7 +329,7 @@ def build_llama_train_graph(cfg: LlamaConfig, micro_batch: int, seq_len: int,
             loss_report = ht.comm(
                 loss, spec._ds({-1: spec.num_devices}, [-1]),
                 name="loss_allreduce")
-        opt = Adam(lr=lr)
+        opt = Adam(lr=lr, zero=zero)
         train_op = opt.minimize(loss)
     finally:
         pop_graph()
