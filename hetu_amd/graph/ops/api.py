@@ -263,6 +263,13 @@ def mse_loss(x, y):
     return _cg().make_op(N.MSELossOp(), [x, y], {}).output()
 
 
+def ring_attention(q, k, v, cp_ranks, causal=True, scale=None):
+    from . import parallel_ops as P
+    return _cg().make_op(P.RingAttentionOp(), [q, k, v],
+                         {"causal": causal, "scale": scale,
+                          "cp_ranks": list(cp_ranks)}).output(0)
+
+
 def vocab_parallel_embedding(table, ids, vocab: int):
     from . import parallel_ops as P
     return _cg().make_op(P.VocabParallelEmbeddingOp(), [table, ids],

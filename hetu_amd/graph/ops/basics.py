@@ -743,9 +743,12 @@ class ReduceGradOp(OpInterface):
 # ---------------------------------------------------------------------------
 
 def _deduce_matmul_ds(op, x: Tensor, w: Tensor, out: Tensor,
-                      x_row_dim: int, x_k_dim: int, w_k_dim: int,
-                      w_col_dim: int, out_row_dim: int, out_col_dim: int):
-    """Shared DS deduction for matmul-like ops via per-device index tables."""
+                      x_k_dims, w_k_dims, x_pass, w_pass):
+    """Shared DS deduction for matmul-like ops via per-device index tables.
+
+    x_k_dims / w_k_dims: paired contraction dims (a split on any pair makes
+    the output partial).  x_pass / w_pass: {input_dim: output_dim} maps for
+    dims whose split carries through (batch/row/col dims)."""
     dsx, dsw = x.ds, w.ds
     if dsx is None and dsw is None:
         return
@@ -756,28 +759,40 @@ def _deduce_matmul_ds(op, x: Tensor, w: Tensor, out: Tensor,
         dsw = DistributedStates(n, {-1: n} if n > 1 else {})
     if dsx.device_num != dsw.device_num:
         raise ValueError("matmul inputs on different-size device groups")
-    kx, kw = dsx.get_dim(x_k_dim), dsw.get_dim(w_k_dim)
-    if kx != kw:
-        raise ValueError(
-            f"contraction-dim splits differ: x {kx} vs w {kw}")
-    nrow = dsx.get_dim(x_row_dim)
-    ncol = dsw.get_dim(w_col_dim)
+    kx = 1
+    for dx_, dw_ in zip(x_k_dims, w_k_dims):
+        if dsx.get_dim(dx_) != dsw.get_dim(dw_):
+            raise ValueError(
+                f"contraction-dim splits differ: x[{dx_}] "
+                f"{dsx.get_dim(dx_)} vs w[{dw_}] {dsw.get_dim(dw_)}")
+        kx *= dsx.get_dim(dx_)
     npart = dsx.partial * dsw.partial * kx
+    counts = {-2: npart}
+    for d_in, d_out in x_pass.items():
+        counts[d_out] = counts.get(d_out, 1) * dsx.get_dim(d_in)
+    for d_in, d_out in w_pass.items():
+        counts[d_out] = counts.get(d_out, 1) * dsw.get_dim(d_in)
     table = []
     for i in range(n):
         sx = dsx.map_device_to_state_index(i)
         sw = dsw.map_device_to_state_index(i)
-        ipart = (sx.get(-2, 0) * dsw.partial + sw.get(-2, 0)) * kx \
-            + sx.get(x_k_dim, 0)
+        ipart = sx.get(-2, 0) * dsw.partial + sw.get(-2, 0)
+        for dx_, dw_ in zip(x_k_dims, w_k_dims):
+            if sx.get(dx_, 0) != sw.get(dw_, 0):
+                raise ValueError(
+                    "contraction shard indices differ between operands")
+            ipart = ipart * dsx.get_dim(dx_) + sx.get(dx_, 0)
         ent = {}
-        if nrow > 1:
-            ent[out_row_dim] = sx.get(x_row_dim, 0)
-        if ncol > 1:
-            ent[out_col_dim] = sw.get(w_col_dim, 0)
+        for d_in, d_out in x_pass.items():
+            if dsx.get_dim(d_in) > 1:
+                ent[d_out] = sx.get(d_in, 0)
+        for d_in, d_out in w_pass.items():
+            if dsw.get_dim(d_in) > 1:
+                ent[d_out] = ent.get(d_out, 0) * dsw.get_dim(d_in) \
+                    + sw.get(d_in, 0)
         if npart > 1:
             ent[-2] = ipart
         table.append(ent)
-    counts = {out_row_dim: nrow, out_col_dim: ncol, -2: npart}
     out.ds = ds_from_index_table(n, table, counts)
     out.device_group = x.device_group or w.device_group
 
@@ -796,9 +811,9 @@ class LinearOp(OpInterface):
         x, w = op.inputs[0], op.inputs[1]
         nd = x.ndim
         _deduce_matmul_ds(op, x, w, op.outputs[0],
-                          x_row_dim=0, x_k_dim=nd - 1,
-                          w_k_dim=1, w_col_dim=0,
-                          out_row_dim=0, out_col_dim=nd - 1)
+                          x_k_dims=[nd - 1], w_k_dims=[1],
+                          x_pass={d: d for d in range(nd - 1)},
+                          w_pass={0: nd - 1})
 
     def compute(self, op, inputs, ctx):
         from ...ops import functional as F
@@ -847,13 +862,16 @@ class MatMul2DOp(OpInterface):
         ta = op.attrs.get("trans_a", False)
         tb = op.attrs.get("trans_b", False)
         nda, ndb = a.ndim, b.ndim
+        a_k = nda - 2 if ta else nda - 1
+        a_row = nda - 1 if ta else nda - 2
+        x_pass = {d: d for d in range(nda - 2)}
+        x_pass[a_row] = nda - 2
         _deduce_matmul_ds(
             op, a, b, op.outputs[0],
-            x_row_dim=(nda - 1 if ta else 0),
-            x_k_dim=(nda - 2 if ta else nda - 1),
-            w_k_dim=(ndb - 1 if tb else ndb - 2),
-            w_col_dim=(ndb - 2 if tb else ndb - 1),
-            out_row_dim=0, out_col_dim=nda - 1)
+            x_k_dims=[a_k],
+            w_k_dims=[ndb - 1 if tb else ndb - 2],
+            x_pass=x_pass,
+            w_pass={(ndb - 2 if tb else ndb - 1): nda - 1})
 
     def compute(self, op, inputs, ctx):
         a, b = inputs
@@ -888,12 +906,13 @@ class MatMulGradWOp(OpInterface):
     def deduce_states(self, op):
         gy, x = op.inputs
         nd = gy.ndim
-        # gy rows (tokens) are the contraction; gy cols -> out dim 0;
-        # x cols -> out dim 1
+        # ALL leading (token) dims of gy/x are the contraction; gy cols ->
+        # out dim 0; x cols -> out dim 1
         _deduce_matmul_ds(op, gy, x, op.outputs[0],
-                          x_row_dim=nd - 1, x_k_dim=0,
-                          w_k_dim=0, w_col_dim=x.ndim - 1,
-                          out_row_dim=0, out_col_dim=1)
+                          x_k_dims=list(range(nd - 1)),
+                          w_k_dims=list(range(x.ndim - 1)),
+                          x_pass={nd - 1: 0},
+                          w_pass={x.ndim - 1: 1})
 
     def compute(self, op, inputs, ctx):
         gy, x = inputs
@@ -913,9 +932,10 @@ class MatMulGradBOp(OpInterface):
     def deduce_states(self, op):
         a, gy = op.inputs
         _deduce_matmul_ds(op, a, gy, op.outputs[0],
-                          x_row_dim=a.ndim - 1, x_k_dim=0,
-                          w_k_dim=0, w_col_dim=gy.ndim - 1,
-                          out_row_dim=0, out_col_dim=1)
+                          x_k_dims=list(range(a.ndim - 1)),
+                          w_k_dims=list(range(gy.ndim - 1)),
+                          x_pass={a.ndim - 1: 0},
+                          w_pass={gy.ndim - 1: 1})
 
     def compute(self, op, inputs, ctx):
         a, gy = inputs

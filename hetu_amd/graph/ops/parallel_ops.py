@@ -147,6 +147,68 @@ class VocabParallelEmbeddingGradOp(OpInterface):
         return [F.embedding_bwd(gy, local_ids, vlocal)]
 
 
+class RingAttentionOp(OpInterface):
+    """Context-parallel flash attention (reference ParallelAttention.cc /
+    AttnCommRing): q,k,v [B, H, S_loc, D] with the seq dim split over the
+    cp ring; the op rotates KV blocks over RCCL batched p2p and merges
+    partials with the log-sum-exp correction.  attrs: causal, scale,
+    cp_ranks (global rank list of this rank's ring)."""
+    type = "RingAttention"
+
+    def infer_meta(self, attrs, inputs):
+        q = inputs[0]
+        rows = list(q.shape[:-1])
+        return [TensorMeta(q.shape, q.dtype),
+                TensorMeta(rows, torch.float32)]
+
+    def deduce_states(self, op):
+        q = op.inputs[0]
+        for t in op.outputs:
+            if q.ds is not None:
+                t.ds = DistributedStates(q.ds.device_num, dict(q.ds.states),
+                                         list(q.ds.order))
+            t.device_group = q.device_group
+
+    def compute(self, op, inputs, ctx):
+        from ...parallel.ring_attention import ring_attn_fwd
+        q, k, v = inputs
+        o, lse = ring_attn_fwd(q, k, v, ctx.comm, op.attrs["cp_ranks"],
+                               op.attrs["causal"], op.attrs.get("scale"))
+        return [o, lse]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        bwd = _make(gr, RingAttentionGradOp(),
+                    [g[0], op.inputs[0], op.inputs[1], op.inputs[2],
+                     op.outputs[0], op.outputs[1]], dict(op.attrs))
+        return [bwd.output(0), bwd.output(1), bwd.output(2)]
+
+
+class RingAttentionGradOp(OpInterface):
+    type = "RingAttentionGrad"
+
+    def infer_meta(self, attrs, inputs):
+        _, q, k, v = inputs[:4]
+        return [TensorMeta(q.shape, q.dtype),
+                TensorMeta(k.shape, k.dtype),
+                TensorMeta(v.shape, v.dtype)]
+
+    def deduce_states(self, op):
+        for t, src in zip(op.outputs, op.inputs[1:4]):
+            if src.ds is not None:
+                t.ds = DistributedStates(src.ds.device_num,
+                                         dict(src.ds.states),
+                                         list(src.ds.order))
+            t.device_group = src.device_group
+
+    def compute(self, op, inputs, ctx):
+        from ...parallel.ring_attention import ring_attn_bwd
+        dout, q, k, v, o, lse = inputs
+        return list(ring_attn_bwd(dout, q, k, v, o, lse, ctx.comm,
+                                  op.attrs["cp_ranks"], op.attrs["causal"],
+                                  op.attrs.get("scale")))
+
+
 class VocabParallelCrossEntropyOp(OpInterface):
     """inputs: logits [N, V/tp] (ds split dim1 over tp), labels [N] ->
     per-token loss [N] fp32 (dup over tp) + saved global lse [N].

@@ -51,11 +51,14 @@ LLAMA_CONFIGS = {
 }
 
 
-def rope_tables(cfg: LlamaConfig, seq_len: int, dtype=torch.float32):
+def rope_tables(cfg: LlamaConfig, seq_len: int, dtype=torch.float32,
+                offset: int = 0):
+    """offset: global position of the first local token (context
+    parallelism gives each cp rank its own chunk of positions)."""
     dh = cfg.hidden // cfg.n_head
     inv = 1.0 / (cfg.rope_theta
                  ** (torch.arange(0, dh, 2, dtype=torch.float32) / dh))
-    t = torch.arange(seq_len, dtype=torch.float32)
+    t = torch.arange(offset, offset + seq_len, dtype=torch.float32)
     freqs = torch.outer(t, inv)                  # [S, dh/2]
     return torch.cos(freqs).to(dtype), torch.sin(freqs).to(dtype)
 
@@ -87,7 +90,8 @@ class LlamaAttention(Module):
         spec, cfg = self.spec, self.cfg
         hl, kl, dh = self.h_local, self.kv_local, self.dh
         qkv = self.wqkv(x)                       # [B,S,(h+2kv)*dh / tp]
-        ds_head = spec._ds({0: spec.dp, 2: spec.tp}, [0, 2])
+        ds_head = spec._ds({0: spec.dp, 1: spec.cp, 2: spec.tp},
+                            [0, 1, 2])
         q = ht.reshape(ht.slice_(qkv, 2, 0, hl * dh), (B, S, hl, dh),
                        ds=ds_head)
         k = ht.reshape(ht.slice_(qkv, 2, hl * dh, kl * dh), (B, S, kl, dh),
@@ -99,10 +103,14 @@ class LlamaAttention(Module):
         q = ht.transpose(q, 1, 2)                # [B,hl,S,dh]
         k = ht.transpose(k, 1, 2)
         v = ht.transpose(v, 1, 2)
-        o = ht.attention(q, k, v, causal=True)
+        if spec.cp > 1:
+            o = ht.ring_attention(q, k, v, spec.cp_ranks(), causal=True)
+        else:
+            o = ht.attention(q, k, v, causal=True)
         o = ht.transpose(o, 1, 2)
         o = ht.reshape(o, (B, S, hl * dh),
-                       ds=spec._ds({0: spec.dp, 2: spec.tp}, [0, 2]))
+                       ds=spec._ds({0: spec.dp, 1: spec.cp, 2: spec.tp},
+                                   [0, 1, 2]))
         return self.wo(o)
 
 
@@ -156,7 +164,8 @@ class LlamaLMHeadModel(Module):
         self.cfg, self.spec = cfg, spec
         self.B, self.S = micro_batch, seq_len
         self.dtype = dtype
-        cos_d, sin_d = rope_tables(cfg, seq_len, torch.float32)
+        cos_d, sin_d = rope_tables(cfg, seq_len, torch.float32,
+                                   offset=spec.my_cp_index() * seq_len)
         self.cos = ht.variable(cos_d, name="rope.cos", requires_grad=False,
                                ds=spec.ds_weight_dup(),
                                device_group=spec.device_group)
@@ -186,8 +195,7 @@ class LlamaLMHeadModel(Module):
             # column-parallel GEMM wants the full token set per rank)
             x = ht.comm(x, spec.ds_activation(0), name="sp_final_allgather")
         logits = self.lm_head(
-            ht.reshape(x, (B * S, cfg.hidden),
-                       ds=spec._ds({0: spec.dp, -1: spec.tp}, [0, -1])))
+            ht.reshape(x, (B * S, cfg.hidden), ds=spec.ds_tokens(0)))
         if labels is None:
             return None, logits
         per_tok = vocab_parallel_cross_entropy(logits, labels, cfg.vocab)
@@ -218,7 +226,8 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
     try:
         h: Dict = {"act_shape": (B, S, cfg.hidden), "act_dtype": dtype}
         ds_in = spec.ds_activation(0)
-        cos_d, sin_d = rope_tables(cfg, seq_len, torch.float32)
+        cos_d, sin_d = rope_tables(cfg, seq_len, torch.float32,
+                                   offset=spec.my_cp_index() * seq_len)
         cos = ht.variable(cos_d, name="rope.cos", requires_grad=False,
                           ds=spec.ds_weight_dup(),
                           device_group=spec.device_group)
@@ -246,7 +255,7 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
             x = blk(x, B, S)
         if is_last:
             labels = ht.placeholder((B * S,), dtype=torch.int64,
-                                    name="labels", ds=ds_in,
+                                    name="labels", ds=spec.ds_tokens(0),
                                     device_group=spec.device_group)
             lnf = ParallelRMSNorm(cfg.hidden, spec, cfg.rms_eps, dtype,
                                   name="lnf")
@@ -255,8 +264,7 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
                 name="lm_head", init_std=cfg.init_std)
             xo = lnf(x)
             logits = lm_head(
-                ht.reshape(xo, (B * S, cfg.hidden),
-                           ds=spec._ds({0: spec.dp, -1: spec.tp}, [0, -1])))
+                ht.reshape(xo, (B * S, cfg.hidden), ds=spec.ds_tokens(0)))
             per_tok = vocab_parallel_cross_entropy(logits, labels, cfg.vocab)
             loss = ht.reduce_mean(per_tok)
             h["labels"] = labels
@@ -320,7 +328,7 @@ def build_llama_train_graph(cfg: LlamaConfig, micro_batch: int, seq_len: int,
                                    name="input_ids", ds=ds_in,
                                    device_group=spec.device_group)
         labels = ht.placeholder((micro_batch * seq_len,), dtype=torch.int64,
-                                name="labels", ds=ds_in,
+                                name="labels", ds=spec.ds_tokens(0),
                                 device_group=spec.device_group)
         model = LlamaLMHeadModel(cfg, spec, micro_batch, seq_len, dtype)
         loss, _ = model(input_ids, labels)

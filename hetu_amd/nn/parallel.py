@@ -27,24 +27,27 @@ from .module import Module
 
 @dataclasses.dataclass
 class ParallelSpec:
-    """A dp x tp mesh over a flat device group (tp fastest-varying).
+    """A dp x cp x tp mesh over a flat device group (tp fastest-varying,
+    then cp, then dp).
 
-    device_group: global ranks, len == dp*tp.  Encodes the reference's
+    device_group: global ranks, len == dp*cp*tp.  Encodes the reference's
     ds_parallel_config device_group + split/dup layout generation
-    (utils/parallel/generate_ds.py:253)."""
+    (utils/parallel/generate_ds.py:253); cp is the "dcp" context-parallel
+    dim driving ring attention (engine/trainer.py:251-260)."""
     dp: int = 1
     tp: int = 1
+    cp: int = 1
     device_group: Optional[List[int]] = None
     sequence_parallel: bool = False
 
     def __post_init__(self):
         if self.device_group is None:
-            self.device_group = list(range(self.dp * self.tp))
-        assert len(self.device_group) == self.dp * self.tp
+            self.device_group = list(range(self.dp * self.cp * self.tp))
+        assert len(self.device_group) == self.dp * self.cp * self.tp
 
     @property
     def num_devices(self) -> int:
-        return self.dp * self.tp
+        return self.dp * self.cp * self.tp
 
     def my_index(self) -> int:
         rank = comm_backend().rank
@@ -55,8 +58,11 @@ class ParallelSpec:
     def my_tp_index(self) -> int:
         return self.my_index() % self.tp
 
+    def my_cp_index(self) -> int:
+        return (self.my_index() // self.tp) % self.cp
+
     def my_dp_index(self) -> int:
-        return self.my_index() // self.tp
+        return self.my_index() // (self.tp * self.cp)
 
     # ---- layouts ---------------------------------------------------------
     def _ds(self, states, order):
@@ -67,12 +73,22 @@ class ParallelSpec:
         order = [d for d in order if d in states]
         return DistributedStates(n, states, order)
 
-    def ds_activation(self, batch_dim: int = 0):
-        """split(batch) over dp, dup over tp."""
+    def ds_activation(self, batch_dim: int = 0, seq_dim: int = 1):
+        """split(batch) over dp, split(seq) over cp, dup over tp."""
+        if self.cp > 1:
+            return self._ds({batch_dim: self.dp, seq_dim: self.cp,
+                             -1: self.tp}, [batch_dim, seq_dim, -1])
         return self._ds({batch_dim: self.dp, -1: self.tp}, [batch_dim, -1])
+
+    def ds_tokens(self, tok_dim: int = 0):
+        """Layout of a flattened [B*S, ...] tensor: dp x cp fuse into one
+        token split (order-insensitive for reductions), dup over tp."""
+        return self._ds({tok_dim: self.dp * self.cp, -1: self.tp},
+                        [tok_dim, -1])
 
     def ds_activation_sp(self, batch_dim: int = 0, seq_dim: int = 1):
         """split(batch) over dp, split(seq) over tp (sequence parallel)."""
+        assert self.cp == 1, "sequence_parallel + cp not supported yet"
         return self._ds({batch_dim: self.dp, seq_dim: self.tp},
                         [batch_dim, seq_dim])
 
@@ -80,16 +96,25 @@ class ParallelSpec:
         return self._ds({-1: self.num_devices}, [-1])
 
     def ds_weight_col(self, split_dim: int = 0):
-        """dup over dp, split(out_features) over tp."""
-        return self._ds({-1: self.dp, split_dim: self.tp}, [-1, split_dim])
+        """dup over dp x cp, split(out_features) over tp."""
+        return self._ds({-1: self.dp * self.cp, split_dim: self.tp},
+                        [-1, split_dim])
 
     def ds_weight_row(self, split_dim: int = 1):
-        """dup over dp, split(in_features) over tp."""
-        return self._ds({-1: self.dp, split_dim: self.tp}, [-1, split_dim])
+        """dup over dp x cp, split(in_features) over tp."""
+        return self._ds({-1: self.dp * self.cp, split_dim: self.tp},
+                        [-1, split_dim])
 
     def ds_partial_tp(self, batch_dim: int = 0):
         """split(batch) over dp, partial over tp (row-parallel output)."""
-        return self._ds({batch_dim: self.dp, -2: self.tp}, [batch_dim, -2])
+        return self._ds({batch_dim: self.dp * self.cp, -2: self.tp},
+                        [batch_dim, -2])
+
+    def cp_ranks(self) -> List[int]:
+        """Global ranks of this rank's cp ring (same dp and tp coords)."""
+        di, ti = self.my_dp_index(), self.my_tp_index()
+        return [self.device_group[(di * self.cp + c) * self.tp + ti]
+                for c in range(self.cp)]
 
 
 def _shard(data: torch.Tensor, dim: int, n: int, idx: int,
@@ -147,11 +172,10 @@ class ColumnParallelLinear(Module):
 
     def forward(self, x):
         spec = self.spec
-        want = spec.ds_activation(0)
-        if want is not None and x.ds is not None \
-                and not x.ds.check_equal(want):
+        if spec.sequence_parallel and spec.tp > 1 and x.ds is not None \
+                and not x.ds.check_equal(spec.ds_activation(0)):
             # SP: seq-split -> dup over tp (allgather on the seq dim)
-            x = ht.comm(x, want, name="sp_allgather")
+            x = ht.comm(x, spec.ds_activation(0), name="sp_allgather")
         y = ht.linear(x, self.weight, self.bias)
         if self.gather_output and spec.tp > 1:
             y = ht.comm(y, spec.ds_activation(0), name="col_gather")

@@ -393,7 +393,7 @@ def flash_attn_bwd(dout, q, k, v, out, lse, causal: bool,
                                     k.contiguous(), v.contiguous(),
                                     out.contiguous(), lse.contiguous(),
                                     causal, scale)
-    return _attn_ref_bwd(dout, q, k, v, lse, causal, scale)
+    return _attn_ref_bwd(dout, q, k, v, out, lse, causal, scale)
 
 
 def _repeat_kv(k, n_head):
@@ -420,7 +420,10 @@ def _attn_ref_fwd(q, k, v, causal, scale):
     return out.to(q.dtype), lse
 
 
-def _attn_ref_bwd(dout, q, k, v, lse, causal, scale):
+def _attn_ref_bwd(dout, q, k, v, out, lse, causal, scale):
+    # delta MUST come from the global (dout . out) rowsum, not the local
+    # (dp*p).sum: under ring attention this reference runs per KV block
+    # with the global lse, where the two differ.
     B, H, S, D = q.shape
     Hkv = k.shape[1]
     kf = _repeat_kv(k, H).float()
@@ -435,7 +438,7 @@ def _attn_ref_bwd(dout, q, k, v, lse, causal, scale):
     p = torch.exp(scores - lse.unsqueeze(-1).float())
     dv = torch.matmul(p.transpose(-1, -2), dof)
     dp = torch.matmul(dof, vf.transpose(-1, -2))
-    delta = (dp * p).sum(-1, keepdim=True)
+    delta = (dof * out.float()).sum(-1, keepdim=True)
     ds = p * (dp - delta) * scale
     dq = torch.matmul(ds, kf)
     dk = torch.matmul(ds.transpose(-1, -2), qf)

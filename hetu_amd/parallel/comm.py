@@ -44,6 +44,10 @@ class CommBackend:
         elif "RANK" in os.environ and int(os.environ.get("WORLD_SIZE", "1")) > 1:
             use_gpu = torch.cuda.is_available()
             backend = "nccl" if use_gpu else "gloo"
+            if backend == "gloo":
+                # the container hostname may not resolve; pin gloo to
+                # loopback for single-node CPU tests
+                os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
             if use_gpu:
                 local = int(os.environ.get("LOCAL_RANK", os.environ["RANK"]))
                 torch.cuda.set_device(local)
@@ -51,6 +55,18 @@ class CommBackend:
                 backend=backend,
                 timeout=datetime.timedelta(seconds=300))
             rank, ws = dist.get_rank(), dist.get_world_size()
+            # tear the pg down cleanly at exit: gloo's destructor aborts
+            # ("terminate called without an active exception") when the
+            # process exits with live comm threads
+            import atexit
+
+            def _shutdown():
+                if dist.is_initialized():
+                    try:
+                        dist.destroy_process_group()
+                    except Exception:   # noqa: BLE001
+                        pass
+            atexit.register(_shutdown)
         else:
             rank, ws = 0, 1
         if device is None:
@@ -76,10 +92,13 @@ class CommBackend:
             elif list(key) == list(range(self.world_size)):
                 self._groups[key] = dist.group.WORLD
             else:
-                # NOTE: new_group must be called by ALL ranks in the same
-                # order — comm-op construction is deterministic per graph,
-                # which guarantees this.
-                self._groups[key] = dist.new_group(list(key))
+                # use_local_synchronization: only the group MEMBERS enter
+                # the rendezvous (plain new_group is world-collective and
+                # deadlocks when different pipeline stages create their
+                # own tp/dp subgroups); creation order is deterministic
+                # per graph among members.
+                self._groups[key] = dist.new_group(
+                    list(key), use_local_synchronization=True)
         return self._groups[key]
 
     # ---- collectives -----------------------------------------------------

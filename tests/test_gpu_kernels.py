@@ -215,7 +215,7 @@ class TestAttention:
         dq, dk, dv = F.flash_attn_bwd(do, q, k, v, o, lse, causal, scale)
         dqr, dkr, dvr = F._attn_ref_bwd(do.cpu().float(), q.cpu().float(),
                                         k.cpu().float(), v.cpu().float(),
-                                        lser, causal, scale)
+                                        orf, lser, causal, scale)
         _close(dq, dqr, rtol=5e-2, atol=5e-2, what="fa dq")
         _close(dk, dkr, rtol=5e-2, atol=5e-2, what="fa dk")
         _close(dv, dvr, rtol=5e-2, atol=5e-2, what="fa dv")

@@ -25,23 +25,25 @@ ws = int(os.environ.get("WORLD_SIZE", "1"))
 rank = int(os.environ.get("RANK", "0"))
 sp = os.environ.get("HETU_TEST_SP", "0") == "1"
 zero = os.environ.get("HETU_TEST_ZERO", "0") == "1"
+cp = int(os.environ.get("HETU_TEST_CP", "1"))
 dp = int(os.environ.get("HETU_TEST_DP", "1"))
-tp = ws // dp
+tp = ws // (dp * cp)
 cfg = LlamaConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
                   ffn_hidden=128, vocab=312, max_seq=16)
-spec = ParallelSpec(dp=dp, tp=tp, sequence_parallel=sp)
-B = 2 * (2 // dp)   # global batch 4 tokens-wise fixed: dp=1 -> B=4? keep 2/dp
+spec = ParallelSpec(dp=dp, tp=tp, cp=cp, sequence_parallel=sp)
 B = 4 // dp
-g, h = build_llama_train_graph(cfg, micro_batch=B, seq_len=16,
+SL = 16 // cp                       # local seq chunk per cp rank
+g, h = build_llama_train_graph(cfg, micro_batch=B, seq_len=SL,
                                dtype=torch.float32, lr=1e-3, spec=spec,
                                zero=zero)
 ctx = prepare_run_context(g, torch.device("cpu"))
 gen = torch.Generator().manual_seed(99)
 ids = torch.randint(0, cfg.vocab, (4, 16), generator=gen)
 labels = torch.randint(0, cfg.vocab, (4 * 16,), generator=gen)
-di = spec.my_dp_index()
-my_ids = ids[di * B:(di + 1) * B]
-my_labels = labels.reshape(4, 16)[di * B:(di + 1) * B].reshape(-1)
+di, ci = spec.my_dp_index(), spec.my_cp_index()
+my_ids = ids[di * B:(di + 1) * B, ci * SL:(ci + 1) * SL]
+my_labels = labels.reshape(4, 16)[di * B:(di + 1) * B,
+                                  ci * SL:(ci + 1) * SL].reshape(-1)
 losses = []
 for i in range(5):
     lv, _ = g.run([h["loss"], h["train_op"]],
@@ -52,7 +54,7 @@ if rank == 0:
 """
 
 
-def _launch(ws, extra_env, port):
+def _launch_once(ws, extra_env, port):
     base_env = dict(os.environ)
     base_env["HETU_REPO"] = REPO
     base_env["MASTER_ADDR"] = "127.0.0.1"
@@ -77,6 +79,15 @@ def _launch(ws, extra_env, port):
                 losses = json.loads(line[len("LOSSES:"):])
     assert losses is not None
     return losses
+
+
+def _launch(ws, extra_env, port):
+    """Retry once: gloo occasionally SIGABRTs in teardown after a clean
+    run (non-deterministic; results already printed)."""
+    try:
+        return _launch_once(ws, extra_env, port)
+    except AssertionError:
+        return _launch_once(ws, extra_env, port + 40)
 
 
 @pytest.fixture(scope="module")
@@ -110,3 +121,11 @@ def test_dp2_zero_matches_single(single_losses):
                        29535)
     assert np.allclose(z_losses, single_losses, rtol=2e-4, atol=1e-4), \
         f"dp2+zero {z_losses} vs single {single_losses}"
+
+
+def test_cp2_ring_attention_matches_single(single_losses):
+    """Context parallelism: cp=2 ring attention over gloo must match the
+    single-process run (seq chunks per rank)."""
+    cp_losses = _launch(2, {"HETU_TEST_DP": "1", "HETU_TEST_CP": "2"}, 29536)
+    assert np.allclose(cp_losses, single_losses, rtol=5e-4, atol=2e-4), \
+        f"cp2 {cp_losses} vs single {single_losses}"
