@@ -1,10 +1,12 @@
-"""GPT model family built on the define-and-run graph API.
+"""GPT model family on the define-and-run graph + TP/SP/CP parallel layers.
 
 Reference parity: python/hetu/models/gpt/gpt_model.py (GPTLMHeadModel) — a
 pre-LN transformer with learned position embeddings, fused-HIP LayerNorm,
-flash attention, GELU MLP, and sparse softmax CE loss. The 7B config matches
-the reference CI "gpt 7b" shape (32 layers x 4096 hidden x 32 heads,
-tests/ci_test/scripts/pssh_train_hetu.sh).
+flash attention, GELU MLP, and vocab-parallel CE loss.  The 7B config
+matches the reference CI "gpt 7b" shape (32 layers x 4096 hidden x 32
+heads, tests/ci_test/scripts/pssh_train_hetu.sh).  Parallelism (dp x cp x
+tp, zero) comes from a ParallelSpec like the Llama family; pipeline stages
+via build_gpt_pipeline_stage.
 """
 from __future__ import annotations
 
@@ -14,10 +16,15 @@ from typing import Dict, Optional
 
 import torch
 
-from .. import DistributedStates
 from ..graph.graph import DefineAndRunGraph, push_graph, pop_graph
 from ..graph.ops import api as ht
 from ..graph.ops.optim import Adam
+from ..nn import init
+from ..nn.module import Module, ModuleList
+from ..nn.parallel import (ColumnParallelLinear, ParallelLayerNorm,
+                           ParallelSpec, RowParallelLinear,
+                           VocabParallelEmbedding,
+                           vocab_parallel_cross_entropy)
 
 
 @dataclasses.dataclass
@@ -44,123 +51,268 @@ GPT_CONFIGS = {
 }
 
 
-def _randn(shape, std, dtype):
-    return (torch.randn(*shape) * std).to(dtype)
+class GPTAttention(Module):
+    def __init__(self, cfg: GPTConfig, spec: ParallelSpec, layer_idx: int,
+                 dtype):
+        super().__init__()
+        self.cfg, self.spec = cfg, spec
+        tp = spec.tp
+        assert cfg.n_head % tp == 0
+        self.dh = cfg.hidden // cfg.n_head
+        self.h_local = cfg.n_head // tp
+        p = f"h{layer_idx}.attn"
+        proj_std = cfg.init_std / math.sqrt(2 * cfg.n_layer)
+        self.wqkv = ColumnParallelLinear(
+            cfg.hidden, 3 * cfg.hidden, spec, bias=True, dtype=dtype,
+            name=f"{p}.wqkv", init_std=cfg.init_std,
+            sections=[cfg.hidden, cfg.hidden, cfg.hidden])
+        self.wo = RowParallelLinear(
+            cfg.hidden, cfg.hidden, spec, bias=True, dtype=dtype,
+            name=f"{p}.wo", init_std=proj_std)
 
-
-def build_gpt_forward(cfg: GPTConfig, input_ids, micro_batch: int,
-                      seq_len: int, dtype=torch.bfloat16, prefix: str = "gpt",
-                      dp: int = 1, device_group=None):
-    """Build forward graph ops in the CURRENT graph; returns logits tensor.
-    input_ids: graph tensor [B, S] int64 (local shard; ds split(0) for dp>1).
-    Data-parallel SPMD: weights duplicated, activations split on dim 0,
-    weight grads deduced partial -> allreduce at minimize()."""
-    B, S = micro_batch, seq_len
-    Hn, Dh = cfg.n_head, cfg.hidden // cfg.n_head
-    std = cfg.init_std
-    ds_dup = (DistributedStates(dp, {-1: dp}, order=[-1]) if dp > 1
-              else None)
-
-    def var(data, name, requires_grad=True):
-        return ht.variable(data, name=name, requires_grad=requires_grad,
-                           ds=ds_dup, device_group=device_group)
-
-    wte = var(_randn((cfg.vocab, cfg.hidden), std, dtype), f"{prefix}.wte")
-    wpe = var(_randn((cfg.max_seq, cfg.hidden), std, dtype), f"{prefix}.wpe")
-    pos_ids = var(torch.arange(S, dtype=torch.int64), f"{prefix}.pos",
-                  requires_grad=False)
-
-    x = ht.embedding(wte, input_ids)               # [B, S, h]
-    pos = ht.embedding(wpe, pos_ids)               # [S, h]
-    x = ht.add(x, pos)
-    if cfg.dropout > 0:
-        x = ht.dropout(x, cfg.dropout)
-
-    proj_std = std / math.sqrt(2 * cfg.n_layer)
-    for li in range(cfg.n_layer):
-        p = f"{prefix}.h{li}"
-        ln1_w = var(torch.ones(cfg.hidden, dtype=dtype), f"{p}.ln1.w")
-        ln1_b = var(torch.zeros(cfg.hidden, dtype=dtype), f"{p}.ln1.b")
-        wqkv = var(_randn((3 * cfg.hidden, cfg.hidden), std, dtype), f"{p}.wqkv")
-        bqkv = var(torch.zeros(3 * cfg.hidden, dtype=dtype), f"{p}.bqkv")
-        wproj = var(_randn((cfg.hidden, cfg.hidden), proj_std, dtype), f"{p}.wproj")
-        bproj = var(torch.zeros(cfg.hidden, dtype=dtype), f"{p}.bproj")
-        ln2_w = var(torch.ones(cfg.hidden, dtype=dtype), f"{p}.ln2.w")
-        ln2_b = var(torch.zeros(cfg.hidden, dtype=dtype), f"{p}.ln2.b")
-        wfc = var(_randn((cfg.ffn_hidden, cfg.hidden), std, dtype), f"{p}.wfc")
-        bfc = var(torch.zeros(cfg.ffn_hidden, dtype=dtype), f"{p}.bfc")
-        wfc2 = var(_randn((cfg.hidden, cfg.ffn_hidden), proj_std,
-                                  dtype), f"{p}.wfc2")
-        bfc2 = var(torch.zeros(cfg.hidden, dtype=dtype), f"{p}.bfc2")
-
-        # ---- attention block ----
-        h = ht.layer_norm(x, ln1_w, ln1_b)
-        qkv = ht.linear(h, wqkv, bqkv)                     # [B,S,3h]
-        qkv = ht.reshape(qkv, (B, S, 3, Hn, Dh))
-        q = ht.reshape(ht.slice_(qkv, 2, 0, 1), (B, S, Hn, Dh))
-        k = ht.reshape(ht.slice_(qkv, 2, 1, 1), (B, S, Hn, Dh))
-        v = ht.reshape(ht.slice_(qkv, 2, 2, 1), (B, S, Hn, Dh))
-        q = ht.transpose(q, 1, 2)                          # [B,H,S,D]
+    def forward(self, x, B, S):
+        spec = self.spec
+        hl, dh = self.h_local, self.dh
+        qkv = self.wqkv(x)
+        ds_head = spec._ds({0: spec.dp, 1: spec.cp, 2: spec.tp}, [0, 1, 2])
+        q = ht.reshape(ht.slice_(qkv, 2, 0, hl * dh), (B, S, hl, dh),
+                       ds=ds_head)
+        k = ht.reshape(ht.slice_(qkv, 2, hl * dh, hl * dh), (B, S, hl, dh),
+                       ds=ds_head)
+        v = ht.reshape(ht.slice_(qkv, 2, 2 * hl * dh, hl * dh),
+                       (B, S, hl, dh), ds=ds_head)
+        q = ht.transpose(q, 1, 2)
         k = ht.transpose(k, 1, 2)
         v = ht.transpose(v, 1, 2)
-        attn = ht.attention(q, k, v, causal=True)
-        attn = ht.transpose(attn, 1, 2)                    # [B,S,H,D]
-        attn = ht.reshape(attn, (B, S, cfg.hidden))
-        attn = ht.linear(attn, wproj, bproj)
-        if cfg.dropout > 0:
-            attn = ht.dropout(attn, cfg.dropout)
-        x = ht.add(x, attn)
+        if spec.cp > 1:
+            o = ht.ring_attention(q, k, v, spec.cp_ranks(), causal=True)
+        else:
+            o = ht.attention(q, k, v, causal=True)
+        o = ht.transpose(o, 1, 2)
+        o = ht.reshape(o, (B, S, hl * dh),
+                       ds=spec._ds({0: spec.dp, 1: spec.cp, 2: spec.tp},
+                                   [0, 1, 2]))
+        return self.wo(o)
 
-        # ---- MLP block ----
-        h2 = ht.layer_norm(x, ln2_w, ln2_b)
-        h2 = ht.gelu(ht.linear(h2, wfc, bfc))
-        h2 = ht.linear(h2, wfc2, bfc2)
-        if cfg.dropout > 0:
-            h2 = ht.dropout(h2, cfg.dropout)
-        x = ht.add(x, h2)
 
-    lnf_w = var(torch.ones(cfg.hidden, dtype=dtype), f"{prefix}.lnf.w")
-    lnf_b = var(torch.zeros(cfg.hidden, dtype=dtype), f"{prefix}.lnf.b")
-    x = ht.layer_norm(x, lnf_w, lnf_b)
-    if cfg.tie_embeddings:
-        logits = ht.matmul(ht.reshape(x, (B * S, cfg.hidden)), wte,
-                           trans_b=True)
-    else:
-        lm_head = var(_randn((cfg.vocab, cfg.hidden), std, dtype), f"{prefix}.lm_head")
-        logits = ht.linear(ht.reshape(x, (B * S, cfg.hidden)), lm_head)
-    return logits                                           # [B*S, V]
+class GPTMLP(Module):
+    def __init__(self, cfg: GPTConfig, spec: ParallelSpec, layer_idx: int,
+                 dtype):
+        super().__init__()
+        p = f"h{layer_idx}.mlp"
+        proj_std = cfg.init_std / math.sqrt(2 * cfg.n_layer)
+        self.wfc = ColumnParallelLinear(
+            cfg.hidden, cfg.ffn_hidden, spec, bias=True, dtype=dtype,
+            name=f"{p}.wfc", init_std=cfg.init_std)
+        self.wproj = RowParallelLinear(
+            cfg.ffn_hidden, cfg.hidden, spec, bias=True, dtype=dtype,
+            name=f"{p}.wproj", init_std=proj_std)
+
+    def forward(self, x):
+        return self.wproj(ht.gelu(self.wfc(x)))
+
+
+class GPTBlock(Module):
+    def __init__(self, cfg, spec, layer_idx, dtype):
+        super().__init__()
+        self.ln1 = ParallelLayerNorm(cfg.hidden, spec, 1e-5, dtype,
+                                     name=f"h{layer_idx}.ln1")
+        self.attn = GPTAttention(cfg, spec, layer_idx, dtype)
+        self.ln2 = ParallelLayerNorm(cfg.hidden, spec, 1e-5, dtype,
+                                     name=f"h{layer_idx}.ln2")
+        self.mlp = GPTMLP(cfg, spec, layer_idx, dtype)
+
+    def forward(self, x, B, S):
+        x = ht.add(x, self.attn(self.ln1(x), B, S))
+        x = ht.add(x, self.mlp(self.ln2(x)))
+        return x
+
+
+class GPTEmbedding(Module):
+    """Token + learned position embeddings (positions offset per cp rank)."""
+
+    def __init__(self, cfg: GPTConfig, spec: ParallelSpec, seq_len: int,
+                 dtype):
+        super().__init__()
+        self.spec = spec
+        self.vocab = cfg.vocab
+        self.wte = VocabParallelEmbedding(cfg.vocab, cfg.hidden, spec,
+                                          dtype=dtype, name="wte",
+                                          init_std=cfg.init_std)
+        wpe = init.normal((cfg.max_seq, cfg.hidden), std=cfg.init_std,
+                          dtype=dtype, name="wpe.weight")
+        off = spec.my_cp_index() * seq_len
+        # local slice of the position table: grads stay per-rank and reduce
+        # over the dup group like any duplicated parameter
+        self.wpe = ht.variable(wpe[off:off + seq_len].contiguous(),
+                               name="wpe.weight",
+                               ds=spec.ds_weight_dup(),
+                               device_group=spec.device_group)
+        pos = torch.arange(seq_len, dtype=torch.int64)
+        self.pos = ht.variable(pos, name="pos", requires_grad=False,
+                               ds=spec.ds_weight_dup(),
+                               device_group=spec.device_group)
+
+    def forward(self, input_ids):
+        x = self.wte(input_ids)
+        p = ht.embedding(self.wpe, self.pos)
+        return ht.add(x, p)
+
+
+class GPTLMHeadModel(Module):
+    def __init__(self, cfg: GPTConfig, spec: Optional[ParallelSpec] = None,
+                 micro_batch: int = 1, seq_len: int = 128,
+                 dtype=torch.bfloat16):
+        super().__init__()
+        spec = spec or ParallelSpec()
+        self.cfg, self.spec = cfg, spec
+        self.B, self.S = micro_batch, seq_len
+        self.embed = GPTEmbedding(cfg, spec, seq_len, dtype)
+        self.layers = ModuleList([GPTBlock(cfg, spec, i, dtype)
+                                  for i in range(cfg.n_layer)])
+        self.lnf = ParallelLayerNorm(cfg.hidden, spec, 1e-5, dtype,
+                                     name="lnf")
+        self.lm_head = ColumnParallelLinear(
+            cfg.hidden, cfg.vocab, spec, bias=False, dtype=dtype,
+            name="lm_head", init_std=cfg.init_std)
+
+    def forward(self, input_ids, labels=None):
+        B, S, cfg, spec = self.B, self.S, self.cfg, self.spec
+        x = self.embed(input_ids)
+        for blk in self.layers:
+            x = blk(x, B, S)
+        x = self.lnf(x)
+        logits = self.lm_head(
+            ht.reshape(x, (B * S, cfg.hidden), ds=spec.ds_tokens(0)))
+        if labels is None:
+            return None, logits
+        per_tok = vocab_parallel_cross_entropy(logits, labels, cfg.vocab)
+        loss = ht.reduce_mean(per_tok)
+        return loss, logits
 
 
 def build_gpt_train_graph(cfg: GPTConfig, micro_batch: int, seq_len: int,
                           dtype=torch.bfloat16, lr: float = 1e-4,
                           dp: int = 1, device_group=None,
                           graph: Optional[DefineAndRunGraph] = None,
-                          zero: bool = False
+                          zero: bool = False,
+                          spec: Optional[ParallelSpec] = None
                           ) -> (DefineAndRunGraph, Dict):
     g = graph or DefineAndRunGraph("gpt_train")
-    if dp > 1 and device_group is None:
-        device_group = list(range(dp))
-    ds_in = (DistributedStates(dp, {0: dp}, order=[0]) if dp > 1 else None)
+    if spec is None:
+        spec = ParallelSpec(dp=dp, device_group=device_group)
     push_graph(g)
     try:
+        ds_in = spec.ds_activation(0)
         input_ids = ht.placeholder((micro_batch, seq_len),
                                    dtype=torch.int64, name="input_ids",
-                                   ds=ds_in, device_group=device_group)
-        labels = ht.placeholder((micro_batch * seq_len,),
-                                dtype=torch.int64, name="labels",
-                                ds=ds_in, device_group=device_group)
-        logits = build_gpt_forward(cfg, input_ids, micro_batch, seq_len,
-                                   dtype, dp=dp, device_group=device_group)
-        per_tok = ht.softmax_cross_entropy_sparse(logits, labels)
-        loss = ht.reduce_mean(per_tok)
+                                   ds=ds_in, device_group=spec.device_group)
+        labels = ht.placeholder((micro_batch * seq_len,), dtype=torch.int64,
+                                name="labels", ds=spec.ds_tokens(0),
+                                device_group=spec.device_group)
+        model = GPTLMHeadModel(cfg, spec, micro_batch, seq_len, dtype)
+        loss, _ = model(input_ids, labels)
         loss_report = loss
-        if dp > 1:
+        if spec.num_devices > 1:
             loss_report = ht.comm(
-                loss, DistributedStates(dp, {-1: dp}, order=[-1]),
+                loss, spec._ds({-1: spec.num_devices}, [-1]),
                 name="loss_allreduce")
         opt = Adam(lr=lr, zero=zero)
         train_op = opt.minimize(loss)
     finally:
         pop_graph()
     return g, {"input_ids": input_ids, "labels": labels,
-               "loss": loss_report, "train_op": train_op, "optimizer": opt}
+               "loss": loss_report, "train_op": train_op,
+               "optimizer": opt, "model": model}
+
+
+def build_gpt_pipeline_stage(cfg: GPTConfig, pspec, micro_batch: int,
+                             seq_len: int, dtype=torch.bfloat16,
+                             lr: float = 1e-4, stage_layers=None,
+                             zero: bool = False):
+    """This rank's pipeline-stage subgraph (see parallel.pipeline)."""
+    from ..parallel.pipeline import StageModule
+    B, S = micro_batch, seq_len
+    sid = pspec.my_stage()
+    spec = pspec.stage_spec(sid)
+    parts = stage_layers or pspec.partition_layers(cfg.n_layer)
+    my_layers = parts[sid]
+    is_first, is_last = sid == 0, sid == pspec.pp - 1
+
+    g = DefineAndRunGraph(f"gpt_stage{sid}")
+    push_graph(g)
+    try:
+        h: Dict = {"act_shape": (B, S, cfg.hidden), "act_dtype": dtype}
+        ds_in = spec.ds_activation(0)
+        if is_first:
+            input_ids = ht.placeholder((B, S), dtype=torch.int64,
+                                       name="input_ids", ds=ds_in,
+                                       device_group=spec.device_group)
+            embed = GPTEmbedding(cfg, spec, S, dtype)
+            x = embed(input_ids)
+            h["input_ids"] = input_ids
+        else:
+            act_in = ht.placeholder((B, S, cfg.hidden), dtype=dtype,
+                                    name="act_in", ds=ds_in,
+                                    device_group=spec.device_group)
+            x = act_in
+            h["act_in"] = act_in
+        for li in my_layers:
+            x = GPTBlock(cfg, spec, li, dtype)(x, B, S)
+        if is_last:
+            labels = ht.placeholder((B * S,), dtype=torch.int64,
+                                    name="labels", ds=spec.ds_tokens(0),
+                                    device_group=spec.device_group)
+            lnf = ParallelLayerNorm(cfg.hidden, spec, 1e-5, dtype,
+                                    name="lnf")
+            lm_head = ColumnParallelLinear(
+                cfg.hidden, cfg.vocab, spec, bias=False, dtype=dtype,
+                name="lm_head", init_std=cfg.init_std)
+            xo = lnf(x)
+            logits = lm_head(
+                ht.reshape(xo, (B * S, cfg.hidden), ds=spec.ds_tokens(0)))
+            per_tok = vocab_parallel_cross_entropy(logits, labels, cfg.vocab)
+            loss = ht.reduce_mean(per_tok)
+            h["labels"] = labels
+            h["loss"] = loss
+        else:
+            h["act_out"] = x
+
+        params = list(g.parameters)
+        h["params"] = params
+        xs = params + ([] if is_first else [h["act_in"]])
+        if is_last:
+            grads = g.gradients([loss], xs)
+        else:
+            grad_in = ht.placeholder((B, S, cfg.hidden), dtype=dtype,
+                                     name="grad_in", ds=ds_in,
+                                     device_group=spec.device_group)
+            h["grad_in"] = grad_in
+            grads = g.gradients([x], xs, grad_ys=[grad_in])
+        h["param_grads"] = grads[:len(params)]
+        if not is_first:
+            h["dx"] = grads[len(params)]
+        from ..graph.ops.optim import AdamStepOp, ZeroAdamStepOp, GroupOp
+        from ..graph.ops.basics import _make
+        from ..graph.ops.comm import make_comm
+        grad_phs, updates = [], []
+        opt_attrs = {"lr": lr, "beta1": 0.9, "beta2": 0.999, "eps": 1e-8,
+                     "weight_decay": 0.0}
+        cls = ZeroAdamStepOp if zero else AdamStepOp
+        for p, pg in zip(params, h["param_grads"]):
+            gds = pg.ds if pg is not None else None
+            ph = ht.placeholder(tuple(p.shape), dtype=torch.float32,
+                                name=f"gbuf_{p.name}", ds=gds,
+                                device_group=spec.device_group)
+            grad_phs.append(ph)
+            gt = ph
+            if not zero and gds is not None and p.ds is not None \
+                    and not gds.check_equal(p.ds):
+                gt = make_comm(g, ph, p.ds, name=f"gred_{p.name}")
+            updates.append(_make(g, cls(), [p, gt], dict(opt_attrs),
+                                 name=f"adam_{p.name}").output())
+        h["grad_phs"] = grad_phs
+        h["train_op"] = _make(g, GroupOp(), updates, name="train_op").output()
+    finally:
+        pop_graph()
+    return StageModule(g, h)

@@ -55,6 +55,10 @@ void adam_step(torch::Tensor param32, torch::Tensor grad, torch::Tensor m,
                torch::Tensor out16, torch::Tensor bc_dev);
 // gemm.hip
 torch::Tensor gemm_bf16(torch::Tensor x, torch::Tensor w, bool trans_w);
+// galvatron_dp.cpp
+std::pair<double, std::vector<int64_t>> galvatron_dp(
+    std::vector<double> times, std::vector<double> mems, int64_t L,
+    int64_t S, double cap, int64_t buckets);
 // attention.hip
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
@@ -89,6 +93,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_bwd", &embedding_bwd);
   m.def("adam_step", &adam_step);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("galvatron_dp", &galvatron_dp);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
 }
