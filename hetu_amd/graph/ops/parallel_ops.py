@@ -254,6 +254,7 @@ class VocabParallelCrossEntropyOp(OpInterface):
 
     def compute(self, op, inputs, ctx):
         logits, labels = inputs
+        labels = labels.reshape(logits.shape[:-1])
         lds = op.inputs[0].ds
         dg = op.inputs[0].device_group
         my = _my_index(ctx, dg)
@@ -310,6 +311,8 @@ class VocabParallelCrossEntropyGradOp(OpInterface):
 
     def compute(self, op, inputs, ctx):
         gy, logits, labels, lse = inputs
+        labels = labels.reshape(logits.shape[:-1])
+        gy = gy.reshape(logits.shape[:-1])
         lds = op.inputs[1].ds
         my = _my_index(ctx, op.inputs[1].device_group)
         vdim = logits.ndim - 1
