@@ -160,12 +160,17 @@ class ColumnParallelLinear(Module):
                                   name=f"{name}.weight",
                                   ds=spec.ds_weight_col(0),
                                   device_group=spec.device_group)
+        if sections is not None and tp > 1:
+            # checkpoint de-interleave metadata (utils/checkpoint)
+            self.weight.shard_sections = list(sections)
         if bias:
             b = init.zeros((out_features,), dtype)
             self.bias = ht.variable(_shard(b, 0, tp, ti, sections),
                                     name=f"{name}.bias",
                                     ds=spec.ds_weight_col(0),
                                     device_group=spec.device_group)
+            if sections is not None and tp > 1:
+                self.bias.shard_sections = list(sections)
         else:
             self.register_parameter("bias", None)
         self.gather_output = gather_output
