@@ -20,7 +20,10 @@ def test_kv_put_get_barrier():
     done = []
 
     def member():
-        kv.barrier("b1", 2)
+        # own client connection: a TCPStore client socket is not
+        # thread-safe to share across concurrently-blocking calls
+        kv2 = KVStore(port=29711, is_server=False, world_size=1)
+        kv2.barrier("b1", 2)
         done.append(1)
     t = threading.Thread(target=member)
     t.start()
