@@ -263,6 +263,10 @@ def mse_loss(x, y):
     return _cg().make_op(N.MSELossOp(), [x, y], {}).output()
 
 
+def check_finite(tensors) -> Tensor:
+    return _cg().make_op(B.CheckFiniteOp(), list(tensors), {}).output()
+
+
 def ring_attention(q, k, v, cp_ranks, causal=True, scale=None):
     from . import parallel_ops as P
     return _cg().make_op(P.RingAttentionOp(), [q, k, v],

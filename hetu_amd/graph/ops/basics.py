@@ -701,6 +701,28 @@ class ReduceOp_(OpInterface):
                       dict(op.attrs)).output()]
 
 
+class CheckFiniteOp(OpInterface):
+    """All-finite flag over N tensors (reference CheckFinite.cu + the AMP
+    inf-check in gradscaler.cc): output fp32 scalar 1.0 iff every element
+    of every input is finite."""
+    type = "CheckFinite"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta([], torch.float32)]
+
+    def deduce_states(self, op):
+        op.outputs[0].ds = None
+        op.outputs[0].device_group = (op.inputs[0].device_group
+                                      if op.inputs else None)
+
+    def compute(self, op, inputs, ctx):
+        dev = inputs[0].device if inputs else "cpu"
+        ok = torch.ones((), device=dev)
+        for t in inputs:
+            ok = ok * torch.isfinite(t).all().to(ok.dtype)
+        return [ok]
+
+
 class ReduceGradOp(OpInterface):
     type = "ReduceGrad"
 

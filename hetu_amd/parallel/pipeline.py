@@ -96,10 +96,13 @@ class PipelineRunner:
     """Drives one optimizer step = M micro-batches through 1F1B."""
 
     def __init__(self, spec: PipelineSpec, stage: StageModule,
-                 device: torch.device, ctx=None):
+                 device: torch.device, ctx=None, scaler=None,
+                 recompute: bool = False):
         self.spec = spec
         self.stage = stage
         self.device = device
+        self.scaler = scaler          # engine.amp.GradScaler or None
+        self.recompute = recompute    # rerun fwd in bwd instead of caching
         self.comm = comm_backend(device)
         self.rank = self.comm.rank
         self.sid = spec.my_stage(self.rank)
@@ -125,8 +128,13 @@ class PipelineRunner:
         feed = dict(micro_batches[i])
         if not self.is_first:
             feed[h["act_in"]] = act
-        cache: Dict[int, torch.Tensor] = {}
         out_t = h["loss"] if self.is_last else h["act_out"]
+        if self.recompute:
+            # activation recompute (reference recompute.cc semantics at
+            # stage granularity): store only the feed; bwd reruns fwd
+            out = self.stage.graph.run([out_t], feed, ctx=self.ctx)
+            return out[0], (feed, None)
+        cache: Dict[int, torch.Tensor] = {}
         out = self.stage.graph.run([out_t], feed, ctx=self.ctx,
                                    keep_values=cache)
         return out[0], (feed, cache)
@@ -220,6 +228,13 @@ class PipelineRunner:
         # mean, so the accumulated grads divide by M
         for b in self.grad_bufs:
             b /= M
+        if self.scaler is not None:
+            self.scaler.unscale_(self.grad_bufs)
+            if not self.scaler.check_and_update(self.grad_bufs):
+                # overflow: skip the update, keep the backed-off scale
+                if self.is_last and losses:
+                    return torch.stack([l.float() for l in losses]).mean()
+                return None
         feed = {ph: buf for ph, buf in zip(h["grad_phs"], self.grad_bufs)}
         self.stage.graph.run([h["train_op"]], feed, ctx=self.ctx)
 
