@@ -26,6 +26,7 @@ class ExecContext:
         self.comm = comm                  # parallel.comm.CommBackend or None
         self.training = training
         self.symbols = {}
+        self.profiler = None              # utils.profiler.OpProfiler
         # stream roles (MI355X: overlap collectives with compute on separate
         # HIP streams; reference used a fixed 16-stream convention
         # hetu/core/stream.h:8-20 — we keep {compute, comm, p2p, h2d} roles)
@@ -122,7 +123,13 @@ class Executor:
                     values[tid] = feeds[tid]
                     continue
                 raise RuntimeError(f"placeholder {op.name} not fed")
-            outs = op.interface.compute(op, ins, ctx)
+            prof = ctx.profiler
+            if prof is not None:
+                tok = prof.begin(op)
+                outs = op.interface.compute(op, ins, ctx)
+                prof.end(tok)
+            else:
+                outs = op.interface.compute(op, ins, ctx)
             for t, v in zip(op.outputs, outs):
                 values[t.id] = v
             if keep_values is not None:

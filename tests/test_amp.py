@@ -71,3 +71,26 @@ def test_pipeline_runner_scaler_skips(tmp_path):
     assert sc.skipped == 1
     for p in stage.graph.parameters:
         assert torch.equal(p.get_data(), before[p.name]), p.name
+
+
+def test_op_profiler():
+    from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.utils.profiler import OpProfiler, MemorySnapshots
+    g = DefineAndRunGraph("prof")
+    push_graph(g)
+    try:
+        a = ht.placeholder((64, 64), name="a")
+        y = ht.gelu(ht.matmul(a, a))
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    prof = OpProfiler(use_events=False)
+    ctx.profiler = prof
+    g.run([y], {a: torch.randn(64, 64)}, ctx=ctx)
+    s = prof.summary()
+    assert "MatMul" in s and "Gelu" in s
+    ms = MemorySnapshots()
+    ms.mark("step0")
+    assert "step0" in ms.report()
