@@ -31,13 +31,12 @@ class OpProfiler:
         if self.use_events and torch.cuda.is_available():
             e0 = torch.cuda.Event(enable_timing=True)
             e0.record()
-            return (op, e0)
-        return (op, time.perf_counter())
+            return (op, e0, True)
+        return (op, time.perf_counter(), False)
 
     def end(self, token):
-        op, t0 = token
-        if isinstance(t0, torch.cuda.Event if torch.cuda.is_available()
-                      else float):
+        op, t0, is_event = token
+        if is_event:
             e1 = torch.cuda.Event(enable_timing=True)
             e1.record()
             self._pending.append((op, t0, e1))
