@@ -58,6 +58,9 @@ def main():
 
     if args.capture != "auto":
         os.environ["HETU_AMD_CAPTURE"] = args.capture
+    # hipGraph capture happens on the 2nd step (optimizer state must exist
+    # before capture); keep it out of the timed region
+    args.warmup = max(args.warmup, 2)
 
     import hetu_amd  # noqa: F401
     from hetu_amd.engine.trainer import Trainer

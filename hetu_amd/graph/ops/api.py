@@ -288,6 +288,140 @@ def vocab_parallel_cross_entropy(logits, labels, vocab: int,
                           "ignore_index": ignore_index}).output(0)
 
 
+# ---- extended families (einsum / vision / losses / manipulation) -----------
+
+def einsum(equation: str, *ts) -> Tensor:
+    from . import extra as E
+    return _cg().make_op(E.EinsumOp(), list(ts),
+                         {"equation": equation}).output()
+
+
+def conv2d(x, w, bias=None, stride=1, padding=0, dilation=1, groups=1):
+    from . import extra as E
+    ins = [x, w] + ([bias] if bias is not None else [])
+    return _cg().make_op(E.Conv2dOp(), ins,
+                         {"stride": stride, "padding": padding,
+                          "dilation": dilation, "groups": groups}).output()
+
+
+def max_pool2d(x, kernel, stride=None, padding=0):
+    from . import extra as E
+    return _cg().make_op(E.MaxPool2dOp(), [x],
+                         {"kernel": kernel, "stride": stride,
+                          "padding": padding}).output()
+
+
+def avg_pool2d(x, kernel, stride=None, padding=0):
+    from . import extra as E
+    return _cg().make_op(E.AvgPool2dOp(), [x],
+                         {"kernel": kernel, "stride": stride,
+                          "padding": padding}).output()
+
+
+def batch_norm(x, w, b, eps=1e-5):
+    from . import extra as E
+    return _cg().make_op(E.BatchNormOp(), [x, w, b], {"eps": eps}).output()
+
+
+def instance_norm(x, eps=1e-5):
+    from . import extra as E
+    return _cg().make_op(E.InstanceNormOp(), [x], {"eps": eps}).output()
+
+
+def interpolate(x, scale=None, size=None, mode="nearest"):
+    from . import extra as E
+    return _cg().make_op(E.InterpolateOp(), [x],
+                         {"scale": scale, "size": size,
+                          "mode": mode}).output()
+
+
+def binary_cross_entropy(x, target, reduction="mean"):
+    from . import extra as E
+    return _cg().make_op(E.BCEOp(), [x, target],
+                         {"reduction": reduction}).output()
+
+
+def kl_div(x, target, reduction="batchmean"):
+    from . import extra as E
+    return _cg().make_op(E.KLDivOp(), [x, target],
+                         {"reduction": reduction}).output()
+
+
+def nll_loss(x, target, reduction="mean", ignore_index=-100):
+    from . import extra as E
+    return _cg().make_op(E.NLLOp(), [x, target],
+                         {"reduction": reduction,
+                          "ignore_index": ignore_index}).output()
+
+
+def where(cond, a, b):
+    from . import extra as E
+    return _cg().make_op(E.WhereOp(), [cond, a, b], {}).output()
+
+
+def triu(a, diagonal=0):
+    from . import extra as E
+    return _cg().make_op(E.TriuOp(), [a], {"diagonal": diagonal}).output()
+
+
+def clamp(a, min=None, max=None):  # noqa: A002
+    from . import extra as E
+    return _cg().make_op(E.ClampOp(), [a], {"min": min, "max": max}).output()
+
+
+def gather(a, dim, index):
+    from . import extra as E
+    return _cg().make_op(E.GatherOp(), [a, index], {"dim": dim}).output()
+
+
+def index_add(a, dim, index, src):
+    from . import extra as E
+    return _cg().make_op(E.IndexAddOp(), [a, index, src],
+                         {"dim": dim}).output()
+
+
+def masked_fill(a, mask, value):
+    from . import extra as E
+    return _cg().make_op(E.MaskedFillOp(), [a, mask],
+                         {"value": value}).output()
+
+
+def pad(a, pad_widths, value=0.0):
+    from . import extra as E
+    return _cg().make_op(E.PadOp(), [a], {"pad": list(pad_widths),
+                                          "value": value}).output()
+
+
+def repeat(a, repeats):
+    from . import extra as E
+    return _cg().make_op(E.RepeatOp(), [a],
+                         {"repeats": list(repeats)}).output()
+
+
+def roll(a, shifts, dims=None):
+    from . import extra as E
+    return _cg().make_op(E.RollOp(), [a], {"shifts": shifts,
+                                           "dims": dims}).output()
+
+
+def onehot(ids, num_classes):
+    from . import extra as E
+    return _cg().make_op(E.OnehotOp(), [ids],
+                         {"num_classes": num_classes}).output()
+
+
+def arange(end, start=0, step=1, dtype=torch.int64):
+    from . import extra as E
+    return _cg().make_op(E.ArangeOp(), [], {"start": start, "end": end,
+                                            "step": step,
+                                            "dtype": dtype}).output()
+
+
+def eye(n, dtype=torch.float32):
+    from . import extra as E
+    return _cg().make_op(E.EyeOp(), [], {"n": n, "dtype": dtype}).output()
+
+
 # ---- comm ------------------------------------------------------------------
 
 def comm(x: Tensor, dst_ds: DistributedStates, name="comm") -> Tensor:

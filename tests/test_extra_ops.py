@@ -1,0 +1,130 @@
+"""Extended op families vs torch references (fwd + grads)."""
+import torch
+
+from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
+from hetu_amd.graph.ops import api as ht
+from hetu_amd.engine.runner import prepare_run_context
+
+
+def _run(build, feeds, wrt=None):
+    g = DefineAndRunGraph("t")
+    push_graph(g)
+    try:
+        phs, out = build()
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    fetches = [out]
+    if wrt is not None:
+        grads = g.gradients([out], [phs[i] for i in wrt])
+        fetches += grads
+    vals = g.run(fetches, dict(zip(phs, feeds)), ctx=ctx)
+    return vals
+
+
+def test_einsum_fwd_bwd():
+    a = torch.randn(4, 5, requires_grad=True)
+    b = torch.randn(5, 6, requires_grad=True)
+
+    def build():
+        x = ht.placeholder((4, 5), name="a")
+        y = ht.placeholder((5, 6), name="b")
+        return [x, y], ht.reduce_sum(ht.einsum("ij,jk->ik", x, y))
+    out, ga, gb = _run(build, [a.detach(), b.detach()], wrt=[0, 1])
+    ref = torch.einsum("ij,jk->ik", a, b).sum()
+    ref.backward()
+    assert torch.allclose(out, ref.detach(), rtol=1e-5)
+    assert torch.allclose(ga, a.grad, rtol=1e-5)
+    assert torch.allclose(gb, b.grad, rtol=1e-5)
+
+
+def test_conv2d_bwd():
+    x = torch.randn(2, 3, 8, 8, requires_grad=True)
+    w = torch.randn(4, 3, 3, 3, requires_grad=True)
+
+    def build():
+        xp = ht.placeholder((2, 3, 8, 8), name="x")
+        wp = ht.placeholder((4, 3, 3, 3), name="w")
+        return [xp, wp], ht.reduce_sum(ht.conv2d(xp, wp, padding=1))
+    out, gx, gw = _run(build, [x.detach(), w.detach()], wrt=[0, 1])
+    ref = torch.nn.functional.conv2d(x, w, padding=1).sum()
+    ref.backward()
+    assert torch.allclose(out, ref.detach(), rtol=1e-4)
+    assert torch.allclose(gx, x.grad, rtol=1e-4, atol=1e-5)
+    assert torch.allclose(gw, w.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_pools_and_norms():
+    x = torch.randn(2, 3, 8, 8)
+
+    def build():
+        xp = ht.placeholder((2, 3, 8, 8), name="x")
+        return [xp], ht.max_pool2d(xp, 2)
+    (out,) = _run(build, [x])
+    assert torch.allclose(out, torch.nn.functional.max_pool2d(x, 2))
+
+    w = torch.ones(3)
+    b = torch.zeros(3)
+
+    def build2():
+        xp = ht.placeholder((2, 3, 8, 8), name="x")
+        wp = ht.placeholder((3,), name="w")
+        bp = ht.placeholder((3,), name="b")
+        return [xp, wp, bp], ht.batch_norm(xp, wp, bp)
+    (out2,) = _run(build2, [x, w, b])
+    ref2 = torch.nn.functional.batch_norm(x, None, None, w, b, training=True)
+    assert torch.allclose(out2, ref2, rtol=1e-4, atol=1e-5)
+
+
+def test_losses():
+    p = torch.rand(8).clamp(0.01, 0.99)
+    t = torch.rand(8).round()
+
+    def build():
+        xp = ht.placeholder((8,), name="x")
+        tp = ht.placeholder((8,), name="t")
+        return [xp, tp], ht.binary_cross_entropy(xp, tp)
+    (out,) = _run(build, [p, t])
+    assert torch.allclose(out, torch.nn.functional.binary_cross_entropy(p, t),
+                          rtol=1e-5)
+
+
+def test_manipulation_ops():
+    x = torch.randn(4, 6)
+
+    def b1():
+        xp = ht.placeholder((4, 6), name="x")
+        return [xp], ht.triu(xp, 1)
+    (out,) = _run(b1, [x])
+    assert torch.equal(out, torch.triu(x, 1))
+
+    def b2():
+        xp = ht.placeholder((4, 6), name="x")
+        return [xp], ht.reduce_sum(ht.clamp(xp, min=-0.5, max=0.5))
+    out, gx = _run(b2, [x], wrt=[0])
+    xr = x.clone().requires_grad_(True)
+    xr.clamp(-0.5, 0.5).sum().backward()
+    assert torch.allclose(gx, xr.grad)
+
+    idx = torch.randint(0, 6, (4, 3))
+
+    def b3():
+        xp = ht.placeholder((4, 6), name="x")
+        ip = ht.placeholder((4, 3), dtype=torch.int64, name="i")
+        return [xp, ip], ht.reduce_sum(ht.gather(xp, 1, ip))
+    out, gx = _run(b3, [x, idx], wrt=[0])
+    xr = x.clone().requires_grad_(True)
+    xr.gather(1, idx).sum().backward()
+    assert torch.allclose(gx, xr.grad)
+
+    def b4():
+        xp = ht.placeholder((4, 6), name="x")
+        return [xp], ht.pad(xp, [1, 2], value=3.0)
+    (out,) = _run(b4, [x])
+    assert out.shape == (4, 9) and float(out[0, 0]) == 3.0
+
+    def b5():
+        xp = ht.placeholder((4, 6), name="x")
+        return [xp], ht.roll(xp, 2, 1)
+    (out,) = _run(b5, [x])
+    assert torch.equal(out, torch.roll(x, 2, 1))
