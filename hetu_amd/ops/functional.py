@@ -480,3 +480,74 @@ def linear(x, w, bias=None, trans_w: bool = True):
     if bias is not None:
         y = y + bias
     return y
+
+
+# ---------------------------------------------------------------------------
+# Blockwise quantization (reference quantization.cu / Quantization.h;
+# bitsandbytes-style fp4/nf4/int8 with per-block absmax)
+# ---------------------------------------------------------------------------
+
+_FP4_CODE = torch.tensor([
+    0.0, 0.0052083333, 0.6666667, 1.0, 0.3333333, 0.5, 0.1666667, 0.25,
+    -0.0, -0.0052083333, -0.6666667, -1.0, -0.3333333, -0.5,
+    -0.1666667, -0.25])
+_NF4_CODE = torch.tensor([
+    -1.0, -0.6961928009986877, -0.5250730514526367, -0.39491748809814453,
+    -0.28444138169288635, -0.18477343022823334, -0.09105003625154495, 0.0,
+    0.07958029955625534, 0.16093020141124725, 0.24611230194568634,
+    0.33791524171829224, 0.44070982933044434, 0.5626170039176941,
+    0.7229568362236023, 1.0])
+
+
+def quantize_blockwise(x: torch.Tensor, qtype: str = "nf4",
+                       blocksize: int = 64):
+    """returns (packed uint8, absmax fp32 [nblocks])."""
+    if _gpu(x):
+        return tuple(ext().quantize_blockwise(x.contiguous(), qtype,
+                                              blocksize))
+    flat = x.float().reshape(-1)
+    n = flat.numel()
+    nblk = (n + blocksize - 1) // blocksize
+    pad = nblk * blocksize - n
+    if pad:
+        flat = torch.cat([flat, flat.new_zeros(pad)])
+    blocks = flat.reshape(nblk, blocksize)
+    absmax = blocks.abs().amax(-1)
+    if qtype == "int8":
+        inv = torch.where(absmax > 0, 127.0 / absmax,
+                          torch.zeros_like(absmax))
+        q = torch.round(blocks * inv.unsqueeze(-1)) + 128
+        return q.reshape(-1)[:n].to(torch.uint8), absmax
+    code = _NF4_CODE if qtype == "nf4" else _FP4_CODE
+    inv = torch.where(absmax > 0, 1.0 / absmax, torch.zeros_like(absmax))
+    norm = blocks * inv.unsqueeze(-1)
+    idx = (norm.unsqueeze(-1) - code).abs().argmin(-1).to(torch.uint8)
+    idx = idx.reshape(-1)[:n]
+    hi, lo = idx[0::2], idx[1::2]
+    return (hi << 4) | lo, absmax
+
+
+def dequantize_blockwise(q: torch.Tensor, absmax: torch.Tensor,
+                         qtype: str, blocksize: int, numel: int,
+                         dtype=torch.float32) -> torch.Tensor:
+    if _gpu(q):
+        return ext().dequantize_blockwise(q, absmax, qtype, blocksize,
+                                          numel, dtype)
+    if qtype == "int8":
+        s = absmax.repeat_interleave(blocksize)[:numel] / 127.0
+        return ((q.float() - 128) * s).to(dtype)
+    code = _NF4_CODE if qtype == "nf4" else _FP4_CODE
+    hi, lo = (q >> 4).long(), (q & 15).long()
+    vals = torch.stack([code[hi], code[lo]], -1).reshape(-1)[:numel]
+    s = absmax.repeat_interleave(blocksize)[:numel]
+    return (vals * s).to(dtype)
+
+
+def matmul_4bit(x: torch.Tensor, qweight: torch.Tensor,
+                absmax: torch.Tensor, qtype: str, blocksize: int,
+                shape) -> torch.Tensor:
+    """y = x @ dequant(W)^T (reference matmul4bit: dequant then GEMM —
+    weight stays 4-bit in HBM, dequant streams through once)."""
+    w = dequantize_blockwise(qweight, absmax, qtype, blocksize,
+                             shape[0] * shape[1], x.dtype).reshape(shape)
+    return x @ w.t()

@@ -55,6 +55,14 @@ void adam_step(torch::Tensor param32, torch::Tensor grad, torch::Tensor m,
                torch::Tensor out16, torch::Tensor bc_dev);
 // gemm.hip
 torch::Tensor gemm_bf16(torch::Tensor x, torch::Tensor w, bool trans_w);
+// quant.hip
+std::vector<torch::Tensor> quantize_blockwise(torch::Tensor x,
+                                              std::string qtype,
+                                              int64_t blocksize);
+torch::Tensor dequantize_blockwise(torch::Tensor q, torch::Tensor absmax,
+                                   std::string qtype, int64_t blocksize,
+                                   int64_t numel,
+                                   torch::ScalarType out_dtype);
 // galvatron_dp.cpp
 std::pair<double, std::vector<int64_t>> galvatron_dp(
     std::vector<double> times, std::vector<double> mems, int64_t L,
@@ -93,6 +101,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_bwd", &embedding_bwd);
   m.def("adam_step", &adam_step);
   m.def("gemm_bf16", &gemm_bf16);
+  m.def("quantize_blockwise", &quantize_blockwise);
+  m.def("dequantize_blockwise", &dequantize_blockwise);
   m.def("galvatron_dp", &galvatron_dp);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);

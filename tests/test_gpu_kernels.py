@@ -233,3 +233,17 @@ class TestAttention:
         orf, _ = F._attn_ref_fwd(q.cpu().float(), k.cpu().float(),
                                  v.cpu().float(), True, scale)
         _close(o, orf, rtol=3e-2, atol=3e-2, what="fa spike")
+
+
+class TestQuant:
+    @pytest.mark.parametrize("qtype", ["int8", "nf4", "fp4"])
+    def test_quant_roundtrip_matches_cpu(self, qtype):
+        torch.manual_seed(0)
+        x = torch.randn(8192, dtype=torch.float32, device=dev())
+        q, am = F.quantize_blockwise(x, qtype, 64)
+        qr, amr = F.quantize_blockwise(x.cpu(), qtype, 64)
+        _close(am, amr, rtol=1e-5, atol=1e-6, what="absmax")
+        # codes may differ at exact midpoints; compare dequantized values
+        y = F.dequantize_blockwise(q, am, qtype, 64, 8192)
+        yr = F.dequantize_blockwise(qr, amr, qtype, 64, 8192)
+        _close(y, yr, rtol=1e-3, atol=2e-2, what=f"{qtype} dequant")
