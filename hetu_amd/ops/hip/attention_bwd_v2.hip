@@ -108,11 +108,12 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dq_kernel(
   auto k_lds = [&](int b) -> char* { return smem + b * KB; };
   auto v_lds = [&](int b) -> char* { return smem + (2 + b) * KB; };
 
-  const int bh = blockIdx.y;
+  // (bh, q-block) grid: spreads causal depths across CUs (see fwd note)
+  const int bh = blockIdx.x;
   const int h = bh % H;
   const int b = bh / H;
   const int hkv = h / (H / Hkv);
-  const int q0 = blockIdx.x * 256;
+  const int q0 = blockIdx.y * 256;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -284,11 +285,12 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
   char* vw_base = smem + 4 * QB;   // per-wave V image, 8 KiB each
   char* pw_base = smem + 4 * QB + 8 * 8192;   // per-wave P'/dS' (2 x 2304 B)
 
-  const int bh = blockIdx.y;
+  // (bh, kv-block) grid: spreads causal depths across CUs (see fwd note)
+  const int bh = blockIdx.x;
   const int h = bh % H;
   const int b = bh / H;
   const int hkv = h / (H / Hkv);
-  const int kb0 = blockIdx.x * 256;
+  const int kb0 = blockIdx.y * 256;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -533,7 +535,7 @@ std::vector<torch::Tensor> fa2_bwd_launch(
     dv16 = torch::empty_like(v);
   }
   {
-    dim3 grid((S + 255) / 256, B * H);
+    dim3 grid(B * H, (S + 255) / 256);
     size_t lds = 4 * (size_t)64 * 256;          // 64 KiB
     hipLaunchKernelGGL(fa2_bwd_dq_kernel<128>, grid, dim3(THREADS), lds,
                        stream, (const bf16*)q.data_ptr(),
@@ -543,7 +545,7 @@ std::vector<torch::Tensor> fa2_bwd_launch(
                        B, H, Hkv, S, Skv, (float)scale, causal);
   }
   {
-    dim3 grid((Skv + 255) / 256, B * H);
+    dim3 grid(B * H, (Skv + 255) / 256);
     size_t lds = 4 * (size_t)32 * 256 + 8 * 8192 + 8 * 4608;
     hipLaunchKernelGGL(fa2_bwd_dkv_kernel<128>, grid, dim3(THREADS), lds,
                        stream, (const bf16*)q.data_ptr(),

@@ -28,7 +28,9 @@ using u32x2 = __attribute__((ext_vector_type(2))) unsigned int;
 
 constexpr int BM = 256;       // q rows per workgroup (8 waves x 32)
 constexpr int BN = 64;        // kv tile
-constexpr int THREADS = 512;
+constexpr int THREADS = BM * 2;   // one wave per 32 q rows
+constexpr int NGLDS = 64 * 256 / (THREADS * 16);  // glds per lane per image
+constexpr int NBUF = 3;       // KV ring depth (counted-vmcnt pipelining)
 
 // K tile LDS image: row-major [64][256B], byte-in-row ^ ((row&15)<<4)
 DEV int kswz(int row, int byte_in_row) {
@@ -51,13 +53,16 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
   constexpr int KBYTES = BN * D * 2;          // 16 KiB
   extern __shared__ __attribute__((aligned(16))) char smem[];
   auto k_lds = [&](int b) -> char* { return smem + b * KBYTES; };
-  auto v_lds = [&](int b) -> char* { return smem + (2 + b) * KBYTES; };
+  auto v_lds = [&](int b) -> char* { return smem + (NBUF + b) * KBYTES; };
 
-  const int bh = blockIdx.y;
+  // grid is (bh, q-block): consecutive blockIdx.x (-> XCD b%8 placement)
+  // walk bh, so every CU samples ALL causal depths — with (q-block, bh)
+  // each CU aliased onto ONE depth and causal ran as slow as non-causal
+  const int bh = blockIdx.x;
   const int h = bh % H;
   const int b = bh / H;
   const int hkv = h / (H / Hkv);
-  const int q0 = blockIdx.x * BM;
+  const int q0 = blockIdx.y * BM;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -101,8 +106,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
 #define ISSUE_GLDS(k0, buf)                                                 \
   do {                                                                      \
     _Pragma("unroll")                                                       \
-    for (int i = 0; i < 2; ++i) {                                           \
-      int pos = (wid * 2 + i) * 1024 + wlane16;                             \
+    for (int i = 0; i < NGLDS; ++i) {                                       \
+      int pos = (wid * NGLDS + i) * 1024 + wlane16;                             \
       int krow = pos >> 8;                                                  \
       int kd = ((pos & 255) ^ ((krow & 15) << 4)) >> 1;                     \
       const bf16* ksrc = Kb + (int64_t)min((k0) + krow, Skv - 1) * D + kd;  \
@@ -136,17 +141,19 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
     n_tiles = min(n_tiles, max(lim, 1));
   }
 
-  // prologue: tile 0 into buffer 0
+  // prologue: tiles 0 and 1 in flight (3-deep ring, counted vmcnt:
+  // per wave each tile is 2*NGLDS glds; waiting vmcnt(2*NGLDS) at the
+  // loop head means "tile t landed, tile t+1 still flying" — the DMA for
+  // t+2 then issues after the barrier and lands under two tiles of MFMAs)
   ISSUE_GLDS(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __builtin_amdgcn_s_barrier();
+  if (n_tiles > 1) ISSUE_GLDS(BN, 1);
 
   for (int t = 0; t < n_tiles; ++t) {
     const int k0 = t * BN;
-    const int cur = t & 1;
-    // issue next tile's DMA before compute; it lands under the MFMAs
-    const bool have_next = (t + 1) < n_tiles;
-    if (have_next) ISSUE_GLDS(k0 + BN, cur ^ 1);
+    const int cur = t % NBUF;
+    asm volatile("s_waitcnt vmcnt(%0)" :: "i"(2 * NGLDS) : "memory");
+    __builtin_amdgcn_s_barrier();
+    if (t + 2 < n_tiles) ISSUE_GLDS(k0 + 2 * BN, (t + 2) % NBUF);
 
     // does this wave have any unmasked key in this tile?
     const bool active = !causal || (k0 <= wave_qmax + diag);
@@ -267,11 +274,6 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
       }
     }
 
-    // drain this wave's DMA, then one barrier per tile: reads of buf cur^1
-    // all completed before the PREVIOUS barrier, so write(t+1)/read(t+1)
-    // are ordered by this one.
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
   }
 
   // ---- epilogue -------------------------------------------------------
@@ -311,8 +313,8 @@ bool fa2_fwd_supported(int D, int S) { return D == 128; }
 void fa2_fwd_launch(const void* q, const void* k, const void* v, void* o,
                     float* lse, int B, int H, int Hkv, int S, int Skv,
                     float scale, bool causal, hipStream_t stream) {
-  dim3 grid((S + BM - 1) / BM, B * H);
-  size_t lds = 4 * (size_t)BN * 128 * 2;   // 64 KiB
+  dim3 grid(B * H, (S + BM - 1) / BM);
+  size_t lds = 2 * NBUF * (size_t)BN * 128 * 2;   // 96 KiB
   hipLaunchKernelGGL(fa2_fwd_kernel<128>, grid, dim3(THREADS), lds, stream,
                      (const bf16*)q, (const bf16*)k, (const bf16*)v,
                      (bf16*)o, lse, B, H, Hkv, S, Skv, scale, causal);
