@@ -77,7 +77,11 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout,
                                           torch::Tensor lse, bool causal,
                                           double scale);
 
+// embed_cache.cpp
+void register_embed_cache(pybind11::module& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  register_embed_cache(m);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);

@@ -29,6 +29,7 @@ SOURCES = [
     "attention_bwd_v2.hip",
     "quant.hip",
     "galvatron_dp.cpp",
+    "embed_cache.cpp",
     "bindings.cpp",
 ]
 
