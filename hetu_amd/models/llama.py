@@ -269,6 +269,7 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
             loss = ht.reduce_mean(per_tok)
             h["labels"] = labels
             h["loss"] = loss
+            h["logits"] = logits
         else:
             h["act_out"] = x
 
@@ -331,7 +332,7 @@ def build_llama_train_graph(cfg: LlamaConfig, micro_batch: int, seq_len: int,
                                 name="labels", ds=spec.ds_tokens(0),
                                 device_group=spec.device_group)
         model = LlamaLMHeadModel(cfg, spec, micro_batch, seq_len, dtype)
-        loss, _ = model(input_ids, labels)
+        loss, logits = model(input_ids, labels)
         loss_report = loss
         if spec.num_devices > 1:
             loss_report = ht.comm(
@@ -342,5 +343,5 @@ def build_llama_train_graph(cfg: LlamaConfig, micro_batch: int, seq_len: int,
     finally:
         pop_graph()
     return g, {"input_ids": input_ids, "labels": labels,
-               "loss": loss_report, "train_op": train_op,
+               "loss": loss_report, "logits": logits, "train_op": train_op,
                "optimizer": opt, "model": model}
