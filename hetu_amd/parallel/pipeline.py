@@ -97,12 +97,17 @@ class PipelineRunner:
 
     def __init__(self, spec: PipelineSpec, stage: StageModule,
                  device: torch.device, ctx=None, scaler=None,
-                 recompute: bool = False):
+                 recompute: bool = False, offload: bool = False):
         self.spec = spec
         self.stage = stage
         self.device = device
         self.scaler = scaler          # engine.amp.GradScaler or None
         self.recompute = recompute    # rerun fwd in bwd instead of caching
+        self.offloader = None
+        if offload and not recompute:
+            from ..engine.offload import ActOffloader
+            # 1F1B holds up to pp in-flight micro-batches on stage 0
+            self.offloader = ActOffloader(device, slots=spec.pp + 1)
         self.comm = comm_backend(device)
         self.rank = self.comm.rank
         self.sid = spec.my_stage(self.rank)
@@ -137,11 +142,18 @@ class PipelineRunner:
         cache: Dict[int, torch.Tensor] = {}
         out = self.stage.graph.run([out_t], feed, ctx=self.ctx,
                                    keep_values=cache)
+        if self.offloader is not None:
+            handle = self.offloader.offload(cache)
+            cache.clear()                 # drop HBM refs
+            return out[0], (feed, handle)
         return out[0], (feed, cache)
 
     def _bwd(self, saved, gin):
         h = self.stage.h
         feed, cache = saved
+        if self.offloader is not None and cache is not None \
+                and not isinstance(cache, dict):
+            cache = self.offloader.fetch(cache)
         if not self.is_last:
             feed = dict(feed)
             feed[h["grad_in"]] = gin
@@ -198,6 +210,9 @@ class PipelineRunner:
                 gin = torch.empty(tuple(out.shape), dtype=out.dtype,
                                   device=self.device)
                 _p2p(self.comm, [(out, self.next)], [(gin, self.next)])
+            if self.offloader is not None and len(pending) > 1 \
+                    and not isinstance(pending[1][1], dict):
+                self.offloader.prefetch(pending[1][1])
             dx = self._bwd(pending.pop(0), gin)
             if self.is_last:
                 losses[-1] = losses[-1].clone()   # loss survives cache free

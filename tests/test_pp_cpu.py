@@ -33,7 +33,9 @@ pspec = PipelineSpec(pp=pp, dp=dp, tp=tp)
 comm = comm_backend(torch.device("cpu"))
 stage = build_llama_pipeline_stage(cfg, pspec, micro_batch=1, seq_len=16,
                                    dtype=torch.float32, lr=1e-3)
-runner = PipelineRunner(pspec, stage, torch.device("cpu"))
+runner = PipelineRunner(pspec, stage, torch.device("cpu"),
+                        offload=bool(int(os.environ.get(
+                            "HETU_TEST_OFFLOAD", "0"))))
 gen = torch.Generator().manual_seed(99)
 ids = torch.randint(0, cfg.vocab, (M, 1, 16), generator=gen)
 labels = torch.randint(0, cfg.vocab, (M, 16), generator=gen)
@@ -150,3 +152,13 @@ def test_pp2_tp2_matches_single(single_losses):
     assert losses is not None
     assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
         f"pp2xtp2 {losses} vs single {single_losses}"
+
+
+def test_pp2_offload_matches_single(single_losses):
+    """Activation CPU offload between fwd and bwd must be numerically
+    invisible (reference activation_cpu_offload.cc semantics)."""
+    losses = _launch(2, {"HETU_TEST_PP": "2", "HETU_TEST_OFFLOAD": "1"},
+                     29543)
+    assert losses is not None
+    assert np.allclose(losses, single_losses, rtol=2e-4, atol=1e-4), \
+        f"pp2+offload {losses} vs single {single_losses}"
