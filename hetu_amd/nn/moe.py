@@ -20,29 +20,49 @@ from .parallel import ParallelSpec
 
 
 class MoEMLP(Module):
-    """Top-K gated expert FFN (gelu).  Experts sharded over the ep group
-    (= the spec's full device group for now); gate weights replicated."""
+    """Gated expert FFN (gelu).  Experts sharded over the ep group
+    (= the spec's full device group for now); gate weights replicated.
+
+    gate_type (reference v1/python/hetu/layers/gates/ — 5 gate families):
+      "topk"   — learned softmax top-k (k=2 == GShard top-2)
+      "switch" — learned top-1 (Switch Transformer)
+      "hash"   — static modulo-hash routing by token position (no gate
+                 params; HashGate semantics for static-shape graphs)
+      "random" — static pseudo-random token->expert assignment (BASE-like
+                 balanced random routing, fixed at build time)
+    """
 
     def __init__(self, hidden: int, ffn_hidden: int, num_experts: int,
                  spec: Optional[ParallelSpec] = None, k: int = 2,
                  capacity_factor: float = 1.25, dtype=torch.float32,
-                 name: str = "moe"):
+                 gate_type: str = "topk", name: str = "moe"):
         super().__init__()
         spec = spec or ParallelSpec()
         self.spec = spec
         self.E = num_experts
+        assert gate_type in ("topk", "switch", "hash", "random")
+        self.gate_type = gate_type
+        if gate_type == "switch":
+            k = 1
+        elif gate_type in ("hash", "random"):
+            k = 1
         self.k = k
+        self.dtype = dtype
+        self.name = name
         self.capacity_factor = capacity_factor
         self.hidden, self.ffn = hidden, ffn_hidden
         P = spec.num_devices
         assert num_experts % P == 0, "experts must divide ep group"
         El = num_experts // P
         me = spec.my_index()
-        w_gate = init.normal((num_experts, hidden), std=0.02, dtype=dtype,
-                             name=f"{name}.gate")
-        self.gate = ht.variable(w_gate, name=f"{name}.gate.weight",
-                                ds=spec.ds_weight_dup(),
-                                device_group=spec.device_group)
+        self.gate = None
+        if gate_type in ("topk", "switch"):
+            w_gate = init.normal((num_experts, hidden), std=0.02,
+                                 dtype=dtype, name=f"{name}.gate")
+            self.gate = ht.variable(w_gate, name=f"{name}.gate.weight",
+                                    ds=spec.ds_weight_dup(),
+                                    device_group=spec.device_group)
+        self._static_probs = {}   # N -> probs variable (hash/random)
         w1 = init.normal((num_experts, hidden, ffn_hidden), std=0.02,
                          dtype=dtype, name=f"{name}.w1")
         w2 = init.normal((num_experts, ffn_hidden, hidden), std=0.02,
@@ -56,14 +76,33 @@ class MoEMLP(Module):
                               name=f"{name}.w2", ds=spec.ds_weight_dup(),
                               device_group=spec.device_group)
 
+    def _static_gate(self, N: int):
+        if N not in self._static_probs:
+            if self.gate_type == "hash":
+                assign = torch.arange(N) % self.E
+            else:                                  # random (fixed at build)
+                g = torch.Generator().manual_seed(hash((self.name, N))
+                                                  & 0x7FFFFFFF)
+                assign = torch.randperm(N, generator=g) % self.E
+            probs = torch.zeros(N, self.E, dtype=self.dtype)
+            probs[torch.arange(N), assign] = 1.0
+            self._static_probs[N] = ht.variable(
+                probs, name=f"{self.name}.static_gate_{N}",
+                requires_grad=False, ds=self.spec.ds_weight_dup(),
+                device_group=self.spec.device_group)
+        return self._static_probs[N]
+
     def forward(self, x):
         """x: [N, h] tokens -> [N, h]."""
         spec = self.spec
         N = x.shape[0]
         P = spec.num_devices
         C = max(1, int(self.capacity_factor * self.k * N / self.E))
-        logits = ht.linear(x, self.gate)          # [N, E]
-        probs = ht.softmax(logits, dim=-1)
+        if self.gate_type in ("topk", "switch"):
+            logits = ht.linear(x, self.gate)      # [N, E]
+            probs = ht.softmax(logits, dim=-1)
+        else:
+            probs = self._static_gate(int(N))
         g = x.graph
         attrs = {"experts": self.E, "capacity": C, "k": self.k,
                  "ep_ranks": list(spec.device_group) if P > 1 else None}

@@ -143,3 +143,34 @@ def test_moe_ep2_matches_single():
     assert single is not None
     assert np.allclose(outs[0], single, rtol=1e-3, atol=1e-4), \
         np.abs(outs[0] - single).max()
+
+
+@pytest.mark.parametrize("gate_type", ["switch", "hash", "random"])
+def test_moe_gate_types_train(gate_type):
+    """Gate families beyond top-k (reference v1 gates/): switch top-1,
+    static hash and random routing all must train."""
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.graph.ops.optim import Adam
+    from hetu_amd.nn.moe import MoEMLP
+    torch.manual_seed(0)
+    g = DefineAndRunGraph("m")
+    push_graph(g)
+    try:
+        x = ht.placeholder((16, 32), name="x")
+        t = ht.placeholder((16, 32), name="t")
+        moe = MoEMLP(32, 64, 4, capacity_factor=100.0, gate_type=gate_type)
+        loss = ht.mse_loss(moe(x), t)
+        op = Adam(lr=1e-3).minimize(loss)
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    xd, td = torch.randn(16, 32), torch.randn(16, 32)
+    losses = []
+    for _ in range(8):
+        lv, _ = g.run([loss, op], {x: xd, t: td}, ctx=ctx)
+        losses.append(float(lv))
+    assert losses[-1] < losses[0], (gate_type, losses)
+    if gate_type in ("hash", "random"):
+        assert moe.gate is None          # no gate params for static routing
