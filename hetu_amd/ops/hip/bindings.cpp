@@ -79,9 +79,12 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout,
 
 // embed_cache.cpp
 void register_embed_cache(pybind11::module& m);
+// dataloader.cpp
+void register_dataloader(pybind11::module& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_embed_cache(m);
+  register_dataloader(m);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);

@@ -30,6 +30,7 @@ SOURCES = [
     "quant.hip",
     "galvatron_dp.cpp",
     "embed_cache.cpp",
+    "dataloader.cpp",
     "bindings.cpp",
 ]
 
