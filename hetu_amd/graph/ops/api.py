@@ -437,3 +437,63 @@ def gradients(ys, xs, grad_ys=None):
     gy_l = None if grad_ys is None else (
         [grad_ys] if isinstance(grad_ys, Tensor) else list(grad_ys))
     return g.gradients(ys_l, list(xs), gy_l)
+
+
+# ---- bulk unary families ----------------------------------------------------
+def _unary_api(op_name):
+    from . import extra as _x
+    from .basics import _make as _mk
+    cls = getattr(_x, f"{op_name}Op")
+
+    def f(a, **attrs):
+        return _mk(_cg(), cls(), [a], attrs, name=op_name.lower()).output()
+    f.__name__ = op_name.lower()
+    return f
+
+
+abs_ = _unary_api("Abs")
+ceil = _unary_api("Ceil")
+floor = _unary_api("Floor")
+round_ = _unary_api("Round")
+sin = _unary_api("Sin")
+cos = _unary_api("Cos")
+reciprocal = _unary_api("Reciprocal")
+leaky_relu = _unary_api("LeakyRelu")
+mish = _unary_api("Mish")
+elu = _unary_api("Elu")
+hardshrink = _unary_api("Hardshrink")
+hardsigmoid = _unary_api("Hardsigmoid")
+hardswish = _unary_api("Hardswish")
+hardtanh = _unary_api("Hardtanh")
+logsigmoid = _unary_api("Logsigmoid")
+softplus = _unary_api("Softplus")
+softshrink = _unary_api("Softshrink")
+
+
+def outer(a, b):
+    from .basics import _make as _mk
+    from .extra import OuterOp
+    return _mk(_cg(), OuterOp(), [a, b], name="outer").output()
+
+
+def dot(a, b):
+    from .basics import _make as _mk
+    from .extra import DotOp
+    return _mk(_cg(), DotOp(), [a, b], name="dot").output()
+
+
+def diagonal(a, offset=0, dim1=0, dim2=1):
+    from .basics import _make as _mk
+    from .extra import DiagonalOp
+    return _mk(_cg(), DiagonalOp(), [a],
+                 {"offset": offset, "dim1": dim1, "dim2": dim2},
+                 name="diagonal").output()
+
+
+def split(a, sections: int, dim: int = 0):
+    """Even split into `sections` along dim (reference Split.cc) as a list
+    of slices."""
+    n = a.shape[dim]
+    assert n % sections == 0, "uneven split"
+    step = n // sections
+    return [slice_(a, dim, i * step, step) for i in range(sections)]

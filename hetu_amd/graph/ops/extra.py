@@ -507,3 +507,82 @@ class EyeOp(OpInterface):
         return [torch.eye(op.attrs["n"],
                           dtype=op.attrs.get("dtype", torch.float32),
                           device=ctx.device)]
+
+
+# ---------------------------------------------------------------------------
+# bulk unary / activation families (reference graph/ops/<Name>.cc each):
+# Abs/Ceil/Floor/Round/Sin/Cos/Reciprocal and the remaining activations
+# LeakyRelu/Mish/Elu/Hardshrink/Hardsigmoid/Hardswish/Hardtanh/Logsigmoid/
+# Softplus/Softshrink.  All route through torch (fused elementwise on GPU);
+# gradients come from the shared autograd-replay op.
+# ---------------------------------------------------------------------------
+def _unary(name, fn_builder, grad=True):
+    cls = type(f"{name}Op", (_AutogradOp,), {
+        "type": name,
+        "fn": lambda self, attrs, _f=fn_builder: _f(attrs),
+        "grad_mask": [True] if grad else [False],
+    })
+    _register(cls)
+    return cls
+
+
+AbsOp = _unary("Abs", lambda a: torch.abs)
+CeilOp = _unary("Ceil", lambda a: torch.ceil)
+FloorOp = _unary("Floor", lambda a: torch.floor)
+RoundOp = _unary("Round", lambda a: torch.round)
+SinOp = _unary("Sin", lambda a: torch.sin)
+CosOp = _unary("Cos", lambda a: torch.cos)
+ReciprocalOp = _unary("Reciprocal", lambda a: torch.reciprocal)
+LeakyReluOp = _unary(
+    "LeakyRelu",
+    lambda a: lambda x: torch.nn.functional.leaky_relu(
+        x, a.get("alpha", 0.01)))
+MishOp = _unary("Mish", lambda a: torch.nn.functional.mish)
+EluOp = _unary(
+    "Elu", lambda a: lambda x: torch.nn.functional.elu(x,
+                                                       a.get("alpha", 1.0)))
+HardshrinkOp = _unary(
+    "Hardshrink",
+    lambda a: lambda x: torch.nn.functional.hardshrink(
+        x, a.get("lambd", 0.5)))
+HardsigmoidOp = _unary("Hardsigmoid", lambda a: torch.nn.functional.hardsigmoid)
+HardswishOp = _unary("Hardswish", lambda a: torch.nn.functional.hardswish)
+HardtanhOp = _unary(
+    "Hardtanh",
+    lambda a: lambda x: torch.nn.functional.hardtanh(
+        x, a.get("min", -1.0), a.get("max", 1.0)))
+LogsigmoidOp = _unary("Logsigmoid", lambda a: torch.nn.functional.logsigmoid)
+SoftplusOp = _unary(
+    "Softplus",
+    lambda a: lambda x: torch.nn.functional.softplus(
+        x, a.get("beta", 1.0)))
+SoftshrinkOp = _unary(
+    "Softshrink",
+    lambda a: lambda x: torch.nn.functional.softshrink(
+        x, a.get("lambd", 0.5)))
+
+
+@_register
+class OuterOp(_AutogradOp):
+    type = "Outer"
+
+    def fn(self, attrs):
+        return torch.outer
+
+
+@_register
+class DotOp(_AutogradOp):
+    type = "Dot"
+
+    def fn(self, attrs):
+        return torch.dot
+
+
+@_register
+class DiagonalOp(_AutogradOp):
+    type = "Diagonal"
+
+    def fn(self, attrs):
+        return lambda x: torch.diagonal(x, attrs.get("offset", 0),
+                                        attrs.get("dim1", 0),
+                                        attrs.get("dim2", 1))

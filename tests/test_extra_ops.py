@@ -128,3 +128,74 @@ def test_manipulation_ops():
         return [xp], ht.roll(xp, 2, 1)
     (out,) = _run(b5, [x])
     assert torch.equal(out, torch.roll(x, 2, 1))
+
+
+def test_bulk_unary_family():
+    """Abs/Ceil/Floor/Round/Sin/Cos/Reciprocal + remaining activations
+    (reference graph/ops/<Name>.cc families) with autograd."""
+    import torch.nn.functional as F
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import (DefineAndRunGraph, pop_graph,
+                                      push_graph)
+    from hetu_amd.graph.ops import api as ht
+    g = DefineAndRunGraph("u")
+    push_graph(g)
+    try:
+        x = ht.placeholder((4, 5), name="x")
+        outs = {"abs": ht.abs_(x), "ceil": ht.ceil(x), "floor": ht.floor(x),
+                "round": ht.round_(x), "sin": ht.sin(x), "cos": ht.cos(x),
+                "recip": ht.reciprocal(x), "lrelu": ht.leaky_relu(x),
+                "mish": ht.mish(x), "elu": ht.elu(x),
+                "hshrink": ht.hardshrink(x), "hsig": ht.hardsigmoid(x),
+                "hswish": ht.hardswish(x), "htanh": ht.hardtanh(x),
+                "lsig": ht.logsigmoid(x), "splus": ht.softplus(x),
+                "sshrink": ht.softshrink(x)}
+        grads = ht.gradients([ht.reduce_sum(outs["mish"])], [x])
+        sp = ht.split(x, 5, dim=1)
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    xd = torch.randn(4, 5).abs() + 0.5
+    keys = list(outs)
+    res = g.run([outs[k] for k in keys] + grads + sp, {x: xd}, ctx=ctx)
+    ref = {"abs": xd.abs(), "ceil": xd.ceil(), "floor": xd.floor(),
+           "round": xd.round(), "sin": xd.sin(), "cos": xd.cos(),
+           "recip": xd.reciprocal(), "lrelu": F.leaky_relu(xd, 0.01),
+           "mish": F.mish(xd), "elu": F.elu(xd),
+           "hshrink": F.hardshrink(xd), "hsig": F.hardsigmoid(xd),
+           "hswish": F.hardswish(xd), "htanh": F.hardtanh(xd),
+           "lsig": F.logsigmoid(xd), "splus": F.softplus(xd),
+           "sshrink": F.softshrink(xd)}
+    for i, k in enumerate(keys):
+        assert torch.allclose(res[i], ref[k], atol=1e-6), k
+    xr = xd.clone().requires_grad_(True)
+    F.mish(xr).sum().backward()
+    assert torch.allclose(res[len(keys)], xr.grad, atol=1e-5)
+    assert torch.allclose(res[len(keys) + 1], xd[:, :1])
+
+
+def test_outer_dot_diagonal():
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import (DefineAndRunGraph, pop_graph,
+                                      push_graph)
+    from hetu_amd.graph.ops import api as ht
+    g = DefineAndRunGraph("odd")
+    push_graph(g)
+    try:
+        a = ht.placeholder((4,), name="a")
+        b = ht.placeholder((4,), name="b")
+        o, dt = ht.outer(a, b), ht.dot(a, b)
+        m = ht.placeholder((3, 3), name="m")
+        d = ht.diagonal(m)
+        go = ht.gradients([ht.reduce_sum(o)], [a])
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    av, bv, mv = torch.randn(4), torch.randn(4), torch.randn(3, 3)
+    ro, rdt, rd, rga = g.run([o, dt, d, go[0]],
+                             {a: av, b: bv, m: mv}, ctx=ctx)
+    assert torch.allclose(ro, torch.outer(av, bv))
+    assert torch.allclose(rdt, torch.dot(av, bv))
+    assert torch.allclose(rd, mv.diagonal())
+    assert torch.allclose(rga, bv.sum().expand(4), atol=1e-6) or \
+        torch.allclose(rga, torch.full((4,), bv.sum().item()), atol=1e-6)
