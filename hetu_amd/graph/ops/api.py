@@ -178,18 +178,48 @@ def reduce_max(a: Tensor, dim=None, keepdim=False) -> Tensor:
 
 # ---- GEMM ------------------------------------------------------------------
 
+# ---- autocast (reference graph/autocast/autocast.cc:39-92: dtype
+# inference inserts DataTransferOp casts around compute ops; here the
+# casts are inserted at op-build time while the context is active) --------
+_AUTOCAST_STACK: List[torch.dtype] = []
+
+
+class autocast:
+    """with ht.autocast(torch.bfloat16): matmul/linear/bmm/attention
+    inputs are cast to the target dtype (fp32 params keep a cast edge, so
+    the grads flow back in fp32 — AMP semantics)."""
+
+    def __init__(self, dtype=torch.bfloat16):
+        self.dtype = dtype
+
+    def __enter__(self):
+        _AUTOCAST_STACK.append(self.dtype)
+        return self
+
+    def __exit__(self, *a):
+        _AUTOCAST_STACK.pop()
+
+
+def _ac(t: Tensor) -> Tensor:
+    if _AUTOCAST_STACK and t.dtype in (torch.float32, torch.float16,
+                                       torch.bfloat16) \
+            and t.dtype != _AUTOCAST_STACK[-1]:
+        return cast(t, _AUTOCAST_STACK[-1])
+    return t
+
+
 def matmul(a: Tensor, b: Tensor, trans_a=False, trans_b=False) -> Tensor:
-    return _cg().make_op(B.MatMul2DOp(), [a, b],
+    return _cg().make_op(B.MatMul2DOp(), [_ac(a), _ac(b)],
                          {"trans_a": trans_a, "trans_b": trans_b}).output()
 
 
 def linear(x: Tensor, w: Tensor, bias: Optional[Tensor] = None) -> Tensor:
-    ins = [x, w] + ([bias] if bias is not None else [])
+    ins = [_ac(x), _ac(w)] + ([_ac(bias)] if bias is not None else [])
     return _cg().make_op(B.LinearOp(), ins, {}).output()
 
 
 def bmm(a: Tensor, b: Tensor) -> Tensor:
-    return _cg().make_op(B.BatchMatMulOp(), [a, b], {}).output()
+    return _cg().make_op(B.BatchMatMulOp(), [_ac(a), _ac(b)], {}).output()
 
 
 # ---- nn --------------------------------------------------------------------
@@ -250,6 +280,7 @@ def rotary(x, cos, sin):
 
 
 def attention(q, k, v, causal=True, scale=None):
+    q, k, v = _ac(q), _ac(k), _ac(v)
     return _cg().make_op(N.AttentionOp(), [q, k, v],
                          {"causal": causal, "scale": scale}).output(0)
 

@@ -94,3 +94,29 @@ def test_op_profiler():
     ms = MemorySnapshots()
     ms.mark("step0")
     assert "step0" in ms.report()
+
+
+def test_autocast_context():
+    """ht.autocast inserts bf16 casts on compute ops; grads return fp32
+    (reference autocast.cc DataTransferOp insertion)."""
+    import torch
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import (DefineAndRunGraph, pop_graph,
+                                      push_graph)
+    from hetu_amd.graph.ops import api as ht
+    g = DefineAndRunGraph("ac")
+    push_graph(g)
+    try:
+        x = ht.placeholder((4, 8), name="x")
+        w = ht.variable(torch.randn(6, 8) * 0.1, name="w")
+        with ht.autocast(torch.bfloat16):
+            y = ht.linear(x, w)
+        loss = ht.reduce_sum(ht.mul(y, y))
+        gs = ht.gradients([loss], [w])
+    finally:
+        pop_graph()
+    assert y.dtype == torch.bfloat16
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    yl, gw = g.run([y, gs[0]], {x: torch.randn(4, 8)}, ctx=ctx)
+    assert yl.dtype == torch.bfloat16
+    assert gw.dtype == torch.float32
