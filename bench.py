@@ -18,7 +18,17 @@ import re
 import sys
 import time
 
-sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+_REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, _REPO)
+
+# hipBLASLt algorithm pinning: if a pre-tuned TunableOp table is committed
+# under tunableop/, use it (tuning off -> zero runtime cost, same math)
+_TUNED = os.path.join(_REPO, "tunableop", "tunableop_results0.csv")
+if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
+    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = os.path.join(
+        _REPO, "tunableop", "tunableop_results%d.csv")
 
 import torch  # noqa: E402
 
