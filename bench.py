@@ -60,7 +60,7 @@ def main():
     ap.add_argument("--micro-batch", type=int, default=0,
                     help="0 = from strategy search")
     ap.add_argument("--global-batch", type=int, default=0,
-                    help="0 = 8 per GPU")
+                    help="0 = 16 per GPU")
     ap.add_argument("--seq-len", type=int, default=2048)
     ap.add_argument("--parallel", default="auto")
     ap.add_argument("--capture", default="auto")
@@ -89,7 +89,7 @@ def main():
 
     cfg = GPT_CONFIGS[args.model]
     S = args.seq_len
-    global_batch = args.global_batch or 8 * ws
+    global_batch = args.global_batch or 16 * ws
 
     # ---- strategy ---------------------------------------------------------
     if args.parallel == "auto":
