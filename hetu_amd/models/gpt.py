@@ -39,6 +39,11 @@ class GPTConfig:
     dropout: float = 0.0
     init_std: float = 0.02
     tie_embeddings: bool = False
+    # MoE (HetuMoE parity): >0 experts switches the MLP to a gated
+    # expert FFN with EP over the device group + (hierarchical) all-to-all
+    moe_experts: int = 0
+    moe_k: int = 2
+    moe_capacity: float = 1.25
 
 
 GPT_CONFIGS = {
@@ -48,6 +53,10 @@ GPT_CONFIGS = {
                          ffn_hidden=16384, vocab=50304, max_seq=2048),
     "gpt3-13b": GPTConfig(n_layer=40, n_head=40, n_kv_head=40, hidden=5120,
                           ffn_hidden=20480, vocab=50304, max_seq=2048),
+    # BASELINE config 4: GPT-MoE 8 x 1.3B experts (HetuMoE hierarchical a2a)
+    "gpt-moe-8x1.3b": GPTConfig(n_layer=24, n_head=16, n_kv_head=16,
+                                hidden=2048, ffn_hidden=8192, vocab=50304,
+                                max_seq=2048, moe_experts=8, moe_k=2),
 }
 
 
@@ -112,6 +121,30 @@ class GPTMLP(Module):
         return self.wproj(ht.gelu(self.wfc(x)))
 
 
+class GPTMoEMLP(Module):
+    """MoE expert FFN in place of the dense MLP (HetuMoE moe_layer.py
+    parity): tokens flatten to [B*S, h], dispatch over the EP group."""
+
+    def __init__(self, cfg: GPTConfig, spec: ParallelSpec, layer_idx: int,
+                 dtype):
+        super().__init__()
+        from ..nn.moe import MoEMLP
+        self.spec = spec
+        self.moe = MoEMLP(cfg.hidden, cfg.ffn_hidden, cfg.moe_experts,
+                          spec=spec if spec.num_devices > 1 else None,
+                          k=cfg.moe_k, capacity_factor=cfg.moe_capacity,
+                          dtype=dtype, name=f"h{layer_idx}.moe")
+        self.hidden = cfg.hidden
+
+    def forward(self, x):
+        B, S = x.shape[0], x.shape[1]
+        flat = ht.reshape(x, (B * S, self.hidden),
+                          ds=self.spec.ds_tokens(0))
+        y = self.moe(flat)
+        return ht.reshape(y, (B, S, self.hidden),
+                          ds=self.spec.ds_activation(0))
+
+
 class GPTBlock(Module):
     def __init__(self, cfg, spec, layer_idx, dtype):
         super().__init__()
@@ -120,7 +153,10 @@ class GPTBlock(Module):
         self.attn = GPTAttention(cfg, spec, layer_idx, dtype)
         self.ln2 = ParallelLayerNorm(cfg.hidden, spec, 1e-5, dtype,
                                      name=f"h{layer_idx}.ln2")
-        self.mlp = GPTMLP(cfg, spec, layer_idx, dtype)
+        if cfg.moe_experts > 0:
+            self.mlp = GPTMoEMLP(cfg, spec, layer_idx, dtype)
+        else:
+            self.mlp = GPTMLP(cfg, spec, layer_idx, dtype)
 
     def forward(self, x, B, S):
         x = ht.add(x, self.attn(self.ln1(x), B, S))

@@ -174,3 +174,24 @@ def test_moe_gate_types_train(gate_type):
     assert losses[-1] < losses[0], (gate_type, losses)
     if gate_type in ("hash", "random"):
         assert moe.gate is None          # no gate params for static routing
+
+
+def test_gpt_moe_model_trains():
+    """GPT-MoE (BASELINE config 4 family): dense trunk + expert MLP."""
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
+    cfg = GPTConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
+                    ffn_hidden=128, vocab=312, max_seq=16, moe_experts=4,
+                    moe_k=2)
+    g, h = build_gpt_train_graph(cfg, micro_batch=2, seq_len=16,
+                                 dtype=torch.float32, lr=1e-3)
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    torch.manual_seed(0)
+    ids = torch.randint(0, 312, (2, 16))
+    lab = torch.randint(0, 312, (32,))
+    losses = []
+    for _ in range(6):
+        lv, _ = g.run([h["loss"], h["train_op"]],
+                      {h["input_ids"]: ids, h["labels"]: lab}, ctx=ctx)
+        losses.append(float(lv))
+    assert losses[-1] < losses[0] - 0.5, losses
