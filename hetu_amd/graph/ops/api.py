@@ -279,6 +279,17 @@ def rotary(x, cos, sin):
     return _cg().make_op(N.RotaryOp(), [x, cos, sin], {}).output()
 
 
+def fused_qkv_attention(qkv, n_head, n_kv_head, head_dim, cos=None,
+                        sin=None, causal=True, scale=None):
+    """Fused attention over the qkv GEMM output (see
+    nnops.FusedQKVAttentionOp) -> o [B, S, n_head*head_dim]."""
+    ins = [qkv] + ([cos, sin] if cos is not None else [])
+    return _cg().make_op(N.FusedQKVAttentionOp(), ins,
+                         {"n_head": n_head, "n_kv_head": n_kv_head,
+                          "head_dim": head_dim, "causal": causal,
+                          "scale": scale}).output()
+
+
 def attention(q, k, v, causal=True, scale=None):
     q, k, v = _ac(q), _ac(k), _ac(v)
     return _cg().make_op(N.AttentionOp(), [q, k, v],

@@ -565,3 +565,62 @@ class MSELossGradOp(OpInterface):
         g, x, y = inputs
         n = x.numel()
         return [(g.float() * 2.0 / n * (x.float() - y.float())).to(x.dtype)]
+
+
+class FusedQKVAttentionOp(OpInterface):
+    """Fused attention straight off the column-parallel qkv GEMM output
+    [B, S, (H + 2*Hkv)*D]: optional in-place RoPE on the q|k sections,
+    then flash attention with strided q/k/v views — none of the
+    slice/reshape/transpose copies of the composed path (reference
+    ParallelAttention.cc packs the same way).  NOTE: mutates its qkv
+    input in place (rotation is linear; backward never needs the
+    pre-rotation values), so qkv must have no other consumer.
+    inputs: qkv[, cos, sin]; attrs: n_head, n_kv_head, head_dim, causal.
+    outputs: o [B, S, H*D], lse [B, H, S] fp32."""
+    type = "FusedQKVAttention"
+
+    def infer_meta(self, attrs, inputs):
+        qkv = inputs[0]
+        B, S, C = qkv.shape
+        H = attrs["n_head"]
+        D = attrs["head_dim"]
+        return [TensorMeta((B, S, H * D), qkv.dtype),
+                TensorMeta((B, H, S), torch.float32)]
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        a = op.attrs
+        cos = inputs[1] if len(inputs) > 1 else None
+        sin = inputs[2] if len(inputs) > 2 else None
+        o, lse = F.fused_qkv_attention_fwd(
+            inputs[0], a["n_head"], a["n_kv_head"], a["head_dim"],
+            cos, sin, a.get("causal", True), a.get("scale"))
+        return [o, lse]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        ins = [g[0], op.inputs[0], op.outputs[0], op.outputs[1]] \
+            + list(op.inputs[1:])
+        bwd = _make(gr, FusedQKVAttentionGradOp(), ins, dict(op.attrs),
+                    name="fused_qkv_attn_grad")
+        return [bwd.output(0)] + [None] * (len(op.inputs) - 1)
+
+
+class FusedQKVAttentionGradOp(OpInterface):
+    type = "FusedQKVAttentionGrad"
+
+    def infer_meta(self, attrs, inputs):
+        qkv = inputs[1]
+        return [TensorMeta(qkv.shape, qkv.dtype)]
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        a = op.attrs
+        dout, qkv, out, lse = inputs[:4]
+        cos = inputs[4] if len(inputs) > 4 else None
+        sin = inputs[5] if len(inputs) > 5 else None
+        dqkv = F.fused_qkv_attention_bwd(
+            dout, qkv, out, lse, a["n_head"], a["n_kv_head"],
+            a["head_dim"], cos, sin, a.get("causal", True),
+            a.get("scale"))
+        return [dqkv]

@@ -11,6 +11,7 @@ via build_gpt_pipeline_stage.
 from __future__ import annotations
 
 import dataclasses
+import os
 import math
 from typing import Dict, Optional
 
@@ -83,6 +84,13 @@ class GPTAttention(Module):
         spec = self.spec
         hl, dh = self.h_local, self.dh
         qkv = self.wqkv(x)
+        if spec.cp == 1 and dh == 128 and os.environ.get(
+                "HETU_AMD_FUSED_ATTN", "1") == "1":
+            # fused path: attention reads q/k/v straight out of the qkv
+            # GEMM output; no slice/transpose copies (the CPU fallback
+            # computes the identical reference math)
+            o = ht.fused_qkv_attention(qkv, hl, hl, dh, causal=True)
+            return self.wo(o)
         ds_head = spec._ds({0: spec.dp, 1: spec.cp, 2: spec.tp}, [0, 1, 2])
         q = ht.reshape(ht.slice_(qkv, 2, 0, hl * dh), (B, S, hl, dh),
                        ds=ds_head)

@@ -9,6 +9,7 @@ ride RCCL over xGMI via CommOp deduction.
 from __future__ import annotations
 
 import dataclasses
+import os
 import math
 from typing import Dict, Optional
 
@@ -90,6 +91,13 @@ class LlamaAttention(Module):
         spec, cfg = self.spec, self.cfg
         hl, kl, dh = self.h_local, self.kv_local, self.dh
         qkv = self.wqkv(x)                       # [B,S,(h+2kv)*dh / tp]
+        if spec.cp == 1 and dh == 128 and os.environ.get(
+                "HETU_AMD_FUSED_ATTN", "1") == "1":
+            # fused path: in-place RoPE on the q|k sections + strided
+            # flash attention, zero layout copies
+            o = ht.fused_qkv_attention(qkv, hl, kl, dh, self.cos,
+                                       self.sin, causal=True)
+            return self.wo(o)
         ds_head = spec._ds({0: spec.dp, 1: spec.cp, 2: spec.tp},
                             [0, 1, 2])
         q = ht.reshape(ht.slice_(qkv, 2, 0, hl * dh), (B, S, hl, dh),

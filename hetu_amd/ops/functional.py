@@ -551,3 +551,71 @@ def matmul_4bit(x: torch.Tensor, qweight: torch.Tensor,
     w = dequantize_blockwise(qweight, absmax, qtype, blocksize,
                              shape[0] * shape[1], x.dtype).reshape(shape)
     return x @ w.t()
+
+
+def fused_qkv_attention_fwd(qkv, n_head: int, n_kv_head: int, head_dim: int,
+                            cos=None, sin=None, causal: bool = True,
+                            scale: Optional[float] = None):
+    """Fused attention over the qkv GEMM output [B, S, (H+2Hkv)*D]:
+    in-place RoPE on the q|k sections (optional), then flash attention
+    reading q/k/v as strided views — zero slice/transpose copies.
+    Mutates qkv (rotation is linear, so backward never needs the
+    pre-rotation values).  Returns (o [B,S,H*D], lse [B,H,S])."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(head_dim)
+    if _gpu(qkv):
+        e = ext()
+        if cos is not None:
+            e.rope_qk_inplace(qkv, cos.contiguous(), sin.contiguous(),
+                              n_head + n_kv_head, head_dim, 1)
+        o, lse = e.flash_attn_fwd_qkv(qkv, n_head, n_kv_head, head_dim,
+                                      causal, scale)
+        return o, lse
+    # CPU reference: same math with views
+    B, S, _ = qkv.shape
+    H, Hkv, D = n_head, n_kv_head, head_dim
+    qkv4 = qkv.view(B, S, H + 2 * Hkv, D)
+    if cos is not None:
+        qkv4[:, :, :H + Hkv] = _rope_ref(qkv4[:, :, :H + Hkv], cos, sin,
+                                         False).to(qkv.dtype)
+    q = qkv4[:, :, :H].permute(0, 2, 1, 3)
+    k = qkv4[:, :, H:H + Hkv].permute(0, 2, 1, 3)
+    v = qkv4[:, :, H + Hkv:].permute(0, 2, 1, 3)
+    o, lse = _attn_ref_fwd(q, k, v, causal, scale)
+    return o.permute(0, 2, 1, 3).reshape(B, S, H * D).contiguous(), lse
+
+
+def fused_qkv_attention_bwd(dout, qkv, out, lse, n_head: int,
+                            n_kv_head: int, head_dim: int, cos=None,
+                            sin=None, causal: bool = True,
+                            scale: Optional[float] = None):
+    """Backward of fused_qkv_attention: returns dqkv [B,S,(H+2Hkv)*D]
+    (RoPE backward applied in place on the dq|dk sections)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(head_dim)
+    if _gpu(qkv):
+        e = ext()
+        dqkv = e.flash_attn_bwd_qkv(dout.contiguous(), qkv,
+                                    out.contiguous(), lse.contiguous(),
+                                    n_head, n_kv_head, head_dim, causal,
+                                    scale)
+        if cos is not None:
+            e.rope_qk_inplace(dqkv, cos.contiguous(), sin.contiguous(),
+                              n_head + n_kv_head, head_dim, -1)
+        return dqkv
+    B, S, _ = qkv.shape
+    H, Hkv, D = n_head, n_kv_head, head_dim
+    qkv4 = qkv.view(B, S, H + 2 * Hkv, D)   # already rotated
+    q = qkv4[:, :, :H].permute(0, 2, 1, 3)
+    k = qkv4[:, :, H:H + Hkv].permute(0, 2, 1, 3)
+    v = qkv4[:, :, H + Hkv:].permute(0, 2, 1, 3)
+    do4 = dout.view(B, S, H, D).permute(0, 2, 1, 3)
+    o4 = out.view(B, S, H, D).permute(0, 2, 1, 3)
+    dq, dk, dv = _attn_ref_bwd(do4, q, k, v, o4, lse, causal, scale)
+    dqkv = torch.cat([dq, dk, dv], dim=1).permute(0, 2, 1, 3) \
+        .reshape(B, S, (H + 2 * Hkv) * D).contiguous()
+    if cos is not None:
+        d4 = dqkv.view(B, S, H + 2 * Hkv, D)
+        d4[:, :, :H + Hkv] = _rope_ref(d4[:, :, :H + Hkv], cos, sin,
+                                       True).to(dqkv.dtype)
+    return dqkv

@@ -546,7 +546,8 @@ __global__ __launch_bounds__(THREADS) void fa_bwd_kernel(
 bool fa2_fwd_supported(int D, int S);
 void fa2_fwd_launch(const void* q, const void* k, const void* v, void* o,
                     float* lse, int B, int H, int Hkv, int S, int Skv,
-                    float scale, bool causal, hipStream_t stream);
+                    float scale, bool causal, hipStream_t stream,
+                    FaStrides sq, FaStrides skv, FaStrides so);
 
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
@@ -565,9 +566,12 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
     return e && e[0] == '1';
   }();
   if (!force_v1 && fa2_fwd_supported(D, S)) {
+    FaStrides sq{(long long)H * S * D, (long long)S * D, (long long)D};
+    FaStrides skv{(long long)Hkv * Skv * D, (long long)Skv * D,
+                  (long long)D};
     fa2_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                    lse.data_ptr<float>(), B, H, Hkv, S, Skv, (float)scale,
-                   causal, stream);
+                   causal, stream, sq, skv, sq);
     return {o, lse};
   }
   dim3 grid((S + BM - 1) / BM, B * H);

@@ -101,7 +101,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dq_kernel(
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
     bf16* __restrict__ dQ, int B, int H, int Hkv, int S, int Skv,
-    float scale, bool causal) {
+    float scale, bool causal, FaStrides sq, FaStrides skv, FaStrides sdo,
+    FaStrides sdq) {
   static_assert(D == 128);
   constexpr int KB = 64 * 256;     // 16 KiB per rm image
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -122,10 +123,10 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dq_kernel(
   const int hi = lane >> 5;
   const int g1 = (lane >> 4) & 1;
 
-  const bf16* Qb = Q + (int64_t)bh * S * D;
-  const bf16* dOb = dO + (int64_t)bh * S * D;
-  const bf16* Kb = K + ((int64_t)(b * Hkv + hkv)) * Skv * D;
-  const bf16* Vb = V + ((int64_t)(b * Hkv + hkv)) * Skv * D;
+  const bf16* Qb = Q + (int64_t)b * sq.bs + (int64_t)h * sq.hs;
+  const bf16* dOb = dO + (int64_t)b * sdo.bs + (int64_t)h * sdo.hs;
+  const bf16* Kb = K + (int64_t)b * skv.bs + (int64_t)hkv * skv.hs;
+  const bf16* Vb = V + (int64_t)b * skv.bs + (int64_t)hkv * skv.hs;
 
   const int my_q = q0 + wid * 32 + iq;
   const int diag = Skv - S;
@@ -135,8 +136,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dq_kernel(
   // Q (scaled) and dO in registers
   bf16x8 qreg[8], doreg[8];
   {
-    const bf16* qrow = Qb + (int64_t)min(my_q, S - 1) * D;
-    const bf16* drow = dOb + (int64_t)min(my_q, S - 1) * D;
+    const bf16* qrow = Qb + (int64_t)min(my_q, S - 1) * sq.rs;
+    const bf16* drow = dOb + (int64_t)min(my_q, S - 1) * sdo.rs;
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {
       ushort8 uq = *reinterpret_cast<const ushort8*>(qrow + kk * 16 + hi * 8);
@@ -160,7 +161,7 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dq_kernel(
       int pos = (wid * 2 + i) * 1024 + wlane16;                             \
       int krow = pos >> 8;                                                  \
       int kd = ((pos & 255) ^ ((krow & 15) << 4)) >> 1;                     \
-      int64_t roff = (int64_t)min((k0) + krow, Skv - 1) * D + kd;           \
+      int64_t roff = (int64_t)min((k0) + krow, Skv - 1) * skv.rs + kd;      \
       __builtin_amdgcn_global_load_lds(                                     \
           (const __attribute__((address_space(1))) void*)(Kb + roff),       \
           (__attribute__((address_space(3))) void*)(k_lds(buf) + pos),      \
@@ -242,7 +243,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dq_kernel(
 
   // ---- epilogue: dQ[my_q][d] = scale * dq^T[d][my_q] -------------------
   if (my_q < S) {
-    bf16* qrow = dQ + ((int64_t)bh * S + my_q) * D;
+    bf16* qrow = dQ + (int64_t)b * sdq.bs + (int64_t)h * sdq.hs
+               + (int64_t)my_q * sdq.rs;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
 #pragma unroll
@@ -276,7 +278,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
     float* __restrict__ dK32, float* __restrict__ dV32,
     bf16* __restrict__ dK16, bf16* __restrict__ dV16,
-    int B, int H, int Hkv, int S, int Skv, float scale, bool causal) {
+    int B, int H, int Hkv, int S, int Skv, float scale, bool causal,
+    FaStrides sq, FaStrides skv, FaStrides sdo, FaStrides sdkv) {
   static_assert(D == 128);
   constexpr int QB = 32 * 256;     // 8 KiB per rm image
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -299,10 +302,10 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
   const int hi = lane >> 5;
   const int g1 = (lane >> 4) & 1;
 
-  const bf16* Qb = Q + (int64_t)bh * S * D;
-  const bf16* dOb = dO + (int64_t)bh * S * D;
-  const bf16* Kb = K + ((int64_t)(b * Hkv + hkv)) * Skv * D;
-  const bf16* Vb = V + ((int64_t)(b * Hkv + hkv)) * Skv * D;
+  const bf16* Qb = Q + (int64_t)b * sq.bs + (int64_t)h * sq.hs;
+  const bf16* dOb = dO + (int64_t)b * sdo.bs + (int64_t)h * sdo.hs;
+  const bf16* Kb = K + (int64_t)b * skv.bs + (int64_t)hkv * skv.hs;
+  const bf16* Vb = V + (int64_t)b * skv.bs + (int64_t)hkv * skv.hs;
   const float* lse_b = LSE + (int64_t)bh * S;
   const float* del_b = DELTA + (int64_t)bh * S;
 
@@ -317,7 +320,7 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
   // (keeps the VGPR budget at 2 waves/SIMD without spills)
   bf16x8 kreg[8];
   {
-    const bf16* krow = Kb + (int64_t)min(my_key, Skv - 1) * D;
+    const bf16* krow = Kb + (int64_t)min(my_key, Skv - 1) * skv.rs;
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {
       ushort8 uk = *reinterpret_cast<const ushort8*>(krow + kk * 16 + hi * 8);
@@ -334,7 +337,7 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
     int vrow = pos >> 8;
     int vd = ((pos & 255) ^ ((vrow & 15) << 4)) >> 1;
     const bf16* vsrc = Vb
-        + (int64_t)min(kb0 + wid * 32 + vrow, Skv - 1) * D + vd;
+        + (int64_t)min(kb0 + wid * 32 + vrow, Skv - 1) * skv.rs + vd;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)vsrc,
         (__attribute__((address_space(3))) void*)(vw_lds + pos), 16, 0, 0);
@@ -348,13 +351,15 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
     int pos = tid * 16;                                                     \
     int qrow = pos >> 8;                                                    \
     int qd = ((pos & 255) ^ ((qrow & 15) << 4)) >> 1;                       \
-    int64_t roff = (int64_t)min((qt0) + qrow, S - 1) * D + qd;              \
+    int qr_ = min((qt0) + qrow, S - 1);                                     \
     __builtin_amdgcn_global_load_lds(                                       \
-        (const __attribute__((address_space(1))) void*)(Qb + roff),         \
+        (const __attribute__((address_space(1))) void*)(                    \
+            Qb + (int64_t)qr_ * sq.rs + qd),                                \
         (__attribute__((address_space(3))) void*)(q_lds(buf) + pos),        \
         16, 0, 0);                                                          \
     __builtin_amdgcn_global_load_lds(                                       \
-        (const __attribute__((address_space(1))) void*)(dOb + roff),        \
+        (const __attribute__((address_space(1))) void*)(                    \
+            dOb + (int64_t)qr_ * sdo.rs + qd),                              \
         (__attribute__((address_space(3))) void*)(do_lds(buf) + pos),       \
         16, 0, 0);                                                          \
   } while (0)
@@ -466,14 +471,17 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
     for (int r = 0; r < 16; ++r) {
       int key = kb0 + wid * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       if (key < Skv) {
-        int64_t off = ((int64_t)(b * Hkv + hkv) * Skv + key) * D
-                    + 32 * dt + iq;
         float dkv_ = dk_acc[dt][r] * scale;
         float dvv_ = dv_acc[dt][r];
         if (gqa) {
-          atomicAdd(dK32 + off, dkv_);
-          atomicAdd(dV32 + off, dvv_);
+          // f32 accumulation buffers stay BHSD-contiguous
+          int64_t off32 = ((int64_t)(b * Hkv + hkv) * Skv + key) * D
+                        + 32 * dt + iq;
+          atomicAdd(dK32 + off32, dkv_);
+          atomicAdd(dV32 + off32, dvv_);
         } else {
+          int64_t off = (int64_t)b * sdkv.bs + (int64_t)hkv * sdkv.hs
+                      + (int64_t)key * sdkv.rs + 32 * dt + iq;
           dK16[off] = (bf16)dkv_;
           dV16[off] = (bf16)dvv_;
         }
@@ -487,7 +495,9 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
 __global__ void fa2_delta_kernel(const bf16* __restrict__ dO,
                                  const bf16* __restrict__ O,
                                  float* __restrict__ delta,
-                                 int64_t rows, int D) {
+                                 int64_t rows, int D, int H, int S,
+                                 FaStrides so) {
+  // delta rows are LSE-ordered (b, h, s); dO/O may be BHSD or BS[HD]
   // D == 128 fast path: 4 lanes per row (32 elems each, 16B vector
   // loads), quad shuffle reduction — pure bandwidth (the old one-block-
   // per-row version left 240/256 lanes idle: 580us vs the ~65us bound).
@@ -497,8 +507,12 @@ __global__ void fa2_delta_kernel(const bf16* __restrict__ dO,
     const int q = tid & 3;
     const int64_t rstride = ((int64_t)gridDim.x * blockDim.x) >> 2;
     for (; row < rows; row += rstride) {
-      const bf16* a = dO + row * 128 + q * 32;
-      const bf16* bb = O + row * 128 + q * 32;
+      const int64_t b = row / ((int64_t)H * S);
+      const int64_t hh = (row / S) % H;
+      const int64_t ss = row % S;
+      const int64_t base = b * so.bs + hh * so.hs + ss * so.rs + q * 32;
+      const bf16* a = dO + base;
+      const bf16* bb = O + base;
       float s = 0.f;
 #pragma unroll
       for (int i = 0; i < 32; i += 8) {
@@ -517,8 +531,12 @@ __global__ void fa2_delta_kernel(const bf16* __restrict__ dO,
   // generic fallback: one lane per row, scalar loads
   for (int64_t row = tid; row < rows;
        row += (int64_t)gridDim.x * blockDim.x) {
-    const bf16* a = dO + row * D;
-    const bf16* bb = O + row * D;
+    const int64_t b = row / ((int64_t)H * S);
+    const int64_t hh = (row / S) % H;
+    const int64_t ss = row % S;
+    const int64_t base = b * so.bs + hh * so.hs + ss * so.rs;
+    const bf16* a = dO + base;
+    const bf16* bb = O + base;
     float s = 0.f;
     for (int i = 0; i < D; ++i) s += (float)a[i] * (float)bb[i];
     delta[row] = s;
@@ -529,21 +547,49 @@ __global__ void fa2_delta_kernel(const bf16* __restrict__ dO,
 
 bool fa2_bwd_supported(int D, int S) { return D == 128; }
 
+namespace {
+// shared core over arbitrary layouts
+std::vector<torch::Tensor> fa2_bwd_core(
+    const bf16* doutp, const bf16* qp, const bf16* kp, const bf16* vp,
+    const bf16* outp, torch::Tensor lse, int B, int H, int Hkv, int S,
+    int Skv, int D, bool causal, float scale, FaStrides sq, FaStrides skv,
+    FaStrides sdo, bf16* dqp, bf16* dk16p, bf16* dv16p, float* dk32p,
+    float* dv32p, FaStrides sdq, FaStrides sdkv, bool gqa) {
+  auto stream = hetu_current_stream();
+  auto delta = torch::empty({B, H, S}, lse.options());
+  {
+    int64_t rows = (int64_t)B * H * S;
+    int grid = (int)std::min<int64_t>((rows * 4 + 255) / 256, 16384);
+    hipLaunchKernelGGL(fa2_delta_kernel, dim3(grid), dim3(256), 0, stream,
+                       doutp, outp, delta.data_ptr<float>(), rows, D, H, S,
+                       sdo);
+  }
+  {
+    dim3 grid(B * H, (S + 255) / 256);
+    size_t lds = 4 * (size_t)64 * 256;          // 64 KiB
+    hipLaunchKernelGGL(fa2_bwd_dq_kernel<128>, grid, dim3(THREADS), lds,
+                       stream, qp, kp, vp, doutp, lse.data_ptr<float>(),
+                       delta.data_ptr<float>(), dqp,
+                       B, H, Hkv, S, Skv, scale, causal, sq, skv, sdo, sdq);
+  }
+  {
+    dim3 grid(B * H, (Skv + 255) / 256);
+    size_t lds = 4 * (size_t)32 * 256 + 8 * 8192 + 8 * 4608;
+    hipLaunchKernelGGL(fa2_bwd_dkv_kernel<128>, grid, dim3(THREADS), lds,
+                       stream, qp, kp, vp, doutp, lse.data_ptr<float>(),
+                       delta.data_ptr<float>(), dk32p, dv32p, dk16p, dv16p,
+                       B, H, Hkv, S, Skv, scale, causal, sq, skv, sdo,
+                       sdkv);
+  }
+  return {};
+}
+}  // namespace
+
 std::vector<torch::Tensor> fa2_bwd_launch(
     torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor out, torch::Tensor lse, bool causal, double scale) {
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   const int Hkv = k.size(1), Skv = k.size(2);
-  auto stream = hetu_current_stream();
-  auto delta = torch::empty({B, H, S}, q.options().dtype(at::kFloat));
-  {
-    int64_t rows = (int64_t)B * H * S;
-    int grid = (int)std::min<int64_t>((rows * 4 + 255) / 256, 16384);
-    hipLaunchKernelGGL(fa2_delta_kernel, dim3(grid), dim3(256), 0, stream,
-                       (const bf16*)dout.data_ptr(),
-                       (const bf16*)out.data_ptr(), delta.data_ptr<float>(),
-                       rows, D);
-  }
   auto dq = torch::empty_like(q);
   const bool gqa = (H != Hkv);
   torch::Tensor dk16, dv16, dk32, dv32;
@@ -554,32 +600,64 @@ std::vector<torch::Tensor> fa2_bwd_launch(
     dk16 = torch::empty_like(k);
     dv16 = torch::empty_like(v);
   }
-  {
-    dim3 grid(B * H, (S + 255) / 256);
-    size_t lds = 4 * (size_t)64 * 256;          // 64 KiB
-    hipLaunchKernelGGL(fa2_bwd_dq_kernel<128>, grid, dim3(THREADS), lds,
-                       stream, (const bf16*)q.data_ptr(),
-                       (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
-                       (const bf16*)dout.data_ptr(), lse.data_ptr<float>(),
-                       delta.data_ptr<float>(), (bf16*)dq.data_ptr(),
-                       B, H, Hkv, S, Skv, (float)scale, causal);
-  }
-  {
-    dim3 grid(B * H, (Skv + 255) / 256);
-    size_t lds = 4 * (size_t)32 * 256 + 8 * 8192 + 8 * 4608;
-    hipLaunchKernelGGL(fa2_bwd_dkv_kernel<128>, grid, dim3(THREADS), lds,
-                       stream, (const bf16*)q.data_ptr(),
-                       (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
-                       (const bf16*)dout.data_ptr(), lse.data_ptr<float>(),
-                       delta.data_ptr<float>(),
-                       gqa ? dk32.data_ptr<float>() : nullptr,
-                       gqa ? dv32.data_ptr<float>() : nullptr,
-                       gqa ? nullptr : (bf16*)dk16.data_ptr(),
-                       gqa ? nullptr : (bf16*)dv16.data_ptr(),
-                       B, H, Hkv, S, Skv, (float)scale, causal);
-  }
+  FaStrides sq{(long long)H * S * D, (long long)S * D, (long long)D};
+  FaStrides skv{(long long)Hkv * Skv * D, (long long)Skv * D,
+                (long long)D};
+  fa2_bwd_core((const bf16*)dout.data_ptr(), (const bf16*)q.data_ptr(),
+               (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
+               (const bf16*)out.data_ptr(), lse, B, H, Hkv, S, Skv, D,
+               causal, (float)scale, sq, skv, sq,
+               (bf16*)dq.data_ptr(),
+               gqa ? nullptr : (bf16*)dk16.data_ptr(),
+               gqa ? nullptr : (bf16*)dv16.data_ptr(),
+               gqa ? dk32.data_ptr<float>() : nullptr,
+               gqa ? dv32.data_ptr<float>() : nullptr, sq, skv, gqa);
   if (gqa) {
     return {dq, dk32.to(k.scalar_type()), dv32.to(v.scalar_type())};
   }
   return {dq, dk16, dv16};
+}
+
+// Fused-QKV backward: dout/out [B,S,H*D], qkv [B,S,(H+2Hkv)*D] ->
+// dqkv [B,S,(H+2Hkv)*D] written in place by the kernels (no slice or
+// transpose copies).
+torch::Tensor flash_attn_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
+                                 torch::Tensor out, torch::Tensor lse,
+                                 int64_t H, int64_t Hkv, int64_t D,
+                                 bool causal, double scale) {
+  TORCH_CHECK(qkv.dim() == 3 && qkv.is_contiguous());
+  TORCH_CHECK(dout.is_contiguous() && out.is_contiguous());
+  const int B = qkv.size(0), S = qkv.size(1);
+  const int64_t C = (H + 2 * Hkv) * D;
+  const bool gqa = (H != Hkv);
+  auto dqkv = torch::empty_like(qkv);
+  torch::Tensor dk32, dv32;
+  float *dk32p = nullptr, *dv32p = nullptr;
+  if (gqa) {
+    dk32 = torch::zeros({B, Hkv, (int64_t)S, D},
+                        qkv.options().dtype(at::kFloat));
+    dv32 = torch::zeros_like(dk32);
+    dk32p = dk32.data_ptr<float>();
+    dv32p = dv32.data_ptr<float>();
+  }
+  FaStrides sqkv{(long long)S * C, (long long)D, (long long)C};
+  FaStrides so{(long long)S * H * D, (long long)D, (long long)H * D};
+  const bf16* base = (const bf16*)qkv.data_ptr();
+  bf16* dbase = (bf16*)dqkv.data_ptr();
+  fa2_bwd_core((const bf16*)dout.data_ptr(), base, base + H * D,
+               base + (H + Hkv) * D, (const bf16*)out.data_ptr(), lse,
+               B, H, Hkv, S, S, D, causal, (float)scale, sqkv, sqkv, so,
+               dbase,
+               gqa ? nullptr : dbase + H * D,
+               gqa ? nullptr : dbase + (H + Hkv) * D,
+               dk32p, dv32p, sqkv, sqkv, gqa);
+  if (gqa) {
+    // scatter the fp32 accumulators into the k/v sections of dqkv
+    auto dkv_view = dqkv.view({B, (int64_t)S, H + 2 * Hkv, D});
+    dkv_view.narrow(2, H, Hkv).copy_(
+        dk32.permute({0, 2, 1, 3}).to(qkv.scalar_type()));
+    dkv_view.narrow(2, H + Hkv, Hkv).copy_(
+        dv32.permute({0, 2, 1, 3}).to(qkv.scalar_type()));
+  }
+  return dqkv;
 }

@@ -48,7 +48,7 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, bf16* __restrict__ O,
     float* __restrict__ LSE, int B, int H, int Hkv, int S, int Skv,
-    float scale, bool causal) {
+    float scale, bool causal, FaStrides sq, FaStrides skv, FaStrides so) {
   static_assert(D == 128, "fa2 fwd: D=128 only");
   constexpr int KBYTES = BN * D * 2;          // 16 KiB
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -70,9 +70,9 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
   const int iq = lane & 31;      // this lane's q row within the wave block
   const int hi = lane >> 5;
 
-  const bf16* Qb = Q + ((int64_t)bh * S) * D;
-  const bf16* Kb = K + ((int64_t)(b * Hkv + hkv)) * Skv * D;
-  const bf16* Vb = V + ((int64_t)(b * Hkv + hkv)) * Skv * D;
+  const bf16* Qb = Q + (int64_t)b * sq.bs + (int64_t)h * sq.hs;
+  const bf16* Kb = K + (int64_t)b * skv.bs + (int64_t)hkv * skv.hs;
+  const bf16* Vb = V + (int64_t)b * skv.bs + (int64_t)hkv * skv.hs;
 
   const int my_q = q0 + wid * 32 + iq;          // global q row (this lane)
   const int diag = Skv - S;                     // causal offset
@@ -81,7 +81,7 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
   // lane holds Q[my_q][16*kk + 8*hi .. +8) for kk = 0..7
   bf16x8 qreg[8];
   {
-    const bf16* qrow = Qb + (int64_t)min(my_q, S - 1) * D;
+    const bf16* qrow = Qb + (int64_t)min(my_q, S - 1) * sq.rs;
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {
       ushort8 u = *reinterpret_cast<const ushort8*>(qrow + kk * 16 + hi * 8);
@@ -110,7 +110,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
       int pos = (wid * NGLDS + i) * 1024 + wlane16;                             \
       int krow = pos >> 8;                                                  \
       int kd = ((pos & 255) ^ ((krow & 15) << 4)) >> 1;                     \
-      const bf16* ksrc = Kb + (int64_t)min((k0) + krow, Skv - 1) * D + kd;  \
+      const bf16* ksrc =                                                    \
+          Kb + (int64_t)min((k0) + krow, Skv - 1) * skv.rs + kd;            \
       __builtin_amdgcn_global_load_lds(                                     \
           (const __attribute__((address_space(1))) void*)ksrc,              \
           (__attribute__((address_space(3))) void*)(k_lds(buf) + pos),      \
@@ -118,7 +119,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
       int st = pos >> 10;                                                   \
       int vkey = (st >> 3) * 32 + ((pos >> 5) & 31);                        \
       int vd = (st & 7) * 16 + ((pos >> 4) & 1) * 8;                        \
-      const bf16* vsrc = Vb + (int64_t)min((k0) + vkey, Skv - 1) * D + vd;  \
+      const bf16* vsrc =                                                    \
+          Vb + (int64_t)min((k0) + vkey, Skv - 1) * skv.rs + vd;            \
       __builtin_amdgcn_global_load_lds(                                     \
           (const __attribute__((address_space(1))) void*)vsrc,              \
           (__attribute__((address_space(3))) void*)(v_lds(buf) + pos),      \
@@ -279,7 +281,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
   // ---- epilogue -------------------------------------------------------
   if (my_q < S) {
     float inv = (l_i > 0.f) ? 1.f / l_i : 0.f;
-    bf16* orow = O + ((int64_t)bh * S + my_q) * D;
+    bf16* orow = O + (int64_t)b * so.bs + (int64_t)h * so.hs
+                 + (int64_t)my_q * so.rs;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
 #pragma unroll
@@ -312,10 +315,36 @@ bool fa2_fwd_supported(int D, int S) { return D == 128; }
 
 void fa2_fwd_launch(const void* q, const void* k, const void* v, void* o,
                     float* lse, int B, int H, int Hkv, int S, int Skv,
-                    float scale, bool causal, hipStream_t stream) {
+                    float scale, bool causal, hipStream_t stream,
+                    FaStrides sq, FaStrides skv, FaStrides so) {
   dim3 grid(B * H, (S + BM - 1) / BM);
   size_t lds = 2 * NBUF * (size_t)BN * 128 * 2;   // 96 KiB
   hipLaunchKernelGGL(fa2_fwd_kernel<128>, grid, dim3(THREADS), lds, stream,
                      (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                     (bf16*)o, lse, B, H, Hkv, S, Skv, scale, causal);
+                     (bf16*)o, lse, B, H, Hkv, S, Skv, scale, causal,
+                     sq, skv, so);
+}
+
+// Fused-QKV entry: qkv [B, S, (H+2*Hkv)*D] straight from the column-
+// parallel GEMM (q|k|v interleaved per token); no slice/transpose copies.
+// Returns (o [B, S, H*D], lse [B, H, S]).
+std::vector<torch::Tensor> flash_attn_fwd_qkv(torch::Tensor qkv, int64_t H,
+                                              int64_t Hkv, int64_t D,
+                                              bool causal, double scale) {
+  TORCH_CHECK(qkv.dim() == 3 && qkv.is_contiguous());
+  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16, "fa2 qkv: bf16 only");
+  TORCH_CHECK(D == 128, "fa2 qkv: D=128 only");
+  const int B = qkv.size(0), S = qkv.size(1);
+  const int64_t C = (H + 2 * Hkv) * D;
+  TORCH_CHECK(qkv.size(2) == C, "qkv width mismatch");
+  auto o = torch::empty({B, (int64_t)S, H * D}, qkv.options());
+  auto lse = torch::empty({B, H, S}, qkv.options().dtype(at::kFloat));
+  auto stream = hetu_current_stream();
+  const bf16* base = (const bf16*)qkv.data_ptr();
+  FaStrides sqkv{(long long)S * C, (long long)D, (long long)C};
+  FaStrides so{(long long)S * H * D, (long long)D, (long long)H * D};
+  fa2_fwd_launch(base, base + H * D, base + (H + Hkv) * D, o.data_ptr(),
+                 lse.data_ptr<float>(), B, H, Hkv, S, S, (float)scale,
+                 causal, stream, sqkv, sqkv, so);
+  return {o, lse};
 }

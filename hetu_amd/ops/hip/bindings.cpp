@@ -76,6 +76,17 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout,
                                           torch::Tensor v, torch::Tensor out,
                                           torch::Tensor lse, bool causal,
                                           double scale);
+// fused-qkv fast path (attention_v2 / attention_bwd_v2)
+std::vector<torch::Tensor> flash_attn_fwd_qkv(torch::Tensor qkv, int64_t H,
+                                              int64_t Hkv, int64_t D,
+                                              bool causal, double scale);
+torch::Tensor flash_attn_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
+                                 torch::Tensor out, torch::Tensor lse,
+                                 int64_t H, int64_t Hkv, int64_t D,
+                                 bool causal, double scale);
+// rope.hip
+void rope_qk_inplace(torch::Tensor qkv, torch::Tensor cs, torch::Tensor sn,
+                     int64_t n_rot, int64_t D, int64_t sign);
 
 // embed_cache.cpp
 void register_embed_cache(pybind11::module& m);
@@ -113,4 +124,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("galvatron_dp", &galvatron_dp);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("flash_attn_fwd_qkv", &flash_attn_fwd_qkv);
+  m.def("flash_attn_bwd_qkv", &flash_attn_bwd_qkv);
+  m.def("rope_qk_inplace", &rope_qk_inplace);
 }

@@ -14,6 +14,13 @@
 
 typedef unsigned int u32;
 typedef unsigned long long u64;
+
+// per-tensor addressing for attention kernels: element strides for batch,
+// head and token.  BHSD contiguous: {H*S*D, S*D, D}; fused-qkv / BS[HD]
+// views: {S*row, D, row} with row = (H + 2*Hkv) * D.
+struct FaStrides {
+  long long bs, hs, rs;
+};
 typedef __hip_bfloat16 bf16;
 
 // ---- vector types ---------------------------------------------------------

@@ -247,3 +247,43 @@ class TestQuant:
         y = F.dequantize_blockwise(q, am, qtype, 64, 8192)
         yr = F.dequantize_blockwise(qr, amr, qtype, 64, 8192)
         _close(y, yr, rtol=1e-3, atol=2e-2, what=f"{qtype} dequant")
+
+
+class TestFusedQKVAttention:
+    """GPU fused path (in-place RoPE + strided FA on the qkv buffer) vs
+    the CPU fp32 composed reference."""
+
+    @pytest.mark.parametrize("cfg", [
+        dict(B=2, H=4, Hkv=4, S=512, rope=True, causal=True),
+        dict(B=1, H=8, Hkv=2, S=512, rope=True, causal=True),    # GQA
+        dict(B=1, H=4, Hkv=4, S=300, rope=False, causal=True),   # ragged
+        dict(B=1, H=2, Hkv=2, S=1024, rope=False, causal=False),
+    ])
+    def test_fused_vs_cpu_ref(self, cfg):
+        B, H, Hkv, S = cfg["B"], cfg["H"], cfg["Hkv"], cfg["S"]
+        D = 128
+        torch.manual_seed(0)
+        C = (H + 2 * Hkv) * D
+        qkv = torch.randn(B, S, C, dtype=torch.bfloat16, device=dev())
+        cos = sin = None
+        if cfg["rope"]:
+            t = torch.arange(S, dtype=torch.float32)
+            inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2).float() / D))
+            fr = torch.outer(t, inv)
+            cos, sin = fr.cos().to(dev()), fr.sin().to(dev())
+        qkv_cpu = qkv.cpu().float()
+        o, lse = F.fused_qkv_attention_fwd(
+            qkv, H, Hkv, D, cos, sin, cfg["causal"], None)
+        cc = cos.cpu() if cos is not None else None
+        ss = sin.cpu() if sin is not None else None
+        orf, lser = F.fused_qkv_attention_fwd(
+            qkv_cpu, H, Hkv, D, cc, ss, cfg["causal"], None)
+        _close(o, orf, rtol=3e-2, atol=3e-2, what="fused fwd")
+        _close(lse, lser, rtol=1e-2, atol=1e-2, what="fused lse")
+        do = torch.randn_like(o)
+        dqkv = F.fused_qkv_attention_bwd(
+            do, qkv, o, lse, H, Hkv, D, cos, sin, cfg["causal"], None)
+        dref = F.fused_qkv_attention_bwd(
+            do.cpu().float(), qkv_cpu, orf, lser, H, Hkv, D, cc, ss,
+            cfg["causal"], None)
+        _close(dqkv, dref, rtol=6e-2, atol=6e-2, what="fused dqkv")

@@ -229,3 +229,48 @@ class TestEager:
             b = ht.variable(torch.ones(3, 3) * 3, name="b")
             c = ht.mul(a, b)
             assert np.allclose(c.get_data().numpy(), 6.0)
+
+
+def test_fused_qkv_attention_matches_composed():
+    """Fused qkv attention (in-place RoPE + strided FA) must equal the
+    composed slice/rope/transpose/attention path, incl. GQA, dh=128."""
+    import os
+    import subprocess
+    import sys
+    code = r"""
+import os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+def run(fused):
+    os.environ["HETU_AMD_FUSED_ATTN"] = "1" if fused else "0"
+    import hetu_amd.models.llama as L
+    from hetu_amd.engine.runner import prepare_run_context
+    torch.manual_seed(0)
+    cfg = L.LlamaConfig(n_layer=2, n_head=4, n_kv_head=2, hidden=512,
+                        ffn_hidden=256, vocab=128, max_seq=32)
+    g, h = L.build_llama_train_graph(cfg, 2, 32, dtype=torch.float32,
+                                     lr=1e-3)
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    gen = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (2, 32), generator=gen)
+    lab = torch.randint(0, 128, (64,), generator=gen)
+    out = []
+    for _ in range(3):
+        lv, _ = g.run([h["loss"], h["train_op"]],
+                      {h["input_ids"]: ids, h["labels"]: lab}, ctx=ctx)
+        out.append(float(lv))
+    return out
+mode = sys.argv[1]
+print("LOSSES:" + repr(run(mode == "fused")))
+"""
+    import re
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    res = {}
+    for mode in ("fused", "composed"):
+        p = subprocess.run([sys.executable, "-c", code, mode],
+                           env={**os.environ, "HETU_REPO": repo},
+                           capture_output=True, text=True, timeout=300)
+        assert p.returncode == 0, p.stderr
+        m = re.search(r"LOSSES:(\[.*\])", p.stdout)
+        res[mode] = eval(m.group(1))  # noqa: S307
+    assert all(abs(a - b) < 2e-4
+               for a, b in zip(res["fused"], res["composed"])), res
