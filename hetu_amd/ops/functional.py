@@ -475,11 +475,17 @@ def linear(x, w, bias=None, trans_w: bool = True):
             if bias is not None:
                 y = y + bias
             return y.reshape(*x.shape[:-1], N)
+    if bias is not None and trans_w:
+        # F.linear hits the hipBLASLt bias-epilogue path (one GEMM, no
+        # separate broadcast-add kernel over the [M, N] output)
+        return torch.nn.functional.linear(x, w.to(x.dtype),
+                                          bias.to(x.dtype))
     wm = w.t() if trans_w else w
-    y = torch.matmul(x, wm.to(x.dtype))
     if bias is not None:
-        y = y + bias
-    return y
+        xs = x.reshape(-1, x.shape[-1])
+        y = torch.addmm(bias.to(x.dtype), xs, wm.to(x.dtype))
+        return y.reshape(*x.shape[:-1], wm.shape[-1])
+    return torch.matmul(x, wm.to(x.dtype))
 
 
 # ---------------------------------------------------------------------------
