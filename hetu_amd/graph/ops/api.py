@@ -539,3 +539,32 @@ def split(a, sections: int, dim: int = 0):
     assert n % sections == 0, "uneven split"
     step = n // sections
     return [slice_(a, dim, i * step, step) for i in range(sections)]
+
+
+def dropout2d(a, p: float, seed: int = 0, offset: int = 0):
+    from .basics import _make as _mk
+    from .extra import Dropout2dOp
+    return _mk(_cg(), Dropout2dOp(), [a],
+               {"p": p, "seed": seed, "offset": offset},
+               name="dropout2d").output()
+
+
+def bool_(a):
+    from .basics import _make as _mk
+    from .extra import BoolOp
+    return _mk(_cg(), BoolOp(), [a], name="bool").output()
+
+
+def range_mask(a, start, end):
+    from .basics import _make as _mk
+    from .extra import RangeMaskOp
+    return _mk(_cg(), RangeMaskOp(), [a], {"start": start, "end": end},
+               name="range_mask").output()
+
+
+def as_strided(a, size, stride, offset=0):
+    from .basics import _make as _mk
+    from .extra import AsStridedOp
+    return _mk(_cg(), AsStridedOp(), [a],
+               {"size": list(size), "stride": list(stride),
+                "offset": offset}, name="as_strided").output()

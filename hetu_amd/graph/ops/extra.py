@@ -586,3 +586,110 @@ class DiagonalOp(_AutogradOp):
         return lambda x: torch.diagonal(x, attrs.get("offset", 0),
                                         attrs.get("dim1", 0),
                                         attrs.get("dim2", 1))
+
+
+class Dropout2dOp(OpInterface):
+    """Channel dropout for [N, C, ...] (reference graph/ops/Dropout2d):
+    one Bernoulli draw per (n, c), deterministic per (seed, offset)."""
+    type = "Dropout2d"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, inputs[0].dtype),
+                TensorMeta(inputs[0].shape[:2], torch.bool)]
+
+    def _mask(self, op, x, ctx):
+        g = torch.Generator(device="cpu").manual_seed(
+            int(op.attrs["seed"]) + int(op.attrs.get("offset", 0)))
+        keep = torch.rand(x.shape[0], x.shape[1],
+                          generator=g) >= op.attrs["p"]
+        return keep.to(x.device)
+
+    def compute(self, op, inputs, ctx):
+        x = inputs[0]
+        p = op.attrs["p"]
+        if not ctx.training or p <= 0.0:
+            return [x, torch.empty(0, dtype=torch.bool, device=x.device)]
+        keep = self._mask(op, x, ctx)
+        shape = list(keep.shape) + [1] * (x.ndim - 2)
+        y = x * keep.reshape(shape).to(x.dtype) / (1.0 - p)
+        return [y, keep]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, Dropout2dGradOp(), [g[0], op.outputs[1]],
+                      dict(op.attrs)).output()]
+
+
+@_register
+class Dropout2dGradOp(OpInterface):
+    type = "Dropout2dGrad"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        gy, keep = inputs
+        p = op.attrs["p"]
+        if not ctx.training or p <= 0.0 or keep.numel() == 0:
+            return [gy]
+        shape = list(keep.shape) + [1] * (gy.ndim - 2)
+        return [gy * keep.reshape(shape).to(gy.dtype) / (1.0 - p)]
+
+
+class BoolOp(OpInterface):
+    """x != 0 -> bool mask (reference graph/ops/Bool.cc)."""
+    type = "Bool"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, torch.bool)]
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0] != 0]
+
+
+class RangeMaskOp(OpInterface):
+    """1.0 where start <= x < end else 0.0 (reference RangeMask.cc —
+    used for vocab-range masking in TP losses)."""
+    type = "RangeMask"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[0].shape, inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        x = inputs[0]
+        lo, hi = op.attrs["start"], op.attrs["end"]
+        return [((x >= lo) & (x < hi)).to(x.dtype)]
+
+
+class AsStridedOp(OpInterface):
+    """View with explicit size/stride (reference AsStrided.cc); grad
+    scatters back via as_strided on a zero buffer."""
+    type = "AsStrided"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(tuple(attrs["size"]), inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        return [torch.as_strided(inputs[0], op.attrs["size"],
+                                 op.attrs["stride"],
+                                 op.attrs.get("offset", 0)).contiguous()]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, AsStridedGradOp(), [g[0], op.inputs[0]],
+                      dict(op.attrs)).output()]
+
+
+@_register
+class AsStridedGradOp(OpInterface):
+    type = "AsStridedGrad"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[1].shape, inputs[1].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        gy, x = inputs
+        dx = torch.zeros_like(x)
+        dx.as_strided(op.attrs["size"], op.attrs["stride"],
+                      op.attrs.get("offset", 0)).add_(gy)
+        return [dx]

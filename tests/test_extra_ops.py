@@ -199,3 +199,34 @@ def test_outer_dot_diagonal():
     assert torch.allclose(rd, mv.diagonal())
     assert torch.allclose(rga, bv.sum().expand(4), atol=1e-6) or \
         torch.allclose(rga, torch.full((4,), bv.sum().item()), atol=1e-6)
+
+
+def test_dropout2d_bool_rangemask_asstrided():
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import (DefineAndRunGraph, pop_graph,
+                                      push_graph)
+    from hetu_amd.graph.ops import api as ht
+    g = DefineAndRunGraph("x")
+    push_graph(g)
+    try:
+        x = ht.placeholder((2, 3, 4), name="x")
+        d = ht.dropout2d(x, 0.5, seed=7)
+        b = ht.bool_(x)
+        r = ht.range_mask(x, 0.0, 1.0)
+        a = ht.as_strided(x, (2, 3), (12, 4), 0)
+        gs = ht.gradients([ht.reduce_sum(a)], [x])
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    ctx.training = True
+    xd = torch.randn(2, 3, 4)
+    rd, rb, rr, ra, rg = g.run([d, b, r, a, gs[0]], {x: xd}, ctx=ctx)
+    assert rb.dtype == torch.bool
+    assert set(rr.unique().tolist()) <= {0.0, 1.0}
+    assert torch.allclose(ra, torch.as_strided(xd, (2, 3), (12, 4)))
+    exp = torch.zeros_like(xd)
+    exp.as_strided((2, 3), (12, 4)).add_(torch.ones(2, 3))
+    assert torch.allclose(rg, exp)
+    m = (rd != 0)
+    assert all(m[n, c].all() or (~m[n, c]).all()
+               for n in range(2) for c in range(3))  # whole-channel drops
