@@ -262,21 +262,40 @@ class VocabParallelCrossEntropyOp(OpInterface):
         tp, tp_idx, vlocal = _shard_info(lds, vdim, my, op.attrs["vocab"])
         ranks = _tp_ranks(lds, dg, vdim, my)
         ignore = op.attrs.get("ignore_index", -100)
-        lf = logits.float()
-        gmax = lf.max(dim=-1).values
-        if tp > 1 and ctx.comm is not None:
-            gmax = ctx.comm.allreduce(gmax, ranks, op="max")
-        gsum = torch.exp(lf - gmax.unsqueeze(-1)).sum(-1)
         vstart = tp_idx * vlocal
-        in_range = (labels >= vstart) & (labels < vstart + vlocal)
-        safe = torch.where(in_range, labels - vstart,
-                           torch.zeros_like(labels))
-        pred = lf.gather(-1, safe.unsqueeze(-1)).squeeze(-1)
-        pred = pred * in_range.to(pred.dtype)
-        if tp > 1 and ctx.comm is not None:
-            both = torch.stack([gsum, pred], dim=0)
-            both = ctx.comm.allreduce(both, ranks, op="sum")
-            gsum, pred = both[0], both[1]
+        if logits.is_cuda:
+            # kernel path: never materializes an fp32 logits copy
+            # (26 GB at the 7B bench shape)
+            from ...ops import functional as F
+            shape = logits.shape[:-1]
+            gmax, pred = F.vocab_parallel_ce_local_stats(
+                logits, labels.reshape(-1), vstart, vstart + vlocal,
+                ignore if ignore is not None else -100)
+            gmax = gmax.reshape(shape)
+            pred = pred.reshape(shape)
+            if tp > 1 and ctx.comm is not None:
+                gmax = ctx.comm.allreduce(gmax, ranks, op="max")
+            gsum = F.ext().vp_sumexp(logits.contiguous(),
+                                     gmax.reshape(-1)).reshape(shape)
+            if tp > 1 and ctx.comm is not None:
+                both = torch.stack([gsum, pred], dim=0)
+                both = ctx.comm.allreduce(both, ranks, op="sum")
+                gsum, pred = both[0], both[1]
+        else:
+            lf = logits.float()
+            gmax = lf.max(dim=-1).values
+            if tp > 1 and ctx.comm is not None:
+                gmax = ctx.comm.allreduce(gmax, ranks, op="max")
+            gsum = torch.exp(lf - gmax.unsqueeze(-1)).sum(-1)
+            in_range = (labels >= vstart) & (labels < vstart + vlocal)
+            safe = torch.where(in_range, labels - vstart,
+                               torch.zeros_like(labels))
+            pred = lf.gather(-1, safe.unsqueeze(-1)).squeeze(-1)
+            pred = pred * in_range.to(pred.dtype)
+            if tp > 1 and ctx.comm is not None:
+                both = torch.stack([gsum, pred], dim=0)
+                both = ctx.comm.allreduce(both, ranks, op="sum")
+                gsum, pred = both[0], both[1]
         lse = torch.log(gsum) + gmax
         loss = lse - pred
         if ignore is not None:
@@ -319,6 +338,13 @@ class VocabParallelCrossEntropyGradOp(OpInterface):
         tp, tp_idx, vlocal = _shard_info(lds, vdim, my, op.attrs["vocab"])
         ignore = op.attrs.get("ignore_index", -100)
         vstart = tp_idx * vlocal
+        if logits.is_cuda:
+            from ...ops import functional as F
+            return [F.ext().vp_ce_bwd(
+                gy.reshape(-1).contiguous(), logits.contiguous(),
+                labels.reshape(-1).contiguous(),
+                lse.reshape(-1).contiguous(), vstart, vstart + vlocal,
+                ignore if ignore is not None else -(1 << 40))]
         sm = torch.exp(logits.float() - lse.unsqueeze(-1))
         in_range = (labels >= vstart) & (labels < vstart + vlocal)
         safe = torch.where(in_range, labels - vstart,

@@ -21,6 +21,8 @@ import torch
 
 _EXT = None
 _EXT_ERR = None
+# norm backward variant: split dx + column-reduce dw/db kernels
+_NORM_V2 = os.environ.get("HETU_AMD_NORM_V2", "0") == "1"
 
 
 def _load_ext():
@@ -82,6 +84,9 @@ def rmsnorm_fwd(x: torch.Tensor, w: torch.Tensor, eps: float
 def rmsnorm_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
                 rstd: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     if _gpu(x):
+        if _NORM_V2:
+            return ext().rmsnorm_bwd2(dy.contiguous(), x.contiguous(),
+                                      w.contiguous(), rstd.contiguous())
         return ext().rmsnorm_bwd(dy.contiguous(), x.contiguous(),
                                  w.contiguous(), rstd.contiguous())
     xf, dyf, wf = x.float(), dy.float(), w.float()
@@ -112,6 +117,10 @@ def layernorm_fwd(x, w, b, eps):
 
 def layernorm_bwd(dy, x, w, mean, rstd):
     if _gpu(x):
+        if _NORM_V2:
+            return ext().layernorm_bwd2(dy.contiguous(), x.contiguous(),
+                                        w.contiguous(), mean.contiguous(),
+                                        rstd.contiguous())
         return ext().layernorm_bwd(dy.contiguous(), x.contiguous(),
                                    w.contiguous(), mean.contiguous(),
                                    rstd.contiguous())

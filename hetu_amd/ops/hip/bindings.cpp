@@ -34,6 +34,10 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
 torch::Tensor softmax_ce_bwd(torch::Tensor dloss, torch::Tensor logits,
                              torch::Tensor labels, torch::Tensor lse,
                              int64_t ignore_index);
+torch::Tensor vp_sumexp(torch::Tensor logits, torch::Tensor gmax);
+torch::Tensor vp_ce_bwd(torch::Tensor gy, torch::Tensor logits,
+                        torch::Tensor labels, torch::Tensor lse,
+                        int64_t vstart, int64_t vend, int64_t ignore);
 std::vector<torch::Tensor> vp_ce_local(torch::Tensor logits,
                                        torch::Tensor labels,
                                        int64_t vocab_start,
@@ -48,6 +52,13 @@ torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p,
 torch::Tensor embedding_fwd(torch::Tensor table, torch::Tensor ids);
 torch::Tensor embedding_bwd(torch::Tensor dy, torch::Tensor ids,
                             int64_t num_rows);
+std::vector<torch::Tensor> layernorm_bwd2(torch::Tensor dy, torch::Tensor x,
+                                          torch::Tensor w,
+                                          torch::Tensor mean,
+                                          torch::Tensor rstd);
+std::vector<torch::Tensor> rmsnorm_bwd2(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor w,
+                                        torch::Tensor rstd);
 // adam.hip
 void adam_step(torch::Tensor param32, torch::Tensor grad, torch::Tensor m,
                torch::Tensor v, double lr, double beta1, double beta2,
@@ -100,6 +111,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("layernorm_bwd2", &layernorm_bwd2);
+  m.def("rmsnorm_bwd2", &rmsnorm_bwd2);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("gelu_fwd", &gelu_fwd);
@@ -113,6 +126,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_ce_fwd", &softmax_ce_fwd);
   m.def("softmax_ce_bwd", &softmax_ce_bwd);
   m.def("vp_ce_local", &vp_ce_local);
+  m.def("vp_sumexp", &vp_sumexp);
+  m.def("vp_ce_bwd", &vp_ce_bwd);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("embedding_fwd", &embedding_fwd);

@@ -174,3 +174,104 @@ std::vector<torch::Tensor> vp_ce_local(torch::Tensor logits,
   row_sizes.pop_back();
   return {lmax.view(row_sizes), picked.view(row_sizes)};
 }
+
+namespace {
+
+// sum of exp(logit - gmax[row]) straight off bf16 (no fp32 logits copy)
+template <typename T>
+__global__ void vp_sumexp_kernel(const T* __restrict__ logits,
+                                 const float* __restrict__ gmax,
+                                 float* __restrict__ gsum,
+                                 int64_t rows, int Vloc) {
+  constexpr int V = VecIO<T>::VEC;
+  __shared__ float smem[16];
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* lr = logits + row * Vloc;
+    const float mx = gmax[row];
+    float s = 0.f;
+    for (int i = threadIdx.x * V; i + V <= Vloc; i += BLOCK * V) {
+      float v[VecIO<T>::VEC];
+      VecIO<T>::load(lr + i, v);
+#pragma unroll
+      for (int j = 0; j < V; ++j) s += __expf(v[j] - mx);
+    }
+    for (int i = (Vloc / V) * V + threadIdx.x; i < Vloc; i += BLOCK)
+      s += __expf((float)lr[i] - mx);
+    s = block_sum(s, smem);
+    if (threadIdx.x == 0) gsum[row] = s;
+  }
+}
+
+// dlogits = gy * (softmax - onehot_local) with the GLOBAL lse; rows whose
+// label is out of this shard still get the softmax term
+template <typename T>
+__global__ void vp_ce_bwd_kernel(const float* __restrict__ gy,
+                                 const T* __restrict__ logits,
+                                 const int64_t* __restrict__ labels,
+                                 const float* __restrict__ lse,
+                                 T* __restrict__ dl, int64_t rows, int Vloc,
+                                 int64_t vstart, int64_t vend,
+                                 int64_t ignore) {
+  constexpr int V = VecIO<T>::VEC;
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* lr = logits + row * Vloc;
+    T* dr = dl + row * Vloc;
+    const int64_t lbl = labels[row];
+    const float g = (lbl == ignore) ? 0.f : gy[row];
+    const float l = lse[row];
+    const int64_t local = (lbl >= vstart && lbl < vend) ? lbl - vstart : -1;
+    for (int i = threadIdx.x * V; i + V <= Vloc; i += BLOCK * V) {
+      float v[VecIO<T>::VEC], o[VecIO<T>::VEC];
+      VecIO<T>::load(lr + i, v);
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        float p = __expf(v[j] - l);
+        o[j] = g * (p - ((i + j == local) ? 1.f : 0.f));
+      }
+      VecIO<T>::store(dr + i, o);
+    }
+    for (int i = (Vloc / V) * V + threadIdx.x; i < Vloc; i += BLOCK) {
+      float p = __expf((float)lr[i] - l);
+      dr[i] = (T)(g * (p - ((i == local) ? 1.f : 0.f)));
+    }
+  }
+}
+
+}  // namespace
+
+torch::Tensor vp_sumexp(torch::Tensor logits, torch::Tensor gmax) {
+  const int Vloc = logits.size(-1);
+  const int64_t rows = logits.numel() / Vloc;
+  auto gsum = torch::empty({rows}, logits.options().dtype(at::kFloat));
+  auto stream = hetu_current_stream();
+  int grid = (int)std::min<int64_t>(rows, 8192);
+  auto gm = gmax.contiguous();
+  DISPATCH_FLOAT(logits, "vp_sumexp", [&] {
+    hipLaunchKernelGGL(vp_sumexp_kernel<scalar_t>, dim3(grid), dim3(BLOCK),
+                       0, stream, (const scalar_t*)logits.data_ptr(),
+                       gm.data_ptr<float>(), gsum.data_ptr<float>(), rows,
+                       Vloc);
+  });
+  return gsum;
+}
+
+torch::Tensor vp_ce_bwd(torch::Tensor gy, torch::Tensor logits,
+                        torch::Tensor labels, torch::Tensor lse,
+                        int64_t vstart, int64_t vend, int64_t ignore) {
+  const int Vloc = logits.size(-1);
+  const int64_t rows = logits.numel() / Vloc;
+  auto dl = torch::empty_like(logits);
+  auto stream = hetu_current_stream();
+  int grid = (int)std::min<int64_t>(rows, 8192);
+  auto gyf = gy.to(at::kFloat).contiguous();
+  auto lc = lse.to(at::kFloat).contiguous();
+  DISPATCH_FLOAT(logits, "vp_ce_bwd", [&] {
+    hipLaunchKernelGGL(vp_ce_bwd_kernel<scalar_t>, dim3(grid), dim3(BLOCK),
+                       0, stream, gyf.data_ptr<float>(),
+                       (const scalar_t*)logits.data_ptr(),
+                       labels.data_ptr<int64_t>(), lc.data_ptr<float>(),
+                       (scalar_t*)dl.data_ptr(), rows, Vloc, vstart, vend,
+                       ignore);
+  });
+  return dl;
+}

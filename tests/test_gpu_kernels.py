@@ -287,3 +287,51 @@ class TestFusedQKVAttention:
             do.cpu().float(), qkv_cpu, orf, lser, H, Hkv, D, cc, ss,
             cfg["causal"], None)
         _close(dqkv, dref, rtol=6e-2, atol=6e-2, what="fused dqkv")
+
+
+class TestVocabParallelCEKernels:
+    def test_vp_sumexp_and_bwd_match_ref(self):
+        torch.manual_seed(0)
+        rows, V = 512, 1024
+        logits = torch.randn(rows, V, dtype=torch.bfloat16, device=dev())
+        labels = torch.randint(0, 4 * V, (rows,), device=dev())
+        labels[::17] = -100
+        lf = logits.cpu().float()
+        gmax = lf.max(-1).values + 0.3   # pretend global max is higher
+        gs = F.ext().vp_sumexp(logits, gmax.to(dev()))
+        ref = torch.exp(lf - gmax[:, None]).sum(-1)
+        _close(gs, ref, rtol=1e-2, atol=1e-3, what="vp_sumexp")
+        # bwd: shard covering vocab [V, 2V)
+        lse = torch.log(ref) + gmax
+        gy = torch.rand(rows)
+        dl = F.ext().vp_ce_bwd(gy.to(dev()), logits, labels,
+                               lse.to(dev()), V, 2 * V, -100)
+        sm = torch.exp(lf - lse[:, None])
+        onehot = torch.zeros_like(sm)
+        lb = labels.cpu()
+        for r in range(rows):
+            if V <= lb[r] < 2 * V:
+                onehot[r, lb[r] - V] = 1.0
+        scale = torch.where(lb == -100, torch.zeros_like(gy), gy)
+        refd = (sm - onehot) * scale[:, None]
+        _close(dl, refd, rtol=3e-2, atol=3e-3, what="vp_ce_bwd")
+
+    def test_norm_bwd_v2_matches_v1(self):
+        torch.manual_seed(1)
+        R, D = 2048, 4096
+        x = torch.randn(R, D, dtype=torch.bfloat16, device=dev())
+        w = torch.randn(D, dtype=torch.bfloat16, device=dev())
+        b = torch.randn(D, dtype=torch.bfloat16, device=dev())
+        dy = torch.randn(R, D, dtype=torch.bfloat16, device=dev())
+        y, mean, rstd = F.layernorm_fwd(x, w, b, 1e-5)
+        a = F.ext().layernorm_bwd(dy, x, w, mean, rstd)
+        v2 = F.ext().layernorm_bwd2(dy, x, w, mean, rstd)
+        for n, (t1, t2) in zip(("dx", "dw", "db"), zip(a, v2)):
+            _close(t2, t1.float(), rtol=2e-2, atol=2e-2,
+                   what=f"ln bwd2 {n}")
+        yr, rs = F.rmsnorm_fwd(x, w, 1e-6)
+        a = F.ext().rmsnorm_bwd(dy, x, w, rs)
+        v2 = F.ext().rmsnorm_bwd2(dy, x, w, rs)
+        for n, (t1, t2) in zip(("dx", "dw"), zip(a, v2)):
+            _close(t2, t1.float(), rtol=2e-2, atol=2e-2,
+                   what=f"rms bwd2 {n}")
