@@ -22,7 +22,7 @@ import torch
 _EXT = None
 _EXT_ERR = None
 # norm backward variant: split dx + column-reduce dw/db kernels
-_NORM_V2 = os.environ.get("HETU_AMD_NORM_V2", "0") == "1"
+_NORM_V2 = os.environ.get("HETU_AMD_NORM_V2", "1") == "1"
 
 
 def _load_ext():
