@@ -74,6 +74,46 @@ DEV int kswz_row(int row, int byte_in_row) {
     dst = f_.v;                                                            \
   } while (0)
 
+// batched TR_RM: all 4 dt fragments (colbyte cb0 + 64*dt) of the same
+// row pair in ONE asm block — 8 tr reads issued back-to-back under a
+// single lgkmcnt wait instead of 4 serialized round trips.
+#define TR_RM4(d0, d1, d2, d3, base_lds, rowb, cb0)                        \
+  do {                                                                     \
+    int r1_ = (rowb) + ((lane >> 2) & 3);                                  \
+    int r2_ = r1_ + 4;                                                     \
+    unsigned b1_ = (unsigned)(uintptr_t)(                                  \
+        (__attribute__((address_space(3))) char*)((base_lds) + r1_ * 256));\
+    unsigned b2_ = (unsigned)(uintptr_t)(                                  \
+        (__attribute__((address_space(3))) char*)((base_lds) + r2_ * 256));\
+    const int m1_ = (r1_ & 15) << 4, m2_ = (r2_ & 15) << 4;                \
+    u32x2 x_[8];                                                           \
+    asm volatile("ds_read_b64_tr_b16 %0, %8\n\t"                           \
+                 "ds_read_b64_tr_b16 %1, %9\n\t"                           \
+                 "ds_read_b64_tr_b16 %2, %10\n\t"                          \
+                 "ds_read_b64_tr_b16 %3, %11\n\t"                          \
+                 "ds_read_b64_tr_b16 %4, %12\n\t"                          \
+                 "ds_read_b64_tr_b16 %5, %13\n\t"                          \
+                 "ds_read_b64_tr_b16 %6, %14\n\t"                          \
+                 "ds_read_b64_tr_b16 %7, %15\n\t"                          \
+                 "s_waitcnt lgkmcnt(0)"                                    \
+                 : "=&v"(x_[0]), "=&v"(x_[1]), "=&v"(x_[2]), "=&v"(x_[3]), \
+                   "=&v"(x_[4]), "=&v"(x_[5]), "=&v"(x_[6]), "=&v"(x_[7])  \
+                 : "v"(b1_ + (((cb0) + 0) ^ m1_)),                         \
+                   "v"(b2_ + (((cb0) + 0) ^ m2_)),                         \
+                   "v"(b1_ + (((cb0) + 64) ^ m1_)),                        \
+                   "v"(b2_ + (((cb0) + 64) ^ m2_)),                        \
+                   "v"(b1_ + (((cb0) + 128) ^ m1_)),                       \
+                   "v"(b2_ + (((cb0) + 128) ^ m2_)),                       \
+                   "v"(b1_ + (((cb0) + 192) ^ m1_)),                       \
+                   "v"(b2_ + (((cb0) + 192) ^ m2_)));                      \
+    __builtin_amdgcn_sched_barrier(0);                                     \
+    union { u32x2 u[2]; bf16x8 v; } f_;                                    \
+    f_.u[0] = x_[0]; f_.u[1] = x_[1]; d0 = f_.v;                           \
+    f_.u[0] = x_[2]; f_.u[1] = x_[3]; d1 = f_.v;                           \
+    f_.u[0] = x_[4]; f_.u[1] = x_[5]; d2 = f_.v;                           \
+    f_.u[0] = x_[6]; f_.u[1] = x_[7]; d3 = f_.v;                           \
+  } while (0)
+
 // tr_b16 read from a 72B-row [32 q][32 key] P'/dS' image (own=key, k=q).
 #define TR_P(dst, base_lds, qb, keybyte)                                   \
   do {                                                                     \
@@ -227,14 +267,17 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dq_kernel(
       for (int ks = 0; ks < 4; ++ks) {
         bf16x8 dsf;
         PACK_FRAG(dsf, ds[ks >> 1], (ks & 1) * 8);
-#pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
-          bf16x8 ktf;
-          TR_RM(ktf, k_lds(cur), 16 * ks + 8 * hi,
-                (32 * dt + 16 * g1 + 4 * (lane & 3)) * 2);
-          dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              ktf, dsf, dq_acc[dt], 0, 0, 0);
-        }
+        bf16x8 kt0, kt1, kt2, kt3;
+        TR_RM4(kt0, kt1, kt2, kt3, k_lds(cur), 16 * ks + 8 * hi,
+               (16 * g1 + 4 * (lane & 3)) * 2);
+        dq_acc[0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            kt0, dsf, dq_acc[0], 0, 0, 0);
+        dq_acc[1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            kt1, dsf, dq_acc[1], 0, 0, 0);
+        dq_acc[2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            kt2, dsf, dq_acc[2], 0, 0, 0);
+        dq_acc[3] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            kt3, dsf, dq_acc[3], 0, 0, 0);
       }
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -445,18 +488,27 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
         bf16x8 ptf, dstf;
         TR_P(ptf, p_lds, 16 * s + 8 * hi, (16 * g1 + 4 * (lane & 3)) * 2);
         TR_P(dstf, ds_lds, 16 * s + 8 * hi, (16 * g1 + 4 * (lane & 3)) * 2);
-#pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
-          bf16x8 dof, qf2;
-          TR_RM(dof, do_lds(cur), 16 * s + 8 * hi,
-                (32 * dt + 16 * g1 + 4 * (lane & 3)) * 2);
-          dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              ptf, dof, dv_acc[dt], 0, 0, 0);
-          TR_RM(qf2, q_lds(cur), 16 * s + 8 * hi,
-                (32 * dt + 16 * g1 + 4 * (lane & 3)) * 2);
-          dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              dstf, qf2, dk_acc[dt], 0, 0, 0);
-        }
+        bf16x8 do0, do1, do2, do3, qa0, qa1, qa2, qa3;
+        TR_RM4(do0, do1, do2, do3, do_lds(cur), 16 * s + 8 * hi,
+               (16 * g1 + 4 * (lane & 3)) * 2);
+        TR_RM4(qa0, qa1, qa2, qa3, q_lds(cur), 16 * s + 8 * hi,
+               (16 * g1 + 4 * (lane & 3)) * 2);
+        dv_acc[0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            ptf, do0, dv_acc[0], 0, 0, 0);
+        dk_acc[0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            dstf, qa0, dk_acc[0], 0, 0, 0);
+        dv_acc[1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            ptf, do1, dv_acc[1], 0, 0, 0);
+        dk_acc[1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            dstf, qa1, dk_acc[1], 0, 0, 0);
+        dv_acc[2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            ptf, do2, dv_acc[2], 0, 0, 0);
+        dk_acc[2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            dstf, qa2, dk_acc[2], 0, 0, 0);
+        dv_acc[3] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            ptf, do3, dv_acc[3], 0, 0, 0);
+        dk_acc[3] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            dstf, qa3, dk_acc[3], 0, 0, 0);
       }
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");

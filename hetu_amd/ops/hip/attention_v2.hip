@@ -249,27 +249,40 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
           pk.u[0] = w0; pk.u[1] = w1; pk.u[2] = w2; pk.u[3] = w3;
           pfrag = pk.v;
         }
+        // Measured tr_b16 semantics (scripts/probe_tr16): lane 16g+4r+c
+        // receives elem j from the address of lane 16g+4j+r, at +2c
+        // bytes.  So lane a supplies the key selected by (a>>2)&3 and the
+        // d-block selected by a&3; its own received d = dbase + (a&31).
+        // voff is linear in the d-subtile index, so all 4 dt fragments
+        // sit at static offsets (+2048*dt) from ONE address — issue all
+        // 8 tr reads back-to-back under a single lgkmcnt wait instead of
+        // 4 serialized round trips (the per-pair wait was the fwd
+        // kernel's biggest stall).
+        const int keyb = 16 * ks + 8 * hi + ((lane >> 2) & 3);
+        const int dbase0 = 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+        // ds ops take a 32-bit LDS offset, not a generic 64-bit pointer
+        unsigned a1 = (unsigned)(uintptr_t)(
+            (__attribute__((address_space(3))) char*)(v_lds(cur) +
+                                                      voff(keyb, dbase0)));
+        u32x2 r1[4], r2[4];
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %8\n\t"
+            "ds_read_b64_tr_b16 %1, %8 offset:128\n\t"
+            "ds_read_b64_tr_b16 %2, %8 offset:2048\n\t"
+            "ds_read_b64_tr_b16 %3, %8 offset:2176\n\t"
+            "ds_read_b64_tr_b16 %4, %8 offset:4096\n\t"
+            "ds_read_b64_tr_b16 %5, %8 offset:4224\n\t"
+            "ds_read_b64_tr_b16 %6, %8 offset:6144\n\t"
+            "ds_read_b64_tr_b16 %7, %8 offset:6272\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(r1[0]), "=&v"(r2[0]), "=&v"(r1[1]), "=&v"(r2[1]),
+              "=&v"(r1[2]), "=&v"(r2[2]), "=&v"(r1[3]), "=&v"(r2[3])
+            : "v"(a1));
+        __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
         for (int dt = 0; dt < 4; ++dt) {
-          // Measured tr_b16 semantics (scripts/probe_tr16): lane 16g+4r+c
-          // receives elem j from the address of lane 16g+4j+r, at +2c
-          // bytes.  So lane a supplies the key selected by (a>>2)&3 and the
-          // d-block selected by a&3; its own received d = dbase + (a&31).
-          const int keyb = 16 * ks + 8 * hi + ((lane >> 2) & 3);
-          const int dbase = 32 * dt + 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
-          // ds ops take a 32-bit LDS offset, not a generic 64-bit pointer
-          unsigned a1 = (unsigned)(uintptr_t)(
-              (__attribute__((address_space(3))) char*)(v_lds(cur) +
-                                                        voff(keyb, dbase)));
-          u32x2 r1, r2;
-          asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
-                       "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
-                       "s_waitcnt lgkmcnt(0)"
-                       : "=&v"(r1), "=&v"(r2)
-                       : "v"(a1));
-          __builtin_amdgcn_sched_barrier(0);
           union { u32x2 u[2]; bf16x8 v; } vf;
-          vf.u[0] = r1; vf.u[1] = r2;
+          vf.u[0] = r1[dt]; vf.u[1] = r2[dt];
           o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               vf.v, pfrag, o_acc[dt], 0, 0, 0);
         }
