@@ -403,14 +403,22 @@ __global__ void norm_bwd_dwdb_kernel(const T* __restrict__ dy,
   const int64_t r1 = min((int64_t)rows, r0 + rows_chunk);
   float dw0 = 0.f, dw1 = 0.f, db0 = 0.f, db1 = 0.f;
   for (int64_t row = r0; row < r1; ++row) {
-    union { unsigned u; T t[2]; } dv, xv;
-    dv.u = *reinterpret_cast<const unsigned*>(dy + row * D + c0);
-    xv.u = *reinterpret_cast<const unsigned*>(x + row * D + c0);
     const float mu = LN ? mean[row] : 0.f;
     const float r = rstd[row];
-    float d0 = (float)dv.t[0], d1 = (float)dv.t[1];
-    dw0 += d0 * (((float)xv.t[0] - mu) * r);
-    dw1 += d1 * (((float)xv.t[1] - mu) * r);
+    float d0, d1, x0, x1;
+    if (sizeof(T) == 2) {
+      // bf16: one dword load covers the column pair
+      union { unsigned u; T t[2]; } dv, xv;
+      dv.u = *reinterpret_cast<const unsigned*>(dy + row * D + c0);
+      xv.u = *reinterpret_cast<const unsigned*>(x + row * D + c0);
+      d0 = (float)dv.t[0]; d1 = (float)dv.t[1];
+      x0 = (float)xv.t[0]; x1 = (float)xv.t[1];
+    } else {
+      d0 = (float)dy[row * D + c0]; d1 = (float)dy[row * D + c0 + 1];
+      x0 = (float)x[row * D + c0]; x1 = (float)x[row * D + c0 + 1];
+    }
+    dw0 += d0 * ((x0 - mu) * r);
+    dw1 += d1 * ((x1 - mu) * r);
     if (LN) { db0 += d0; db1 += d1; }
   }
   atomicAdd(dw_accum + c0, dw0);
