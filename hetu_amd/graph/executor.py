@@ -58,6 +58,7 @@ class Executor:
         self.graph = graph
         self._plan_pool: Dict = {}
         self.ctx: Optional[ExecContext] = None
+        self._inflight: Dict[int, "torch.cuda.Event"] = {}
 
     def bind_context(self, ctx: ExecContext):
         self.ctx = ctx
@@ -123,8 +124,41 @@ class Executor:
                     values[tid] = feeds[tid]
                     continue
                 raise RuntimeError(f"placeholder {op.name} not fed")
+            # gradient all-reduces ride the comm stream so they overlap
+            # with the rest of backward (reference: grad-buffer bucket
+            # reduction on the comm stream, executable_graph.cc:1756);
+            # consumers wait via recorded events
+            comm_async = (ctx.stream("comm") is not None
+                          and op.name.startswith("grad_allreduce"))
+            # any input still in flight on the comm stream: make the
+            # compute stream wait before using it
+            for t in op.inputs:
+                ev = self._inflight.pop(t.id, None) if not comm_async \
+                    else None
+                if ev is not None:
+                    cur = torch.cuda.current_stream(ctx.device)
+                    cur.wait_event(ev)
+                    v = values.get(t.id)
+                    if isinstance(v, torch.Tensor) and v.is_cuda and \
+                            not torch.cuda.is_current_stream_capturing():
+                        v.record_stream(cur)
             prof = ctx.profiler
-            if prof is not None:
+            if comm_async:
+                cs = ctx.stream("comm")
+                ev_in = torch.cuda.Event()
+                ev_in.record(torch.cuda.current_stream(ctx.device))
+                cs.wait_event(ev_in)
+                with torch.cuda.stream(cs):
+                    outs = op.interface.compute(op, ins, ctx)
+                    if not torch.cuda.is_current_stream_capturing():
+                        for v in ins:
+                            if isinstance(v, torch.Tensor) and v.is_cuda:
+                                v.record_stream(cs)
+                    ev_out = torch.cuda.Event()
+                    ev_out.record(cs)
+                for t in op.outputs:
+                    self._inflight[t.id] = ev_out
+            elif prof is not None:
                 tok = prof.begin(op)
                 outs = op.interface.compute(op, ins, ctx)
                 prof.end(tok)
@@ -140,6 +174,14 @@ class Executor:
             for t in op.inputs:
                 if plan.last_use.get(t.id) == i and t.id in values:
                     del values[t.id]
+
+        # join any comm-stream work not consumed by an op (e.g. fetched
+        # tensors) back into the compute stream
+        if self._inflight:
+            cur = torch.cuda.current_stream(ctx.device)
+            for ev in self._inflight.values():
+                cur.wait_event(ev)
+            self._inflight.clear()
 
         out: List[torch.Tensor] = []
         for t in fetches:
