@@ -16,6 +16,12 @@ import torch
 from .op import Op
 from .tensor import Tensor
 
+import os
+
+# gradient-allreduce / backward overlap on the comm stream (disable with
+# HETU_AMD_COMM_OVERLAP=0 if a RCCL/capture combination misbehaves)
+_COMM_OVERLAP = os.environ.get("HETU_AMD_COMM_OVERLAP", "1") == "1" 
+
 
 class ExecContext:
     """Per-run execution context handed to every op's compute()."""
@@ -128,7 +134,7 @@ class Executor:
             # with the rest of backward (reference: grad-buffer bucket
             # reduction on the comm stream, executable_graph.cc:1756);
             # consumers wait via recorded events
-            comm_async = (ctx.stream("comm") is not None
+            comm_async = (_COMM_OVERLAP and ctx.stream("comm") is not None
                           and op.name.startswith("grad_allreduce"))
             # any input still in flight on the comm stream: make the
             # compute stream wait before using it
