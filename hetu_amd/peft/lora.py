@@ -85,3 +85,32 @@ class LoRALinear(Module):
         b = self.B.get_data().float()
         w = self.base.weight.get_data()
         w -= (self.scaling * (b @ a)).to(w.dtype)
+
+
+class QLinear(torch.nn.Module):
+    """4-bit-quantized frozen linear (QLoRA base layer): weight stored as
+    packed nf4/fp4/int8 blocks, dequantized through the blockwise kernel
+    into the GEMM (reference graph/ops/Quantization.h matmul4bit).  Pair
+    with LoRALinear for QLoRA fine-tuning."""
+
+    def __init__(self, weight: torch.Tensor, bias=None, qtype: str = "nf4",
+                 blocksize: int = 64):
+        super().__init__()
+        from ..ops import functional as F
+        self.out_features, self.in_features = weight.shape
+        self.qtype, self.blocksize = qtype, blocksize
+        q, amax = F.quantize_blockwise(
+            weight.reshape(-1).contiguous(), qtype, blocksize)
+        self.register_buffer("qweight", q)
+        self.register_buffer("absmax", amax)
+        if bias is not None:
+            self.register_buffer("bias", bias.detach().clone())
+        else:
+            self.bias = None
+
+    def forward(self, x):
+        from ..ops import functional as F
+        y = F.matmul_4bit(x, self.qweight, self.absmax, self.qtype,
+                          self.blocksize,
+                          (self.out_features, self.in_features))
+        return y + self.bias if self.bias is not None else y

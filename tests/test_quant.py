@@ -37,3 +37,27 @@ def test_matmul_4bit():
     ref = x @ w.t()
     rel = (y - ref).norm() / ref.norm()
     assert rel < 0.1, rel
+
+
+def test_matmul_4bit_and_qlinear():
+    """4-bit matmul path (reference Quantization.h matmul4bit): y from the
+    packed weight must match y from the dequantized weight exactly, and
+    stay close to the fp32 original."""
+    import torch
+
+    from hetu_amd.ops import functional as F
+    from hetu_amd.peft.lora import QLinear
+    torch.manual_seed(0)
+    w = torch.randn(32, 64)
+    x = torch.randn(4, 64)
+    q, amax = F.quantize_blockwise(w.reshape(-1), "nf4", 64)
+    y = F.matmul_4bit(x, q, amax, "nf4", 64, (32, 64))
+    wd = F.dequantize_blockwise(q, amax, "nf4", 64, 32 * 64,
+                                torch.float32).reshape(32, 64)
+    assert torch.allclose(y, x @ wd.t(), atol=1e-5)
+    rel = (y - x @ w.t()).abs().max() / w.abs().max()
+    assert rel < 0.5           # nf4 quantization noise, not garbage
+    ql = QLinear(w, bias=torch.randn(32), qtype="nf4")
+    y2 = ql(x)
+    assert y2.shape == (4, 32)
+    assert torch.allclose(y2, x @ wd.t() + ql.bias, atol=1e-5)
