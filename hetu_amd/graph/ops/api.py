@@ -290,6 +290,12 @@ def fused_qkv_attention(qkv, n_head, n_kv_head, head_dim, cos=None,
                           "scale": scale}).output()
 
 
+def varlen_attention(q, k, v, cu_seqlens, causal=True, scale=None):
+    """Packed-varlen attention: q/k/v [T, H, D], cu_seqlens [n+1]."""
+    return _cg().make_op(N.VarlenAttentionOp(), [q, k, v, cu_seqlens],
+                         {"causal": causal, "scale": scale}).output()
+
+
 def attention(q, k, v, causal=True, scale=None):
     q, k, v = _ac(q), _ac(k), _ac(v)
     return _cg().make_op(N.AttentionOp(), [q, k, v],

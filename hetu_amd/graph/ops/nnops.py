@@ -624,3 +624,47 @@ class FusedQKVAttentionGradOp(OpInterface):
             a["head_dim"], cos, sin, a.get("causal", True),
             a.get("scale"))
         return [dqkv]
+
+
+class VarlenAttentionOp(OpInterface):
+    """Packed-varlen flash attention: q/k/v [T, H, D] + cu_seqlens [n+1]
+    (reference ParallelAttention.cc packed path)."""
+    type = "VarlenAttention"
+
+    def infer_meta(self, attrs, inputs):
+        q = inputs[0]
+        T, H, D = q.shape
+        return [TensorMeta(q.shape, q.dtype),
+                TensorMeta((H, T), torch.float32)]
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        o, lse = F.varlen_attention_fwd(inputs[0], inputs[1], inputs[2],
+                                        inputs[3],
+                                        op.attrs.get("causal", True),
+                                        op.attrs.get("scale"))
+        return [o, lse]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        bwd = _make(gr, VarlenAttentionGradOp(),
+                    [g[0], op.inputs[0], op.inputs[1], op.inputs[2],
+                     op.outputs[0], op.outputs[1], op.inputs[3]],
+                    dict(op.attrs))
+        return [bwd.output(0), bwd.output(1), bwd.output(2), None]
+
+
+class VarlenAttentionGradOp(OpInterface):
+    type = "VarlenAttentionGrad"
+
+    def infer_meta(self, attrs, inputs):
+        _, q, k, v = inputs[:4]
+        return [TensorMeta(q.shape, q.dtype), TensorMeta(k.shape, k.dtype),
+                TensorMeta(v.shape, v.dtype)]
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        dout, q, k, v, out, lse, cu = inputs
+        return list(F.varlen_attention_bwd(dout, q, k, v, out, lse, cu,
+                                           op.attrs.get("causal", True),
+                                           op.attrs.get("scale")))

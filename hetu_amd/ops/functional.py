@@ -634,3 +634,53 @@ def fused_qkv_attention_bwd(dout, qkv, out, lse, n_head: int,
         d4[:, :, :H + Hkv] = _rope_ref(d4[:, :, :H + Hkv], cos, sin,
                                        True).to(dqkv.dtype)
     return dqkv
+
+
+def varlen_attention_fwd(q, k, v, cu_seqlens, causal: bool = True,
+                         scale: Optional[float] = None):
+    """Packed-varlen attention (reference ParallelAttention.cc packed
+    path): q/k/v [T, H, D] with cu_seqlens [n+1] int delimiting the packed
+    sequences; each segment attends only within itself.  Runs the flash
+    kernel per segment (B=1 view) — segments in one packed bin are
+    length-bucketed by data/bucket.py so the launches stay few and fat.
+    Returns (out [T, H, D], lse [H, T])."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    T, H, D = q.shape
+    out = torch.empty_like(q)
+    lse = torch.empty(H, T, dtype=torch.float32, device=q.device)
+    cu = cu_seqlens.tolist()
+    for s0, s1 in zip(cu[:-1], cu[1:]):
+        if s1 <= s0:
+            continue
+        qs = q[s0:s1].permute(1, 0, 2).unsqueeze(0)
+        ks = k[s0:s1].permute(1, 0, 2).unsqueeze(0)
+        vs = v[s0:s1].permute(1, 0, 2).unsqueeze(0)
+        o, l = flash_attn_fwd(qs.contiguous(), ks.contiguous(),
+                              vs.contiguous(), causal, scale)
+        out[s0:s1] = o[0].permute(1, 0, 2)
+        lse[:, s0:s1] = l[0]
+    return out, lse
+
+
+def varlen_attention_bwd(dout, q, k, v, out, lse, cu_seqlens,
+                         causal: bool = True,
+                         scale: Optional[float] = None):
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    dq = torch.empty_like(q)
+    dk = torch.empty_like(k)
+    dv = torch.empty_like(v)
+    cu = cu_seqlens.tolist()
+    for s0, s1 in zip(cu[:-1], cu[1:]):
+        if s1 <= s0:
+            continue
+        args = [t[s0:s1].permute(1, 0, 2).unsqueeze(0).contiguous()
+                for t in (dout, q, k, v, out)]
+        ls = lse[:, s0:s1].unsqueeze(0).contiguous()
+        dqs, dks, dvs = flash_attn_bwd(args[0], args[1], args[2], args[3],
+                                       args[4], ls, causal, scale)
+        dq[s0:s1] = dqs[0].permute(1, 0, 2)
+        dk[s0:s1] = dks[0].permute(1, 0, 2)
+        dv[s0:s1] = dvs[0].permute(1, 0, 2)
+    return dq, dk, dv

@@ -230,3 +230,49 @@ def test_dropout2d_bool_rangemask_asstrided():
     m = (rd != 0)
     assert all(m[n, c].all() or (~m[n, c]).all()
                for n in range(2) for c in range(3))  # whole-channel drops
+
+
+def test_varlen_attention_matches_per_segment_ref():
+    """Packed-varlen attention: segments attend only within themselves
+    (reference ParallelAttention packed path)."""
+    import math
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import (DefineAndRunGraph, pop_graph,
+                                      push_graph)
+    from hetu_amd.graph.ops import api as ht
+    torch.manual_seed(0)
+    T, H, D = 48, 2, 64
+    cu = torch.tensor([0, 16, 40, 48], dtype=torch.int32)
+    g = DefineAndRunGraph("vl")
+    push_graph(g)
+    try:
+        q = ht.placeholder((T, H, D), name="q")
+        k = ht.placeholder((T, H, D), name="k")
+        v = ht.placeholder((T, H, D), name="v")
+        c = ht.placeholder((4,), dtype=torch.int32, name="cu")
+        o = ht.varlen_attention(q, k, v, c)
+        gs = ht.gradients([ht.reduce_sum(o)], [q, k, v])
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    qd = torch.randn(T, H, D)
+    kd = torch.randn(T, H, D)
+    vd = torch.randn(T, H, D)
+    res = g.run([o] + gs, {q: qd, k: kd, v: vd, c: cu}, ctx=ctx)
+    ref = torch.empty(T, H, D)
+    for s0, s1 in zip(cu[:-1].tolist(), cu[1:].tolist()):
+        qs = qd[s0:s1].permute(1, 0, 2)
+        ks = kd[s0:s1].permute(1, 0, 2)
+        vs = vd[s0:s1].permute(1, 0, 2)
+        S = s1 - s0
+        sc = (qs @ ks.transpose(-1, -2)) / math.sqrt(D)
+        mask = torch.ones(S, S, dtype=torch.bool).tril()
+        sc = sc.masked_fill(~mask, float("-inf"))
+        ref[s0:s1] = (torch.softmax(sc, -1) @ vs).permute(1, 0, 2)
+    assert (res[0] - ref).abs().max() < 1e-5
+    # cross-segment isolation: perturbing segment 0 must not change the
+    # grads of segment 1's tokens
+    qd2 = qd.clone()
+    qd2[:16] += 1.0
+    res2 = g.run(gs, {q: qd2, k: kd, v: vd, c: cu}, ctx=ctx)
+    assert torch.allclose(res[2][16:40], res2[1][16:40], atol=1e-5)
