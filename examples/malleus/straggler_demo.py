@@ -35,7 +35,7 @@ def main():
     t = profile_rank_speed(device) * slowdown
     speeds = gather_speeds(comm, t)
     stragglers = detect_stragglers(speeds, threshold=1.5)
-    shares = rebalance_micro_batches(speeds, total=8 * ws)
+    shares = rebalance_micro_batches(speeds, total_mb=8 * ws)
     if rank == 0:
         print(f"speeds={['%.4f' % s for s in speeds]} "
               f"stragglers={stragglers} shares={shares}")
