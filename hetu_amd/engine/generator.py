@@ -160,3 +160,100 @@ class LlamaGenerator:
             logits = self._forward(nxt.unsqueeze(1), cache, pos)
             pos += 1
         return torch.cat(out, dim=1)
+
+
+class GPTGenerator:
+    """KV-cached generation for the GPT family (learned positions,
+    LayerNorm, biased linears, gelu MLP)."""
+
+    def __init__(self, cfg, state: Dict[str, torch.Tensor],
+                 device: Optional[torch.device] = None,
+                 dtype: torch.dtype = torch.float32):
+        self.cfg = cfg
+        self.device = device or torch.device(
+            "cuda" if torch.cuda.is_available() else "cpu")
+        self.dtype = dtype
+        self.w = {k: v.to(self.device, dtype) for k, v in state.items()}
+        self.scale = 1.0 / math.sqrt(cfg.hidden // cfg.n_head)
+
+    def _lin(self, x, name):
+        y = x @ self.w[f"{name}.weight"].t()
+        b = self.w.get(f"{name}.bias")
+        return y + b if b is not None else y
+
+    def _attn(self, x, layer, kcache, vcache, pos0):
+        cfg = self.cfg
+        B, S, _ = x.shape
+        H = cfg.n_head
+        dh = cfg.hidden // H
+        qkv = self._lin(x, f"h{layer}.attn.wqkv")
+        q, k, v = qkv.chunk(3, -1)
+        q = q.view(B, S, H, dh).permute(0, 2, 1, 3)
+        kcache[layer][:, :, pos0:pos0 + S] = \
+            k.view(B, S, H, dh).permute(0, 2, 1, 3)
+        vcache[layer][:, :, pos0:pos0 + S] = \
+            v.view(B, S, H, dh).permute(0, 2, 1, 3)
+        kk = kcache[layer][:, :, :pos0 + S]
+        vv = vcache[layer][:, :, :pos0 + S]
+        if S > 1 and q.is_cuda and dh == 128 and q.dtype == torch.bfloat16:
+            o, _ = F.flash_attn_fwd(q.contiguous(), kk.contiguous(),
+                                    vv.contiguous(), True, self.scale)
+        else:
+            scores = (q.float() @ kk.float().transpose(-1, -2)) * self.scale
+            if S > 1:
+                Skv = pos0 + S
+                mask = torch.ones(S, Skv, dtype=torch.bool,
+                                  device=x.device).tril(diagonal=Skv - S)
+                scores = scores.masked_fill(~mask, float("-inf"))
+            o = (torch.softmax(scores, -1) @ vv.float()).to(x.dtype)
+        o = o.permute(0, 2, 1, 3).reshape(B, S, H * dh)
+        return self._lin(o, f"h{layer}.attn.wo")
+
+    def _forward(self, ids, kcache, vcache, pos0):
+        cfg = self.cfg
+        S = ids.shape[1]
+        pos = torch.arange(pos0, pos0 + S, device=self.device)
+        x = self.w["wte.weight"][ids] + self.w["wpe.weight"][pos]
+        for i in range(cfg.n_layer):
+            h = F.layernorm_fwd(x, self.w[f"h{i}.ln1.weight"],
+                                self.w[f"h{i}.ln1.bias"], 1e-5)[0]
+            x = x + self._attn(h, i, kcache, vcache, pos0)
+            h = F.layernorm_fwd(x, self.w[f"h{i}.ln2.weight"],
+                                self.w[f"h{i}.ln2.bias"], 1e-5)[0]
+            h = F.gelu_fwd(self._lin(h, f"h{i}.mlp.wfc"))
+            x = x + self._lin(h, f"h{i}.mlp.wproj")
+        x = F.layernorm_fwd(x, self.w["lnf.weight"], self.w["lnf.bias"],
+                            1e-5)[0]
+        return x[:, -1] @ self.w["lm_head.weight"].t()
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.0, top_k: int = 0,
+                 seed: int = 0) -> torch.Tensor:
+        cfg = self.cfg
+        ids = input_ids.to(self.device)
+        B, S = ids.shape
+        dh = cfg.hidden // cfg.n_head
+        L = S + max_new_tokens
+        kc = torch.zeros(cfg.n_layer, B, cfg.n_head, L, dh,
+                         device=self.device, dtype=self.dtype)
+        vc = torch.zeros_like(kc)
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        logits = self._forward(ids, kc, vc, 0)
+        out = [ids]
+        pos = S
+        for _ in range(max_new_tokens):
+            if temperature <= 0:
+                nxt = logits.argmax(-1)
+            else:
+                lg = logits.float() / temperature
+                if top_k:
+                    kth = lg.topk(top_k, dim=-1).values[:, -1:]
+                    lg = lg.masked_fill(lg < kth, float("-inf"))
+                nxt = torch.multinomial(torch.softmax(lg, -1).cpu(), 1,
+                                        generator=gen).squeeze(-1) \
+                    .to(self.device)
+            out.append(nxt.unsqueeze(1))
+            logits = self._forward(nxt.unsqueeze(1), kc, vc, pos)
+            pos += 1
+        return torch.cat(out, dim=1)

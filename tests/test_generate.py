@@ -68,3 +68,27 @@ def test_sampling_and_eos(tmp_path):
     out2 = gen.generate(prompt, max_new_tokens=8, temperature=0.8,
                         top_k=20, seed=5)
     assert torch.equal(out, out2)          # seeded sampling is reproducible
+
+
+def test_gpt_incremental_matches_full_forward(tmp_path):
+    """GPT KV-cache decode must equal full-sequence recomputation."""
+    import torch as T
+
+    from hetu_amd.engine.generator import GPTGenerator
+    from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
+    cfg = GPTConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
+                    ffn_hidden=96, vocab=157, max_seq=64)
+    T.manual_seed(2)
+    g, h = build_gpt_train_graph(cfg, micro_batch=1, seq_len=16,
+                                 dtype=T.float32)
+    state = {p.name.split(":")[0]: p.get_data() for p in g.parameters}
+    gen = GPTGenerator(cfg, state, device=T.device("cpu"))
+    prompt = T.randint(0, 157, (2, 10))
+    out = gen.generate(prompt, max_new_tokens=6, temperature=0.0)
+    assert out.shape == (2, 16)
+    # re-run with the full prefix at once: next greedy token must agree
+    dh = cfg.hidden // cfg.n_head
+    kc = T.zeros(cfg.n_layer, 2, cfg.n_head, 20, dh)
+    vc = T.zeros_like(kc)
+    logits_full = gen._forward(out[:, :15], kc, vc, 0)
+    assert T.equal(logits_full.argmax(-1), out[:, 15])
