@@ -126,9 +126,18 @@ class PipelineRunner:
             for p in self.params]
         self._act_shape = tuple(h["act_shape"])
         self._act_dtype = h.get("act_dtype", torch.bfloat16)
+        # per-micro-batch memory snapshots (reference CUDAProfiler
+        # GetCurrMemoryInfo at MEMORY_PROFILE_LEVEL=MICRO_BATCH)
+        self.mem_snapshots = None
+        import os as _os
+        if _os.environ.get("HETU_AMD_MEM_PROFILE", "0") == "1":
+            from ..utils.profiler import MemorySnapshots
+            self.mem_snapshots = MemorySnapshots()
 
     # ---- fwd / bwd over the stage graph ---------------------------------
     def _fwd(self, i, micro_batches, act):
+        if self.mem_snapshots is not None:
+            self.mem_snapshots.mark(f"fwd_mb{i}", self.device)
         h = self.stage.h
         feed = dict(micro_batches[i])
         if not self.is_first:
@@ -149,6 +158,8 @@ class PipelineRunner:
         return out[0], (feed, cache)
 
     def _bwd(self, saved, gin):
+        if self.mem_snapshots is not None:
+            self.mem_snapshots.mark("bwd_mb", self.device)
         h = self.stage.h
         feed, cache = saved
         if self.offloader is not None and cache is not None \

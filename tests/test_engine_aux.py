@@ -41,3 +41,38 @@ def test_config_yaml(tmp_path):
     cfg = TrainingConfig.from_yaml(str(p))
     assert cfg.model == "gpt2-345m" and cfg.tp == 2
     assert cfg.dtype() == torch.bfloat16
+
+
+def test_pipeline_memory_snapshots(monkeypatch):
+    """HETU_AMD_MEM_PROFILE=1 records a snapshot per micro-batch fwd/bwd
+    (reference per-micro-batch CUDAProfiler memory info)."""
+    import subprocess
+    import sys
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = r"""
+import os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+os.environ["HETU_AMD_MEM_PROFILE"] = "1"
+from hetu_amd.models.llama import LlamaConfig, build_llama_pipeline_stage
+from hetu_amd.parallel.pipeline import PipelineSpec, PipelineRunner
+cfg = LlamaConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
+                  ffn_hidden=96, vocab=211, max_seq=16)
+pspec = PipelineSpec(pp=1, dp=1, tp=1)
+stage = build_llama_pipeline_stage(cfg, pspec, micro_batch=1, seq_len=16,
+                                   dtype=torch.float32, lr=1e-3)
+runner = PipelineRunner(pspec, stage, torch.device("cpu"))
+h = stage.h
+mbs = [{h["input_ids"]: torch.randint(0, 211, (1, 16)),
+        h["labels"]: torch.randint(0, 211, (16,))} for _ in range(3)]
+runner.step(mbs)
+assert runner.mem_snapshots is not None
+rep = runner.mem_snapshots.report()
+assert "fwd_mb0" in rep and "bwd_mb" in rep, rep
+print("SNAP OK")
+"""
+    p = subprocess.run([sys.executable, "-c", code],
+                       env={**os.environ, "HETU_REPO": repo},
+                       capture_output=True, text=True, timeout=300)
+    assert p.returncode == 0 and "SNAP OK" in p.stdout, \
+        f"{p.stdout}\n{p.stderr}"
