@@ -1041,4 +1041,10 @@ class BatchMatMulTNOp(OpInterface):
         a, b = inputs
         if op.attrs["mode"] == "nt":
             return [torch.matmul(a, b.transpose(-1, -2))]
+        # TN via an explicit transpose copy: the strided-batched TN path
+        # in hipBLASLt memory-faults on the MoE expert-grad shapes
+        # (batch 8, [640, 8192]^T x [640, 2048] bf16) — reproduced and
+        # isolated on MI355X; the contiguous-NN route is fault-free
+        if a.is_cuda:
+            return [torch.matmul(a.transpose(-1, -2).contiguous(), b)]
         return [torch.matmul(a.transpose(-1, -2), b)]
