@@ -1038,13 +1038,21 @@ class BatchMatMulTNOp(OpInterface):
         return [TensorMeta(shape, a.dtype)]
 
     def compute(self, op, inputs, ctx):
+        # On GPU both transposed batched modes materialize the transpose
+        # and run the NN path: the hipBLASLt strided-batched NT kernel
+        # reads ~1 MB past the end of the transposed operand on the MoE
+        # grad shapes ([8,640,2048] @ [8,2048,8192]^T bf16) — isolated on
+        # MI355X via serialized repro (fault address = b.ptr + numel*2 +
+        # 1MB; both inputs clone cleanly).  Normally the overread lands in
+        # allocator slack; at a mapping boundary it faults.  The NN route
+        # (same shapes as the forward expert bmm) is fault-free.
         a, b = inputs
         if op.attrs["mode"] == "nt":
-            return [torch.matmul(a, b.transpose(-1, -2))]
-        # TN via an explicit transpose copy: the strided-batched TN path
-        # in hipBLASLt memory-faults on the MoE expert-grad shapes
-        # (batch 8, [640, 8192]^T x [640, 2048] bf16) — reproduced and
-        # isolated on MI355X; the contiguous-NN route is fault-free
+            bt = b.transpose(-1, -2)
+            if a.is_cuda:
+                bt = bt.contiguous()
+            return [torch.matmul(a, bt)]
+        at = a.transpose(-1, -2)
         if a.is_cuda:
-            return [torch.matmul(a.transpose(-1, -2).contiguous(), b)]
-        return [torch.matmul(a.transpose(-1, -2), b)]
+            at = at.contiguous()
+        return [torch.matmul(at, b)]
