@@ -60,7 +60,9 @@ class Trainer:
                 ctx=self.ctx)
             self._loss_out = loss
         self._cuda_graph = g
-        self._step += 1   # the capture pass itself performed one step
+        # NOTE: stream capture only RECORDS the step — nothing executes —
+        # so capture does not count as a training step; the caller replays
+        # immediately after
 
     def replay(self):
         self._step += 1
@@ -74,7 +76,10 @@ class Trainer:
             if self.want_capture and self._step >= 1:
                 try:
                     self.capture(feed)
-                    return self._loss_out
+                    # run the recorded step for real (capture executed
+                    # nothing; without this the capture call would
+                    # silently skip one update and return stale loss)
+                    return self.replay()
                 except Exception as e:  # noqa: BLE001
                     print(f"[hetu_amd] hipGraph capture failed, running "
                           f"eager: {e}")
