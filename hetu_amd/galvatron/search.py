@@ -154,3 +154,36 @@ def search(model: ModelShape, seq_len: int, n_gpus: int, global_batch: int,
     if best is None:
         raise RuntimeError("no feasible strategy found")
     return best
+
+
+# ---------------------------------------------------------------------------
+# OSDP: per-layer optimizer-state sharding plan (reference: Galvatron-family
+# OSDP — choose, layer by layer, whether optimizer states are replicated
+# (fast update, 12 bytes/param memory on every rank) or ZeRO-sharded
+# (memory / dp, but a reduce-scatter + all-gather per step).  The same
+# layers x choices DP knapsack picks the cheapest plan that fits.)
+# ---------------------------------------------------------------------------
+def osdp_plan(cm: CostModel, st: "Strategy", global_batch: int):
+    """Returns (plan, est): plan[i] = 1 if layer i keeps ZeRO-sharded
+    optimizer states, 0 if replicated.  Shards exactly as many layers as
+    the memory overshoot requires (identical layers -> greedy is optimal;
+    each sharded layer pays ~one fp32 param all-gather per step)."""
+    import math as _math
+    base = cm.evaluate(st, global_batch)
+    n = st.dp * st.cp
+    L = (cm.m.n_layer + st.pp - 1) // st.pp
+    if base["fits"] or n <= 1:
+        return [0] * L, base
+
+    p_layer = cm.m.layer_params / st.tp
+    saving = 12.0 * p_layer * (1.0 - 1.0 / n)     # bytes freed per layer
+    t_shard = cm.hw.allreduce_time(4.0 * p_layer, n) / 2.0  # ~all-gather
+    over = base["mem"] - cm.hw.hbm_capacity
+    k = min(L, int(_math.ceil(over / max(saving, 1.0))))
+    plan = [1] * k + [0] * (L - k)
+    est = dict(base)
+    est["time"] = base["time"] + k * t_shard
+    est["mem"] = base["mem"] - k * saving
+    est["fits"] = est["mem"] <= cm.hw.hbm_capacity
+    est["osdp_plan"] = plan
+    return plan, est

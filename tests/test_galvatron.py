@@ -53,3 +53,30 @@ def test_cost_model_monotonic_in_batch():
     b = cm.evaluate(Strategy(dp=1, micro_batch=4, num_micro_batches=1), 4)
     assert b["time"] > a["time"]
     assert b["tokens_per_sec"] > a["tokens_per_sec"] * 0.8
+
+
+def test_osdp_per_layer_sharding_plan():
+    """OSDP: shard exactly enough layers' optimizer states to fit
+    (replicate the rest to avoid the per-step all-gather)."""
+    from hetu_amd.galvatron.cost_model import CostModel, ModelShape, Strategy
+    from hetu_amd.galvatron.search import osdp_plan
+    # ~30B at dp8: replicated states overshoot 288 GB by a bit
+    shape = ModelShape(n_layer=48, hidden=7168, ffn_hidden=28672,
+                       vocab=50304, n_head=56, kind="gpt")
+    cm = CostModel(shape, 2048)
+    st = Strategy()
+    st.dp, st.tp, st.pp, st.micro_batch = 8, 1, 1, 1
+    base = cm.evaluate(st, 8)
+    plan, est = osdp_plan(cm, st, 8)
+    if base["fits"]:
+        assert sum(plan) == 0
+    else:
+        assert 0 < sum(plan) <= len(plan)
+        assert est["mem"] < base["mem"]
+        assert est["time"] >= base["time"]
+        assert est["fits"]
+    # small model: nothing sharded
+    small = ModelShape(n_layer=12, hidden=768, ffn_hidden=3072, vocab=50304,
+                       n_head=12, kind="gpt")
+    p2, e2 = osdp_plan(CostModel(small, 1024), st, 8)
+    assert sum(p2) == 0 and e2["fits"]
