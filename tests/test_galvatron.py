@@ -60,9 +60,9 @@ def test_osdp_per_layer_sharding_plan():
     (replicate the rest to avoid the per-step all-gather)."""
     from hetu_amd.galvatron.cost_model import CostModel, ModelShape, Strategy
     from hetu_amd.galvatron.search import osdp_plan
-    # ~30B at dp8: replicated states overshoot 288 GB by a bit
-    shape = ModelShape(n_layer=48, hidden=7168, ffn_hidden=28672,
-                       vocab=50304, n_head=56, kind="gpt")
+    # ~22B at dp8: replicated states overshoot HBM; partial shard fits
+    shape = ModelShape(n_layer=40, hidden=6144, ffn_hidden=24576,
+                       vocab=50304, n_head=48, kind="gpt")
     cm = CostModel(shape, 2048)
     st = Strategy()
     st.dp, st.tp, st.pp, st.micro_batch = 8, 1, 1, 1
