@@ -276,3 +276,32 @@ def test_varlen_attention_matches_per_segment_ref():
     qd2[:16] += 1.0
     res2 = g.run(gs, {q: qd2, k: kd, v: vd, c: cu}, ctx=ctx)
     assert torch.allclose(res[2][16:40], res2[1][16:40], atol=1e-5)
+
+
+def test_non_contiguous_inputs():
+    """Ops must accept non-contiguous (transposed/sliced) feeds
+    (reference tests/test_non_contig_ops.py)."""
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import (DefineAndRunGraph, pop_graph,
+                                      push_graph)
+    from hetu_amd.graph.ops import api as ht
+    g = DefineAndRunGraph("nc")
+    push_graph(g)
+    try:
+        x = ht.placeholder((8, 16), name="x")
+        w = ht.placeholder((4, 16), name="w")
+        y = ht.linear(x, w)
+        s = ht.softmax(y)
+        r = ht.reduce_sum(ht.mul(s, s))
+        grads = ht.gradients([r], [x])
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    base_x = torch.randn(16, 8).t()           # transposed view
+    base_w = torch.randn(16, 8)[:, ::2].t()   # strided slice view
+    assert not base_x.is_contiguous() and not base_w.is_contiguous()
+    out = g.run([s, grads[0]], {x: base_x, w: base_w}, ctx=ctx)
+    ref = torch.softmax(base_x.contiguous()
+                        @ base_w.contiguous().t(), dim=-1)
+    assert torch.allclose(out[0], ref, atol=1e-6)
+    assert out[1].shape == (8, 16)
