@@ -143,6 +143,12 @@ def search(model: ModelShape, seq_len: int, n_gpus: int, global_batch: int,
                                               zero=zero))
     for st in cands:
         st2, res = _recompute_plan(cm, st, global_batch)
+        if not res["fits"] and not st2.zero and st2.dp * st2.cp > 1:
+            # OSDP: shard just enough layers' optimizer states before
+            # falling back to full ZeRO / rejecting the candidate
+            plan, res2 = osdp_plan(cm, st2, global_batch)
+            if res2["fits"]:
+                res = res2
         if not res["fits"]:
             continue
         if verbose:
