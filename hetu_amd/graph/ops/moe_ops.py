@@ -50,27 +50,32 @@ def _a2a(comm, ranks, x):
 def _route(probs: torch.Tensor, E: int, C: int, K: int):
     """Greedy capacity routing.  Returns pos [E, C] int64 (-1 pad),
     combine scatter info: slot_of [N, K] int64 (flat index into E*C, -1 if
-    dropped), topk idx [N, K], weights [N, K]."""
+    dropped), topk idx [N, K], weights [N, K].
+
+    Fully static shapes: no nonzero/boolean-mask gathers, so the routing
+    is hipGraph-capturable and never syncs the host (dropped tokens
+    scatter into a sacrificial slot E*C instead)."""
     N = probs.shape[0]
+    dev = probs.device
     w, idx = probs.topk(K, dim=-1)                      # [N, K]
-    pos = torch.full((E, C), -1, dtype=torch.int64, device=probs.device)
-    slot_of = torch.full((N, K), -1, dtype=torch.int64,
-                         device=probs.device)
-    # position within expert queue per (token, k): rank among tokens
-    # choosing that expert (first-come order, GShard style)
+    # pos with one extra dummy slot at index E*C for dropped tokens
+    pos_fl = torch.full((E * C + 1,), -1, dtype=torch.int64, device=dev)
+    slot_of = torch.full((N, K), -1, dtype=torch.int64, device=dev)
+    toks = torch.arange(N, dtype=torch.int64, device=dev)
+    base = torch.zeros(E, dtype=torch.int64, device=dev)
     for k in range(K):
         e = idx[:, k]                                   # [N]
         onehot = torch.nn.functional.one_hot(e, E)      # [N, E]
-        # priority: tokens already queued from earlier k slots
-        base = (pos >= 0).sum(-1)                       # [E]
         order = onehot.cumsum(0) * onehot               # 1-based rank
         q = (order.gather(1, e.unsqueeze(1)).squeeze(1) - 1) + base[e]
         keep = q < C
-        tok = torch.nonzero(keep, as_tuple=False).squeeze(1)
-        if tok.numel():
-            flat = e[tok] * C + q[tok]
-            pos.view(-1)[flat] = tok
-            slot_of[tok, k] = flat
+        flat = torch.where(keep, e * C + q,
+                           torch.full_like(q, E * C))  # dummy if dropped
+        pos_fl.scatter_(0, flat, toks)
+        slot_of[:, k] = torch.where(keep, flat,
+                                    torch.full_like(flat, -1))
+        base = base + onehot.sum(0).clamp(max=C)             if False else (pos_fl[:E * C].reshape(E, C) >= 0).sum(-1)
+    pos = pos_fl[:E * C].reshape(E, C)
     wk = torch.where(slot_of >= 0, w, torch.zeros_like(w))
     return pos, slot_of, idx, wk
 
@@ -110,9 +115,9 @@ class MoEDispatchOp(OpInterface):
         P = len(ranks)
         El = E // P
         pos, slot_of, idx, wk = _route(probs.float(), E, C, K)
-        send = x.new_zeros(E * C, x.shape[-1])
-        valid = pos.view(-1) >= 0
-        send[valid] = x[pos.view(-1)[valid]]
+        # static-shape gather: dropped slots read token 0 and are masked
+        pf = pos.view(-1)
+        send = x[pf.clamp(min=0)] * (pf >= 0).unsqueeze(-1).to(x.dtype)
         if P > 1:
             # [E*C, h] = [P, El*C, h] blocks by destination rank
             recv = _a2a(ctx.comm, ranks, send)
@@ -160,8 +165,9 @@ class MoEDispatchGradOp(OpInterface):
         else:
             gsend = g_exp.reshape(E * C, -1)
         dx = torch.zeros_like(x)
-        valid = pos.view(-1) >= 0
-        dx.index_add_(0, pos.view(-1)[valid], gsend[valid].to(x.dtype))
+        pf = pos.view(-1)
+        dx.index_add_(0, pf.clamp(min=0),
+                      (gsend * (pf >= 0).unsqueeze(-1)).to(x.dtype))
         # dprobs: combine_w = probs.gather(topk) masked -> scatter g_w
         dprobs = torch.zeros_like(probs)
         if g_w is not None:
@@ -207,11 +213,10 @@ class MoECombineOp(OpInterface):
         N = wk.shape[0]
         y = eo.new_zeros(N, eo.shape[-1])
         for k in range(K):
-            ok = slot_of[:, k] >= 0
-            tok = torch.nonzero(ok, as_tuple=False).squeeze(1)
-            if tok.numel():
-                rows = recv[slot_of[tok, k]]
-                y[tok] += rows * wk[tok, k].unsqueeze(-1).to(rows.dtype)
+            sl = slot_of[:, k]
+            rows = recv[sl.clamp(min=0)]
+            # wk is already 0 for dropped slots (masked at routing)
+            y += rows * wk[:, k].unsqueeze(-1).to(rows.dtype)
         return [y]
 
     def gradient(self, op, g):
@@ -239,23 +244,23 @@ class MoECombineGradOp(OpInterface):
         ranks = _ep_ranks(op, ctx)
         P = len(ranks)
         El = E // P
-        # d_recv [E*C, h]: rows scattered from gy * w
-        d_recv = gy.new_zeros(E * C, gy.shape[-1])
+        # d_recv [E*C, h]: rows scattered from gy * w (slot E*C is the
+        # sacrificial target for dropped entries — static shapes)
+        h = gy.shape[-1]
+        d_fl = gy.new_zeros(E * C + 1, h)
         for k in range(K):
-            ok = slot_of[:, k] >= 0
-            tok = torch.nonzero(ok, as_tuple=False).squeeze(1)
-            if tok.numel():
-                d_recv[slot_of[tok, k]] = \
-                    gy[tok] * wk[tok, k].unsqueeze(-1).to(gy.dtype)
-        # d_combine_w: dot(gy[token], recv[slot])
+            sl = slot_of[:, k]
+            tgt = torch.where(sl >= 0, sl, torch.full_like(sl, E * C))
+            d_fl.scatter_(0, tgt.unsqueeze(-1).expand(-1, h),
+                          gy * wk[:, k].unsqueeze(-1).to(gy.dtype))
+        d_recv = d_fl[:E * C]
+        # d_combine_w: dot(gy[token], recv[slot]); 0 where dropped
         recv = MoECombineOp._gather_back(eo, pos, ranks, ctx, E, C, P, El)
         dwk = torch.zeros_like(wk)
         for k in range(K):
-            ok = slot_of[:, k] >= 0
-            tok = torch.nonzero(ok, as_tuple=False).squeeze(1)
-            if tok.numel():
-                dwk[tok, k] = (gy[tok].float()
-                               * recv[slot_of[tok, k]].float()).sum(-1)
+            sl = slot_of[:, k]
+            dot = (gy.float() * recv[sl.clamp(min=0)].float()).sum(-1)
+            dwk[:, k] = dot * (sl >= 0).to(dot.dtype)
         # forward a2a of d_recv to expert layout
         if P > 1:
             recv2 = _a2a(ctx.comm, ranks, d_recv)
