@@ -64,12 +64,16 @@ class Graph:
         self.ops: List[Op] = []
         self.parameters: List[Tensor] = []
         self._tensor_by_id: Dict[int, Tensor] = {}
+        self._rc_scope_stack: List[int] = []   # active recompute scopes
 
     # ---- op construction -------------------------------------------------
     def make_op(self, interface: OpInterface, inputs: Sequence[Tensor],
                 attrs: Dict, name: str = "",
                 ds_list=None, device_group=None) -> Op:
         op = Op(interface, list(inputs), attrs, name=name, graph=self)
+        if self._rc_scope_stack and interface.type not in (
+                "Placeholder", "Variable", "Constant"):
+            op.attrs["_rc_scope"] = self._rc_scope_stack[-1]
         metas = interface.infer_meta(op.attrs, op.inputs)
         for i, meta in enumerate(metas):
             t = Tensor(op, i, meta, graph=self)
@@ -88,6 +92,92 @@ class Graph:
 
     def _post_make_op(self, op: Op):
         pass
+
+    # ---- activation recompute (reference recompute.cc:23-318: per-op
+    # recompute flags duplicate the forward subgraph into the backward) ---
+    class _RecomputeScope:
+        def __init__(self, graph, idx):
+            self.graph, self.idx = graph, idx
+
+        def __enter__(self):
+            self.graph._rc_scope_stack.append(self.idx)
+            return self
+
+        def __exit__(self, *a):
+            self.graph._rc_scope_stack.pop()
+
+    def recompute_scope(self, idx: int):
+        """Ops built inside `with g.recompute_scope(i)` form recompute
+        scope i: after autodiff, apply_recompute() clones the scope's ops
+        that backward needs and rewires backward to the clones, so the
+        scope's internal activations free right after forward."""
+        return Graph._RecomputeScope(self, idx)
+
+    def apply_recompute(self, watermark: int) -> int:
+        """watermark = len(self.ops) when autodiff started (ops at index
+        >= watermark are backward/optimizer ops).  Returns the number of
+        cloned ops.  Boundary tensors (consumed by another scope or
+        unscoped forward ops) are kept, exactly like torch checkpointing
+        keeps block inputs."""
+        fwd_ops = self.ops[:watermark]
+        bwd_ops = self.ops[watermark:]
+        scope_of = {op.id: op.attrs.get("_rc_scope") for op in fwd_ops}
+        # forward consumers per tensor
+        fwd_consumers: Dict[int, List[Op]] = {}
+        for op in fwd_ops:
+            for t in op.inputs:
+                fwd_consumers.setdefault(t.id, []).append(op)
+        # internal tensors of each scope: produced in scope s and only
+        # consumed (in forward) within scope s
+        internal: Dict[int, int] = {}   # tensor id -> scope
+        for op in fwd_ops:
+            s_id = scope_of.get(op.id)
+            if s_id is None:
+                continue
+            for t in op.outputs:
+                cons = fwd_consumers.get(t.id, [])
+                if all(scope_of.get(c.id) == s_id for c in cons):
+                    internal[t.id] = s_id
+        # which internal tensors does backward reference?
+        needed: Dict[int, set] = {}     # scope -> set of tensor ids
+        for op in bwd_ops:
+            for t in op.inputs:
+                s_id = internal.get(t.id)
+                if s_id is not None:
+                    needed.setdefault(s_id, set()).add(t.id)
+        total = 0
+        for s_id, tids in sorted(needed.items()):
+            # ops of the scope whose outputs are transitively needed
+            scope_ops = [op for op in fwd_ops
+                         if scope_of.get(op.id) == s_id]
+            need_ops: List[Op] = []
+            need_t = set(tids)
+            for op in reversed(scope_ops):
+                if any(t.id in need_t for t in op.outputs):
+                    need_ops.append(op)
+                    for t in op.inputs:
+                        if t.id in internal and internal[t.id] == s_id:
+                            need_t.add(t.id)
+            need_ops.reverse()
+            # clone in topo (= creation) order
+            mapping: Dict[int, Tensor] = {}
+            for op in need_ops:
+                new_in = [mapping.get(t.id, t) for t in op.inputs]
+                attrs = {k: v for k, v in op.attrs.items()
+                         if k != "_rc_scope"}
+                new_op = self.make_op(type(op.interface)(), new_in, attrs,
+                                      name=op.name + "_rc")
+                for old_t, new_t in zip(op.outputs, new_op.outputs):
+                    new_t.ds = old_t.ds
+                    new_t.device_group = old_t.device_group
+                    mapping[old_t.id] = new_t
+                total += 1
+            # rewire backward references to the clones
+            for op in bwd_ops:
+                for i, t in enumerate(op.inputs):
+                    if t.id in mapping:
+                        op.inputs[i] = mapping[t.id]
+        return total
 
     # ---- topology --------------------------------------------------------
     def topo_sort(self, fetches: Iterable[Tensor]) -> List[Op]:

@@ -208,6 +208,8 @@ class Optimizer:
                  ) -> Tensor:
         graph = loss.graph
         params = params if params is not None else list(graph.parameters)
+        watermark = len(graph.ops)
+        has_scopes = any("_rc_scope" in op.attrs for op in graph.ops)
         grads = graph.gradients([loss], params)
         updates = []
         for p, g in zip(params, grads):
@@ -221,7 +223,12 @@ class Optimizer:
                 g = make_comm(graph, g, p.ds, name=f"grad_allreduce_{p.name}")
             updates.append(self._make_update(graph, p, g))
         self.update_ops = updates
-        return _make(graph, GroupOp(), updates, name="train_op").output()
+        out = _make(graph, GroupOp(), updates, name="train_op").output()
+        if has_scopes:
+            # duplicate the marked forward subgraphs into the backward
+            # (reference recompute.cc semantics, op granularity)
+            graph.apply_recompute(watermark)
+        return out
 
 
 class SGD(Optimizer):

@@ -206,10 +206,11 @@ class GPTEmbedding(Module):
 class GPTLMHeadModel(Module):
     def __init__(self, cfg: GPTConfig, spec: Optional[ParallelSpec] = None,
                  micro_batch: int = 1, seq_len: int = 128,
-                 dtype=torch.bfloat16):
+                 dtype=torch.bfloat16, recompute: bool = False):
         super().__init__()
         spec = spec or ParallelSpec()
         self.cfg, self.spec = cfg, spec
+        self.recompute = recompute
         self.B, self.S = micro_batch, seq_len
         self.embed = GPTEmbedding(cfg, spec, seq_len, dtype)
         self.layers = ModuleList([GPTBlock(cfg, spec, i, dtype)
@@ -221,10 +222,15 @@ class GPTLMHeadModel(Module):
             name="lm_head", init_std=cfg.init_std)
 
     def forward(self, input_ids, labels=None):
+        import contextlib
         B, S, cfg, spec = self.B, self.S, self.cfg, self.spec
         x = self.embed(input_ids)
-        for blk in self.layers:
-            x = blk(x, B, S)
+        g = x.graph
+        for i, blk in enumerate(self.layers):
+            cm = g.recompute_scope(i) if self.recompute \
+                else contextlib.nullcontext()
+            with cm:
+                x = blk(x, B, S)
         x = self.lnf(x)
         logits = self.lm_head(
             ht.reshape(x, (B * S, cfg.hidden), ds=spec.ds_tokens(0)))
@@ -254,7 +260,8 @@ def build_gpt_train_graph(cfg: GPTConfig, micro_batch: int, seq_len: int,
         labels = ht.placeholder((micro_batch * seq_len,), dtype=torch.int64,
                                 name="labels", ds=spec.ds_tokens(0),
                                 device_group=spec.device_group)
-        model = GPTLMHeadModel(cfg, spec, micro_batch, seq_len, dtype)
+        model = GPTLMHeadModel(cfg, spec, micro_batch, seq_len, dtype,
+                               recompute=recompute)
         loss, _ = model(input_ids, labels)
         loss_report = loss
         if spec.num_devices > 1:

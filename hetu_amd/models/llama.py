@@ -166,10 +166,11 @@ class LlamaLMHeadModel(Module):
 
     def __init__(self, cfg: LlamaConfig, spec: Optional[ParallelSpec] = None,
                  micro_batch: int = 1, seq_len: int = 128,
-                 dtype=torch.bfloat16):
+                 dtype=torch.bfloat16, recompute: bool = False):
         super().__init__()
         spec = spec or ParallelSpec()
         self.cfg, self.spec = cfg, spec
+        self.recompute = recompute
         self.B, self.S = micro_batch, seq_len
         self.dtype = dtype
         cos_d, sin_d = rope_tables(cfg, seq_len, torch.float32,
@@ -193,10 +194,15 @@ class LlamaLMHeadModel(Module):
             name="lm_head", init_std=cfg.init_std)
 
     def forward(self, input_ids, labels=None):
+        import contextlib
         B, S, cfg, spec = self.B, self.S, self.cfg, self.spec
         x = self.wte(input_ids)
-        for blk in self.layers:
-            x = blk(x, B, S)
+        g = x.graph
+        for i, blk in enumerate(self.layers):
+            cm = g.recompute_scope(i) if self.recompute \
+                else contextlib.nullcontext()
+            with cm:
+                x = blk(x, B, S)
         x = self.lnf(x)
         if spec.sequence_parallel and spec.tp > 1:
             # gather the seq shards back before the LM head (the head's
@@ -326,7 +332,7 @@ def build_llama_train_graph(cfg: LlamaConfig, micro_batch: int, seq_len: int,
                             dtype=torch.bfloat16, lr: float = 1e-4,
                             spec: Optional[ParallelSpec] = None,
                             graph: Optional[DefineAndRunGraph] = None,
-                            zero: bool = False
+                            zero: bool = False, recompute: bool = False
                             ) -> (DefineAndRunGraph, Dict):
     g = graph or DefineAndRunGraph("llama_train")
     spec = spec or ParallelSpec()
@@ -339,7 +345,8 @@ def build_llama_train_graph(cfg: LlamaConfig, micro_batch: int, seq_len: int,
         labels = ht.placeholder((micro_batch * seq_len,), dtype=torch.int64,
                                 name="labels", ds=spec.ds_tokens(0),
                                 device_group=spec.device_group)
-        model = LlamaLMHeadModel(cfg, spec, micro_batch, seq_len, dtype)
+        model = LlamaLMHeadModel(cfg, spec, micro_batch, seq_len, dtype,
+                                 recompute=recompute)
         loss, logits = model(input_ids, labels)
         loss_report = loss
         if spec.num_devices > 1:
