@@ -246,7 +246,8 @@ def build_gpt_train_graph(cfg: GPTConfig, micro_batch: int, seq_len: int,
                           dp: int = 1, device_group=None,
                           graph: Optional[DefineAndRunGraph] = None,
                           zero: bool = False,
-                          spec: Optional[ParallelSpec] = None
+                          spec: Optional[ParallelSpec] = None,
+                          recompute: bool = False
                           ) -> (DefineAndRunGraph, Dict):
     g = graph or DefineAndRunGraph("gpt_train")
     if spec is None:
