@@ -574,3 +574,17 @@ def as_strided(a, size, stride, offset=0):
     return _mk(_cg(), AsStridedOp(), [a],
                {"size": list(size), "stride": list(stride),
                 "offset": offset}, name="as_strided").output()
+
+
+def mat_dot(a, b):
+    """out[i, j] = a[i, j] * b[i] (reference MatDot.cc)."""
+    from .basics import _make as _mk
+    from .extra import MatDotOp
+    return _mk(_cg(), MatDotOp(), [a, b], name="mat_dot").output()
+
+
+def dynamic_concat(ts, dim: int = 0):
+    from .basics import _make as _mk
+    from .extra import DynamicConcatOp
+    return _mk(_cg(), DynamicConcatOp(), list(ts), {"dim": dim},
+               name="dynamic_concat").output()

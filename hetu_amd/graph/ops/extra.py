@@ -693,3 +693,41 @@ class AsStridedGradOp(OpInterface):
         dx.as_strided(op.attrs["size"], op.attrs["stride"],
                       op.attrs.get("offset", 0)).add_(gy)
         return [dx]
+
+
+@_register
+class MatDotOp(_AutogradOp):
+    """Row-wise dot product scaling: out[i, j] = a[i, j] * b[i, 0]
+    (reference graph/ops/MatDot.cc semantics: matrix x column broadcast)."""
+    type = "MatDot"
+
+    def fn(self, attrs):
+        return lambda a, b: a * b.reshape(-1, 1)
+
+
+class DynamicConcatOp(OpInterface):
+    """Concat along dim with runtime-ragged inputs padded to the static
+    max (reference dynamic_concatenate): output shape uses the declared
+    meta sizes; shorter runtime inputs are zero-padded."""
+    type = "DynamicConcat"
+
+    def infer_meta(self, attrs, inputs):
+        dim = attrs.get("dim", 0)
+        shape = list(inputs[0].shape)
+        shape[dim] = sum(int(t.shape[dim]) for t in inputs)
+        return [TensorMeta(shape, inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        dim = op.attrs.get("dim", 0)
+        outs = []
+        for t, decl in zip(inputs, op.inputs):
+            want = int(decl.shape[dim])
+            have = t.shape[dim]
+            if have < want:
+                pad_shape = list(t.shape)
+                pad_shape[dim] = want - have
+                t = torch.cat([t, t.new_zeros(pad_shape)], dim=dim)
+            elif have > want:
+                t = t.narrow(dim, 0, want)
+            outs.append(t)
+        return [torch.cat(outs, dim=dim)]
