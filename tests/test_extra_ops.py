@@ -305,3 +305,32 @@ def test_non_contiguous_inputs():
                         @ base_w.contiguous().t(), dim=-1)
     assert torch.allclose(out[0], ref, atol=1e-6)
     assert out[1].shape == (8, 16)
+
+
+def test_matdot_and_dynamic_concat():
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import (DefineAndRunGraph, pop_graph,
+                                      push_graph)
+    from hetu_amd.graph.ops import api as ht
+    g = DefineAndRunGraph("md")
+    push_graph(g)
+    try:
+        a = ht.placeholder((4, 5), name="a")
+        b = ht.placeholder((4,), name="b")
+        y = ht.mat_dot(a, b)
+        gs = ht.gradients([ht.reduce_sum(y)], [a, b])
+        c1 = ht.placeholder((2, 3), name="c1")
+        c2 = ht.placeholder((4, 3), name="c2")
+        cc = ht.dynamic_concat([c1, c2], dim=0)
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    av, bv = torch.randn(4, 5), torch.randn(4)
+    r, ga, gb, rc = g.run(
+        [y, gs[0], gs[1], cc],
+        {a: av, b: bv, c1: torch.ones(1, 3), c2: torch.ones(4, 3)},
+        ctx=ctx)
+    assert torch.allclose(r, av * bv[:, None])
+    assert torch.allclose(ga, bv[:, None].expand(4, 5))
+    assert torch.allclose(gb, av.sum(-1))
+    assert rc.shape == (6, 3) and rc[1].abs().sum() == 0
