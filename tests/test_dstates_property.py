@@ -1,0 +1,81 @@
+"""Property-based tests of the DistributedStates algebra (hypothesis):
+partition completeness, index-map consistency, shape roundtrips —
+invariants the reference's distributed_states.cc relies on everywhere."""
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from hetu_amd.parallel.dstates import DistributedStates, ds_from_index_table
+
+
+def _ds_strategy():
+    """Random valid ds: device_num = product of split factors x dup."""
+
+    @st.composite
+    def build(draw):
+        factors = draw(st.lists(st.sampled_from([2, 2, 3, 4]),
+                                min_size=0, max_size=3))
+        dims = draw(st.permutations([0, 1, 2]))
+        states = {}
+        order = []
+        for f, d in zip(factors, dims):
+            states[d] = f
+            order.append(d)
+        dup = draw(st.sampled_from([1, 2, 3]))
+        if dup > 1 or not states:
+            states[-1] = max(dup, 1) if (dup > 1 or not states) else 1
+            if -1 in states and states[-1] > 0:
+                order.append(-1)
+        n = 1
+        for v in states.values():
+            n *= v
+        return DistributedStates(n, states, order)
+    return build()
+
+
+@settings(max_examples=200, deadline=None)
+@given(_ds_strategy())
+def test_partition_covers_global_exactly(ds):
+    """Union of every device's local_slice = the global tensor; slices of
+    devices in the same dup group coincide, others tile disjointly."""
+    shape = (12, 12, 12)
+    full = torch.arange(12 ** 3).reshape(shape)
+    counts = torch.zeros(shape, dtype=torch.int64)
+    for dev in range(ds.device_num):
+        sl = ds.local_slice(shape, dev)
+        counts[sl] += 1
+    # every element covered exactly dup times
+    assert (counts == ds.dup).all(), (ds, counts.unique())
+
+
+@settings(max_examples=200, deadline=None)
+@given(_ds_strategy())
+def test_local_global_shape_roundtrip(ds):
+    shape = (12, 12, 12)
+    loc = ds.local_shape(shape)
+    assert tuple(ds.global_shape(loc)) == shape
+
+
+@settings(max_examples=200, deadline=None)
+@given(_ds_strategy())
+def test_index_table_roundtrip(ds):
+    """ds -> per-device state index table -> ds reconstructs the same
+    partition (same local slices for every device)."""
+    table = [ds.map_device_to_state_index(i) for i in range(ds.device_num)]
+    counts = {d: ds.get_dim(d) for d in ds.split_dims()}
+    ds2 = ds_from_index_table(ds.device_num, [
+        {d: t.get(d, 0) for d in counts} for t in table], counts)
+    shape = (12, 12, 12)
+    for dev in range(ds.device_num):
+        assert ds.local_slice(shape, dev) == ds2.local_slice(shape, dev), \
+            (ds, ds2, dev)
+
+
+@settings(max_examples=100, deadline=None)
+@given(_ds_strategy())
+def test_group_devices_along_partitions_devices(ds):
+    for d in ds.split_dims():
+        groups = ds.group_devices_along(d)
+        flat = sorted(x for g in groups for x in g)
+        assert flat == list(range(ds.device_num))
+        assert all(len(g) == ds.get_dim(d) for g in groups)
