@@ -79,3 +79,62 @@ def test_group_devices_along_partitions_devices(ds):
         flat = sorted(x for g in groups for x in g)
         assert flat == list(range(ds.device_num))
         assert all(len(g) == ds.get_dim(d) for g in groups)
+
+
+@settings(max_examples=150, deadline=None)
+@given(_ds_strategy(), st.integers(0, 2), st.sampled_from(
+    ["allgather", "reducescatter", "slice_dup", "allreduce"]))
+def test_comm_kind_deduction_consistency(ds, dim, move):
+    """deduce_comm_kind must recognize the canonical transitions built
+    from any source layout (reference Communication.h DoDeduceStates)."""
+    from hetu_amd.graph.ops.comm import deduce_comm_kind
+    n = ds.device_num
+    states = dict(ds.states)
+    order = list(ds.order)
+
+    def mk(st_, od_):
+        return DistributedStates(n, st_, od_)
+
+    if move == "allreduce":
+        if -2 in states or ds.dup == n or n == 1:
+            return
+        # partial over everything -> pure dup
+        src = mk({-2: n}, [-2])
+        dst = mk({-1: n}, [-1])
+        assert deduce_comm_kind(src, dst)[0] == "allreduce"
+    elif move == "allgather":
+        if ds.get_dim(dim) <= 1:
+            return
+        # split(dim) folds into dup
+        k = states.pop(dim)
+        states[-1] = states.get(-1, 1) * k
+        od = [-1 if d == dim else d for d in order]
+        dedup = []
+        for d in od:
+            if d not in dedup:
+                dedup.append(d)
+        dst = mk(states, dedup)
+        kind, d = deduce_comm_kind(ds, dst)
+        assert kind == "allgather" and d == dim, (ds, dst, kind)
+    elif move == "reducescatter":
+        if -2 in states or ds.dup <= 1 or dim in states:
+            return
+        # build a partial source, dst moves partial into split(dim)
+        k = states.pop(-1)
+        src_states = dict(states)
+        src_states[-2] = k
+        src = mk(src_states, [d for d in order if d != -1] + [-2])
+        dst_states = dict(states)
+        dst_states[dim] = k
+        dst = mk(dst_states, [d for d in order if d != -1] + [dim])
+        kind, d = deduce_comm_kind(src, dst)
+        assert kind == "reducescatter" and d == dim, (src, dst, kind)
+    else:  # slice_dup: dup splits into a new dim
+        if ds.dup <= 1 or dim in states:
+            return
+        k = states.pop(-1)
+        dst_states = dict(states)
+        dst_states[dim] = k
+        dst = mk(dst_states, [d for d in order if d != -1] + [dim])
+        kind, _ = deduce_comm_kind(ds, dst)
+        assert kind == "slice", (ds, dst, kind)
