@@ -138,3 +138,34 @@ def test_comm_kind_deduction_consistency(ds, dim, move):
         dst = mk(dst_states, [d for d in order if d != -1] + [dim])
         kind, _ = deduce_comm_kind(ds, dst)
         assert kind == "slice", (ds, dst, kind)
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.lists(st.integers(1, 120), min_size=1, max_size=40),
+       st.integers(0, 10_000))
+def test_pack_data_preserves_tokens(lengths, seed):
+    """Packing invariants (reference Bucket.pack_data): every token lands
+    in exactly one bin at the cu_seqlens offsets, no segment crosses
+    max_seqlen, alignment respected."""
+    from hetu_amd.data.bucket import Bucket
+    torch.manual_seed(seed)
+    b = Bucket(max_seqlen=128, pad_token=-1, alignment=16)
+    seqs = [torch.randint(0, 1000, (n,)) for n in lengths]
+    for s in seqs:
+        b.add(s)
+    tokens, cus = b.pack_data()
+    assert tokens.shape[1] == 128
+    # collect back every packed segment
+    seen = []
+    for bin_i, cu in enumerate(cus):
+        cu = cu.tolist()
+        assert cu[-1] <= 128
+        for s0, s1 in zip(cu[:-1], cu[1:]):
+            assert s0 % 16 == 0
+            seg = tokens[bin_i, s0:s1]
+            real = seg[seg != -1]
+            seen.append(real)
+    # multiset of sequences must match (order-independent)
+    got = sorted([tuple(t.tolist()) for t in seen])
+    want = sorted([tuple(s.tolist()) for s in seqs])
+    assert got == want
