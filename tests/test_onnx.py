@@ -62,3 +62,49 @@ def test_onnx_embedding_and_matmul(tmp_path):
     y1, = g.run([y], {ids: idv}, ctx=ctx1)
     y2, = g2.run([outputs[0]], {inputs["ids"]: idv}, ctx=ctx2)
     assert torch.allclose(y1, y2, atol=1e-5)
+
+
+def test_onnx_roundtrip_fuzz():
+    """Random MLP-ish graphs roundtrip through ONNX bytes with identical
+    outputs (40 cases)."""
+    from hetu_amd.engine.runner import prepare_run_context
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    ACTS = ["gelu", "relu", "sigmoid", "tanh"]
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.lists(st.tuples(st.integers(2, 24),
+                              st.integers(0, len(ACTS) - 1)),
+                    min_size=1, max_size=4),
+           st.integers(0, 9999))
+    def run(widths, seed):
+        torch.manual_seed(seed)
+        g = DefineAndRunGraph("f")
+        push_graph(g)
+        try:
+            d = 8
+            x = ht.placeholder((3, d), name="x")
+            cur = x
+            for li, (w_out, ai) in enumerate(widths):
+                w = ht.variable(torch.randn(w_out, d) * 0.3,
+                                name=f"w{li}")
+                bvar = ht.variable(torch.randn(w_out) * 0.1,
+                                   name=f"b{li}")
+                cur = getattr(ht, ACTS[ai])(ht.linear(cur, w, bvar))
+                d = w_out
+            y = ht.softmax(cur, dim=-1)
+        finally:
+            pop_graph()
+        blob = export_onnx(g, [y])
+        g2, inputs, outputs = import_onnx(blob)
+        xd = torch.randn(3, 8)
+        ctx1 = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+        ctx2 = prepare_run_context(g2, torch.device("cpu"),
+                                   use_comm=False)
+        y1, = g.run([y], {x: xd}, ctx=ctx1)
+        y2, = g2.run([outputs[0]], {list(inputs.values())[0]: xd},
+                     ctx=ctx2)
+        assert torch.allclose(y1, y2, atol=1e-5)
+
+    run()
