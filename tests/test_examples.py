@@ -42,3 +42,12 @@ def test_malleus_example_runs():
         capture_output=True, text=True, timeout=300)
     assert p.returncode == 0, f"{p.stdout}\n{p.stderr}"
     assert "shares=" in p.stdout
+
+
+def test_sft_example_runs():
+    p = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples", "sft",
+                                      "sft_train.py")],
+        capture_output=True, text=True, timeout=600)
+    assert p.returncode == 0, f"{p.stdout}\n{p.stderr}"
+    assert "loss" in p.stdout
