@@ -98,6 +98,11 @@ torch::Tensor flash_attn_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
 // rope.hip
 void rope_qk_inplace(torch::Tensor qkv, torch::Tensor cs, torch::Tensor sn,
                      int64_t n_rot, int64_t D, int64_t sign);
+// attention_v3.hip (experimental round-2 candidate)
+std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q,
+                                             torch::Tensor k,
+                                             torch::Tensor v, bool causal,
+                                             double scale);
 
 // embed_cache.cpp
 void register_embed_cache(pybind11::module& m);
@@ -142,4 +147,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd_qkv", &flash_attn_fwd_qkv);
   m.def("flash_attn_bwd_qkv", &flash_attn_bwd_qkv);
   m.def("rope_qk_inplace", &rope_qk_inplace);
+  m.def("flash_attn_fwd_v3", &flash_attn_fwd_v3);
 }

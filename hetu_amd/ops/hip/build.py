@@ -26,6 +26,7 @@ SOURCES = [
     "gemm.hip",
     "attention.hip",
     "attention_v2.hip",
+    "attention_v3.hip",
     "attention_bwd_v2.hip",
     "quant.hip",
     "galvatron_dp.cpp",

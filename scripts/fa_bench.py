@@ -33,6 +33,13 @@ for causal in (True, False):
     o, lse = ext.flash_attn_fwd(q, kk, vv, causal, scale)
     t = bench(lambda: ext.flash_attn_fwd(q, kk, vv, causal, scale))
     print(f"fwd causal={causal}: {t*1e3:.3f} ms  {flops/t/1e12:.0f} TF/s")
+    if hasattr(ext, "flash_attn_fwd_v3"):
+        o3, lse3 = ext.flash_attn_fwd_v3(q, kk, vv, causal, scale)
+        err = (o3.float() - o.float()).abs().max().item()
+        el = (lse3 - lse).abs().max().item()
+        t3 = bench(lambda: ext.flash_attn_fwd_v3(q, kk, vv, causal, scale))
+        print(f"fwd v3 causal={causal}: {t3*1e3:.3f} ms  "
+              f"{flops/t3/1e12:.0f} TF/s  (err {err:.3e} lse {el:.3e})")
     do = torch.randn_like(o)
     tb = bench(lambda: ext.flash_attn_bwd(do, q, kk, vv, o, lse, causal,
                                           scale), iters=10)
