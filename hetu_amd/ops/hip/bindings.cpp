@@ -103,6 +103,13 @@ std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q,
                                              torch::Tensor k,
                                              torch::Tensor v, bool causal,
                                              double scale);
+std::vector<torch::Tensor> flash_attn_bwd_v3(torch::Tensor dout,
+                                             torch::Tensor q,
+                                             torch::Tensor k,
+                                             torch::Tensor v,
+                                             torch::Tensor out,
+                                             torch::Tensor lse,
+                                             bool causal, double scale);
 
 // embed_cache.cpp
 void register_embed_cache(pybind11::module& m);
@@ -148,4 +155,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_bwd_qkv", &flash_attn_bwd_qkv);
   m.def("rope_qk_inplace", &rope_qk_inplace);
   m.def("flash_attn_fwd_v3", &flash_attn_fwd_v3);
+  m.def("flash_attn_bwd_v3", &flash_attn_bwd_v3);
 }

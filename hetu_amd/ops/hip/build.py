@@ -27,6 +27,7 @@ SOURCES = [
     "attention.hip",
     "attention_v2.hip",
     "attention_v3.hip",
+    "attention_bwd_v3.hip",
     "attention_bwd_v2.hip",
     "quant.hip",
     "galvatron_dp.cpp",

@@ -44,3 +44,13 @@ for causal in (True, False):
     tb = bench(lambda: ext.flash_attn_bwd(do, q, kk, vv, o, lse, causal,
                                           scale), iters=10)
     print(f"bwd causal={causal}: {tb*1e3:.3f} ms  {2.5*flops/tb/1e12:.0f} TF/s")
+    if hasattr(ext, "flash_attn_bwd_v3") and H == Hkv:
+        ref = ext.flash_attn_bwd(do, q, kk, vv, o, lse, causal, scale)
+        g3 = ext.flash_attn_bwd_v3(do, q, kk, vv, o, lse, causal, scale)
+        errs = [(a.float() - b.float()).abs().max().item()
+                for a, b in zip(g3, ref)]
+        t3 = bench(lambda: ext.flash_attn_bwd_v3(do, q, kk, vv, o, lse,
+                                                 causal, scale), iters=10)
+        print(f"bwd v3 causal={causal}: {t3*1e3:.3f} ms  "
+              f"{2.5*flops/t3/1e12:.0f} TF/s  (err dq/dk/dv "
+              f"{errs[0]:.2e}/{errs[1]:.2e}/{errs[2]:.2e})")
