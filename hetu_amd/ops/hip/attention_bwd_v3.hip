@@ -8,6 +8,9 @@
 //       causal fwd 325 -> 494 TF/s together with the grid order).
 //   B2. walking-pointer glds source addressing with a clamped fallback
 //       for the ragged tail (v2 recomputes min()*stride per tile).
+//   B3. launch_bounds occupancy 1: at 96/148 KiB LDS only one block fits
+//       per CU anyway, so the 2-block register budget of v2 only forces
+//       spills (v2's own dkv could take the same change — round 2).
 // Non-GQA (H == Hkv), BHSD contiguous, D=128 only.
 #include <torch/extension.h>
 #include "ext_stream.h"
@@ -104,7 +107,7 @@ DEV int kswz_row3(int row, int byte_in_row) {
 // dq kernel v3: 8 waves x 32 q rows; 3-ring KV staging, counted vmcnt.
 // ===========================================================================
 template <int D>
-__global__ __launch_bounds__(THREADS, 2) void fa3_bwd_dq_kernel(
+__global__ __launch_bounds__(THREADS, 1) void fa3_bwd_dq_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
@@ -311,7 +314,7 @@ __global__ __launch_bounds__(THREADS, 2) void fa3_bwd_dq_kernel(
 // dkv kernel v3: 8 waves x 32 keys; 3-ring Q/dO staging, counted vmcnt.
 // ===========================================================================
 template <int D>
-__global__ __launch_bounds__(THREADS, 2) void fa3_bwd_dkv_kernel(
+__global__ __launch_bounds__(THREADS, 1) void fa3_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
@@ -320,9 +323,11 @@ __global__ __launch_bounds__(THREADS, 2) void fa3_bwd_dkv_kernel(
   static_assert(D == 128);
   constexpr int QB = 32 * 256;     // 8 KiB per rm image
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  // dkv uses a 2-buffer ring with issue-at-end + counted vmcnt (the 3rd
+  // buffer's live state spilled past the 2-waves/SIMD VGPR budget)
   auto q_lds = [&](int bb) -> char* { return smem + bb * QB; };
-  auto do_lds = [&](int bb) -> char* { return smem + (NBUF + bb) * QB; };
-  char* vw_base = smem + 2 * NBUF * QB;
+  auto do_lds = [&](int bb) -> char* { return smem + (2 + bb) * QB; };
+  char* vw_base = smem + 4 * QB;
   char* pw_base = vw_base + 8 * 8192;
 
   const int bh = blockIdx.x;
@@ -384,25 +389,11 @@ __global__ __launch_bounds__(THREADS, 2) void fa3_bwd_dkv_kernel(
   if (causal) t_start = max(0, (kb0 - diag) / 32);
   const int wave_kmin = kb0 + wid * 32;
 
-  const bf16* qp = Qb + (int64_t)(t_start * 32 + qrow_c) * D + qd_c;
-  const bf16* dop = dOb + (int64_t)(t_start * 32 + qrow_c) * D + qd_c;
-  constexpr int64_t QSTEP = 32 * D;
-
-#define DKV3_FAST(buf)                                                      \
-  do {                                                                      \
-    __builtin_amdgcn_global_load_lds(                                       \
-        (const __attribute__((address_space(1))) void*)qp,                  \
-        (__attribute__((address_space(3))) void*)(q_lds(buf) + pos0),       \
-        16, 0, 0);                                                          \
-    __builtin_amdgcn_global_load_lds(                                       \
-        (const __attribute__((address_space(1))) void*)dop,                 \
-        (__attribute__((address_space(3))) void*)(do_lds(buf) + pos0),      \
-        16, 0, 0);                                                          \
-    qp += QSTEP;                                                            \
-    dop += QSTEP;                                                           \
-  } while (0)
-
-#define DKV3_CLAMPED(qt0, buf)                                              \
+  // dkv keeps v2-style clamped addressing (1 slot/lane, the address math
+  // is cheap); the v3 win here is the 3-ring counted-vmcnt pipeline —
+  // the kernel sits exactly at the 2-waves/SIMD VGPR budget, so no
+  // walking pointers
+#define DKV3_ISSUE(qt0, buf)                                                \
   do {                                                                      \
     int qr_ = min((qt0) + qrow_c, S - 1);                                   \
     __builtin_amdgcn_global_load_lds(                                       \
@@ -415,14 +406,6 @@ __global__ __launch_bounds__(THREADS, 2) void fa3_bwd_dkv_kernel(
             dOb + (int64_t)qr_ * D + qd_c),                                 \
         (__attribute__((address_space(3))) void*)(do_lds(buf) + pos0),      \
         16, 0, 0);                                                          \
-    qp += QSTEP;                                                            \
-    dop += QSTEP;                                                           \
-  } while (0)
-
-#define DKV3_ISSUE(qt0, buf)                                                \
-  do {                                                                      \
-    if ((qt0) + 32 <= S) DKV3_FAST(buf);                                    \
-    else DKV3_CLAMPED(qt0, buf);                                            \
   } while (0)
 
   f32x16 dk_acc[4], dv_acc[4];
@@ -436,12 +419,11 @@ __global__ __launch_bounds__(THREADS, 2) void fa3_bwd_dkv_kernel(
 
   for (int t = t_start; t < n_q_tiles; ++t) {
     const int qt0 = t * 32;
-    const int cur = (t - t_start) % NBUF;
-    // allow the newest tile's 2 loads in flight (the V image has long
-    // landed once the first wait retires)
+    const int cur = (t - t_start) & 1;
+    // allow the newest issue's 2 loads to stay in flight; the V image
+    // and tile t have landed once this retires
     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    if (t + 2 < n_q_tiles) DKV3_ISSUE(qt0 + 64, (t - t_start + 2) % NBUF);
 
     const bool active = !causal || (qt0 + 31 >= wave_kmin - diag);
     if (active) {
@@ -522,10 +504,12 @@ __global__ __launch_bounds__(THREADS, 2) void fa3_bwd_dkv_kernel(
             dstf, qa3, dk_acc[3], 0, 0, 0);
       }
     }
+    // all waves done reading buffer `cur` -> refill it with tile t+2
+    // (the DMA then lands under tile t+1's compute)
+    __builtin_amdgcn_s_barrier();
+    if (t + 2 < n_q_tiles) DKV3_ISSUE(qt0 + 64, cur);
   }
 #undef DKV3_ISSUE
-#undef DKV3_FAST
-#undef DKV3_CLAMPED
 
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt) {
@@ -604,7 +588,7 @@ std::vector<torch::Tensor> flash_attn_bwd_v3(torch::Tensor dout,
   }
   {
     dim3 grid(B * H, (Skv + 255) / 256);
-    size_t lds = 2 * NBUF * (size_t)32 * 256 + 8 * 8192 + 8 * 4608;
+    size_t lds = 4 * (size_t)32 * 256 + 8 * 8192 + 8 * 4608;
     hipLaunchKernelGGL(fa3_bwd_dkv_kernel<128>, grid, dim3(THREADS), lds,
                        stream, (const bf16*)q.data_ptr(),
                        (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
