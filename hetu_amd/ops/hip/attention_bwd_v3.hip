@@ -227,6 +227,7 @@ __global__ __launch_bounds__(THREADS, 1) void fa3_bwd_dq_kernel(
 
   DQ3_ISSUE(0, 0);
   if (n_tiles > 1) DQ3_ISSUE(64, 1);
+  else DQ3_CLAMPED(0, 1);     // duplicate: keeps the counted wait sound
 
   for (int t = 0; t < n_tiles; ++t) {
     const int k0 = t * 64;
@@ -416,6 +417,7 @@ __global__ __launch_bounds__(THREADS, 1) void fa3_bwd_dkv_kernel(
 
   DKV3_ISSUE(t_start * 32, 0);
   if (t_start + 1 < n_q_tiles) DKV3_ISSUE(t_start * 32 + 32, 1);
+  else DKV3_ISSUE(t_start * 32, 1);   // duplicate keeps the wait sound
 
   for (int t = t_start; t < n_q_tiles; ++t) {
     const int qt0 = t * 32;

@@ -148,7 +148,11 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
   // loop head means "tile t landed, tile t+1 still flying" — the DMA for
   // t+2 then issues after the barrier and lands under two tiles of MFMAs)
   ISSUE_GLDS(0, 0);
+  // the counted wait below assumes a full tile's loads may trail behind
+  // the newest ISSUE; with a single tile, issue a harmless duplicate so
+  // tile 0 is guaranteed landed when its compute starts
   if (n_tiles > 1) ISSUE_GLDS(BN, 1);
+  else ISSUE_GLDS(0, 1);
 
   for (int t = 0; t < n_tiles; ++t) {
     const int k0 = t * BN;

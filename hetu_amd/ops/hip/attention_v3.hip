@@ -182,6 +182,7 @@ __global__ __launch_bounds__(THREADS, 2) void fa3_fwd_kernel(
 
   ISSUE3(0, 0);
   if (n_tiles > 1) ISSUE3(BN, 1);
+  else ISSUE_CLAMPED(0, 1);   // duplicate: keeps the counted wait sound
 
   for (int t = 0; t < n_tiles; ++t) {
     const int k0 = t * BN;
