@@ -315,7 +315,8 @@ __global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dq_kernel(
 // dkv kernel: 8 waves x 32 keys (exclusive); loops over 32-q tiles.
 // ===========================================================================
 template <int D>
-__global__ __launch_bounds__(THREADS, 2) void fa2_bwd_dkv_kernel(
+// 132 KiB LDS -> occupancy 1 by LDS; see the fwd note on launch bounds
+__global__ __launch_bounds__(THREADS, 1) void fa2_bwd_dkv_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, const bf16* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,

@@ -44,7 +44,9 @@ DEV int voff(int key, int d) {
 }
 
 template <int D>  // D = head dim (128)
-__global__ __launch_bounds__(THREADS, 2) void fa2_fwd_kernel(
+// 96 KiB LDS -> one workgroup per CU regardless; a 2-block register cap
+// would only force tighter allocation for a block that cannot schedule
+__global__ __launch_bounds__(THREADS, 1) void fa2_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, bf16* __restrict__ O,
     float* __restrict__ LSE, int B, int H, int Hkv, int S, int Skv,

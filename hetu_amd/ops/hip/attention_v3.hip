@@ -37,7 +37,7 @@ DEV int voff3(int key, int d) {
 }
 
 template <int D>
-__global__ __launch_bounds__(THREADS, 2) void fa3_fwd_kernel(
+__global__ __launch_bounds__(THREADS, 1) void fa3_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, bf16* __restrict__ O,
     float* __restrict__ LSE, int B, int H, int Hkv, int S, int Skv,
