@@ -696,9 +696,37 @@ class ReduceOp_(OpInterface):
         gr = _g(op.outputs[0])
         mode = op.attrs["mode"]
         if mode not in ("sum", "mean"):
-            raise NotImplementedError("grad for max/min reduce")
+            # max/min: subgradient routed to the extremal positions
+            # (split across ties, matching the mask/count convention)
+            return [_make(gr, ReduceExtremumGradOp(),
+                          [g[0], op.inputs[0], op.outputs[0]],
+                          dict(op.attrs)).output()]
         return [_make(gr, ReduceGradOp(), [g[0], op.inputs[0]],
                       dict(op.attrs)).output()]
+
+
+class ReduceExtremumGradOp(OpInterface):
+    """dx for max/min reduce: gy spread over argext positions, ties
+    sharing equally."""
+    type = "ReduceExtremumGrad"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[1].shape, inputs[1].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        gy, x, y = inputs
+        dim = op.attrs.get("dim")
+        keepdim = op.attrs.get("keepdim", False)
+        xf = x.float()
+        if dim is None:
+            mask = (xf == y.float()).to(xf.dtype)
+            return [(mask / mask.sum().clamp(min=1)
+                     * gy.float()).to(x.dtype)]
+        ye = y.float() if keepdim else y.float().unsqueeze(dim)
+        ge = gy.float() if keepdim else gy.float().unsqueeze(dim)
+        mask = (xf == ye).to(xf.dtype)
+        return [(mask / mask.sum(dim, keepdim=True).clamp(min=1)
+                 * ge).to(x.dtype)]
 
 
 class CheckFiniteOp(OpInterface):
@@ -908,12 +936,28 @@ class MatMul2DOp(OpInterface):
         a, b = op.inputs
         ta = op.attrs.get("trans_a", False)
         tb = op.attrs.get("trans_b", False)
-        if ta or tb:
-            raise NotImplementedError("grad for transposed matmul")
         gy = g[0]
-        da = _make(gr, MatMul2DOp(), [gy, b],
-                   {"trans_a": False, "trans_b": True}).output()
-        db = _make(gr, MatMulGradBOp(), [a, gy]).output()
+        if not ta and not tb:
+            da = _make(gr, MatMul2DOp(), [gy, b],
+                       {"trans_a": False, "trans_b": True}).output()
+            db = _make(gr, MatMulGradBOp(), [a, gy]).output()
+            return [da, db]
+        # y = op_a(a) @ op_b(b): standard transposed-matmul adjoints
+        # (2-D operands; the batched leading-dim case routes through the
+        # untransposed path above)
+        if not ta and tb:       # y = a  @ b^T
+            da = _make(gr, MatMul2DOp(), [gy, b], {}).output()
+            db = _make(gr, MatMul2DOp(), [gy, a],
+                       {"trans_a": True}).output()
+        elif ta and not tb:     # y = a^T @ b
+            da = _make(gr, MatMul2DOp(), [b, gy],
+                       {"trans_b": True}).output()
+            db = _make(gr, MatMul2DOp(), [a, gy], {}).output()
+        else:                   # y = a^T @ b^T
+            da = _make(gr, MatMul2DOp(), [b, gy],
+                       {"trans_a": True, "trans_b": True}).output()
+            db = _make(gr, MatMul2DOp(), [gy, a],
+                       {"trans_a": True, "trans_b": True}).output()
         return [da, db]
 
 

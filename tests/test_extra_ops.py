@@ -334,3 +334,47 @@ def test_matdot_and_dynamic_concat():
     assert torch.allclose(ga, bv[:, None].expand(4, 5))
     assert torch.allclose(gb, av.sum(-1))
     assert rc.shape == (6, 3) and rc[1].abs().sum() == 0
+
+
+def test_transposed_matmul_and_extremum_reduce_grads():
+    """Gradients for all four matmul transpose modes and max-reduce
+    (previously unimplemented corners)."""
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import (DefineAndRunGraph, pop_graph,
+                                      push_graph)
+    from hetu_amd.graph.ops import api as ht
+    torch.manual_seed(0)
+    for ta, tb in [(False, True), (True, False), (True, True)]:
+        g = DefineAndRunGraph("t")
+        push_graph(g)
+        try:
+            A = ht.placeholder((4, 5) if not ta else (5, 4), name="a")
+            Bp = ht.placeholder((6, 5) if tb else (5, 6), name="b")
+            y = ht.matmul(A, Bp, trans_a=ta, trans_b=tb)
+            gs = ht.gradients([ht.reduce_sum(ht.mul(y, y))], [A, Bp])
+        finally:
+            pop_graph()
+        ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+        av = torch.randn(*([4, 5] if not ta else [5, 4])) \
+            .requires_grad_(True)
+        bv = torch.randn(*([6, 5] if tb else [5, 6])).requires_grad_(True)
+        res = g.run(gs, {A: av.detach(), Bp: bv.detach()}, ctx=ctx)
+        aa = av.t() if ta else av
+        bb = bv.t() if tb else bv
+        (aa @ bb).pow(2).sum().backward()
+        assert torch.allclose(res[0], av.grad, atol=1e-5), (ta, tb)
+        assert torch.allclose(res[1], bv.grad, atol=1e-5), (ta, tb)
+
+    g = DefineAndRunGraph("m")
+    push_graph(g)
+    try:
+        x = ht.placeholder((3, 4), name="x")
+        y = ht.reduce_max(x, dim=1)
+        gs = ht.gradients([ht.reduce_sum(ht.mul(y, y))], [x])
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    xv = torch.randn(3, 4).requires_grad_(True)
+    r, = g.run(gs, {x: xv.detach()}, ctx=ctx)
+    xv.amax(1).pow(2).sum().backward()
+    assert torch.allclose(r, xv.grad, atol=1e-5)
