@@ -74,9 +74,6 @@ class CommOp(OpInterface):
             op.attrs["kind"] = ("identity", None)
         else:
             op.attrs["kind"] = deduce_comm_kind(src, dst)
-            if op.attrs["kind"][0] == "generic":
-                raise NotImplementedError(
-                    f"generic redistribution {src} -> {dst} not supported yet")
         op.outputs[0].ds = dst
         op.outputs[0].device_group = op.inputs[0].device_group
 
@@ -127,6 +124,22 @@ class CommOp(OpInterface):
             idx = src.map_device_to_state_index(my).get(d, 0)
             out.narrow(d, idx * x.shape[d], x.shape[d]).copy_(x)
             return [out]
+        if kind == "generic":
+            # total fallback for arbitrary layout transitions (reference
+            # Communication.h falls back to gather+redistribute too):
+            # reduce partial, gather every split dim to the full tensor,
+            # then slice down to the destination layout.  Costs one full
+            # materialization — fine for the rare resharding edges the
+            # faster kinds do not cover.
+            out = x
+            if src.partial > 1:
+                ranks = _ranks(dg, src.group_devices_along(-2), my)
+                out = ctx.comm.allreduce(out, ranks)
+            for d in src.split_dims():
+                ranks = _ranks(dg, src.group_devices_along(d), my)
+                out = ctx.comm.allgather(out, ranks, dim=d)
+            gshape = tuple(out.shape)
+            return [out[dst.local_slice(gshape, my)].contiguous()]
         raise NotImplementedError(kind)
 
     def gradient(self, op, g):
