@@ -150,3 +150,38 @@ def mcmc_search(shape: ModelShape, seq_len: int, world: int,
         if c < best_c:
             best, best_c = cand, c
     return best, best_c
+
+
+def hetero_pipeline_partition(layer_costs: Sequence[float], pp: int,
+                              stage_speeds: Sequence[float]
+                              ) -> Tuple[List[int], float]:
+    """Malleus-style heterogeneous pipeline partition: stage i's wall time
+    is its layer-cost sum divided by its (relative) speed; the bottleneck
+    partition assigns fewer layers to slower stages.  Returns (layer
+    counts per stage, bottleneck time)."""
+    n = len(layer_costs)
+    pp = min(pp, n)
+    assert len(stage_speeds) == pp
+    prefix = [0.0]
+    for c in layer_costs:
+        prefix.append(prefix[-1] + c)
+    INF = float("inf")
+    best = [[INF] * (n + 1) for _ in range(pp + 1)]
+    cut = [[0] * (n + 1) for _ in range(pp + 1)]
+    best[0][0] = 0.0
+    for s in range(1, pp + 1):
+        spd = max(stage_speeds[s - 1], 1e-9)
+        for j in range(1, n + 1):
+            for i in range(s - 1, j):
+                stage = (prefix[j] - prefix[i]) / spd
+                v = max(best[s - 1][i], stage)
+                if v < best[s][j]:
+                    best[s][j] = v
+                    cut[s][j] = i
+    counts = []
+    j = n
+    for s in range(pp, 0, -1):
+        i = cut[s][j]
+        counts.append(j - i)
+        j = i
+    return counts[::-1], best[pp][n]

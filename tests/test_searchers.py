@@ -42,3 +42,15 @@ def test_mcmc_finds_feasible_strategy():
     bad.tp, bad.dp, bad.pp, bad.micro_batch = 8, 1, 1, 1
     cm = CostModel(shape, 2048)
     assert c <= cm.evaluate(bad, 64)["time"] * 1.001
+
+
+def test_hetero_pipeline_partition_shifts_load():
+    """A slow stage gets fewer layers (Malleus hetero-pp resolution)."""
+    from hetu_amd.galvatron.searchers import hetero_pipeline_partition
+    costs = [1.0] * 12
+    even, t_even = hetero_pipeline_partition(costs, 3, [1.0, 1.0, 1.0])
+    assert even == [4, 4, 4]
+    skew, t_skew = hetero_pipeline_partition(costs, 3, [1.0, 0.5, 1.0])
+    assert sum(skew) == 12
+    assert skew[1] < skew[0] and skew[1] < skew[2], skew
+    assert t_skew >= t_even
