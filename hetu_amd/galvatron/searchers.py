@@ -160,7 +160,13 @@ def hetero_pipeline_partition(layer_costs: Sequence[float], pp: int,
     partition assigns fewer layers to slower stages.  Returns (layer
     counts per stage, bottleneck time)."""
     n = len(layer_costs)
-    pp = min(pp, n)
+    if pp > n:
+        # fewer layers than stages: keep the fastest n stages (the DP below
+        # would have no valid assignment for an empty stage)
+        order = sorted(range(pp), key=lambda i: -stage_speeds[i])[:n]
+        keep = sorted(order)
+        stage_speeds = [stage_speeds[i] for i in keep]
+        pp = n
     assert len(stage_speeds) == pp
     prefix = [0.0]
     for c in layer_costs:

@@ -139,7 +139,15 @@ class CommOp(OpInterface):
                 ranks = _ranks(dg, src.group_devices_along(d), my)
                 out = ctx.comm.allgather(out, ranks, dim=d)
             gshape = tuple(out.shape)
-            return [out[dst.local_slice(gshape, my)].contiguous()]
+            out = out[dst.local_slice(gshape, my)].contiguous()
+            if dst.partial > 1 and \
+                    dst.map_device_to_state_index(my).get(-2, 0) != 0:
+                # adjoint-of-slice convention (like the zeropad kind): only
+                # the partial-index-0 rank carries the value, the rest hold
+                # zeros, so a later partial-sum reduction is exact instead
+                # of overcounting by the partial factor.
+                out = torch.zeros_like(out)
+            return [out]
         raise NotImplementedError(kind)
 
     def gradient(self, op, g):

@@ -292,7 +292,15 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
         # ---- backward fetch set ----
         xs = params + ([] if is_first else [h["act_in"]])
         if is_last:
-            grads = g.gradients([loss], xs)
+            # d-loss seed placeholder so an fp16 GradScaler can scale the
+            # backward (PipelineRunner feeds scaler.scale, or 1.0); grads
+            # are unscaled after accumulation (engine/amp.py unscale_).
+            loss_seed = ht.placeholder(tuple(loss.shape),
+                                       dtype=torch.float32,
+                                       name="loss_seed", ds=loss.ds,
+                                       device_group=spec.device_group)
+            h["loss_seed"] = loss_seed
+            grads = g.gradients([loss], xs, grad_ys=[loss_seed])
         else:
             grad_in = ht.placeholder((B, S, cfg.hidden), dtype=dtype,
                                      name="grad_in", ds=ds_in,

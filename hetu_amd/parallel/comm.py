@@ -179,6 +179,19 @@ class CommBackend:
             for r in reqs:
                 r.wait()
 
+    def allgather_object(self, obj, ranks: Optional[List[int]] = None):
+        """All-gather arbitrary picklable objects (used for checkpoint
+        shard-index coordination, not the hot path)."""
+        if not dist.is_initialized():
+            return [obj]
+        ranks = ranks or list(range(self.world_size))
+        if len(ranks) <= 1:
+            return [obj]
+        grp = self.group(ranks)
+        out = [None] * len(ranks)
+        dist.all_gather_object(out, obj, group=grp)
+        return out
+
     def barrier(self):
         if dist.is_initialized():
             dist.barrier()

@@ -168,6 +168,15 @@ class PipelineRunner:
         if not self.is_last:
             feed = dict(feed)
             feed[h["grad_in"]] = gin
+        elif "loss_seed" in h:
+            # scale the backward seed (reference gradscaler.cc semantics):
+            # fp16 grads flow through the stage graph multiplied by
+            # scaler.scale so they stay above the fp16 underflow floor;
+            # unscale_ divides the accumulated fp32 buffers back down.
+            feed = dict(feed)
+            scale = self.scaler.scale if self.scaler is not None else 1.0
+            feed[h["loss_seed"]] = torch.tensor(
+                float(scale), dtype=torch.float32, device=self.device)
         fetches = ([] if self.is_first else [h["dx"]]) + h["param_grads"]
         res = self.stage.graph.run(fetches, feed, ctx=self.ctx,
                                    seed_values=cache)

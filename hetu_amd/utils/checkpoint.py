@@ -83,66 +83,100 @@ def _is_writer(p, comm: Optional[CommBackend]) -> bool:
     return st.get(-1, 0) == 0
 
 
-def save_model(params: List, path: str, comm: Optional[CommBackend] = None,
-               optimizer_states: Optional[Dict] = None,
-               max_shard_bytes: int = 8 << 30) -> None:
-    """params: graph parameter tensors (with .ds/.device_group/.name).
-    Writes model-XXXXX-of-YYYYY.safetensors + model.safetensors.index.json
-    (HF layout).  optimizer_states: {name: {m, v, step, master}} saved to
-    optim-<rank-set>.safetensors alongside."""
-    from safetensors.torch import save_file
-    comm = comm or comm_backend()
-    os.makedirs(path, exist_ok=True)
+def _snapshot_model(params: List, comm: Optional[CommBackend],
+                    optimizer_states: Optional[Dict],
+                    max_shard_bytes: int):
+    """Synchronous phase of a checkpoint save: run the gather collectives,
+    copy every shard to CPU, and coordinate the global HF shard numbering
+    (one allgather_object).  Returns a plan that phase 2 writes with pure
+    file IO — safe to run on a background thread while training continues
+    (no live tensors, no collectives: reference model_saver.py copies state
+    before handing off to its thread too)."""
     mine: Dict[str, torch.Tensor] = {}
     for p in params:
         data = p.get_data()
         full = _gather_global(p, data, comm)
         if _is_writer(p, comm):
-            mine[p.name.split(":")[0]] = full.cpu()
-    # shard my tensors into files; writer set is disjoint across ranks for
-    # pp (different params) and dup groups (leader only)
+            mine[p.name.split(":")[0]] = full.detach().cpu().clone()
+    # partition my tensors into shards; writer set is disjoint across ranks
+    # for pp (different params) and dup groups (leader only)
     rank = comm.rank if comm else 0
-    files: List[Dict[str, torch.Tensor]] = []
+    shards: List[Dict[str, torch.Tensor]] = []
     cur: Dict[str, torch.Tensor] = {}
     size = 0
     for name, t in mine.items():
         nb = t.numel() * t.element_size()
         if cur and size + nb > max_shard_bytes:
-            files.append(cur)
+            shards.append(cur)
             cur, size = {}, 0
         cur[name] = t
         size += nb
     if cur:
-        files.append(cur)
-    index = {"metadata": {}, "weight_map": {}}
-    for i, f in enumerate(files):
-        fname = f"model-r{rank:03d}-{i:05d}.safetensors"
-        save_file(f, os.path.join(path, fname))
-        for name in f:
-            index["weight_map"][name] = fname
-    with open(os.path.join(path, f"index-r{rank:03d}.json"), "w") as fh:
-        json.dump(index, fh)
-    if comm:
-        comm.barrier()
-    # merge rank indexes into the canonical one (rank 0)
-    if rank == 0:
-        merged = {"metadata": {}, "weight_map": {}}
-        for fn in sorted(os.listdir(path)):
-            if fn.startswith("index-r") and fn.endswith(".json"):
-                with open(os.path.join(path, fn)) as fh:
-                    merged["weight_map"].update(json.load(fh)["weight_map"])
-        with open(os.path.join(path, "model.safetensors.index.json"),
-                  "w") as fh:
-            json.dump(merged, fh, indent=1)
+        shards.append(cur)
+    # coordinate global `model-XXXXX-of-YYYYY.safetensors` numbering
+    # (HF convention, reference ht_safetensors WEIGHTS_NAME-{i+1}-of-{n})
+    meta_mine = [([n for n in s],
+                  sum(t.numel() * t.element_size() for t in s.values()))
+                 for s in shards]
+    all_meta = comm.allgather_object(meta_mine) if comm else [meta_mine]
+    n_total = sum(len(m) for m in all_meta)
+    weight_map: Dict[str, str] = {}
+    total_size = 0
+    gi = 0
+    my_names: List[str] = []
+    for r, m in enumerate(all_meta):
+        for names, nbytes in m:
+            if n_total == 1:
+                fname = "model.safetensors"
+            else:
+                fname = f"model-{gi + 1:05d}-of-{n_total:05d}.safetensors"
+            for n in names:
+                weight_map[n] = fname
+            total_size += nbytes
+            if r == rank:
+                my_names.append(fname)
+            gi += 1
+    index = {"metadata": {"total_size": total_size},
+             "weight_map": weight_map}
+    opt_cpu = None
     if optimizer_states:
-        of = {}
+        opt_cpu = {}
         for name, st in optimizer_states.items():
             for k, t in st.items():
                 if isinstance(t, torch.Tensor):
-                    of[f"{name}.{k}"] = t.detach().cpu()
+                    opt_cpu[f"{name}.{k}"] = t.detach().cpu().clone()
                 else:
-                    of[f"{name}.{k}"] = torch.tensor(float(t))
-        save_file(of, os.path.join(path, f"optim-r{rank:03d}.safetensors"))
+                    opt_cpu[f"{name}.{k}"] = torch.tensor(float(t))
+    return shards, my_names, index, opt_cpu, rank, n_total
+
+
+def _write_snapshot(path: str, plan) -> None:
+    """Pure-file-IO phase 2 of a checkpoint save (background-safe)."""
+    from safetensors.torch import save_file
+    shards, my_names, index, opt_cpu, rank, n_total = plan
+    os.makedirs(path, exist_ok=True)
+    for f, fname in zip(shards, my_names):
+        save_file(f, os.path.join(path, fname))
+    if rank == 0 and n_total > 1:
+        with open(os.path.join(path, "model.safetensors.index.json"),
+                  "w") as fh:
+            json.dump(index, fh, indent=1)
+    if opt_cpu is not None:
+        save_file(opt_cpu, os.path.join(path,
+                                        f"optim-r{rank:03d}.safetensors"))
+
+
+def save_model(params: List, path: str, comm: Optional[CommBackend] = None,
+               optimizer_states: Optional[Dict] = None,
+               max_shard_bytes: int = 8 << 30) -> None:
+    """params: graph parameter tensors (with .ds/.device_group/.name).
+    Writes model-XXXXX-of-YYYYY.safetensors + model.safetensors.index.json
+    (HF layout, metadata.total_size populated; a single shard is written as
+    plain model.safetensors).  optimizer_states: {name: {m, v, step,
+    master}} saved to optim-r<rank>.safetensors alongside."""
+    comm = comm or comm_backend()
+    plan = _snapshot_model(params, comm, optimizer_states, max_shard_bytes)
+    _write_snapshot(path, plan)
     if comm:
         comm.barrier()
 
@@ -153,8 +187,15 @@ def load_model(params: List, path: str,
     """Loads global tensors and slices each down to this rank's shard."""
     from safetensors import safe_open
     comm = comm or comm_backend()
-    with open(os.path.join(path, "model.safetensors.index.json")) as fh:
-        index = json.load(fh)["weight_map"]
+    idx_path = os.path.join(path, "model.safetensors.index.json")
+    if os.path.exists(idx_path):
+        with open(idx_path) as fh:
+            index = json.load(fh)["weight_map"]
+    else:
+        # single-file checkpoint (HF convention: no index)
+        with safe_open(os.path.join(path, "model.safetensors"),
+                       framework="pt") as f:
+            index = {k: "model.safetensors" for k in f.keys()}
     missing = []
     handles: Dict[str, "safe_open"] = {}
     for p in params:
@@ -231,10 +272,14 @@ class AsyncSaver:
 
     def save(self, params, path, comm=None, optimizer_states=None):
         self.wait()
-        # snapshot shards on the calling thread (device state is live)
+        # snapshot (gathers + CPU copies + index coordination) happens HERE
+        # on the calling thread, while device state is consistent and no
+        # training collective can interleave; only file writing goes to the
+        # background thread.
+        comm = comm or comm_backend()
+        plan = _snapshot_model(params, comm, optimizer_states, 8 << 30)
         self._thread = threading.Thread(
-            target=save_model, args=(params, path, comm),
-            kwargs={"optimizer_states": optimizer_states}, daemon=True)
+            target=_write_snapshot, args=(path, plan), daemon=True)
         self._thread.start()
 
     def wait(self):

@@ -73,6 +73,32 @@ def test_pipeline_runner_scaler_skips(tmp_path):
         assert torch.equal(p.get_data(), before[p.name]), p.name
 
 
+def test_pipeline_scaler_grad_parity():
+    """Scaled backward seed + unscale_ must reproduce the unscaled grads
+    exactly in fp32 (verifies the loss_seed wiring)."""
+    from hetu_amd.models.llama import LlamaConfig, build_llama_pipeline_stage
+    from hetu_amd.parallel.pipeline import PipelineRunner, PipelineSpec
+    cfg = LlamaConfig(n_layer=1, n_head=2, n_kv_head=2, hidden=32,
+                      ffn_hidden=64, vocab=64, max_seq=8)
+    pspec = PipelineSpec(pp=1)
+    grads = {}
+    for scale in (None, 1024.0):
+        torch.manual_seed(7)
+        stage = build_llama_pipeline_stage(cfg, pspec, 1, 8,
+                                           dtype=torch.float32, lr=0.0)
+        sc = GradScaler(init_scale=scale) if scale else None
+        runner = PipelineRunner(pspec, stage, torch.device("cpu"), scaler=sc)
+        h = stage.h
+        torch.manual_seed(11)
+        ids = torch.randint(0, 64, (1, 8))
+        labels = torch.randint(0, 64, (8,))
+        runner.step([{h["input_ids"]: ids, h["labels"]: labels}])
+        grads[scale] = [b.clone() for b in runner.grad_bufs]
+    for a, b in zip(grads[None], grads[1024.0]):
+        assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
+        assert a.abs().sum() > 0      # not trivially zero
+
+
 def test_op_profiler():
     from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
     from hetu_amd.graph.ops import api as ht
