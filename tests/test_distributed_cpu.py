@@ -163,3 +163,61 @@ def test_generic_resharding_two_ranks():
         out, err = p.communicate(timeout=300)
         ok = (p.returncode in (0, -6)) and "GENOK" in out
         assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
+
+
+GENERIC_PARTIAL_WORKER = r"""
+import os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+from hetu_amd.parallel.comm import comm_backend
+from hetu_amd.parallel.dstates import DistributedStates
+from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
+from hetu_amd.graph.ops import api as ht
+from hetu_amd.engine.runner import prepare_run_context
+comm = comm_backend()
+rank, ws = comm.rank, comm.world_size
+dg = tuple(range(ws))
+src = DistributedStates(ws, {-1: ws}, [-1])    # dup
+dst = DistributedStates(ws, {-2: ws}, [-2])    # -> partial (generic, adjoint
+                                               # of allreduce via gradient)
+from hetu_amd.graph.ops.comm import deduce_comm_kind
+assert deduce_comm_kind(src, dst)[0] == "generic"
+g = DefineAndRunGraph("gp"); push_graph(g)
+try:
+    x = ht.placeholder((4, 6), name="x", ds=src, device_group=dg)
+    y = ht.comm(x, dst)
+    # downstream partial-sum semantics: reducing over the partial group must
+    # give the value exactly once, not ws times
+    z = ht.comm(y, DistributedStates(ws, {-1: ws}, [-1]))
+finally: pop_graph()
+ctx = prepare_run_context(g, torch.device("cpu"))
+full = torch.arange(24.0).reshape(4, 6)
+yl, = g.run([y], {x: full.clone()}, ctx=ctx)
+# dst partial: only partial-index-0 rank carries the value
+if rank == 0:
+    assert torch.equal(yl, full), (rank, yl)
+else:
+    assert torch.equal(yl, torch.zeros_like(yl)), (rank, yl)
+zl, = g.run([z], {x: full.clone()}, ctx=ctx)
+assert torch.equal(zl, full), (rank, zl)
+print("GENPOK")
+"""
+
+
+def test_generic_resharding_partial_destination():
+    """split(dim0) -> partial: generic fallback must zero non-leader ranks
+    so a later partial reduction does not overcount."""
+    import subprocess
+    import sys
+    procs = []
+    env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29639", "GLOO_SOCKET_IFNAME": "lo"}
+    for r in range(2):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="2", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, "-c",
+                                       GENERIC_PARTIAL_WORKER], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=300)
+        ok = (p.returncode in (0, -6)) and "GENPOK" in out
+        assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
