@@ -181,28 +181,54 @@ class Graph:
 
     # ---- topology --------------------------------------------------------
     def topo_sort(self, fetches: Iterable[Tensor]) -> List[Op]:
-        """Reverse-reachable subgraph from fetches, in topological order."""
-        visited: Dict[int, bool] = {}
-        order: List[Op] = []
+        """Reverse-reachable subgraph from fetches, in topological order.
 
-        def visit(op: Op):
-            state = visited.get(op.id)
-            if state is True:
-                return
-            if state is False:
-                raise RuntimeError(f"cycle detected at op {op.name}")
-            visited[op.id] = False
+        Kahn's algorithm with a min-heap keyed by each op's scheduling key
+        (creation id by default).  An op may carry attrs["_sched_key"] to
+        be emitted as soon as possible after its dependencies — the grad
+        bucket all-reduce ops use this so the collective is issued right
+        after the last gradient of the bucket, overlapping with the rest
+        of backward (reference AllReduceCoalesce placement)."""
+        import heapq
+        reach: Dict[int, Op] = {}
+        stack = [t.producer for t in fetches if t.producer is not None]
+        while stack:
+            op = stack.pop()
+            if op.id in reach:
+                continue
+            reach[op.id] = op
             for t in op.inputs:
                 if t.producer is not None:
-                    visit(t.producer)
-            for dep in op.in_deps:
-                visit(dep)
-            visited[op.id] = True
-            order.append(op)
+                    stack.append(t.producer)
+            stack.extend(op.in_deps)
 
-        for t in fetches:
-            if t.producer is not None:
-                visit(t.producer)
+        indeg: Dict[int, int] = {}
+        consumers: Dict[int, List[int]] = {}
+        for op in reach.values():
+            deps = {t.producer.id for t in op.inputs
+                    if t.producer is not None}
+            deps.update(d.id for d in op.in_deps)
+            indeg[op.id] = len(deps)
+            for d in deps:
+                consumers.setdefault(d, []).append(op.id)
+
+        def key(op: Op) -> float:
+            return op.attrs.get("_sched_key", float(op.id))
+
+        heap = [(key(reach[oid]), oid) for oid, n in indeg.items() if n == 0]
+        heapq.heapify(heap)
+        order: List[Op] = []
+        while heap:
+            _, oid = heapq.heappop(heap)
+            op = reach[oid]
+            order.append(op)
+            for cid in consumers.get(oid, ()):
+                indeg[cid] -= 1
+                if indeg[cid] == 0:
+                    heapq.heappush(heap, (key(reach[cid]), cid))
+        if len(order) != len(reach):
+            bad = [reach[oid].name for oid, n in indeg.items() if n > 0]
+            raise RuntimeError(f"cycle detected among ops {bad[:5]}")
         return order
 
     # ---- autodiff (reference graph.cc:117 Gradients) ---------------------

@@ -180,6 +180,90 @@ class ZeroAdamStepOp(OptimizerUpdateOp):
         return [torch.zeros((), device=param.device)]
 
 
+class GradBucketOp(OpInterface):
+    """Coalesced gradient all-reduce: N partial grads -> one flat-buffer
+    RCCL all-reduce -> N reduced grads (reference AllReduceCoalesce,
+    impl/communication/nccl_comm_group.cu:273-301).
+
+    xGMI note: ring collectives are per-link bound (7 x ~153 GB/s), so few
+    LARGE collectives beat many small ones; buckets are sized by
+    HETU_AMD_BUCKET_MB (default 100).  The op carries attrs["_sched_key"]
+    so topo_sort emits it right after the bucket's last gradient and the
+    executor's comm-stream path overlaps it with the rest of backward."""
+    type = "GradAllReduceBucket"
+    is_comm = True
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(list(t.shape), t.dtype) for t in inputs]
+
+    def deduce_states(self, op):
+        for out, dst in zip(op.outputs, op.attrs["dst_dss"]):
+            out.ds = dst
+            out.device_group = op.inputs[0].device_group
+
+    def compute(self, op, inputs, ctx):
+        if ctx.comm is None:
+            return list(inputs)
+        from .comm import _my_index, _ranks
+        src = op.inputs[0].ds
+        dg = op.inputs[0].device_group
+        my = _my_index(ctx, dg)
+        ranks = _ranks(dg, src.group_devices_along(-2), my)
+        if len(ranks) <= 1:
+            return list(inputs)
+        if len(inputs) == 1:
+            return [ctx.comm.allreduce(inputs[0], ranks)]
+        flat = torch._utils._flatten_dense_tensors(tuple(inputs))
+        flat = ctx.comm.allreduce(flat, ranks)
+        return list(torch._utils._unflatten_dense_tensors(flat,
+                                                          tuple(inputs)))
+
+
+def make_grad_buckets(graph, pairs, bucket_bytes: Optional[int] = None,
+                      name_prefix: str = "grad_allreduce_bucket"):
+    """pairs: [(param, partial_grad)] needing reduction to the param layout.
+    Groups them into flat-buffer buckets (compatible comm group + dtype),
+    ordered by backward readiness (reverse creation order), and returns
+    {param_id: reduced_grad}."""
+    import os as _os
+    if bucket_bytes is None:
+        bucket_bytes = int(_os.environ.get("HETU_AMD_BUCKET_MB", "100")) << 20
+    # backward computes late-layer grads first; pairs arrive in forward
+    # (parameter) order, so reversed order approximates readiness order
+    ordered = list(reversed(pairs))
+    buckets: Dict = {}
+    for p, g in ordered:
+        key = (str(g.ds), tuple(p.device_group or ()), g.dtype)
+        nb = 1
+        for s in g.shape:
+            nb *= int(s)
+        nb *= g.dtype.itemsize if hasattr(g.dtype, "itemsize") else \
+            torch.empty(0, dtype=g.dtype).element_size()
+        lst = buckets.setdefault(key, [[]])
+        cur = lst[-1]
+        cur_bytes = sum(b for _, _, b in cur)
+        if cur and cur_bytes + nb > bucket_bytes:
+            cur = []
+            lst.append(cur)
+        cur.append((p, g, nb))
+    out: Dict[int, Tensor] = {}
+    bi = 0
+    for key, lst in buckets.items():
+        for items in lst:
+            ps = [p for p, _, _ in items]
+            gs = [g for _, g, _ in items]
+            attrs = {"dst_dss": [p.ds for p in ps],
+                     "_sched_key": max(
+                         (g.producer.id for g in gs
+                          if g.producer is not None), default=0) + 0.5}
+            op = _make(graph, GradBucketOp(), gs, attrs,
+                       name=f"{name_prefix}{bi}")
+            for p, t in zip(ps, op.outputs):
+                out[p.id] = t
+            bi += 1
+    return out
+
+
 class GroupOp(OpInterface):
     """Join node over update ops (reference ops/group.cc)."""
     type = "Group"
@@ -211,16 +295,28 @@ class Optimizer:
         watermark = len(graph.ops)
         has_scopes = any("_rc_scope" in op.attrs for op in graph.ops)
         grads = graph.gradients([loss], params)
+        # parameter-grad reduction: partial (over dp) -> param layout, via
+        # coalesced flat-buffer buckets overlapped with backward.  Under
+        # ZeRO the update op itself reduce-scatters the partial grad
+        # (COMPUTE_OPTIMIZE_BRIDGE semantics), so no comm here.
+        pend = []
+        for p, g in zip(params, grads):
+            if (g is not None and not self.zero and g.ds is not None
+                    and p.ds is not None and not g.ds.check_equal(p.ds)
+                    and g.ds.check_allreduce(p.ds)):
+                pend.append((p, g))
+        reduced = make_grad_buckets(graph, pend) if pend else {}
         updates = []
         for p, g in zip(params, grads):
             if g is None:
                 continue
-            # parameter-grad reduction: partial (over dp) -> param layout.
-            # Under ZeRO the update op itself reduce-scatters the partial
-            # grad (COMPUTE_OPTIMIZE_BRIDGE semantics), so no comm here.
-            if (not self.zero and g.ds is not None and p.ds is not None
+            if p.id in reduced:
+                g = reduced[p.id]
+            elif (not self.zero and g.ds is not None and p.ds is not None
                     and not g.ds.check_equal(p.ds)):
-                g = make_comm(graph, g, p.ds, name=f"grad_allreduce_{p.name}")
+                # non-allreduce reshard (rare): keep the per-tensor CommOp
+                g = make_comm(graph, g, p.ds,
+                              name=f"grad_allreduce_{p.name}")
             updates.append(self._make_update(graph, p, g))
         self.update_ops = updates
         out = _make(graph, GroupOp(), updates, name="train_op").output()

@@ -352,22 +352,32 @@ def build_gpt_pipeline_stage(cfg: GPTConfig, pspec, micro_batch: int,
         h["param_grads"] = grads[:len(params)]
         if not is_first:
             h["dx"] = grads[len(params)]
-        from ..graph.ops.optim import AdamStepOp, ZeroAdamStepOp, GroupOp
+        from ..graph.ops.optim import (AdamStepOp, ZeroAdamStepOp, GroupOp,
+                                       make_grad_buckets)
         from ..graph.ops.basics import _make
         from ..graph.ops.comm import make_comm
         grad_phs, updates = [], []
         opt_attrs = {"lr": lr, "beta1": 0.9, "beta2": 0.999, "eps": 1e-8,
                      "weight_decay": 0.0}
         cls = ZeroAdamStepOp if zero else AdamStepOp
+        phs, pend = [], []
         for p, pg in zip(params, h["param_grads"]):
             gds = pg.ds if pg is not None else None
             ph = ht.placeholder(tuple(p.shape), dtype=torch.float32,
                                 name=f"gbuf_{p.name}", ds=gds,
                                 device_group=spec.device_group)
             grad_phs.append(ph)
-            gt = ph
+            phs.append((p, ph))
             if not zero and gds is not None and p.ds is not None \
-                    and not gds.check_equal(p.ds):
+                    and not gds.check_equal(p.ds) \
+                    and gds.check_allreduce(p.ds):
+                pend.append((p, ph))
+        reduced = make_grad_buckets(g, pend, name_prefix="gred_bucket") \
+            if pend else {}
+        for p, ph in phs:
+            gt = reduced.get(p.id, ph)
+            if gt is ph and not zero and ph.ds is not None \
+                    and p.ds is not None and not ph.ds.check_equal(p.ds):
                 gt = make_comm(g, ph, p.ds, name=f"gred_{p.name}")
             updates.append(_make(g, cls(), [p, gt], dict(opt_attrs),
                                  name=f"adam_{p.name}").output())

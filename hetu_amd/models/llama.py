@@ -311,21 +311,31 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
         if not is_first:
             h["dx"] = grads[len(params)]
         # ---- update graph: grad placeholders -> (dp allreduce) -> Adam ---
-        from ..graph.ops.optim import AdamStepOp, GroupOp
+        from ..graph.ops.optim import (AdamStepOp, GroupOp,
+                                       make_grad_buckets)
         from ..graph.ops.basics import _make
         from ..graph.ops.comm import make_comm
         grad_phs, updates = [], []
         opt_attrs = {"lr": lr, "beta1": 0.9, "beta2": 0.999, "eps": 1e-8,
                      "weight_decay": 0.0}
+        phs, pend = [], []
         for p, pg in zip(params, h["param_grads"]):
             gds = pg.ds if pg is not None else None
             ph = ht.placeholder(tuple(p.shape), dtype=torch.float32,
                                 name=f"gbuf_{p.name}", ds=gds,
                                 device_group=spec.device_group)
             grad_phs.append(ph)
-            gt = ph
+            phs.append((p, ph))
             if gds is not None and p.ds is not None \
-                    and not gds.check_equal(p.ds):
+                    and not gds.check_equal(p.ds) \
+                    and gds.check_allreduce(p.ds):
+                pend.append((p, ph))
+        reduced = make_grad_buckets(g, pend, name_prefix="gred_bucket") \
+            if pend else {}
+        for p, ph in phs:
+            gt = reduced.get(p.id, ph)
+            if gt is ph and ph.ds is not None and p.ds is not None \
+                    and not ph.ds.check_equal(p.ds):
                 gt = make_comm(g, ph, p.ds, name=f"gred_{p.name}")
             updates.append(_make(g, AdamStepOp(), [p, gt], dict(opt_attrs),
                                  name=f"adam_{p.name}").output())

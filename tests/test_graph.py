@@ -274,3 +274,29 @@ print("LOSSES:" + repr(run(mode == "fused")))
         res[mode] = eval(m.group(1))  # noqa: S307
     assert all(abs(a - b) < 2e-4
                for a, b in zip(res["fused"], res["composed"])), res
+
+
+def test_grad_bucket_construction_and_placement():
+    """minimize() coalesces dp partial grads into flat-buffer bucket ops
+    placed right after their last gradient in the topo (overlap with
+    backward), reference AllReduceCoalesce."""
+    import torch
+    from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
+    cfg = GPTConfig(n_layer=2, n_head=2, hidden=64, ffn_hidden=128,
+                    vocab=128, max_seq=16)
+    g, h = build_gpt_train_graph(cfg, 2, 16, dtype=torch.float32,
+                                 dp=2, device_group=(0, 1))
+    buckets = [op for op in g.ops if op.type == "GradAllReduceBucket"]
+    nparams = len(list(g.parameters))
+    assert buckets, "no grad buckets created for dp=2"
+    assert len(buckets) < nparams, (len(buckets), nparams)
+    # every param with a partial grad routes through a bucket
+    topo = g.topo_sort([h["train_op"]])
+    pos = {op.id: i for i, op in enumerate(topo)}
+    for b in buckets:
+        bp = pos[b.id]
+        last_in = max(pos[t.producer.id] for t in b.inputs)
+        # emitted after its last input, and before unrelated later buckets'
+        # dependency chains complete: within a small window of the last grad
+        assert bp > last_in
+        assert bp - last_in <= 3, (b.name, bp, last_in)
