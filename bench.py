@@ -64,6 +64,8 @@ def main():
     ap.add_argument("--seq-len", type=int, default=2048)
     ap.add_argument("--parallel", default="auto")
     ap.add_argument("--capture", default="auto")
+    ap.add_argument("--allow-cpu", action="store_true",
+                    help="CI smoke only: run the same code path on CPU/gloo")
     args = ap.parse_args()
 
     if args.capture != "auto":
@@ -81,11 +83,13 @@ def main():
     from hetu_amd.galvatron.cost_model import ModelShape
     from hetu_amd.galvatron.search import search
 
-    assert torch.cuda.is_available(), "bench requires a GPU"
+    on_gpu = torch.cuda.is_available()
+    assert on_gpu or args.allow_cpu, "bench requires a GPU"
     comm = comm_backend()
     rank, ws = comm.rank, comm.world_size
     device = comm.device
-    torch.cuda.set_device(device)
+    if on_gpu:
+        torch.cuda.set_device(device)
 
     cfg = GPT_CONFIGS[args.model]
     S = args.seq_len
@@ -93,10 +97,21 @@ def main():
 
     # ---- strategy ---------------------------------------------------------
     if args.parallel == "auto":
-        shape = ModelShape(n_layer=cfg.n_layer, hidden=cfg.hidden,
-                           ffn_hidden=cfg.ffn_hidden, vocab=cfg.vocab,
-                           n_head=cfg.n_head, kind="gpt")
-        st, est = search(shape, S, ws, global_batch)
+        if rank == 0:
+            shape = ModelShape(n_layer=cfg.n_layer, hidden=cfg.hidden,
+                               ffn_hidden=cfg.ffn_hidden, vocab=cfg.vocab,
+                               n_head=cfg.n_head, kind="gpt")
+            st, est = search(shape, S, ws, global_batch)
+        else:
+            st, est = None, None
+        if ws > 1:
+            # every rank must execute the SAME strategy: rank 0 searches,
+            # the choice is broadcast (the search is deterministic, but a
+            # divergent pick would deadlock the collectives)
+            import torch.distributed as dist
+            box = [st]
+            dist.broadcast_object_list(box, src=0)
+            st = box[0]
     else:
         st = parse_strategy(args.parallel, ws)
         est = None
@@ -162,12 +177,14 @@ def main():
         loss = step_fn(i)
 
     comm.barrier()
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for i in range(args.steps):
         loss = step_fn(i)
     comm.barrier()
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
     t1 = time.perf_counter()
 
     elapsed = t1 - t0

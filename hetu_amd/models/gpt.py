@@ -58,6 +58,9 @@ GPT_CONFIGS = {
     "gpt-moe-8x1.3b": GPTConfig(n_layer=24, n_head=16, n_kv_head=16,
                                 hidden=2048, ffn_hidden=8192, vocab=50304,
                                 max_seq=2048, moe_experts=8, moe_k=2),
+    # CI-only config for the multi-process CPU smoke of the bench path
+    "gpt-tiny": GPTConfig(n_layer=2, n_head=2, n_kv_head=2, hidden=64,
+                          ffn_hidden=128, vocab=128, max_seq=64),
 }
 
 

@@ -80,3 +80,24 @@ def test_osdp_per_layer_sharding_plan():
                        n_head=12, kind="gpt")
     p2, e2 = osdp_plan(CostModel(small, 1024), st, 8)
     assert sum(p2) == 0 and e2["fits"]
+
+
+def test_search_ws_expectations_7b():
+    """Pre-committed expectations for the driver's scaling run: the 7B
+    auto-search at ws=1/2/4/8 must be deterministic and pick the dp-only
+    strategy (pp is python-bound, tp loses to dp at this size)."""
+    from hetu_amd.galvatron.cost_model import ModelShape
+    from hetu_amd.galvatron.search import search
+    from hetu_amd.models.gpt import GPT_CONFIGS
+    cfg = GPT_CONFIGS["gpt3-7b"]
+    shape = ModelShape(n_layer=cfg.n_layer, hidden=cfg.hidden,
+                       ffn_hidden=cfg.ffn_hidden, vocab=cfg.vocab,
+                       n_head=cfg.n_head, kind="gpt")
+    for ws in (1, 2, 4, 8):
+        st, est = search(shape, 2048, ws, 16 * ws)
+        assert st.name() == f"dp{ws}", (ws, st.name())
+        assert st.micro_batch * (16 * ws) // (st.dp * st.micro_batch) == 16
+        assert est["fits"] and est["mem"] < 288e9
+        # run twice: deterministic
+        st2, _ = search(shape, 2048, ws, 16 * ws)
+        assert st2.name() == st.name() and st2.micro_batch == st.micro_batch
