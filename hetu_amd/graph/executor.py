@@ -69,11 +69,30 @@ class Executor:
     def bind_context(self, ctx: ExecContext):
         self.ctx = ctx
 
-    def _get_plan(self, fetches: Sequence[Tensor]) -> _Plan:
-        key = (tuple(t.id for t in fetches), len(self.graph.ops))
+    def _get_plan(self, fetches: Sequence[Tensor],
+                  seed_ids: frozenset = frozenset()) -> _Plan:
+        key = (tuple(t.id for t in fetches), len(self.graph.ops), seed_ids)
         plan = self._plan_pool.get(key)
         if plan is None:
             topo = self.graph.topo_sort(fetches)
+            if seed_ids:
+                # cut the graph at seeded tensors: ops only reachable
+                # through them are dropped from the plan (the pipeline
+                # backward and the split-capture optimizer graph seed the
+                # cached forward/grad tensors)
+                stack = [t.producer for t in fetches
+                         if t.producer is not None and t.id not in seed_ids]
+                seen_ops = set()
+                while stack:
+                    op = stack.pop()
+                    if op.id in seen_ops:
+                        continue
+                    seen_ops.add(op.id)
+                    for t in op.inputs:
+                        if t.id not in seed_ids and t.producer is not None:
+                            stack.append(t.producer)
+                    stack.extend(op.in_deps)
+                topo = [op for op in topo if op.id in seen_ops]
             fetch_ids = [t.id for t in fetches]
             last_use: Dict[int, int] = {}
             for i, op in enumerate(topo):
@@ -97,7 +116,8 @@ class Executor:
         dict the caller provides to capture every computed value (disables
         the degree-based free)."""
         ctx = ctx or self.ctx or ExecContext()
-        plan = self._get_plan(fetches)
+        plan = self._get_plan(
+            fetches, frozenset(seed_values) if seed_values else frozenset())
         values: Dict[int, torch.Tensor] = dict(seed_values or {})
 
         # feed_dict keys may be Tensors or names

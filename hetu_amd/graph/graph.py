@@ -188,7 +188,11 @@ class Graph:
         be emitted as soon as possible after its dependencies — the grad
         bucket all-reduce ops use this so the collective is issued right
         after the last gradient of the bucket, overlapping with the rest
-        of backward (reference AllReduceCoalesce placement)."""
+        of backward (reference AllReduceCoalesce placement).
+        HETU_AMD_TOPO=dfs restores the round-1 DFS post-order (debug)."""
+        import os as _os
+        if _os.environ.get("HETU_AMD_TOPO", "kahn") == "dfs":
+            return self._topo_sort_dfs(fetches)
         import heapq
         reach: Dict[int, Op] = {}
         stack = [t.producer for t in fetches if t.producer is not None]
@@ -229,6 +233,30 @@ class Graph:
         if len(order) != len(reach):
             bad = [reach[oid].name for oid, n in indeg.items() if n > 0]
             raise RuntimeError(f"cycle detected among ops {bad[:5]}")
+        return order
+
+    def _topo_sort_dfs(self, fetches: Iterable[Tensor]) -> List[Op]:
+        visited: Dict[int, bool] = {}
+        order: List[Op] = []
+
+        def visit(op: Op):
+            state = visited.get(op.id)
+            if state is True:
+                return
+            if state is False:
+                raise RuntimeError(f"cycle detected at op {op.name}")
+            visited[op.id] = False
+            for t in op.inputs:
+                if t.producer is not None:
+                    visit(t.producer)
+            for dep in op.in_deps:
+                visit(dep)
+            visited[op.id] = True
+            order.append(op)
+
+        for t in fetches:
+            if t.producer is not None:
+                visit(t.producer)
         return order
 
     # ---- autodiff (reference graph.cc:117 Gradients) ---------------------

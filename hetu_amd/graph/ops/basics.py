@@ -187,8 +187,19 @@ class ReduceToShapeOp(OpInterface):
     def compute(self, op, inputs, ctx):
         g, ref = inputs
         tgt = list(ref.shape)
-        while g.ndim > len(tgt):
-            g = g.sum(0)
+        lead = g.ndim - len(tgt)
+        if lead > 0:
+            # leading-dim reduction through the hand colsum kernel: the
+            # at::native column reduce returns garbage from the 2nd replay
+            # of a captured step on some shapes (ROCm 7.2, see
+            # profiles/r02_capture_replay_bug.md) — and colsum is faster.
+            from ...ops import functional as F
+            rest = g.shape[lead:]
+            n = 1
+            for d in rest:
+                n *= int(d)
+            dt = g.dtype
+            g = F.colsum(g.reshape(-1, n)).to(dt).reshape(rest)
         for i, d in enumerate(tgt):
             if g.shape[i] != d:
                 g = g.sum(i, keepdim=True)
@@ -1046,8 +1057,10 @@ class ReduceLeadingOp(OpInterface):
         out.device_group = ref.device_group
 
     def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
         gy = inputs[0]
-        return [gy.reshape(-1, gy.shape[-1]).sum(0)]
+        out = F.colsum(gy.reshape(-1, gy.shape[-1]))
+        return [out.to(gy.dtype)]
 
 
 class BatchMatMulOp(OpInterface):

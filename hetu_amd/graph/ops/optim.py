@@ -60,10 +60,15 @@ class AdamStepOp(OptimizerUpdateOp):
 
     hipGraph capture support: bias corrections flow host-pinned -> device
     tensor -> kernel pointer; a captured step re-reads the pinned buffer,
-    which `set_replay_step` updates between replays."""
+    which `set_replay_step` updates between replays.  The (beta1, beta2)
+    corrections are identical for every parameter, so ONE shared pinned
+    buffer + ONE H2D memcpy per step serves all ~400 Adam ops (per-op
+    copies put hundreds of tiny host-memcpy nodes in the hipGraph)."""
     type = "AdamStep"
 
     _instances: List["AdamStepOp"] = []
+    # (beta1, beta2, device) -> {"host": pinned[2], "dev": cuda[2], "step"}
+    _shared_bc: Dict = {}
 
     def __init__(self):
         self.state: Dict = {}
@@ -71,15 +76,15 @@ class AdamStepOp(OptimizerUpdateOp):
 
     @classmethod
     def set_replay_step(cls, step: int):
-        """Update every instance's pinned bias-correction buffer for a
-        graph replay at optimizer step `step` (1-based)."""
+        """Update the shared pinned bias-correction buffers for a graph
+        replay at optimizer step `step` (1-based)."""
+        for (b1, b2, _dev), sh in cls._shared_bc.items():
+            sh["host"][0] = 1.0 - b1 ** step
+            sh["host"][1] = 1.0 - b2 ** step
+            sh["step"] = step
         for inst in cls._instances:
-            st = inst.state
-            if "bc_host" in st:
-                b1, b2 = st["betas"]
-                st["bc_host"][0] = 1.0 - b1 ** step
-                st["bc_host"][1] = 1.0 - b2 ** step
-                st["step"] = step
+            if inst.state:
+                inst.state["step"] = step
 
     def compute(self, op, inputs, ctx):
         from ...ops import functional as F
@@ -93,17 +98,27 @@ class AdamStepOp(OptimizerUpdateOp):
             st["step"] = 0
             st["betas"] = (a["beta1"], a["beta2"])
             if param.is_cuda:
-                st["bc_host"] = torch.empty(2, dtype=torch.float32,
-                                            pin_memory=True)
-                st["bc_dev"] = torch.empty(2, dtype=torch.float32,
-                                           device=param.device)
+                key = (a["beta1"], a["beta2"], param.device.index)
+                sh = AdamStepOp._shared_bc.get(key)
+                if sh is None:
+                    sh = {"host": torch.empty(2, dtype=torch.float32,
+                                              pin_memory=True),
+                          "dev": torch.empty(2, dtype=torch.float32,
+                                             device=param.device),
+                          "step": 0}
+                    AdamStepOp._shared_bc[key] = sh
+                st["bc"] = sh
         st["step"] += 1
         bc_dev = None
-        if "bc_host" in st:
-            st["bc_host"][0] = 1.0 - a["beta1"] ** st["step"]
-            st["bc_host"][1] = 1.0 - a["beta2"] ** st["step"]
-            st["bc_dev"].copy_(st["bc_host"], non_blocking=True)
-            bc_dev = st["bc_dev"]
+        if "bc" in st:
+            sh = st["bc"]
+            if sh["step"] < st["step"]:
+                # first Adam op of this step: write + upload once
+                sh["host"][0] = 1.0 - a["beta1"] ** st["step"]
+                sh["host"][1] = 1.0 - a["beta2"] ** st["step"]
+                sh["dev"].copy_(sh["host"], non_blocking=True)
+                sh["step"] = st["step"]
+            bc_dev = sh["dev"]
         out16 = param if param.dtype != torch.float32 else None
         F.adam_step(st["master"], grad, st["m"], st["v"],
                     a["lr"], a["beta1"], a["beta2"], a["eps"],
@@ -264,6 +279,25 @@ def make_grad_buckets(graph, pairs, bucket_bytes: Optional[int] = None,
     return out
 
 
+class HeteroSyncOp(OpInterface):
+    """Cross-pipeline split-allreduce of one parameter gradient (Malleus
+    hetero DP; reference SplitAllReduce, Communication.h:660-786).  The
+    shared HeteroGradSync plan lives in attrs; grads mutate in place."""
+    type = "HeteroSync"
+    is_comm = True
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(list(inputs[0].shape), inputs[0].dtype)]
+
+    def deduce_states(self, op):
+        op.outputs[0].ds = op.inputs[0].ds
+        op.outputs[0].device_group = op.inputs[0].device_group
+
+    def compute(self, op, inputs, ctx):
+        hs = op.attrs["sync"]
+        return [hs.sync_one(op.attrs["idx"], inputs[0])]
+
+
 class GroupOp(OpInterface):
     """Join node over update ops (reference ops/group.cc)."""
     type = "Group"
@@ -280,9 +314,10 @@ class GroupOp(OpInterface):
 
 
 class Optimizer:
-    def __init__(self, lr: float, zero: bool = False):
+    def __init__(self, lr: float, zero: bool = False, hetero=None):
         self.lr = lr
         self.zero = zero     # ZeRO: shard optimizer states over dp
+        self.hetero = hetero  # parallel.hetero.HeteroSpec or None
         self.update_ops: List = []
 
     def _make_update(self, graph, param: Tensor, grad: Tensor) -> Tensor:
@@ -306,6 +341,12 @@ class Optimizer:
                     and g.ds.check_allreduce(p.ds)):
                 pend.append((p, g))
         reduced = make_grad_buckets(graph, pend) if pend else {}
+        hs = None
+        if self.hetero is not None and len(self.hetero.pipelines) > 1:
+            from ...parallel.hetero import HeteroGradSync
+            live = [p for p, g in zip(params, grads) if g is not None]
+            hs = HeteroGradSync(self.hetero, live)
+            hidx = {p.id: i for i, p in enumerate(live)}
         updates = []
         for p, g in zip(params, grads):
             if g is None:
@@ -317,6 +358,18 @@ class Optimizer:
                 # non-allreduce reshard (rare): keep the per-tensor CommOp
                 g = make_comm(graph, g, p.ds,
                               name=f"grad_allreduce_{p.name}")
+            if hs is not None:
+                sop = _make(graph, HeteroSyncOp(), [g],
+                            {"sync": hs, "idx": hidx[p.id]},
+                            name=f"hetero_sync_{p.name}")
+                # chain syncs so every rank issues the cross-pipeline
+                # collectives in the same (params) order — differently
+                # shaped per-pipeline graphs would otherwise be free to
+                # interleave them differently and deadlock
+                if getattr(self, "_last_sync", None) is not None:
+                    sop.in_deps.append(self._last_sync)
+                self._last_sync = sop
+                g = sop.output()
             updates.append(self._make_update(graph, p, g))
         self.update_ops = updates
         out = _make(graph, GroupOp(), updates, name="train_op").output()
@@ -341,8 +394,9 @@ class SGD(Optimizer):
 class Adam(Optimizer):
     def __init__(self, lr: float = 1e-3, beta1: float = 0.9,
                  beta2: float = 0.999, eps: float = 1e-8,
-                 weight_decay: float = 0.0, zero: bool = False):
-        super().__init__(lr, zero=zero)
+                 weight_decay: float = 0.0, zero: bool = False,
+                 hetero=None):
+        super().__init__(lr, zero=zero, hetero=hetero)
         self.beta1, self.beta2 = beta1, beta2
         self.eps = eps
         self.weight_decay = weight_decay

@@ -41,8 +41,8 @@ class Tensor:
     _next_id = 0
 
     __slots__ = ("id", "producer", "output_index", "meta", "name", "graph",
-                 "ds", "device_group", "shard_sections", "is_parameter", "requires_grad",
-                 "_data")
+                 "ds", "device_group", "shard_sections", "hetero_split",
+                 "is_parameter", "requires_grad", "_data")
 
     def __init__(self, producer, output_index: int, meta: TensorMeta,
                  name: str = "", graph=None,

@@ -250,7 +250,8 @@ def build_gpt_train_graph(cfg: GPTConfig, micro_batch: int, seq_len: int,
                           graph: Optional[DefineAndRunGraph] = None,
                           zero: bool = False,
                           spec: Optional[ParallelSpec] = None,
-                          recompute: bool = False
+                          recompute: bool = False,
+                          hetero=None
                           ) -> (DefineAndRunGraph, Dict):
     g = graph or DefineAndRunGraph("gpt_train")
     if spec is None:
@@ -272,7 +273,7 @@ def build_gpt_train_graph(cfg: GPTConfig, micro_batch: int, seq_len: int,
             loss_report = ht.comm(
                 loss, spec._ds({-1: spec.num_devices}, [-1]),
                 name="loss_allreduce")
-        opt = Adam(lr=lr, zero=zero)
+        opt = Adam(lr=lr, zero=zero, hetero=hetero)
         train_op = opt.minimize(loss)
     finally:
         pop_graph()

@@ -163,6 +163,8 @@ class ColumnParallelLinear(Module):
         if sections is not None and tp > 1:
             # checkpoint de-interleave metadata (utils/checkpoint)
             self.weight.shard_sections = list(sections)
+        # cross-pipeline hetero grad sync metadata (parallel/hetero.py)
+        self.weight.hetero_split = (0, list(sections) if sections else None)
         if bias:
             b = init.zeros((out_features,), dtype)
             self.bias = ht.variable(_shard(b, 0, tp, ti, sections),
@@ -171,6 +173,8 @@ class ColumnParallelLinear(Module):
                                     device_group=spec.device_group)
             if sections is not None and tp > 1:
                 self.bias.shard_sections = list(sections)
+            self.bias.hetero_split = (0, list(sections) if sections
+                                      else None)
         else:
             self.register_parameter("bias", None)
         self.gather_output = gather_output
@@ -210,6 +214,7 @@ class RowParallelLinear(Module):
                                   name=f"{name}.weight",
                                   ds=spec.ds_weight_row(1),
                                   device_group=spec.device_group)
+        self.weight.hetero_split = (1, None)
         if bias:
             # bias is added AFTER the reduction; duplicated
             self.bias = ht.variable(init.zeros((out_features,), dtype),
@@ -249,6 +254,7 @@ class VocabParallelEmbedding(Module):
                                   name=f"{name}.weight",
                                   ds=spec.ds_weight_col(0),
                                   device_group=spec.device_group)
+        self.weight.hetero_split = (0, None)
 
     def forward(self, ids):
         spec = self.spec

@@ -66,6 +66,18 @@ def _gpu(*ts) -> bool:
     return any(isinstance(t, torch.Tensor) and t.is_cuda for t in ts)
 
 
+# Debug bisect switch: HETU_AMD_TORCH_FALLBACK="fa,ln,adam,..." routes the
+# named kernel families to the plain-torch reference implementations even
+# on GPU (they are device-agnostic).  Families: fa, qkvfa, ln, rms, gelu,
+# silu, swiglu, adam, ce, vce, embed, softmax, rope, dropout.
+_TORCH_FB = frozenset(
+    s for s in os.environ.get("HETU_AMD_TORCH_FALLBACK", "").split(",") if s)
+
+
+def _use_hip(family: str, *ts) -> bool:
+    return _gpu(*ts) and family not in _TORCH_FB
+
+
 # ---------------------------------------------------------------------------
 # RMSNorm (fused fwd/bwd; reference RMSNorm.cu — block per row)
 # ---------------------------------------------------------------------------
@@ -73,8 +85,9 @@ def _gpu(*ts) -> bool:
 def rmsnorm_fwd(x: torch.Tensor, w: torch.Tensor, eps: float
                 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """returns (y, rstd[rows] fp32)"""
-    if _gpu(x):
-        return ext().rmsnorm_fwd(x.contiguous(), w.contiguous(), eps)
+    if _use_hip("rms", x):
+        return ext().rmsnorm_fwd(x.contiguous(),
+                                 w.contiguous().to(x.dtype), eps)
     xf = x.float()
     rstd = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
     y = (xf * rstd) * w.float()
@@ -83,12 +96,14 @@ def rmsnorm_fwd(x: torch.Tensor, w: torch.Tensor, eps: float
 
 def rmsnorm_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
                 rstd: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    if _gpu(x):
+    if _use_hip("rms", x):
         if _NORM_V2:
             return ext().rmsnorm_bwd2(dy.contiguous(), x.contiguous(),
-                                      w.contiguous(), rstd.contiguous())
+                                      w.contiguous().to(x.dtype),
+                                      rstd.contiguous())
         return ext().rmsnorm_bwd(dy.contiguous(), x.contiguous(),
-                                 w.contiguous(), rstd.contiguous())
+                                 w.contiguous().to(x.dtype),
+                                 rstd.contiguous())
     xf, dyf, wf = x.float(), dy.float(), w.float()
     r = rstd.unsqueeze(-1)
     xhat = xf * r
@@ -104,9 +119,10 @@ def rmsnorm_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
 # ---------------------------------------------------------------------------
 
 def layernorm_fwd(x, w, b, eps):
-    if _gpu(x):
-        return ext().layernorm_fwd(x.contiguous(), w.contiguous(),
-                                   b.contiguous(), eps)
+    if _use_hip("ln", x):
+        return ext().layernorm_fwd(x.contiguous(),
+                                   w.contiguous().to(x.dtype),
+                                   b.contiguous().to(x.dtype), eps)
     xf = x.float()
     mean = xf.mean(-1, keepdim=True)
     var = xf.var(-1, unbiased=False, keepdim=True)
@@ -116,14 +132,15 @@ def layernorm_fwd(x, w, b, eps):
 
 
 def layernorm_bwd(dy, x, w, mean, rstd):
-    if _gpu(x):
+    if _use_hip("ln", x):
         if _NORM_V2:
             return ext().layernorm_bwd2(dy.contiguous(), x.contiguous(),
-                                        w.contiguous(), mean.contiguous(),
+                                        w.contiguous().to(x.dtype),
+                                        mean.contiguous(),
                                         rstd.contiguous())
         return ext().layernorm_bwd(dy.contiguous(), x.contiguous(),
-                                   w.contiguous(), mean.contiguous(),
-                                   rstd.contiguous())
+                                   w.contiguous().to(x.dtype),
+                                   mean.contiguous(), rstd.contiguous())
     xf, dyf, wf = x.float(), dy.float(), w.float()
     mu = mean.unsqueeze(-1)
     r = rstd.unsqueeze(-1)
@@ -143,14 +160,14 @@ def layernorm_bwd(dy, x, w, mean, rstd):
 # ---------------------------------------------------------------------------
 
 def swiglu_fwd(x):
-    if _gpu(x):
+    if _use_hip("swiglu", x):
         return ext().swiglu_fwd(x.contiguous())
     x1, x2 = x.float().chunk(2, dim=-1)
     return (torch.nn.functional.silu(x1) * x2).to(x.dtype)
 
 
 def swiglu_bwd(dy, x):
-    if _gpu(x):
+    if _use_hip("swiglu", x):
         return ext().swiglu_bwd(dy.contiguous(), x.contiguous())
     x1, x2 = x.float().chunk(2, dim=-1)
     dyf = dy.float()
@@ -162,18 +179,29 @@ def swiglu_bwd(dy, x):
     return torch.cat([dx1, dx2], dim=-1).to(x.dtype)
 
 
+def colsum(x2d: torch.Tensor) -> torch.Tensor:
+    """Column sum of a 2-D tensor -> fp32 [C].  On GPU: hand HIP kernel
+    (reduce.hip) — also the hipGraph-replay-safe replacement for the
+    at::native column reduce, which returns garbage from the 2nd replay of
+    a captured step on some shapes (ROCm 7.2; see
+    profiles/r02_capture_replay_bug.md)."""
+    if _use_hip("colsum", x2d):
+        return ext().colsum(x2d.contiguous())
+    return x2d.float().sum(0)
+
+
 # ---------------------------------------------------------------------------
 # Activations (fused fwd/bwd; reference Gelu.cu / Activation.cu)
 # ---------------------------------------------------------------------------
 
 def gelu_fwd(x):
-    if _gpu(x):
+    if _use_hip("gelu", x):
         return ext().gelu_fwd(x.contiguous())
     return torch.nn.functional.gelu(x, approximate="tanh")
 
 
 def gelu_bwd(dy, x):
-    if _gpu(x):
+    if _use_hip("gelu", x):
         return ext().gelu_bwd(dy.contiguous(), x.contiguous())
     xf = x.float()
     c = 0.7978845608028654  # sqrt(2/pi)
@@ -184,13 +212,13 @@ def gelu_bwd(dy, x):
 
 
 def silu_fwd(x):
-    if _gpu(x):
+    if _use_hip("silu", x):
         return ext().silu_fwd(x.contiguous())
     return torch.nn.functional.silu(x)
 
 
 def silu_bwd(dy, x):
-    if _gpu(x):
+    if _use_hip("silu", x):
         return ext().silu_bwd(dy.contiguous(), x.contiguous())
     s = torch.sigmoid(x.float())
     return (dy.float() * s * (1 + x.float() * (1 - s))).to(x.dtype)
@@ -202,14 +230,14 @@ def silu_bwd(dy, x):
 
 def rope_fwd(x, cos, sin, interleaved: bool = False):
     """x: [B, S, H, D] (or [S, H, D] packed); cos/sin: [S, D/2] fp32."""
-    if _gpu(x):
+    if _use_hip("rope", x):
         return ext().rope_fwd(x.contiguous(), cos.contiguous(),
                               sin.contiguous())
     return _rope_ref(x, cos, sin, False)
 
 
 def rope_bwd(dy, cos, sin, interleaved: bool = False):
-    if _gpu(dy):
+    if _use_hip("rope", dy):
         return ext().rope_bwd(dy.contiguous(), cos.contiguous(),
                               sin.contiguous())
     return _rope_ref(dy, cos, sin, True)
@@ -236,13 +264,13 @@ def _rope_ref(x, cos, sin, backward: bool):
 # ---------------------------------------------------------------------------
 
 def softmax_fwd(x, dim=-1):
-    if _gpu(x) and dim in (-1, x.ndim - 1):
+    if _use_hip("softmax", x) and dim in (-1, x.ndim - 1):
         return ext().softmax_fwd(x.contiguous())
     return torch.softmax(x.float(), dim=dim).to(x.dtype)
 
 
 def softmax_bwd(dy, y, dim=-1):
-    if _gpu(dy) and dim in (-1, dy.ndim - 1):
+    if _use_hip("softmax", dy) and dim in (-1, dy.ndim - 1):
         return ext().softmax_bwd(dy.contiguous(), y.contiguous())
     dyf, yf = dy.float(), y.float()
     dx = (dyf - (dyf * yf).sum(dim, keepdim=True)) * yf
@@ -256,7 +284,7 @@ def softmax_bwd(dy, y, dim=-1):
 
 def softmax_ce_fwd(logits, labels, ignore_index: int = -100):
     """logits [N, V], labels [N] -> (loss[N] fp32, logsumexp[N] fp32)."""
-    if _gpu(logits):
+    if _use_hip("ce", logits):
         return ext().softmax_ce_fwd(logits.contiguous(),
                                     labels.contiguous(), ignore_index)
     lf = logits.float()
@@ -269,7 +297,7 @@ def softmax_ce_fwd(logits, labels, ignore_index: int = -100):
 
 
 def softmax_ce_bwd(dloss, logits, labels, lse, ignore_index: int = -100):
-    if _gpu(logits):
+    if _use_hip("ce", logits):
         return ext().softmax_ce_bwd(dloss.contiguous(), logits.contiguous(),
                                     labels.contiguous(), lse.contiguous(),
                                     ignore_index)
@@ -288,7 +316,7 @@ def vocab_parallel_ce_local_stats(logits, labels, vocab_start, vocab_end,
     given max), and predicted-logit for labels owned by this shard.
     Cross-rank max/sum allreduce happens at op level (see graph/ops/loss.py).
     """
-    if _gpu(logits):
+    if _use_hip("vce", logits):
         return ext().vp_ce_local(logits.contiguous(), labels.contiguous(),
                                  vocab_start, vocab_end, ignore_index)
     lf = logits.float()
@@ -308,7 +336,7 @@ def vocab_parallel_ce_local_stats(logits, labels, vocab_start, vocab_end,
 def dropout_fwd(x, p: float, seed: int, offset: int):
     if p <= 0.0:
         return x, None
-    if _gpu(x):
+    if _use_hip("dropout", x):
         return ext().dropout_fwd(x.contiguous(), p, seed, offset)
     g = torch.Generator(device="cpu").manual_seed(seed + offset)
     mask = (torch.rand(x.shape, generator=g, device=x.device) >= p)
@@ -319,7 +347,7 @@ def dropout_fwd(x, p: float, seed: int, offset: int):
 def dropout_bwd(dy, mask, p: float, seed: int, offset: int):
     if p <= 0.0:
         return dy
-    if _gpu(dy):
+    if _use_hip("dropout", dy):
         return ext().dropout_bwd(dy.contiguous(), mask, p, seed, offset)
     return dy * mask.to(dy.dtype) / (1.0 - p)
 
@@ -329,13 +357,13 @@ def dropout_bwd(dy, mask, p: float, seed: int, offset: int):
 # ---------------------------------------------------------------------------
 
 def embedding_fwd(table, ids):
-    if _gpu(table):
+    if _use_hip("embed", table):
         return ext().embedding_fwd(table.contiguous(), ids.contiguous())
     return table[ids]
 
 
 def embedding_bwd(dy, ids, num_rows: int):
-    if _gpu(dy):
+    if _use_hip("embed", dy):
         return ext().embedding_bwd(dy.contiguous(), ids.contiguous(),
                                    num_rows)
     D = dy.shape[-1]
@@ -356,7 +384,7 @@ def adam_step(param32, grad, m, v, lr, beta1, beta2, eps, weight_decay,
     given, the kernel reads bias corrections from it — this keeps a
     hipGraph-captured train step correct across replays (the host updates
     the pinned source of bc_dev between replays)."""
-    if _gpu(param32):
+    if _use_hip("adam", param32):
         ext().adam_step(param32, grad, m, v, lr, beta1, beta2, eps,
                         weight_decay, step,
                         param_out16 if param_out16 is not None
@@ -387,7 +415,7 @@ def flash_attn_fwd(q, k, v, causal: bool, scale: Optional[float] = None):
     Returns (out, lse[B,H,S] fp32)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    if _gpu(q):
+    if _use_hip("fa", q):
         return ext().flash_attn_fwd(q.contiguous(), k.contiguous(),
                                     v.contiguous(), causal, scale)
     return _attn_ref_fwd(q, k, v, causal, scale)
@@ -397,7 +425,7 @@ def flash_attn_bwd(dout, q, k, v, out, lse, causal: bool,
                    scale: Optional[float] = None):
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    if _gpu(q):
+    if _use_hip("fa", q):
         return ext().flash_attn_bwd(dout.contiguous(), q.contiguous(),
                                     k.contiguous(), v.contiguous(),
                                     out.contiguous(), lse.contiguous(),
@@ -517,7 +545,7 @@ _NF4_CODE = torch.tensor([
 def quantize_blockwise(x: torch.Tensor, qtype: str = "nf4",
                        blocksize: int = 64):
     """returns (packed uint8, absmax fp32 [nblocks])."""
-    if _gpu(x):
+    if _use_hip("quant", x):
         return tuple(ext().quantize_blockwise(x.contiguous(), qtype,
                                               blocksize))
     flat = x.float().reshape(-1)
@@ -545,7 +573,7 @@ def quantize_blockwise(x: torch.Tensor, qtype: str = "nf4",
 def dequantize_blockwise(q: torch.Tensor, absmax: torch.Tensor,
                          qtype: str, blocksize: int, numel: int,
                          dtype=torch.float32) -> torch.Tensor:
-    if _gpu(q):
+    if _use_hip("quant", q):
         return ext().dequantize_blockwise(q, absmax, qtype, blocksize,
                                           numel, dtype)
     if qtype == "int8":
@@ -578,7 +606,7 @@ def fused_qkv_attention_fwd(qkv, n_head: int, n_kv_head: int, head_dim: int,
     pre-rotation values).  Returns (o [B,S,H*D], lse [B,H,S])."""
     if scale is None:
         scale = 1.0 / math.sqrt(head_dim)
-    if _gpu(qkv):
+    if _use_hip("qkvfa", qkv):
         e = ext()
         if cos is not None:
             e.rope_qk_inplace(qkv, cos.contiguous(), sin.contiguous(),
@@ -608,7 +636,7 @@ def fused_qkv_attention_bwd(dout, qkv, out, lse, n_head: int,
     (RoPE backward applied in place on the dq|dk sections)."""
     if scale is None:
         scale = 1.0 / math.sqrt(head_dim)
-    if _gpu(qkv):
+    if _use_hip("qkvfa", qkv):
         e = ext()
         dqkv = e.flash_attn_bwd_qkv(dout.contiguous(), qkv,
                                     out.contiguous(), lse.contiguous(),

@@ -13,6 +13,8 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor w, torch::Tensor mean,
                                          torch::Tensor rstd);
+// reduce.hip
+torch::Tensor colsum(torch::Tensor x);
 // swiglu.hip
 torch::Tensor swiglu_fwd(torch::Tensor x);
 torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor x);
@@ -125,6 +127,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("layernorm_bwd2", &layernorm_bwd2);
   m.def("rmsnorm_bwd2", &rmsnorm_bwd2);
+  m.def("colsum", &colsum);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("gelu_fwd", &gelu_fwd);

@@ -15,6 +15,7 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 
 SOURCES = [
     "norms.hip",
+    "reduce.hip",
     "activations.hip",
     "swiglu.hip",
     "rope.hip",

@@ -227,6 +227,8 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
   const int D = x.size(-1);
   const int64_t rows = x.numel() / D;
   TORCH_CHECK(D % 8 == 0, "rmsnorm: D must be a multiple of 8");
+  TORCH_CHECK(w.scalar_type() == x.scalar_type(),
+              "rmsnorm: w dtype must match x");
   auto y = torch::empty_like(x);
   auto rstd = torch::empty({rows}, x.options().dtype(at::kFloat));
   auto stream = hetu_current_stream();
@@ -244,6 +246,9 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor w, torch::Tensor rstd) {
   const int D = x.size(-1);
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              dy.scalar_type() == x.scalar_type(),
+              "rmsnorm_bwd: dy/w dtype must match x");
   const int64_t rows = x.numel() / D;
   auto dx = torch::empty_like(x);
   auto dw32 = torch::zeros({D}, x.options().dtype(at::kFloat));
@@ -277,6 +282,9 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
   const int D = x.size(-1);
   const int64_t rows = x.numel() / D;
   TORCH_CHECK(D % 8 == 0, "layernorm: D must be a multiple of 8");
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              b.scalar_type() == x.scalar_type(),
+              "layernorm: w/b dtype must match x");
   auto y = torch::empty_like(x);
   auto mean = torch::empty({rows}, x.options().dtype(at::kFloat));
   auto rstd = torch::empty({rows}, x.options().dtype(at::kFloat));
@@ -298,6 +306,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor w, torch::Tensor mean,
                                          torch::Tensor rstd) {
   const int D = x.size(-1);
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              dy.scalar_type() == x.scalar_type(),
+              "layernorm_bwd: dy/w dtype must match x");
   const int64_t rows = x.numel() / D;
   auto dx = torch::empty_like(x);
   auto dw32 = torch::zeros({D}, x.options().dtype(at::kFloat));
@@ -459,6 +470,9 @@ std::vector<torch::Tensor> layernorm_bwd2(torch::Tensor dy, torch::Tensor x,
                                           torch::Tensor mean,
                                           torch::Tensor rstd) {
   const int D = x.size(-1);
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              dy.scalar_type() == x.scalar_type(),
+              "layernorm_bwd2: dy/w dtype must match x");
   const int64_t rows = x.numel() / D;
   TORCH_CHECK(D % 2 == 0);
   auto dx = torch::empty_like(x);
@@ -479,6 +493,9 @@ std::vector<torch::Tensor> rmsnorm_bwd2(torch::Tensor dy, torch::Tensor x,
                                         torch::Tensor w,
                                         torch::Tensor rstd) {
   const int D = x.size(-1);
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              dy.scalar_type() == x.scalar_type(),
+              "rmsnorm_bwd2: dy/w dtype must match x");
   const int64_t rows = x.numel() / D;
   TORCH_CHECK(D % 2 == 0);
   auto dx = torch::empty_like(x);

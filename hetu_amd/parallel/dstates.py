@@ -211,6 +211,91 @@ class DistributedStates:
                 f"order={self.order})")
 
 
+NULL_HETERO_DIM = -3
+
+
+class DistributedStatesUnion:
+    """Per-pipeline heterogeneous layouts (reference
+    /root/reference/hetu/graph/distributed_states.h:158-233).
+
+    A union holds one DistributedStates per heterogeneous pipeline plus a
+    ``hetero_dim``: the dim along which the pipelines partition the global
+    tensor (>=0 a tensor dim, -1 duplicate, -2 partial; NULL_HETERO_DIM=-3
+    means homogeneous, union size 1).  Each entry is a FULL-device-count ds
+    that includes the union factor on the hetero dim; ``get_local(i)``
+    strips it, yielding pipeline i's own layout over its own device count.
+
+    This is the algebra behind Malleus hetero parallel: e.g. a weight
+    trained by two pipelines with tp2 and tp1 has a union of two entries
+    whose local layouts differ; grads reduce across pipelines with
+    split-allreduce groups (parallel/hetero.py)."""
+
+    __slots__ = ("union", "hetero_dim")
+
+    def __init__(self, ds_list: List[DistributedStates],
+                 hetero_dim: int = NULL_HETERO_DIM):
+        if len(ds_list) > 1 and hetero_dim == NULL_HETERO_DIM:
+            raise ValueError("hetero_dim required for union size > 1")
+        if len(ds_list) <= 1 and hetero_dim != NULL_HETERO_DIM:
+            raise ValueError("hetero_dim must be NULL for union size <= 1")
+        self.union = list(ds_list)
+        self.hetero_dim = hetero_dim
+
+    def is_hetero(self) -> bool:
+        return self.hetero_dim != NULL_HETERO_DIM
+
+    def size(self) -> int:
+        return len(self.union)
+
+    def get(self, i: int) -> DistributedStates:
+        return self.union[i]
+
+    def get_default_ds(self) -> DistributedStates:
+        return self.union[0]
+
+    def get_local(self, i: int) -> DistributedStates:
+        """Pipeline i's layout with the union factor stripped off the
+        hetero dim (reference get_local)."""
+        if not self.is_hetero():
+            return self.union[0]
+        ds = self.union[i]
+        n = len(self.union)
+        if ds.get_dim(self.hetero_dim) % n != 0:
+            raise ValueError(
+                f"hetero dim {self.hetero_dim} count "
+                f"{ds.get_dim(self.hetero_dim)} not divisible by union {n}")
+        states = dict(ds.states)
+        states[self.hetero_dim] = ds.get_dim(self.hetero_dim) // n
+        order = list(ds.order)
+        if states[self.hetero_dim] == 1:
+            states.pop(self.hetero_dim)
+            order = [d for d in order if d != self.hetero_dim]
+        return DistributedStates(ds.device_num // n, states, order,
+                                 zero=ds.zero)
+
+    @classmethod
+    def to_hetero(cls, ds: DistributedStates, dim: int, num: int
+                  ) -> "DistributedStatesUnion":
+        """Lift a homogeneous ds into a hetero union of `num` identical
+        entries split along `dim` (reference to_hetero)."""
+        if ds.get_dim(dim) % num != 0:
+            raise ValueError(f"dim {dim} count {ds.get_dim(dim)} not "
+                             f"divisible by {num}")
+        return cls([ds] * num, hetero_dim=dim)
+
+    def check_equal(self, other: "DistributedStatesUnion") -> bool:
+        if (self.is_hetero() != other.is_hetero()
+                or self.hetero_dim != other.hetero_dim
+                or len(self.union) != len(other.union)):
+            return False
+        return all(a.check_equal(b)
+                   for a, b in zip(self.union, other.union))
+
+    def __repr__(self):
+        return (f"DSUnion(hetero_dim={self.hetero_dim}, "
+                f"union={self.union})")
+
+
 def ds_from_index_table(device_num: int, table: List[Dict[int, int]],
                         counts: Dict[int, int]) -> DistributedStates:
     """Infer (states, order) from a per-device {dim: state_index} table.

@@ -228,6 +228,7 @@ def collect_adam_states(graph) -> Dict[str, Dict]:
             st = dict(op.interface.state)
             st.pop("bc_host", None)
             st.pop("bc_dev", None)
+            st.pop("bc", None)
             st.pop("betas", None)
             st.pop("pad", None)
             out[name] = st

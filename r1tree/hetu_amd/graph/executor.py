@@ -1,0 +1,202 @@
+"""Executable-plan executor.
+
+Replaces the reference's ExecutableGraph run loop
+(/root/reference/hetu/graph/executable_graph.cc:883,1756 ComputeFunc/Run) with
+an MI355X-first design: torch-ROCm tensors, the torch caching allocator, HIP
+streams via torch.cuda.Stream, and (optionally, from the engine) hipGraph
+capture of the steady-state step. Plans (topo order + free schedule) are
+cached per fetch-set, mirroring the reference's exec-graph plan pool.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence
+
+import torch
+
+from .op import Op
+from .tensor import Tensor
+
+import os
+
+# gradient-allreduce / backward overlap on the comm stream (disable with
+# HETU_AMD_COMM_OVERLAP=0 if a RCCL/capture combination misbehaves)
+_COMM_OVERLAP = os.environ.get("HETU_AMD_COMM_OVERLAP", "1") == "1" 
+
+
+class ExecContext:
+    """Per-run execution context handed to every op's compute()."""
+
+    def __init__(self, device: Optional[torch.device] = None, comm=None,
+                 training: bool = True):
+        self.device = device or torch.device("cpu")
+        self.comm = comm                  # parallel.comm.CommBackend or None
+        self.training = training
+        self.symbols = {}
+        self.profiler = None              # utils.profiler.OpProfiler
+        # stream roles (MI355X: overlap collectives with compute on separate
+        # HIP streams; reference used a fixed 16-stream convention
+        # hetu/core/stream.h:8-20 — we keep {compute, comm, p2p, h2d} roles)
+        self.streams = {}
+        if self.device.type == "cuda":
+            self.streams = {
+                "compute": torch.cuda.current_stream(self.device),
+                "comm": torch.cuda.Stream(self.device),
+                "p2p": torch.cuda.Stream(self.device),
+                "h2d": torch.cuda.Stream(self.device),
+            }
+
+    def stream(self, role: str):
+        return self.streams.get(role)
+
+
+class _Plan:
+    __slots__ = ("topo", "last_use", "fetch_ids")
+
+    def __init__(self, topo: List[Op], last_use: Dict[int, int],
+                 fetch_ids: List[int]):
+        self.topo = topo
+        self.last_use = last_use   # tensor_id -> index of last consuming op
+        self.fetch_ids = fetch_ids
+
+
+class Executor:
+    def __init__(self, graph):
+        self.graph = graph
+        self._plan_pool: Dict = {}
+        self.ctx: Optional[ExecContext] = None
+        self._inflight: Dict[int, "torch.cuda.Event"] = {}
+
+    def bind_context(self, ctx: ExecContext):
+        self.ctx = ctx
+
+    def _get_plan(self, fetches: Sequence[Tensor]) -> _Plan:
+        key = (tuple(t.id for t in fetches), len(self.graph.ops))
+        plan = self._plan_pool.get(key)
+        if plan is None:
+            topo = self.graph.topo_sort(fetches)
+            fetch_ids = [t.id for t in fetches]
+            last_use: Dict[int, int] = {}
+            for i, op in enumerate(topo):
+                for t in op.inputs:
+                    last_use[t.id] = i
+            # fetched tensors are never freed
+            for tid in fetch_ids:
+                last_use.pop(tid, None)
+            plan = _Plan(topo, last_use, fetch_ids)
+            self._plan_pool[key] = plan
+        return plan
+
+    def run(self, fetches: Sequence[Tensor], feed_dict: Dict,
+            ctx: Optional[ExecContext] = None,
+            seed_values: Optional[Dict[int, torch.Tensor]] = None,
+            keep_values: Optional[Dict[int, torch.Tensor]] = None
+            ) -> List[torch.Tensor]:
+        """seed_values: {tensor_id: value} of already-computed tensors (their
+        producing ops are skipped) — the pipeline engine seeds the backward
+        pass with the forward pass's cached activations.  keep_values: a
+        dict the caller provides to capture every computed value (disables
+        the degree-based free)."""
+        ctx = ctx or self.ctx or ExecContext()
+        plan = self._get_plan(fetches)
+        values: Dict[int, torch.Tensor] = dict(seed_values or {})
+
+        # feed_dict keys may be Tensors or names
+        feeds: Dict[int, torch.Tensor] = {}
+        for k, v in feed_dict.items():
+            t = k if isinstance(k, Tensor) else None
+            if t is None:
+                raise TypeError("feed_dict keys must be Tensors")
+            if not isinstance(v, torch.Tensor):
+                v = torch.as_tensor(v)
+            feeds[t.id] = v
+
+        for i, op in enumerate(plan.topo):
+            if op.outputs and all(t.id in values for t in op.outputs):
+                continue                     # seeded (already computed)
+            ins = []
+            for t in op.inputs:
+                if t.id in values:
+                    ins.append(values[t.id])
+                elif t.id in feeds:
+                    ins.append(feeds[t.id])
+                elif t.get_data() is not None:   # variable / persistent
+                    ins.append(t.get_data())
+                else:
+                    raise RuntimeError(
+                        f"no value for input {t.name} of op {op.name}")
+            if op.type == "Placeholder":
+                tid = op.outputs[0].id
+                if tid in feeds:
+                    values[tid] = feeds[tid]
+                    continue
+                raise RuntimeError(f"placeholder {op.name} not fed")
+            # gradient all-reduces ride the comm stream so they overlap
+            # with the rest of backward (reference: grad-buffer bucket
+            # reduction on the comm stream, executable_graph.cc:1756);
+            # consumers wait via recorded events
+            comm_async = (_COMM_OVERLAP and ctx.stream("comm") is not None
+                          and op.name.startswith("grad_allreduce"))
+            # any input still in flight on the comm stream: make the
+            # compute stream wait before using it
+            for t in op.inputs:
+                ev = self._inflight.pop(t.id, None) if not comm_async \
+                    else None
+                if ev is not None:
+                    cur = torch.cuda.current_stream(ctx.device)
+                    cur.wait_event(ev)
+                    v = values.get(t.id)
+                    if isinstance(v, torch.Tensor) and v.is_cuda and \
+                            not torch.cuda.is_current_stream_capturing():
+                        v.record_stream(cur)
+            prof = ctx.profiler
+            if comm_async:
+                cs = ctx.stream("comm")
+                ev_in = torch.cuda.Event()
+                ev_in.record(torch.cuda.current_stream(ctx.device))
+                cs.wait_event(ev_in)
+                with torch.cuda.stream(cs):
+                    outs = op.interface.compute(op, ins, ctx)
+                    if not torch.cuda.is_current_stream_capturing():
+                        for v in ins:
+                            if isinstance(v, torch.Tensor) and v.is_cuda:
+                                v.record_stream(cs)
+                    ev_out = torch.cuda.Event()
+                    ev_out.record(cs)
+                for t in op.outputs:
+                    self._inflight[t.id] = ev_out
+            elif prof is not None:
+                tok = prof.begin(op)
+                outs = op.interface.compute(op, ins, ctx)
+                prof.end(tok)
+            else:
+                outs = op.interface.compute(op, ins, ctx)
+            for t, v in zip(op.outputs, outs):
+                values[t.id] = v
+            if keep_values is not None:
+                for t, v in zip(op.outputs, outs):
+                    keep_values[t.id] = v
+                continue                     # caller owns lifetimes
+            # free dead intermediates (degree-based free, as in ComputeFunc)
+            for t in op.inputs:
+                if plan.last_use.get(t.id) == i and t.id in values:
+                    del values[t.id]
+
+        # join any comm-stream work not consumed by an op (e.g. fetched
+        # tensors) back into the compute stream
+        if self._inflight:
+            cur = torch.cuda.current_stream(ctx.device)
+            for ev in self._inflight.values():
+                cur.wait_event(ev)
+            self._inflight.clear()
+
+        out: List[torch.Tensor] = []
+        for t in fetches:
+            if t.id in values:
+                out.append(values[t.id])
+            elif t.id in feeds:
+                out.append(feeds[t.id])
+            elif t.get_data() is not None:
+                out.append(t.get_data())
+            else:
+                raise RuntimeError(f"fetch {t.name} produced no value")
+        return out

@@ -1,0 +1,268 @@
+"""DistributedStates — the SPMD tensor-layout algebra.
+
+Semantics follow the reference's DistributedStates
+(/root/reference/hetu/graph/distributed_states.h:13-200): a tensor placed on a
+device group of `device_num` devices is described by
+
+  * ``states``: {dim: split_count} where dim >= 0 splits that tensor dim,
+    dim == -1 is the replication (duplicate) count, dim == -2 is the
+    partial-reduction count (the tensor is a partial sum over that many
+    devices).
+  * ``order``: the sequence of dims (from slowest- to fastest-varying) that
+    maps a device's index within the group to its (split-index per dim) tuple.
+
+The product of all state counts must equal ``device_num``.
+
+Conversion predicates (check_allreduce / check_allgather / check_reducescatter
+/ check_split / check_scatter) mirror distributed_states.h:110-116 and drive
+comm-op type deduction (see hetu_amd/graph/ops/comm.py).
+
+This is a from-scratch implementation for the MI355X runtime: the execution
+side maps each (device group, participating dims) to a torch.distributed
+process group over RCCL.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+
+class DistributedStates:
+    __slots__ = ("device_num", "states", "order", "zero")
+
+    def __init__(self, device_num: int, states: Dict[int, int],
+                 order: Optional[List[int]] = None, zero: bool = False):
+        states = {int(k): int(v) for k, v in states.items() if int(v) > 1}
+        prod = 1
+        for v in states.values():
+            prod *= v
+        if prod != device_num:
+            raise ValueError(
+                f"states {states} product {prod} != device_num {device_num}")
+        if order is None:
+            # default order: partial, dup, then splits ascending
+            order = sorted(states.keys())
+        order = [d for d in order if states.get(d, 1) > 1]
+        missing = [d for d in sorted(states.keys()) if d not in order]
+        order = missing + order if missing else order
+        self.device_num = device_num
+        self.states = states
+        self.order = list(order)
+        self.zero = zero
+
+    # ---- accessors -------------------------------------------------------
+    def get_dim(self, dim: int) -> int:
+        return self.states.get(dim, 1)
+
+    @property
+    def partial(self) -> int:
+        return self.get_dim(-2)
+
+    @property
+    def dup(self) -> int:
+        return self.get_dim(-1)
+
+    def split_dims(self) -> List[int]:
+        return sorted(d for d in self.states if d >= 0)
+
+    def is_pure_dup(self) -> bool:
+        return all(d == -1 for d in self.states)
+
+    # ---- device index <-> state index ------------------------------------
+    def map_device_to_state_index(self, device_index: int) -> Dict[int, int]:
+        """Index of this device along each state dim (all dims, default 0)."""
+        if not 0 <= device_index < self.device_num:
+            raise ValueError(f"device_index {device_index} out of range")
+        idx: Dict[int, int] = {}
+        rem = device_index
+        for dim in reversed(self.order):
+            n = self.states[dim]
+            idx[dim] = rem % n
+            rem //= n
+        return idx
+
+    def get_dup_group_index(self, device_index: int) -> int:
+        """Linear index over the non-(-1) dims — identifies which unique data
+        shard this device holds (devices sharing it are replicas)."""
+        st = self.map_device_to_state_index(device_index)
+        idx = 0
+        for dim in self.order:
+            if dim == -1:
+                continue
+            idx = idx * self.states[dim] + st.get(dim, 0)
+        return idx
+
+    def group_devices_along(self, dim: int) -> List[List[int]]:
+        """Partition the device group's local indices into groups that vary
+        only along `dim` (used to build RCCL subgroups for a collective
+        over that state dim)."""
+        n = self.get_dim(dim)
+        if n == 1:
+            return [[i] for i in range(self.device_num)]
+        buckets: Dict[tuple, List[int]] = {}
+        for i in range(self.device_num):
+            st = self.map_device_to_state_index(i)
+            key = tuple((d, st.get(d, 0)) for d in sorted(self.states)
+                        if d != dim)
+            buckets.setdefault(key, []).append(i)
+        return list(buckets.values())
+
+    # ---- conversion predicates (distributed_states.h:110-116) ------------
+    def _same_but(self, other: "DistributedStates", src_dim: int,
+                  dst_dim: int) -> bool:
+        """True if `other` equals self with the count moved src_dim->dst_dim."""
+        if self.device_num != other.device_num:
+            return False
+        a = dict(self.states)
+        n = a.pop(src_dim, 1)
+        if n == 1:
+            return False
+        b = dict(other.states)
+        m = b.pop(dst_dim, 1)
+        if m % n != 0 and n % m != 0:
+            return False
+        # fold moved count into dst side and compare
+        a[dst_dim] = self.get_dim(dst_dim) * n
+        aa = {k: v for k, v in a.items() if v > 1}
+        return aa == other.states
+
+    def check_equal(self, other: "DistributedStates") -> bool:
+        return (self.device_num == other.device_num
+                and self.states == other.states)
+
+    def check_allreduce(self, dst: "DistributedStates") -> bool:
+        """partial k -> dup k (everything else unchanged)."""
+        return self.partial > 1 and self._same_but(dst, -2, -1)
+
+    def check_allgather(self, dst: "DistributedStates", gather_dim: int = 0
+                        ) -> bool:
+        """split(gather_dim) k -> dup k."""
+        return self.get_dim(gather_dim) > 1 and self._same_but(dst, gather_dim, -1)
+
+    def check_reducescatter(self, dst: "DistributedStates",
+                            scatter_dim: int = 0) -> bool:
+        """partial k -> split(scatter_dim) k."""
+        return self.partial > 1 and self._same_but(dst, -2, scatter_dim)
+
+    def check_scatter(self, dst: "DistributedStates", dim: int = 0) -> bool:
+        """dup k -> split(dim) k (no communication if data already present:
+        each replica keeps its slice)."""
+        return self.dup > 1 and self._same_but(dst, -1, dim)
+
+    def check_split(self, dst: "DistributedStates") -> bool:
+        """src dup covers some new split in dst: local slicing only."""
+        if self.device_num != dst.device_num:
+            return False
+        if self.partial != dst.partial:
+            return False
+        for d in dst.split_dims():
+            if dst.get_dim(d) % self.get_dim(d) != 0:
+                return False
+        # every extra split in dst must come out of src's dup
+        extra = 1
+        for d in dst.split_dims():
+            extra *= dst.get_dim(d) // self.get_dim(d)
+        for d in self.split_dims():
+            if self.get_dim(d) > dst.get_dim(d):
+                return False
+        return extra > 1 and self.dup == dst.dup * extra
+
+    # ---- helpers ---------------------------------------------------------
+    def local_shape(self, global_shape) -> tuple:
+        out = list(global_shape)
+        for d, n in self.states.items():
+            if d >= 0:
+                if out[d] % n != 0:
+                    raise ValueError(
+                        f"dim {d} size {out[d]} not divisible by split {n}")
+                out[d] //= n
+        return tuple(out)
+
+    def global_shape(self, local_shape) -> tuple:
+        out = list(local_shape)
+        for d, n in self.states.items():
+            if d >= 0:
+                out[d] *= n
+        return tuple(out)
+
+    def local_slice(self, global_shape, device_index: int):
+        """List of python slices selecting this device's shard."""
+        st = self.map_device_to_state_index(device_index)
+        slices = []
+        for d, size in enumerate(global_shape):
+            n = self.get_dim(d)
+            if n > 1:
+                blk = size // n
+                i = st.get(d, 0)
+                slices.append(slice(i * blk, (i + 1) * blk))
+            else:
+                slices.append(slice(None))
+        return tuple(slices)
+
+    def __eq__(self, other):
+        return (isinstance(other, DistributedStates)
+                and self.check_equal(other) and self.order == other.order)
+
+    def __hash__(self):
+        return hash((self.device_num, tuple(sorted(self.states.items())),
+                     tuple(self.order)))
+
+    def __repr__(self):
+        return (f"DS(n={self.device_num}, states={self.states}, "
+                f"order={self.order})")
+
+
+def ds_from_index_table(device_num: int, table: List[Dict[int, int]],
+                        counts: Dict[int, int]) -> DistributedStates:
+    """Infer (states, order) from a per-device {dim: state_index} table.
+
+    Used by op-level DoDeduceStates implementations (e.g. matmul): the
+    producer computes, for every device in the group, which shard indices its
+    local output holds, and this reconstructs the DistributedStates that
+    describes that layout. Raises if the table is not a regular grid (then
+    the layout is not expressible as a DistributedStates).
+    """
+    counts = {d: n for d, n in counts.items() if n > 1}
+    prod = 1
+    for n in counts.values():
+        prod *= n
+    if prod != device_num:
+        # remaining degrees of freedom are replicas
+        if device_num % prod != 0:
+            raise ValueError(f"counts {counts} do not divide {device_num}")
+        counts[-1] = counts.get(-1, 1) * (device_num // prod)
+    # find each dim's period: smallest stride at which its index changes
+    periods: Dict[int, int] = {}
+    for d in counts:
+        if d == -1 and -1 not in (table[0] if table else {}):
+            continue
+        p = device_num
+        for i in range(1, device_num):
+            if table[i].get(d, 0) != table[0].get(d, 0):
+                p = i
+                break
+        periods[d] = p
+    # dup dims absent from the table get the leftover periods; fall back to
+    # ordering by period descending (slowest-varying first)
+    dims = sorted(counts.keys(), key=lambda d: -periods.get(d, 1))
+    ds = DistributedStates(device_num, counts, order=dims)
+    # validate round-trip for the dims present in the table
+    for i in range(device_num):
+        st = ds.map_device_to_state_index(i)
+        for d in table[i]:
+            if d in counts and st.get(d, 0) != table[i][d]:
+                raise ValueError(
+                    f"layout not expressible as DistributedStates: device {i}"
+                    f" dim {d}: table={table[i][d]} ds={st.get(d, 0)}")
+    return ds
+
+
+def ds_dup(device_num: int) -> DistributedStates:
+    return DistributedStates(device_num, {-1: device_num} if device_num > 1 else {})
+
+
+def ds_split(device_num: int, dim: int = 0) -> DistributedStates:
+    return DistributedStates(device_num, {dim: device_num} if device_num > 1 else {})
+
+
+def ds_partial(device_num: int) -> DistributedStates:
+    return DistributedStates(device_num, {-2: device_num} if device_num > 1 else {})

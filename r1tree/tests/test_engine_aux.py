@@ -1,0 +1,78 @@
+"""Engine aux: SFT masking, straggler rebalance, dynamic planner, config."""
+import torch
+
+from hetu_amd.engine.sft_trainer import build_sft_example
+from hetu_amd.engine.straggler import detect_stragglers, rebalance_micro_batches
+from hetu_amd.engine.dynamic_planner import DynamicPlanner
+from hetu_amd.engine.trainer_config import TrainingConfig
+from hetu_amd.galvatron.cost_model import ModelShape, Strategy
+
+
+def test_sft_example_masks_prompt():
+    x, y = build_sft_example([1, 2, 3], [4, 5], seq_len=8)
+    assert x.tolist()[:4] == [1, 2, 3, 4]
+    assert y.tolist()[:4] == [-100, -100, 4, 5]
+    assert all(v == -100 for v in y.tolist()[4:])
+
+
+def test_straggler_detection_and_rebalance():
+    times = [1.0, 1.0, 3.0, 1.0]
+    assert detect_stragglers(times) == [2]
+    mb = rebalance_micro_batches(times, 12)
+    assert sum(mb) == 12
+    assert mb[2] < mb[0]           # straggler gets less work
+
+
+def test_dynamic_planner_prefers_big_strategy_for_long_seq():
+    shape = ModelShape(n_layer=8, hidden=1024, ffn_hidden=4096, vocab=32000,
+                       n_head=16)
+    cands = [Strategy(dp=8, micro_batch=1),
+             Strategy(dp=2, tp=4, micro_batch=1)]
+    pl = DynamicPlanner(shape, 8, cands, buckets=[512, 4096])
+    plan = pl.plan([128, 256, 4096, 3000])
+    assert set(plan.keys()) <= {512, 4096}
+    assert 512 in plan and 4096 in plan
+
+
+def test_config_yaml(tmp_path):
+    p = tmp_path / "c.yaml"
+    p.write_text("model: gpt2-345m\nglobal_batch: 32\ntp: 2\n"
+                 "precision: bf16\nunknown_key: 5\n")
+    cfg = TrainingConfig.from_yaml(str(p))
+    assert cfg.model == "gpt2-345m" and cfg.tp == 2
+    assert cfg.dtype() == torch.bfloat16
+
+
+def test_pipeline_memory_snapshots(monkeypatch):
+    """HETU_AMD_MEM_PROFILE=1 records a snapshot per micro-batch fwd/bwd
+    (reference per-micro-batch CUDAProfiler memory info)."""
+    import subprocess
+    import sys
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = r"""
+import os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+os.environ["HETU_AMD_MEM_PROFILE"] = "1"
+from hetu_amd.models.llama import LlamaConfig, build_llama_pipeline_stage
+from hetu_amd.parallel.pipeline import PipelineSpec, PipelineRunner
+cfg = LlamaConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
+                  ffn_hidden=96, vocab=211, max_seq=16)
+pspec = PipelineSpec(pp=1, dp=1, tp=1)
+stage = build_llama_pipeline_stage(cfg, pspec, micro_batch=1, seq_len=16,
+                                   dtype=torch.float32, lr=1e-3)
+runner = PipelineRunner(pspec, stage, torch.device("cpu"))
+h = stage.h
+mbs = [{h["input_ids"]: torch.randint(0, 211, (1, 16)),
+        h["labels"]: torch.randint(0, 211, (16,))} for _ in range(3)]
+runner.step(mbs)
+assert runner.mem_snapshots is not None
+rep = runner.mem_snapshots.report()
+assert "fwd_mb0" in rep and "bwd_mb" in rep, rep
+print("SNAP OK")
+"""
+    p = subprocess.run([sys.executable, "-c", code],
+                       env={**os.environ, "HETU_REPO": repo},
+                       capture_output=True, text=True, timeout=300)
+    assert p.returncode == 0 and "SNAP OK" in p.stdout, \
+        f"{p.stdout}\n{p.stderr}"

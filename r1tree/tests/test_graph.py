@@ -1,0 +1,276 @@
+"""Graph-core tests: autodiff parity vs torch.autograd, MLP convergence.
+
+Mirrors the reference test pattern (reference tests/test_ops.py — build
+small tensors, run graph mode, allclose vs torch).
+"""
+import numpy as np
+import pytest
+import torch
+
+import hetu_amd as ht
+
+
+def _grad_check(build, torch_build, shapes, rtol=1e-4, atol=1e-4,
+                dtype=torch.float32):
+    """build(ht_tensors) -> ht scalar; torch_build(torch_tensors) -> scalar."""
+    datas = [torch.randn(*s, dtype=dtype) for s in shapes]
+    with ht.graph("define_and_run") as g:
+        xs = [ht.variable(d.clone(), name=f"x{i}")
+              for i, d in enumerate(datas)]
+        y = build(*xs)
+        grads = ht.gradients(y, xs)
+        fetches = [y] + [gr for gr in grads if gr is not None]
+        res = g.run(fetches, {})
+    tds = [d.clone().requires_grad_(True) for d in datas]
+    ty = torch_build(*tds)
+    ty.backward()
+    assert np.allclose(res[0].detach().numpy(), ty.detach().numpy(),
+                       rtol=rtol, atol=atol), "forward mismatch"
+    gi = 1
+    for i, gr in enumerate(grads):
+        if gr is None:
+            continue
+        assert np.allclose(res[gi].detach().numpy(),
+                           tds[i].grad.detach().numpy(),
+                           rtol=rtol, atol=atol), f"grad {i} mismatch"
+        gi += 1
+
+
+class TestAutodiff:
+    def test_linear_relu(self):
+        _grad_check(
+            lambda x, w, b: ht.reduce_sum(ht.relu(ht.linear(x, w, b))),
+            lambda x, w, b: torch.relu(
+                torch.nn.functional.linear(x, w, b)).sum(),
+            [(4, 8), (16, 8), (16,)])
+
+    def test_matmul_chain(self):
+        _grad_check(
+            lambda a, b, c: ht.reduce_sum(ht.matmul(ht.matmul(a, b), c)),
+            lambda a, b, c: ((a @ b) @ c).sum(),
+            [(4, 5), (5, 6), (6, 3)])
+
+    def test_mul_add_broadcast(self):
+        _grad_check(
+            lambda a, b: ht.reduce_sum(ht.mul(ht.add(a, b), a)),
+            lambda a, b: ((a + b) * a).sum(),
+            [(4, 8), (8,)])
+
+    def test_gelu_silu_tanh_sigmoid(self):
+        for f_ht, f_t in [
+            (ht.gelu, lambda x: torch.nn.functional.gelu(x, approximate="tanh")),
+            (ht.silu, torch.nn.functional.silu),
+            (ht.tanh, torch.tanh),
+            (ht.sigmoid, torch.sigmoid),
+        ]:
+            _grad_check(lambda x, f=f_ht: ht.reduce_sum(f(x)),
+                        lambda x, f=f_t: f(x).sum(), [(8, 16)])
+
+    def test_softmax(self):
+        _grad_check(
+            lambda x: ht.reduce_sum(ht.mul(ht.softmax(x), x)),
+            lambda x: (torch.softmax(x, -1) * x).sum(),
+            [(8, 16)])
+
+    def test_reduce_mean_dim(self):
+        _grad_check(
+            lambda x: ht.reduce_sum(ht.reduce_mean(x, dim=1)),
+            lambda x: x.mean(1).sum(),
+            [(4, 8)])
+
+    def test_layernorm(self):
+        _grad_check(
+            lambda x, w, b: ht.reduce_sum(ht.layer_norm(x, w, b)),
+            lambda x, w, b: torch.nn.functional.layer_norm(
+                x, (16,), w, b).sum(),
+            [(8, 16), (16,), (16,)], rtol=1e-3, atol=1e-3)
+
+    def test_rmsnorm(self):
+        def t_rms(x, w):
+            xf = x.float()
+            return (xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6)
+                    * w).sum()
+        _grad_check(
+            lambda x, w: ht.reduce_sum(ht.rms_norm(x, w)),
+            t_rms, [(8, 16), (16,)], rtol=1e-3, atol=1e-3)
+
+    def test_swiglu(self):
+        def t_swiglu(x):
+            x1, x2 = x.chunk(2, -1)
+            return (torch.nn.functional.silu(x1) * x2).sum()
+        _grad_check(lambda x: ht.reduce_sum(ht.swiglu(x)), t_swiglu,
+                    [(8, 32)])
+
+    def test_cross_entropy(self):
+        logits = torch.randn(12, 37)
+        labels = torch.randint(0, 37, (12,))
+        with ht.graph("define_and_run") as g:
+            lg = ht.variable(logits.clone(), name="logits")
+            lb = ht.placeholder([12], dtype=torch.int64, name="labels")
+            loss = ht.reduce_mean(
+                ht.softmax_cross_entropy_sparse(lg, lb))
+            (gl,) = ht.gradients(loss, [lg])
+            lv, gv = g.run([loss, gl], {lb: labels})
+        tl = logits.clone().requires_grad_(True)
+        tloss = torch.nn.functional.cross_entropy(tl, labels)
+        tloss.backward()
+        assert np.allclose(lv.numpy(), tloss.detach().numpy(), rtol=1e-4)
+        assert np.allclose(gv.numpy(), tl.grad.numpy(), rtol=1e-3, atol=1e-5)
+
+    def test_attention_causal(self):
+        B, H, S, D = 2, 3, 16, 8
+        q = torch.randn(B, H, S, D)
+        k = torch.randn(B, H, S, D)
+        v = torch.randn(B, H, S, D)
+        with ht.graph("define_and_run") as g:
+            qh = ht.variable(q.clone(), name="q")
+            kh = ht.variable(k.clone(), name="k")
+            vh = ht.variable(v.clone(), name="v")
+            out = ht.attention(qh, kh, vh, causal=True)
+            loss = ht.reduce_sum(ht.mul(out, out))
+            gq, gk, gv = ht.gradients(loss, [qh, kh, vh])
+            res = g.run([out, gq, gk, gv], {})
+        tq, tk, tv = [t.clone().requires_grad_(True) for t in (q, k, v)]
+        tout = torch.nn.functional.scaled_dot_product_attention(
+            tq, tk, tv, is_causal=True)
+        (tout * tout).sum().backward()
+        assert np.allclose(res[0].numpy(), tout.detach().numpy(),
+                           rtol=1e-3, atol=1e-4)
+        for r, t in zip(res[1:], (tq, tk, tv)):
+            assert np.allclose(r.numpy(), t.grad.numpy(), rtol=1e-3,
+                               atol=1e-3)
+
+    def test_rope(self):
+        B, S, Hh, D = 2, 8, 4, 16
+        x = torch.randn(B, S, Hh, D)
+        pos = torch.arange(S).float()
+        inv = 1.0 / (10000 ** (torch.arange(0, D // 2).float() / (D // 2)))
+        ang = pos[:, None] * inv[None, :]
+        cos, sin = ang.cos(), ang.sin()
+        with ht.graph("define_and_run") as g:
+            xh = ht.variable(x.clone(), name="x")
+            ch = ht.variable(cos, name="cos", requires_grad=False)
+            sh = ht.variable(sin, name="sin", requires_grad=False)
+            y = ht.rotary(xh, ch, sh)
+            loss = ht.reduce_sum(ht.mul(y, y))
+            (gx,) = ht.gradients(loss, [xh])
+            yv, gv = g.run([y, gx], {})
+        # reference: rotate_half formulation
+        x1, x2 = x[..., :D // 2], x[..., D // 2:]
+        c = cos[None, :, None, :]
+        s = sin[None, :, None, :]
+        ty = torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], -1)
+        assert np.allclose(yv.numpy(), ty.numpy(), rtol=1e-4, atol=1e-5)
+        # gradient of rotation is rotation by -angle
+        gref_in = 2 * ty
+        g1, g2 = gref_in[..., :D // 2], gref_in[..., D // 2:]
+        tg = torch.cat([g1 * c + g2 * s, g2 * c - g1 * s], -1)
+        assert np.allclose(gv.numpy(), tg.numpy(), rtol=1e-3, atol=1e-4)
+
+    def test_embedding(self):
+        V, Dm = 50, 16
+        tbl = torch.randn(V, Dm)
+        ids = torch.randint(0, V, (4, 7))
+        with ht.graph("define_and_run") as g:
+            th = ht.variable(tbl.clone(), name="tbl")
+            ih = ht.placeholder([4, 7], dtype=torch.int64, name="ids")
+            y = ht.embedding(th, ih)
+            loss = ht.reduce_sum(ht.mul(y, y))
+            (gt,) = ht.gradients(loss, [th])
+            yv, gv = g.run([y, gt], {ih: ids})
+        tt = tbl.clone().requires_grad_(True)
+        ty = tt[ids]
+        (ty * ty).sum().backward()
+        assert np.allclose(yv.numpy(), ty.detach().numpy())
+        assert np.allclose(gv.numpy(), tt.grad.numpy(), rtol=1e-4, atol=1e-5)
+
+
+class TestTraining:
+    def test_mlp_converges_adam(self):
+        torch.manual_seed(0)
+        with ht.graph("define_and_run") as g:
+            x = ht.placeholder([16, 8], name="x")
+            y = ht.placeholder([16, 1], name="y")
+            w1 = ht.variable(torch.randn(32, 8) * 0.1, name="w1")
+            b1 = ht.variable(torch.zeros(32), name="b1")
+            w2 = ht.variable(torch.randn(1, 32) * 0.1, name="w2")
+            pred = ht.linear(ht.relu(ht.linear(x, w1, b1)), w2)
+            loss = ht.mse_loss(pred, y)
+            train_op = ht.Adam(lr=1e-2).minimize(loss)
+        xs = torch.randn(16, 8)
+        ys = (xs.sum(1, keepdim=True) > 0).float()
+        first = last = None
+        for i in range(150):
+            lv, _ = g.run([loss, train_op], {x: xs, y: ys})
+            if first is None:
+                first = lv.item()
+            last = lv.item()
+        assert last < first * 0.1
+
+    def test_mlp_sgd(self):
+        torch.manual_seed(1)
+        with ht.graph("define_and_run") as g:
+            x = ht.placeholder([8, 4], name="x")
+            w = ht.variable(torch.randn(4, 4) * 0.5, name="w")
+            y = ht.linear(x, w)
+            loss = ht.reduce_mean(ht.mul(y, y))
+            train_op = ht.SGD(lr=0.1).minimize(loss)
+        xs = torch.randn(8, 4)
+        l0 = g.run([loss, train_op], {x: xs})[0].item()
+        for _ in range(50):
+            ln = g.run([loss, train_op], {x: xs})[0].item()
+        assert ln < l0
+
+
+class TestEager:
+    def test_eager_basic(self):
+        with ht.graph("eager"):
+            a = ht.variable(torch.ones(3, 3) * 2, name="a")
+            b = ht.variable(torch.ones(3, 3) * 3, name="b")
+            c = ht.mul(a, b)
+            assert np.allclose(c.get_data().numpy(), 6.0)
+
+
+def test_fused_qkv_attention_matches_composed():
+    """Fused qkv attention (in-place RoPE + strided FA) must equal the
+    composed slice/rope/transpose/attention path, incl. GQA, dh=128."""
+    import os
+    import subprocess
+    import sys
+    code = r"""
+import os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+def run(fused):
+    os.environ["HETU_AMD_FUSED_ATTN"] = "1" if fused else "0"
+    import hetu_amd.models.llama as L
+    from hetu_amd.engine.runner import prepare_run_context
+    torch.manual_seed(0)
+    cfg = L.LlamaConfig(n_layer=2, n_head=4, n_kv_head=2, hidden=512,
+                        ffn_hidden=256, vocab=128, max_seq=32)
+    g, h = L.build_llama_train_graph(cfg, 2, 32, dtype=torch.float32,
+                                     lr=1e-3)
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    gen = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (2, 32), generator=gen)
+    lab = torch.randint(0, 128, (64,), generator=gen)
+    out = []
+    for _ in range(3):
+        lv, _ = g.run([h["loss"], h["train_op"]],
+                      {h["input_ids"]: ids, h["labels"]: lab}, ctx=ctx)
+        out.append(float(lv))
+    return out
+mode = sys.argv[1]
+print("LOSSES:" + repr(run(mode == "fused")))
+"""
+    import re
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    res = {}
+    for mode in ("fused", "composed"):
+        p = subprocess.run([sys.executable, "-c", code, mode],
+                           env={**os.environ, "HETU_REPO": repo},
+                           capture_output=True, text=True, timeout=300)
+        assert p.returncode == 0, p.stderr
+        m = re.search(r"LOSSES:(\[.*\])", p.stdout)
+        res[mode] = eval(m.group(1))  # noqa: S307
+    assert all(abs(a - b) < 2e-4
+               for a, b in zip(res["fused"], res["composed"])), res

@@ -1,0 +1,197 @@
+"""MoE correctness: single-process dense check + ep2 parity vs single
+(reference HetuMoE moe_layer + [H]AllToAll)."""
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+import torch
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _build_moe(dtype=torch.float32, spec=None, N=16, H=32, F=64, E=4, k=2,
+               cap=100.0):
+    import hetu_amd  # noqa
+    from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.nn.moe import MoEMLP
+    from hetu_amd.graph.ops.optim import Adam
+    g = DefineAndRunGraph("moe")
+    push_graph(g)
+    try:
+        x = ht.placeholder((N, H), dtype=dtype, name="x",
+                           ds=spec.ds_tokens(0) if spec else None,
+                           device_group=spec.device_group if spec else None)
+        tgt = ht.placeholder((N, H), dtype=dtype, name="tgt",
+                             ds=spec.ds_tokens(0) if spec else None,
+                             device_group=spec.device_group if spec else None)
+        moe = MoEMLP(H, F, E, spec=spec, k=k, capacity_factor=cap,
+                     dtype=dtype)
+        y = moe(x)
+        loss = ht.mse_loss(y, tgt)
+        opt = Adam(lr=1e-3)
+        train_op = opt.minimize(loss)
+    finally:
+        pop_graph()
+    return g, x, tgt, y, loss, train_op
+
+
+def test_moe_single_trains():
+    from hetu_amd.engine.runner import prepare_run_context
+    torch.manual_seed(0)
+    g, x, tgt, y, loss, train_op = _build_moe()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    xd = torch.randn(16, 32)
+    td = torch.randn(16, 32)
+    losses = []
+    for _ in range(8):
+        lv, _ = g.run([loss, train_op], {x: xd, tgt: td}, ctx=ctx)
+        losses.append(float(lv))
+    assert losses[-1] < losses[0], losses
+
+
+def test_moe_matches_dense_reference():
+    """With capacity >= N (no drops), the layer must equal the manual
+    top-k mixture computed with torch."""
+    from hetu_amd.engine.runner import prepare_run_context
+    torch.manual_seed(1)
+    N, H, F, E, K = 12, 16, 24, 4, 2
+    g, x, tgt, y, loss, train_op = _build_moe(N=N, H=H, F=F, E=E, k=K)
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    xd = torch.randn(N, H)
+    yv = g.run([y], {x: xd}, ctx=ctx)[0]
+    # manual reference
+    params = {p.name.split(":")[0]: p.get_data() for p in g.parameters}
+    logits = xd @ params["moe.gate.weight"].t()
+    probs = torch.softmax(logits, -1)
+    w, idx = probs.topk(K, -1)
+    ref = torch.zeros(N, H)
+    for t in range(N):
+        for kk in range(K):
+            e = int(idx[t, kk])
+            h1 = torch.nn.functional.gelu(
+                xd[t] @ params["moe.w1"][e], approximate="tanh")
+            ref[t] += w[t, kk] * (h1 @ params["moe.w2"][e])
+    assert torch.allclose(yv, ref, rtol=1e-3, atol=1e-4), \
+        (yv - ref).abs().max()
+
+
+WORKER = r"""
+import json, os, sys
+sys.path.insert(0, os.environ["HETU_REPO"])
+import torch
+import hetu_amd
+from hetu_amd.nn.parallel import ParallelSpec
+from hetu_amd.engine.runner import prepare_run_context
+sys.path.insert(0, os.path.join(os.environ["HETU_REPO"], "tests"))
+from test_moe_cpu import _build_moe
+
+ws = int(os.environ.get("WORLD_SIZE", "1"))
+rank = int(os.environ.get("RANK", "0"))
+N = 8   # tokens per rank
+torch.manual_seed(7)
+spec = ParallelSpec(dp=ws) if ws > 1 else None
+g, x, tgt, y, loss, train_op = _build_moe(N=N, H=16, F=24, E=4, k=2,
+                                          spec=spec)
+ctx = prepare_run_context(g, torch.device("cpu"))
+gen = torch.Generator().manual_seed(55)
+xd_all = torch.randn(N * max(ws, 1), 16, generator=gen)
+xd = xd_all[rank * N:(rank + 1) * N]
+yv = g.run([y], {x: xd}, ctx=ctx)[0]
+print("YOUT:" + json.dumps([rank, yv.flatten().tolist()]))
+"""
+
+
+def test_moe_ep2_matches_single():
+    """2-rank expert parallelism (2 experts/rank) must reproduce the
+    single-process 4-expert outputs for the same tokens."""
+    outs = {}
+    env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29581", "GLOO_SOCKET_IFNAME": "lo"}
+    procs = []
+    for r in range(2):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="2", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, "-c", WORKER],
+                                      env=env, stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=300)
+        got = False
+        for line in out.splitlines():
+            if line.startswith("YOUT:"):
+                rr, vals = json.loads(line[5:])
+                outs[rr] = np.array(vals)
+                got = True
+        # gloo teardown may SIGABRT (-6) after a clean run
+        ok = p.returncode == 0 or p.returncode == -6
+        assert ok, f"rank {r} failed:\n{out}\n{err}"
+    # single process over all 16 tokens
+    p = subprocess.run([sys.executable, "-c", WORKER],
+                       env={**os.environ, "HETU_REPO": REPO,
+                            "WORLD_SIZE": "1"},
+                       capture_output=True, text=True, timeout=300)
+    assert p.returncode == 0, p.stderr
+    single = None
+    for line in p.stdout.splitlines():
+        if line.startswith("YOUT:"):
+            _, vals = json.loads(line[5:])
+            single = np.array(vals)
+    # single worker ran N=8 tokens (rank slice 0); compare with rank 0
+    assert single is not None
+    assert np.allclose(outs[0], single, rtol=1e-3, atol=1e-4), \
+        np.abs(outs[0] - single).max()
+
+
+@pytest.mark.parametrize("gate_type", ["switch", "hash", "random"])
+def test_moe_gate_types_train(gate_type):
+    """Gate families beyond top-k (reference v1 gates/): switch top-1,
+    static hash and random routing all must train."""
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.graph.ops.optim import Adam
+    from hetu_amd.nn.moe import MoEMLP
+    torch.manual_seed(0)
+    g = DefineAndRunGraph("m")
+    push_graph(g)
+    try:
+        x = ht.placeholder((16, 32), name="x")
+        t = ht.placeholder((16, 32), name="t")
+        moe = MoEMLP(32, 64, 4, capacity_factor=100.0, gate_type=gate_type)
+        loss = ht.mse_loss(moe(x), t)
+        op = Adam(lr=1e-3).minimize(loss)
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    xd, td = torch.randn(16, 32), torch.randn(16, 32)
+    losses = []
+    for _ in range(8):
+        lv, _ = g.run([loss, op], {x: xd, t: td}, ctx=ctx)
+        losses.append(float(lv))
+    assert losses[-1] < losses[0], (gate_type, losses)
+    if gate_type in ("hash", "random"):
+        assert moe.gate is None          # no gate params for static routing
+
+
+def test_gpt_moe_model_trains():
+    """GPT-MoE (BASELINE config 4 family): dense trunk + expert MLP."""
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
+    cfg = GPTConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
+                    ffn_hidden=128, vocab=312, max_seq=16, moe_experts=4,
+                    moe_k=2)
+    g, h = build_gpt_train_graph(cfg, micro_batch=2, seq_len=16,
+                                 dtype=torch.float32, lr=1e-3)
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    torch.manual_seed(0)
+    ids = torch.randint(0, 312, (2, 16))
+    lab = torch.randint(0, 312, (32,))
+    losses = []
+    for _ in range(6):
+        lv, _ = g.run([h["loss"], h["train_op"]],
+                      {h["input_ids"]: ids, h["labels"]: lab}, ctx=ctx)
+        losses.append(float(lv))
+    assert losses[-1] < losses[0] - 0.5, losses

@@ -1,0 +1,164 @@
+"""Pipeline-parallel correctness on CPU (gloo): pp=2 1F1B with 4 micro
+batches must match single-process training on the same data (grads averaged
+over micro-batches).  Mirrors the reference CI's dp_tp_pp configs at
+miniature scale (tests/ci_test/ds_parallel_config/)."""
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = r"""
+import json, os, sys
+sys.path.insert(0, os.environ["HETU_REPO"])
+import torch
+import hetu_amd as ht
+from hetu_amd.models.llama import LlamaConfig, build_llama_pipeline_stage
+from hetu_amd.parallel.pipeline import PipelineSpec, PipelineRunner
+from hetu_amd.parallel.comm import comm_backend
+
+ws = int(os.environ.get("WORLD_SIZE", "1"))
+rank = int(os.environ.get("RANK", "0"))
+pp = int(os.environ.get("HETU_TEST_PP", str(ws)))
+tp = int(os.environ.get("HETU_TEST_TP", "1"))
+dp = ws // (pp * tp)
+M = 4
+cfg = LlamaConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
+                  ffn_hidden=128, vocab=312, max_seq=16)
+pspec = PipelineSpec(pp=pp, dp=dp, tp=tp)
+comm = comm_backend(torch.device("cpu"))
+stage = build_llama_pipeline_stage(cfg, pspec, micro_batch=1, seq_len=16,
+                                   dtype=torch.float32, lr=1e-3)
+runner = PipelineRunner(pspec, stage, torch.device("cpu"),
+                        offload=bool(int(os.environ.get(
+                            "HETU_TEST_OFFLOAD", "0"))))
+gen = torch.Generator().manual_seed(99)
+ids = torch.randint(0, cfg.vocab, (M, 1, 16), generator=gen)
+labels = torch.randint(0, cfg.vocab, (M, 16), generator=gen)
+h = stage.h
+losses = []
+for step in range(4):
+    mbs = []
+    for m in range(M):
+        feed = {}
+        if "input_ids" in h:
+            feed[h["input_ids"]] = ids[m]
+        if "labels" in h:
+            feed[h["labels"]] = labels[m]
+        mbs.append(feed)
+    loss = runner.step(mbs)
+    if loss is not None:
+        losses.append(float(loss))
+if losses and rank == pspec.world - 1:
+    print("LOSSES:" + json.dumps(losses))
+"""
+
+SINGLE = r"""
+import json, os, sys
+sys.path.insert(0, os.environ["HETU_REPO"])
+import torch
+from hetu_amd.models.llama import LlamaConfig, build_llama_pipeline_stage
+from hetu_amd.parallel.pipeline import PipelineSpec, PipelineRunner
+
+M = 4
+cfg = LlamaConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
+                  ffn_hidden=128, vocab=312, max_seq=16)
+pspec = PipelineSpec(pp=1, dp=1, tp=1)
+stage = build_llama_pipeline_stage(cfg, pspec, micro_batch=1, seq_len=16,
+                                   dtype=torch.float32, lr=1e-3)
+runner = PipelineRunner(pspec, stage, torch.device("cpu"))
+gen = torch.Generator().manual_seed(99)
+ids = torch.randint(0, cfg.vocab, (M, 1, 16), generator=gen)
+labels = torch.randint(0, cfg.vocab, (M, 16), generator=gen)
+h = stage.h
+losses = []
+for step in range(4):
+    mbs = [{h["input_ids"]: ids[m], h["labels"]: labels[m].reshape(-1)}
+           for m in range(M)]
+    losses.append(float(runner.step(mbs)))
+print("LOSSES:" + json.dumps(losses))
+"""
+
+
+def _launch_once(ws, extra_env, port, script=None):
+    base_env = dict(os.environ)
+    base_env["HETU_REPO"] = REPO
+    base_env["MASTER_ADDR"] = "127.0.0.1"
+    base_env["MASTER_PORT"] = str(port)
+    base_env.update(extra_env)
+    procs = []
+    for rank in range(ws):
+        env = dict(base_env)
+        if ws > 1:
+            env["RANK"] = str(rank)
+            env["WORLD_SIZE"] = str(ws)
+            env["LOCAL_RANK"] = str(rank)
+        procs.append(subprocess.Popen(
+            [sys.executable, "-c", script or WORKER], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    losses = None
+    for rank, p in enumerate(procs):
+        out, err = p.communicate(timeout=300)
+        # gloo teardown may SIGABRT (-6) after a clean run
+        ok = p.returncode in (0, -6)
+        assert ok, f"rank {rank} failed:\n{out}\n{err}"
+        for line in out.splitlines():
+            if line.startswith("LOSSES:"):
+                losses = json.loads(line[len("LOSSES:"):])
+    return losses
+
+
+# WORKER feeds labels flattened? fix: labels[m] is [16] then reshape needed
+WORKER = WORKER.replace('feed[h["labels"]] = labels[m]',
+                        'feed[h["labels"]] = labels[m].reshape(-1)')
+
+
+def _launch(ws, extra_env, port, script=None):
+    """Retry once: gloo occasionally SIGABRTs in teardown after a clean
+    run (non-deterministic; results already printed)."""
+    try:
+        return _launch_once(ws, extra_env, port, script)
+    except AssertionError:
+        return _launch_once(ws, extra_env, port + 40, script)
+
+
+@pytest.fixture(scope="module")
+def single_losses():
+    p = subprocess.run([sys.executable, "-c", SINGLE],
+                       env={**os.environ, "HETU_REPO": REPO},
+                       capture_output=True, text=True, timeout=300)
+    assert p.returncode == 0, f"single failed:\n{p.stdout}\n{p.stderr}"
+    for line in p.stdout.splitlines():
+        if line.startswith("LOSSES:"):
+            return json.loads(line[len("LOSSES:"):])
+    raise AssertionError("no losses")
+
+
+def test_pp2_matches_single(single_losses):
+    pp_losses = _launch(2, {"HETU_TEST_PP": "2"}, 29541)
+    assert pp_losses is not None
+    assert np.allclose(pp_losses, single_losses, rtol=2e-4, atol=1e-4), \
+        f"pp2 {pp_losses} vs single {single_losses}"
+    assert pp_losses[-1] < pp_losses[0]
+
+
+def test_pp2_tp2_matches_single(single_losses):
+    """4-rank grid: pp=2 x tp=2."""
+    losses = _launch(4, {"HETU_TEST_PP": "2", "HETU_TEST_TP": "2"}, 29542)
+    assert losses is not None
+    assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
+        f"pp2xtp2 {losses} vs single {single_losses}"
+
+
+def test_pp2_offload_matches_single(single_losses):
+    """Activation CPU offload between fwd and bwd must be numerically
+    invisible (reference activation_cpu_offload.cc semantics)."""
+    losses = _launch(2, {"HETU_TEST_PP": "2", "HETU_TEST_OFFLOAD": "1"},
+                     29543)
+    assert losses is not None
+    assert np.allclose(losses, single_losses, rtol=2e-4, atol=1e-4), \
+        f"pp2+offload {losses} vs single {single_losses}"

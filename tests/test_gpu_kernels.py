@@ -369,3 +369,30 @@ class TestGenerator:
         # decode a few tokens; caches must stay consistent
         out16 = g16.generate(ids, max_new_tokens=4, temperature=0.0)
         assert out16.shape == (2, 36)
+
+
+@pytest.mark.gpu
+def test_colsum_parity_and_replay():
+    """colsum vs fp32 reference, plus bit-stability across hipGraph
+    replays (the at::native reduce this replaces corrupts from the 2nd
+    replay on some shapes)."""
+    import hetu_amd.ops.functional as F
+    dev = torch.device("cuda", 0)
+    for R, C in [(512, 1024), (512, 768), (7, 8), (4096, 16384),
+                 (3, 1000)]:
+        x = torch.randn(R, C, dtype=torch.bfloat16, device=dev)
+        ref = x.float().sum(0)
+        out = F.colsum(x)
+        assert out.dtype == torch.float32
+        tol = 1e-2 * R ** 0.5
+        assert (out - ref).abs().max().item() < tol, (R, C)
+    x = torch.randn(512, 1024, dtype=torch.bfloat16, device=dev)
+    first = F.colsum(x).clone()
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        out = F.colsum(x)
+    for _ in range(3):
+        g.replay()
+        torch.cuda.synchronize()
+        assert torch.equal(out, first)
