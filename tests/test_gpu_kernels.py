@@ -384,7 +384,7 @@ def test_colsum_parity_and_replay():
         ref = x.float().sum(0)
         out = F.colsum(x)
         assert out.dtype == torch.float32
-        tol = 1e-2 * R ** 0.5
+        tol = 3e-2 * R ** 0.5 + 1e-3
         assert (out - ref).abs().max().item() < tol, (R, C)
     x = torch.randn(512, 1024, dtype=torch.bfloat16, device=dev)
     first = F.colsum(x).clone()
