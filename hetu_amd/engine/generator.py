@@ -54,8 +54,13 @@ class LlamaGenerator:
         import os
 
         from safetensors import safe_open
-        with open(os.path.join(path, "model.safetensors.index.json")) as fh:
-            index = json.load(fh)["weight_map"]
+        idx = os.path.join(path, "model.safetensors.index.json")
+        if os.path.exists(idx):
+            with open(idx) as fh:
+                index = json.load(fh)["weight_map"]
+        else:
+            # single-shard HF convention: plain model.safetensors, no index
+            index = {"": "model.safetensors"}
         state = {}
         for fn in sorted(set(index.values())):
             with safe_open(os.path.join(path, fn), framework="pt") as f:
