@@ -179,6 +179,31 @@ class CommBackend:
             for r in reqs:
                 r.wait()
 
+    def ensure_groups(self, rank_lists: List[List[int]]):
+        """WORLD-COLLECTIVE eager creation of subgroups, in a deterministic
+        order, called by every rank (members and non-members alike).
+
+        Needed for heterogeneous layouts: the lazy `group()` path uses
+        use_local_synchronization, whose hashed rendezvous name includes
+        this process's group-creation COUNT — ranks in differently-shaped
+        pipelines have different counts and would rendezvous on different
+        store keys (hang).  Eager world-collective creation keeps every
+        rank's counter aligned."""
+        if not dist.is_initialized():
+            return
+        seen = set()
+        ordered = []
+        for ranks in rank_lists:
+            key = tuple(sorted(ranks))
+            if len(key) > 1 and key not in seen and key not in self._groups:
+                seen.add(key)
+                ordered.append(key)
+        for key in sorted(ordered):
+            if list(key) == list(range(self.world_size)):
+                self._groups[key] = dist.group.WORLD
+            else:
+                self._groups[key] = dist.new_group(list(key))
+
     def allgather_object(self, obj, ranks: Optional[List[int]] = None):
         """All-gather arbitrary picklable objects (used for checkpoint
         shard-index coordination, not the hot path)."""

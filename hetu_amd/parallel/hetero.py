@@ -132,6 +132,23 @@ class HeteroGradSync:
         self.pi = spec.my_pipeline(self.rank)
         self.my_spec = spec.pipelines[self.pi]
         self.plans = [self._plan(p) for p in params]
+        # eagerly create every comm group ANY rank's sync will use, in the
+        # same order on every rank (see CommBackend.ensure_groups) — the
+        # union must be computed from the spec, not from this rank's view
+        groups = []
+        for plan in self.plans:
+            if plan is None:
+                continue
+            if plan["kind"] == "dup":
+                groups.append(plan["reps"])
+            else:
+                groups.extend(grp["reps"] for grp in plan["groups"])
+        for sp in spec.pipelines:
+            groups.append(list(sp.device_group))       # dup-param broadcast
+            for t in range(sp.tp):                     # split-param dups
+                _, dups = self._rep_and_dup(sp, t)
+                groups.append(dups)
+        self.comm.ensure_groups(groups)
 
     # ---- plan construction (once) ---------------------------------------
     def _rep_and_dup(self, spec: ParallelSpec, tp_idx: int):

@@ -1,0 +1,185 @@
+"""Heterogeneous data parallelism (Malleus): DistributedStatesUnion algebra
++ 3-rank exact-parity — pipeline A (tp2, ranks 0/1) and pipeline B (tp1,
+rank 2) train the same model on a 2:1 batch split and must match the
+single-process run exactly (reference distributed_states.h:158-233,
+SplitAllReduce Communication.h:660-786)."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_ds_union_algebra():
+    from hetu_amd.parallel.dstates import (DistributedStates,
+                                           DistributedStatesUnion,
+                                           NULL_HETERO_DIM)
+    ds = DistributedStates(4, {0: 2, -1: 2}, [0, -1])
+    homo = DistributedStatesUnion([ds])
+    assert not homo.is_hetero()
+    assert homo.get_local(0) is ds
+    # lift to a 2-entry hetero union along dim 0
+    u = DistributedStatesUnion.to_hetero(ds, 0, 2)
+    assert u.is_hetero() and u.hetero_dim == 0 and u.size() == 2
+    loc = u.get_local(0)
+    assert loc.device_num == 2
+    assert loc.get_dim(0) == 1 and loc.dup == 2
+    # hetero over the dup dim (grad unions)
+    d2 = DistributedStates(4, {-1: 4}, [-1])
+    u2 = DistributedStatesUnion([d2, d2], hetero_dim=-1)
+    assert u2.get_local(1).device_num == 2
+    assert u2.get_local(1).dup == 2
+    with pytest.raises(ValueError):
+        DistributedStatesUnion([ds, ds])          # union>1 needs hetero_dim
+    with pytest.raises(ValueError):
+        DistributedStatesUnion([ds], hetero_dim=0)
+    assert u.check_equal(DistributedStatesUnion.to_hetero(ds, 0, 2))
+    assert not u.check_equal(homo)
+    assert NULL_HETERO_DIM == -3
+
+
+def test_hetero_spec_batch_split_and_union():
+    from hetu_amd.nn.parallel import ParallelSpec
+    from hetu_amd.parallel.hetero import HeteroSpec
+    spec = HeteroSpec(pipelines=[
+        ParallelSpec(dp=1, tp=2, device_group=[0, 1]),
+        ParallelSpec(dp=1, tp=1, device_group=[2])],
+        weights=[2 / 3, 1 / 3])
+    assert spec.micro_batches(3) == [2, 1]
+    assert spec.micro_batches(6) == [4, 2]
+    assert spec.my_pipeline(rank=1) == 0
+    assert spec.my_pipeline(rank=2) == 1
+
+    class P:
+        shape = (8, 4)
+        hetero_split = (0, None)
+        ds = None
+    u = spec.param_union(P())
+    assert u.is_hetero() and u.hetero_dim == -1
+    assert u.get(0).get_dim(0) == 2      # tp2 pipeline splits dim 0
+    assert u.get(1).device_num == 1
+
+
+HETERO_WORKER = r"""
+import os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+from hetu_amd.parallel.comm import comm_backend
+from hetu_amd.nn.parallel import ParallelSpec
+from hetu_amd.parallel.hetero import HeteroSpec
+from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
+from hetu_amd.engine.runner import prepare_run_context
+comm = comm_backend()
+rank = comm.rank
+specA = ParallelSpec(dp=1, tp=2, device_group=[0, 1])
+specB = ParallelSpec(dp=1, tp=1, device_group=[2])
+hs = HeteroSpec(pipelines=[specA, specB], weights=[2/3, 1/3])
+pi = hs.my_pipeline()
+my_spec = hs.pipelines[pi]
+mb = hs.micro_batches(3)[pi]          # A: 2 rows, B: 1 row
+cfg = GPTConfig(n_layer=2, n_head=2, n_kv_head=2, hidden=32,
+                ffn_hidden=64, vocab=64, max_seq=8)
+S = 8
+g, h = build_gpt_train_graph(cfg, micro_batch=mb, seq_len=S,
+                             dtype=torch.float32, lr=1e-2,
+                             spec=my_spec, hetero=hs)
+ctx = prepare_run_context(g, torch.device("cpu"))
+gen = torch.Generator().manual_seed(99)
+ids_full = torch.randint(0, 64, (3, S), generator=gen)
+lab_full = torch.randint(0, 64, (3, S), generator=gen)
+rows = slice(0, 2) if pi == 0 else slice(2, 3)
+losses = []
+for step in range(3):
+    lv, _ = g.run([h["loss"], h["train_op"]],
+                  {h["input_ids"]: ids_full[rows],
+                   h["labels"]: lab_full[rows].reshape(-1)}, ctx=ctx)
+    losses.append(float(lv))
+print(f"HLOSS:{rank}:{losses}")
+if rank == 2:
+    # tp1 pipeline holds full params: dump a couple for parity checking
+    out = {}
+    for p in g.parameters:
+        if p.name.startswith(("h0.attn.wqkv.weight", "h0.mlp.wfc.weight",
+                              "wte.weight", "lnf.weight")):
+            out[p.name] = p.get_data().double().sum().item()
+    import json as _json
+    print("HPAR:" + _json.dumps(out))
+"""
+
+SINGLE_WORKER = r"""
+import os, sys, torch, json
+sys.path.insert(0, os.environ["HETU_REPO"])
+from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
+from hetu_amd.engine.runner import prepare_run_context
+cfg = GPTConfig(n_layer=2, n_head=2, n_kv_head=2, hidden=32,
+                ffn_hidden=64, vocab=64, max_seq=8)
+S = 8
+g, h = build_gpt_train_graph(cfg, micro_batch=3, seq_len=S,
+                             dtype=torch.float32, lr=1e-2)
+ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+gen = torch.Generator().manual_seed(99)
+ids_full = torch.randint(0, 64, (3, S), generator=gen)
+lab_full = torch.randint(0, 64, (3, S), generator=gen)
+losses = []
+for step in range(3):
+    lv, _ = g.run([h["loss"], h["train_op"]],
+                  {h["input_ids"]: ids_full,
+                   h["labels"]: lab_full.reshape(-1)}, ctx=ctx)
+    losses.append(float(lv))
+print(f"SLOSS:{losses}")
+out = {}
+for p in g.parameters:
+    if p.name.startswith(("h0.attn.wqkv.weight", "h0.mlp.wfc.weight",
+                          "wte.weight", "lnf.weight")):
+        out[p.name] = p.get_data().double().sum().item()
+print("SPAR:" + json.dumps(out))
+"""
+
+
+def test_hetero_tp2_tp1_exact_parity():
+    env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29687", "GLOO_SOCKET_IFNAME": "lo"}
+    procs = []
+    for r in range(3):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="3", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, "-c", HETERO_WORKER],
+                                      env=env, stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    hloss = {}
+    hpar = None
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=600)
+        assert p.returncode in (0, -6), f"rank {r}: {out}\n{err}"
+        for ln in out.splitlines():
+            if ln.startswith("HLOSS:"):
+                _, rr, ls = ln.split(":", 2)
+                hloss[int(rr)] = json.loads(ls)
+            elif ln.startswith("HPAR:"):
+                hpar = json.loads(ln[len("HPAR:"):])
+    assert hpar is not None and 0 in hloss and 2 in hloss
+
+    env = dict(env0)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    sp = subprocess.run([sys.executable, "-c", SINGLE_WORKER], env=env,
+                        capture_output=True, text=True, timeout=600)
+    assert sp.returncode == 0, sp.stderr
+    sloss = spar = None
+    for ln in sp.stdout.splitlines():
+        if ln.startswith("SLOSS:"):
+            sloss = json.loads(ln[len("SLOSS:"):])
+        elif ln.startswith("SPAR:"):
+            spar = json.loads(ln[len("SPAR:"):])
+
+    # weighted pipeline losses == global loss, every step
+    for i in range(3):
+        mix = (2 / 3) * hloss[0][i] + (1 / 3) * hloss[2][i]
+        assert abs(mix - sloss[i]) < 1e-5, (i, mix, sloss[i])
+    # params after 3 hetero steps match the single-process params (small
+    # fp32 summation-order drift amplified by Adam's rsqrt is expected;
+    # the per-step loss parity above already pins the update semantics)
+    for k, v in spar.items():
+        assert abs(hpar[k] - v) < 5e-4, (k, hpar[k], v)
