@@ -62,9 +62,10 @@ class ElasticController:
                 for r in alive:
                     if r in votes:
                         continue
+                    key = f"elastic/vote/{self.epoch}/{r}"
                     try:
-                        v = self.kv.get(f"elastic/vote/{self.epoch}/{r}")
-                        votes[r] = int(v)
+                        if self.kv.has(key):
+                            votes[r] = int(self.kv.get(key))
                     except Exception:  # noqa: BLE001
                         pass
                 time.sleep(0.02)
@@ -100,9 +101,13 @@ class ElasticWorker:
         if status == VOTING and self._voted_epoch < epoch:
             self.kv.put(f"elastic/vote/{epoch}/{self.rank}", completed_step)
             self._voted_epoch = epoch
-        # wait for the published plan of this epoch
+        # wait for the published plan of this epoch (non-fatal timeout:
+        # the caller polls again)
         key = f"elastic/plan/{epoch}"
-        self.kv.wait([key], timeout_s=60.0)
+        try:
+            self.kv.wait([key], timeout_s=30.0)
+        except Exception:  # noqa: BLE001
+            return None
         plan = self.kv.get(key)
         if isinstance(plan, str):
             plan = json.loads(plan)

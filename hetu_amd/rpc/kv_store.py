@@ -45,6 +45,9 @@ class KVStore:
                 return v
         return v
 
+    def has(self, key: str) -> bool:
+        return self.store.check([key])
+
     def wait(self, keys: List[str], timeout_s: Optional[float] = None):
         if timeout_s is not None:
             self.store.wait(keys, timedelta(seconds=timeout_s))

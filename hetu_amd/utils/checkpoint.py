@@ -236,10 +236,16 @@ def collect_adam_states(graph) -> Dict[str, Dict]:
 
 
 def load_adam_states(graph, path: str,
-                     comm: Optional[CommBackend] = None) -> int:
+                     comm: Optional[CommBackend] = None,
+                     rank_override: Optional[int] = None) -> int:
+    """rank_override: read another rank's optim shard (pure-dp elastic
+    recovery: every rank holds identical states, survivors read rank 0's
+    file after a reshape)."""
     from safetensors import safe_open
     comm = comm or comm_backend()
     rank = comm.rank if comm else 0
+    if rank_override is not None:
+        rank = rank_override
     fp = os.path.join(path, f"optim-r{rank:03d}.safetensors")
     if not os.path.exists(fp):
         return 0
