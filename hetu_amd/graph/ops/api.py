@@ -290,6 +290,14 @@ def fused_qkv_attention(qkv, n_head, n_kv_head, head_dim, cos=None,
                           "scale": scale}).output()
 
 
+def fused_mlp(x, wfc, b1, wproj, b2=None):
+    """Epilogue-fused transformer MLP (see nnops.FusedMLPOp):
+    y = gelu(x @ wfc^T + b1) @ wproj^T (+ b2)."""
+    ins = [x, wfc, b1, wproj] + ([b2] if b2 is not None else [])
+    return _cg().make_op(N.FusedMLPOp(), ins,
+                         {"with_b2": b2 is not None}).output()
+
+
 def varlen_attention(q, k, v, cu_seqlens, causal=True, scale=None):
     """Packed-varlen attention: q/k/v [T, H, D], cu_seqlens [n+1]."""
     return _cg().make_op(N.VarlenAttentionOp(), [q, k, v, cu_seqlens],

@@ -668,3 +668,109 @@ class VarlenAttentionGradOp(OpInterface):
         return list(F.varlen_attention_bwd(dout, q, k, v, out, lse, cu,
                                            op.attrs.get("causal", True),
                                            op.attrs.get("scale")))
+
+
+class FusedMLPOp(OpInterface):
+    """Transformer MLP block with hipBLASLt epilogue fusion:
+    y = gelu(x @ wfc^T + b1) @ wproj^T (+ b2 when attrs["with_b2"]).
+    gelu rides the fc GEMM (GELU_AUX_BIAS); backward's dgelu and b1-grad
+    ride the dgrad GEMM (DGELU_BGRAD) — no standalone gelu/reduce kernels
+    (reference keeps Gelu.cu + Reduce.cu around cuBLAS).
+    inputs: x [B, S, H], wfc [F, H], b1 [F], wproj [Ho, F][, b2 [Ho]];
+    outputs: y [B, S, Ho], a (gelu out, saved), aux (pre-gelu, saved).
+    tp>1 layouts should use the composed path (the fused op is the
+    single-device / dp hot path)."""
+    type = "FusedMLP"
+
+    def infer_meta(self, attrs, inputs):
+        x, wfc = inputs[0], inputs[1]
+        wproj = inputs[3]
+        B, S = x.shape[0], x.shape[1]
+        F = wfc.shape[0]
+        Ho = wproj.shape[0]
+        return [TensorMeta((B, S, Ho), x.dtype),
+                TensorMeta((B * S, F), x.dtype),
+                TensorMeta((B * S, F), x.dtype)]
+
+    def deduce_states(self, op):
+        x = op.inputs[0]
+        for out in op.outputs:
+            out.ds = x.ds
+            out.device_group = x.device_group
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        x, wfc, b1, wproj = inputs[:4]
+        b2 = inputs[4] if len(inputs) > 4 else None
+        B, S, H = x.shape
+        x2 = x.reshape(B * S, H)
+        a, aux = F.linear_gelu_aux(x2, wfc, b1)
+        y = F.linear(a, wproj, b2)
+        return [y.reshape(B, S, wproj.shape[0]), a, aux]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        ins = [g[0], op.inputs[0], op.inputs[1], op.inputs[3],
+               op.outputs[1], op.outputs[2]]
+        bwd = _make(gr, FusedMLPGradOp(),
+                    ins, dict(op.attrs), name="fused_mlp_grad")
+        grads = [bwd.output(0), bwd.output(1), bwd.output(2),
+                 bwd.output(3)]
+        if len(op.inputs) > 4:
+            grads.append(bwd.output(4))
+        return grads
+
+
+class FusedMLPGradOp(OpInterface):
+    """inputs: dy, x, wfc, wproj, a, aux; outputs: dx, dwfc, db1, dwproj
+    [, db2]."""
+    type = "FusedMLPGrad"
+
+    def infer_meta(self, attrs, inputs):
+        dy, x, wfc, wproj = inputs[:4]
+        outs = [TensorMeta(x.shape, x.dtype),
+                TensorMeta(wfc.shape, wfc.dtype),
+                TensorMeta((wfc.shape[0],), wfc.dtype),
+                TensorMeta(wproj.shape, wproj.dtype)]
+        if attrs.get("with_b2"):
+            outs.append(TensorMeta((wproj.shape[0],), wproj.dtype))
+        return outs
+
+    def deduce_states(self, op):
+        x = op.inputs[1]
+        op.outputs[0].ds = x.ds
+        if x.ds is not None:
+            # weight/bias grads: partial over the token-split (dp) dims,
+            # dup over the rest (same rule as LayerNormGradOp)
+            n = x.ds.device_num
+            npart = x.ds.partial
+            for d in x.ds.split_dims():
+                npart *= x.ds.get_dim(d)
+            states = {}
+            if npart > 1:
+                states[-2] = npart
+            if n // max(npart, 1) > 1:
+                states[-1] = n // max(npart, 1)
+            wds = DistributedStates(n, states)
+            for out in op.outputs[1:]:
+                out.ds = wds
+        for out in op.outputs:
+            out.device_group = x.device_group
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        dy, x, wfc, wproj, a, aux = inputs
+        B, S, H = x.shape
+        Ho = wproj.shape[0]
+        dy2 = dy.reshape(B * S, Ho)
+        x2 = x.reshape(B * S, H)
+        # dwproj = dy^T @ a; da->dgelu(+db1) fused; dwfc = dh^T @ x;
+        # dx = dh @ wfc
+        dwproj = torch.matmul(dy2.t(), a).to(wproj.dtype)
+        dh, db1 = F.dgelu_bgrad(dy2, wproj, aux)
+        dwfc = torch.matmul(dh.t(), x2).to(wfc.dtype)
+        dx = torch.matmul(dh, wfc).reshape(B, S, H)
+        outs = [dx, dwfc, db1, dwproj]
+        if op.attrs.get("with_b2"):
+            outs.append(F.colsum(dy2).to(dy.dtype))
+        return outs

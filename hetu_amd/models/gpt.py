@@ -129,6 +129,12 @@ class GPTMLP(Module):
             name=f"{p}.wproj", init_std=proj_std)
 
     def forward(self, x):
+        if self.wfc.spec.tp == 1 and os.environ.get(
+                "HETU_AMD_FUSED_MLP", "1") == "1":
+            # hipBLASLt epilogue fusion (single-device/dp hot path):
+            # gelu rides the fc GEMM, dgelu+b1-grad ride the dgrad GEMM
+            return ht.fused_mlp(x, self.wfc.weight, self.wfc.bias,
+                                self.wproj.weight, self.wproj.bias)
         return self.wproj(ht.gelu(self.wfc(x)))
 
 

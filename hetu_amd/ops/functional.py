@@ -712,3 +712,30 @@ def varlen_attention_bwd(dout, q, k, v, out, lse, cu_seqlens,
         dk[s0:s1] = dks[0].permute(1, 0, 2)
         dv[s0:s1] = dvs[0].permute(1, 0, 2)
     return dq, dk, dv
+
+
+# ---------------------------------------------------------------------------
+# hipBLASLt epilogue-fused MLP pieces (ltgemm.cpp): gelu folded into the
+# fc GEMM (GELU_AUX_BIAS) and dgelu+bias-grad into the backward GEMM
+# (DGELU_BGRAD).  CPU path = the exact reference composition.
+# ---------------------------------------------------------------------------
+
+def linear_gelu_aux(x2d, w, b):
+    """returns (gelu(x@w^T+b), pre-gelu aux)."""
+    if _use_hip("ltgemm", x2d):
+        return tuple(ext().lt_linear_gelu_aux(x2d.contiguous(),
+                                              w.contiguous(),
+                                              b.contiguous()))
+    h = torch.nn.functional.linear(x2d, w.to(x2d.dtype), b.to(x2d.dtype))
+    return gelu_fwd(h), h
+
+
+def dgelu_bgrad(dy2d, w, aux):
+    """returns (dgelu(dy@w, aux), colsum(dgelu(...)))."""
+    if _use_hip("ltgemm", dy2d):
+        return tuple(ext().lt_dgelu_bgrad(dy2d.contiguous(),
+                                          w.contiguous(),
+                                          aux.contiguous()))
+    da = torch.matmul(dy2d, w.to(dy2d.dtype))
+    dh = gelu_bwd(da, aux)
+    return dh, colsum(dh).to(dh.dtype)

@@ -15,6 +15,12 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor rstd);
 // reduce.hip
 torch::Tensor colsum(torch::Tensor x);
+// ltgemm.cpp
+std::vector<torch::Tensor> lt_linear_gelu_aux(torch::Tensor x,
+                                              torch::Tensor w,
+                                              torch::Tensor b);
+std::vector<torch::Tensor> lt_dgelu_bgrad(torch::Tensor dy, torch::Tensor w,
+                                          torch::Tensor aux);
 // swiglu.hip
 torch::Tensor swiglu_fwd(torch::Tensor x);
 torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor x);
@@ -128,6 +134,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd2", &layernorm_bwd2);
   m.def("rmsnorm_bwd2", &rmsnorm_bwd2);
   m.def("colsum", &colsum);
+  m.def("lt_linear_gelu_aux", &lt_linear_gelu_aux);
+  m.def("lt_dgelu_bgrad", &lt_dgelu_bgrad);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("gelu_fwd", &gelu_fwd);

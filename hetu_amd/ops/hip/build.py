@@ -31,6 +31,7 @@ SOURCES = [
     "attention_bwd_v3.hip",
     "attention_bwd_v2.hip",
     "quant.hip",
+    "ltgemm.cpp",
     "galvatron_dp.cpp",
     "embed_cache.cpp",
     "dataloader.cpp",
@@ -87,7 +88,7 @@ def build(verbose: bool = True) -> str:
         objs.append(obj)
     link = ["hipcc", "-shared", "-fPIC", "-o", out] + objs + [
         f"-L{libdir}", "-ltorch", "-ltorch_cpu", "-lc10",
-        "-ltorch_python", "-lamdhip64",
+        "-ltorch_python", "-lamdhip64", "-lhipblaslt",
         f"-Wl,-rpath,{libdir}",
     ]
     hiplibs = [l for l in ("torch_hip", "c10_hip")

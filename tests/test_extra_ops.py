@@ -378,3 +378,43 @@ def test_transposed_matmul_and_extremum_reduce_grads():
     r, = g.run(gs, {x: xv.detach()}, ctx=ctx)
     xv.amax(1).pow(2).sum().backward()
     assert torch.allclose(r, xv.grad, atol=1e-5)
+
+
+def test_fused_mlp_parity():
+    """FusedMLPOp (hipBLASLt-epilogue path on GPU; exact composition on
+    CPU) vs the composed linear+gelu+linear graph: forward and all grads."""
+    import torch
+    from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.engine.runner import prepare_run_context
+    torch.manual_seed(3)
+    B, S, H, F = 2, 8, 16, 32
+    xw = torch.randn(B, S, H)
+    wfc = torch.randn(F, H) * 0.2
+    b1 = torch.randn(F) * 0.1
+    wproj = torch.randn(H, F) * 0.2
+    b2 = torch.randn(H) * 0.1
+    outs = {}
+    for mode in ("fused", "composed"):
+        g = DefineAndRunGraph(mode)
+        push_graph(g)
+        try:
+            x = ht.placeholder((B, S, H), name="x")
+            vfc = ht.variable(wfc.clone(), name="wfc")
+            v1 = ht.variable(b1.clone(), name="b1")
+            vpr = ht.variable(wproj.clone(), name="wproj")
+            v2 = ht.variable(b2.clone(), name="b2")
+            if mode == "fused":
+                y = ht.fused_mlp(x, vfc, v1, vpr, v2)
+            else:
+                y = ht.add(ht.linear(ht.gelu(ht.linear(x, vfc, v1)), vpr),
+                           v2)
+            loss = ht.reduce_sum(ht.mul(y, y))
+            gs = ht.gradients([loss], [x, vfc, v1, vpr, v2])
+        finally:
+            pop_graph()
+        ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+        vals = g.run([y] + gs, {x: xw.clone()}, ctx=ctx)
+        outs[mode] = vals
+    for i, (a, b) in enumerate(zip(outs["fused"], outs["composed"])):
+        assert torch.allclose(a, b, atol=1e-4), (i, (a - b).abs().max())
