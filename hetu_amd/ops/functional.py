@@ -720,22 +720,45 @@ def varlen_attention_bwd(dout, q, k, v, out, lse, cu_seqlens,
 # (DGELU_BGRAD).  CPU path = the exact reference composition.
 # ---------------------------------------------------------------------------
 
+_LT_OK = [None]     # None = untested; False = no algos on this build
+
+
+def _lt_available() -> bool:
+    return _LT_OK[0] is not False
+
+
 def linear_gelu_aux(x2d, w, b):
     """returns (gelu(x@w^T+b), pre-gelu aux)."""
-    if _use_hip("ltgemm", x2d):
-        return tuple(ext().lt_linear_gelu_aux(x2d.contiguous(),
-                                              w.contiguous(),
-                                              b.contiguous()))
+    if _use_hip("ltgemm", x2d) and _lt_available():
+        try:
+            out = tuple(ext().lt_linear_gelu_aux(x2d.contiguous(),
+                                                 w.contiguous(),
+                                                 b.contiguous()))
+            _LT_OK[0] = True
+            return out
+        except RuntimeError as e:
+            if "no algo" not in str(e):
+                raise
+            # this hipBLASLt build has no GELU_AUX solutions: compose
+            # (bias still fused via F.linear; gelu = hand HIP kernel)
+            _LT_OK[0] = False
+            print("[hetu_amd] hipBLASLt GELU_AUX epilogue unavailable, "
+                  "using composed MLP path")
     h = torch.nn.functional.linear(x2d, w.to(x2d.dtype), b.to(x2d.dtype))
     return gelu_fwd(h), h
 
 
 def dgelu_bgrad(dy2d, w, aux):
     """returns (dgelu(dy@w, aux), colsum(dgelu(...)))."""
-    if _use_hip("ltgemm", dy2d):
-        return tuple(ext().lt_dgelu_bgrad(dy2d.contiguous(),
-                                          w.contiguous(),
-                                          aux.contiguous()))
+    if _use_hip("ltgemm", dy2d) and _LT_OK[0] is True:
+        try:
+            return tuple(ext().lt_dgelu_bgrad(dy2d.contiguous(),
+                                              w.contiguous(),
+                                              aux.contiguous()))
+        except RuntimeError as e:
+            if "no algo" not in str(e):
+                raise
+            _LT_OK[0] = False
     da = torch.matmul(dy2d, w.to(dy2d.dtype))
     dh = gelu_bwd(da, aux)
     return dh, colsum(dh).to(dh.dtype)
