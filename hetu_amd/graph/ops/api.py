@@ -323,10 +323,18 @@ def check_finite(tensors) -> Tensor:
     return _cg().make_op(B.CheckFiniteOp(), list(tensors), {}).output()
 
 
-def ring_attention(q, k, v, cp_ranks, causal=True, scale=None):
+def ring_attention(q, k, v, cp_ranks, causal=True, scale=None,
+                   split=None):
+    """split: NORMAL (contiguous chunks) or SYM (zigzag [head|tail]
+    halves, causal-load-balanced); default from
+    HETU_AMD_ATTN_SPLIT (reference HETU_PARALLEL_ATTN_SPLIT_PATTERN)."""
+    import os as _os
     from . import parallel_ops as P
+    if split is None:
+        split = _os.environ.get("HETU_AMD_ATTN_SPLIT", "NORMAL")
     return _cg().make_op(P.RingAttentionOp(), [q, k, v],
                          {"causal": causal, "scale": scale,
+                          "split": split,
                           "cp_ranks": list(cp_ranks)}).output(0)
 
 

@@ -170,10 +170,19 @@ class RingAttentionOp(OpInterface):
             t.device_group = q.device_group
 
     def compute(self, op, inputs, ctx):
-        from ...parallel.ring_attention import ring_attn_fwd
+        from ...parallel.ring_attention import (ring_attn_fwd,
+                                                ring_attn_fwd_sym)
         q, k, v = inputs
-        o, lse = ring_attn_fwd(q, k, v, ctx.comm, op.attrs["cp_ranks"],
-                               op.attrs["causal"], op.attrs.get("scale"))
+        if op.attrs.get("split", "NORMAL") == "SYM" \
+                and op.attrs.get("causal", True) and ctx.comm is not None \
+                and len(op.attrs["cp_ranks"]) > 1:
+            o, lse = ring_attn_fwd_sym(q, k, v, ctx.comm,
+                                       op.attrs["cp_ranks"],
+                                       op.attrs.get("scale"))
+        else:
+            o, lse = ring_attn_fwd(q, k, v, ctx.comm, op.attrs["cp_ranks"],
+                                   op.attrs["causal"],
+                                   op.attrs.get("scale"))
         return [o, lse]
 
     def gradient(self, op, g):
@@ -202,8 +211,15 @@ class RingAttentionGradOp(OpInterface):
             t.device_group = src.device_group
 
     def compute(self, op, inputs, ctx):
-        from ...parallel.ring_attention import ring_attn_bwd
+        from ...parallel.ring_attention import (ring_attn_bwd,
+                                                ring_attn_bwd_sym)
         dout, q, k, v, o, lse = inputs
+        if op.attrs.get("split", "NORMAL") == "SYM" \
+                and op.attrs.get("causal", True) and ctx.comm is not None \
+                and len(op.attrs["cp_ranks"]) > 1:
+            return list(ring_attn_bwd_sym(dout, q, k, v, o, lse, ctx.comm,
+                                          op.attrs["cp_ranks"],
+                                          op.attrs.get("scale")))
         return list(ring_attn_bwd(dout, q, k, v, o, lse, ctx.comm,
                                   op.attrs["cp_ranks"], op.attrs["causal"],
                                   op.attrs.get("scale")))

@@ -88,6 +88,77 @@ def ring_attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return o, lse
 
 
+def _sym_subblocks(my_pos: int, src_pos: int, n: int):
+    """Sub-block schedule for the SYM (zigzag) split: rank p owns global
+    chunks (p, 2n-1-p) as its [head | tail] halves.  Yields
+    (q_half, kv_half, causal_diag) for every visible pair — causal work is
+    (2n+1) chunk-pairs for EVERY rank (reference STRIPE/SYM split,
+    ParallelAttention.cc:196-204; data side: bucket.generate_cp_pack_data).
+    """
+    mine = (my_pos, 2 * n - 1 - my_pos)
+    src = (src_pos, 2 * n - 1 - src_pos)
+    for qh, qc in enumerate(mine):
+        for kh, kc in enumerate(src):
+            if kc < qc:
+                yield qh, kh, False
+            elif kc == qc:
+                yield qh, kh, True
+
+
+def ring_attn_fwd_sym(q, k, v, comm, ranks, scale=None):
+    """Causal ring attention with the SYM split: local seq = [head|tail]
+    halves of the global sequence (chunks p and 2n-1-p)."""
+    n = len(ranks)
+    half = q.shape[2] // 2
+    my_pos = ranks.index(comm.rank)
+    B, H, S, D = q.shape
+    o = torch.zeros(B, H, S, D, dtype=torch.float32, device=q.device)
+    lse = torch.full((B, H, S), float("-inf"), dtype=torch.float32,
+                     device=q.device)
+    kv_k, kv_v = k, v
+    src_pos = my_pos
+    for step in range(n):
+        for qh, kh, diag in _sym_subblocks(my_pos, src_pos, n):
+            qs = slice(qh * half, (qh + 1) * half)
+            ks = slice(kh * half, (kh + 1) * half)
+            ob, lb = F.flash_attn_fwd(q[:, :, qs], kv_k[:, :, ks],
+                                      kv_v[:, :, ks], diag, scale)
+            om, lm = _merge(o[:, :, qs].to(q.dtype), lse[:, :, qs], ob, lb)
+            o[:, :, qs] = om.float()
+            lse[:, :, qs] = lm
+        if step < n - 1:
+            kv_k, kv_v = _ring_exchange(comm, ranks, my_pos, [kv_k, kv_v])
+            src_pos = (src_pos - 1) % n
+    return o.to(q.dtype), lse
+
+
+def ring_attn_bwd_sym(dout, q, k, v, o, lse, comm, ranks, scale=None):
+    n = len(ranks)
+    half = q.shape[2] // 2
+    my_pos = ranks.index(comm.rank)
+    dq = torch.zeros_like(q, dtype=torch.float32)
+    dk_acc = torch.zeros_like(k, dtype=torch.float32)
+    dv_acc = torch.zeros_like(v, dtype=torch.float32)
+    kv_k, kv_v = k, v
+    src_pos = my_pos
+    for step in range(n):
+        for qh, kh, diag in _sym_subblocks(my_pos, src_pos, n):
+            qs = slice(qh * half, (qh + 1) * half)
+            ks = slice(kh * half, (kh + 1) * half)
+            dqb, dkb, dvb = F.flash_attn_bwd(
+                dout[:, :, qs], q[:, :, qs], kv_k[:, :, ks],
+                kv_v[:, :, ks], o[:, :, qs], lse[:, :, qs], diag, scale)
+            dq[:, :, qs] += dqb.float()
+            dk_acc[:, :, ks] += dkb.float()
+            dv_acc[:, :, ks] += dvb.float()
+        if step < n - 1:
+            kv_k, kv_v, dk_acc, dv_acc = _ring_exchange(
+                comm, ranks, my_pos, [kv_k, kv_v, dk_acc, dv_acc])
+            src_pos = (src_pos - 1) % n
+    dk_acc, dv_acc = _ring_exchange(comm, ranks, my_pos, [dk_acc, dv_acc])
+    return dq.to(q.dtype), dk_acc.to(k.dtype), dv_acc.to(v.dtype)
+
+
 def ring_attn_bwd(dout: torch.Tensor, q: torch.Tensor, k: torch.Tensor,
                   v: torch.Tensor, o: torch.Tensor, lse: torch.Tensor,
                   comm: Optional[CommBackend], ranks: List[int],
