@@ -226,9 +226,15 @@ class GPTLMHeadModel(Module):
                                   for i in range(cfg.n_layer)])
         self.lnf = ParallelLayerNorm(cfg.hidden, spec, 1e-5, dtype,
                                      name="lnf")
-        self.lm_head = ColumnParallelLinear(
-            cfg.hidden, cfg.vocab, spec, bias=False, dtype=dtype,
-            name="lm_head", init_std=cfg.init_std)
+        if cfg.tie_embeddings:
+            # tied wte/lm_head: the SAME graph tensor feeds both the
+            # embedding lookup and the output projection; autodiff sums
+            # the two gradient paths (reference shared-weight semantics)
+            self.lm_head = None
+        else:
+            self.lm_head = ColumnParallelLinear(
+                cfg.hidden, cfg.vocab, spec, bias=False, dtype=dtype,
+                name="lm_head", init_std=cfg.init_std)
 
     def forward(self, input_ids, labels=None):
         import contextlib
@@ -241,8 +247,11 @@ class GPTLMHeadModel(Module):
             with cm:
                 x = blk(x, B, S)
         x = self.lnf(x)
-        logits = self.lm_head(
-            ht.reshape(x, (B * S, cfg.hidden), ds=spec.ds_tokens(0)))
+        xr = ht.reshape(x, (B * S, cfg.hidden), ds=spec.ds_tokens(0))
+        if self.lm_head is None:
+            logits = ht.linear(xr, self.embed.wte.weight)
+        else:
+            logits = self.lm_head(xr)
         if labels is None:
             return None, logits
         per_tok = vocab_parallel_cross_entropy(logits, labels, cfg.vocab)
@@ -313,6 +322,8 @@ def build_gpt_pipeline_stage(cfg: GPTConfig, pspec, micro_batch: int,
             embed = GPTEmbedding(cfg, spec, S, dtype)
             x = embed(input_ids)
             h["input_ids"] = input_ids
+            if cfg.tie_embeddings:
+                h["tied_name"] = embed.wte.weight.name
         else:
             act_in = ht.placeholder((B, S, cfg.hidden), dtype=dtype,
                                     name="act_in", ds=ds_in,
@@ -327,12 +338,26 @@ def build_gpt_pipeline_stage(cfg: GPTConfig, pspec, micro_batch: int,
                                     device_group=spec.device_group)
             lnf = ParallelLayerNorm(cfg.hidden, spec, 1e-5, dtype,
                                     name="lnf")
-            lm_head = ColumnParallelLinear(
-                cfg.hidden, cfg.vocab, spec, bias=False, dtype=dtype,
-                name="lm_head", init_std=cfg.init_std)
             xo = lnf(x)
-            logits = lm_head(
-                ht.reshape(xo, (B * S, cfg.hidden), ds=spec.ds_tokens(0)))
+            xr = ht.reshape(xo, (B * S, cfg.hidden), ds=spec.ds_tokens(0))
+            if cfg.tie_embeddings and is_first:
+                # pp == 1: the one stage holds wte — tie directly
+                logits = ht.linear(xr, embed.wte.weight)
+            elif cfg.tie_embeddings:
+                # shared wte/lm_head across first/last stage: the last
+                # stage holds its own copy (identical per-name init);
+                # PipelineRunner p2p-sums the two stages' grads each step
+                # (reference executable_graph.cc:929-933 shared-weight p2p)
+                tied_w = VocabParallelEmbedding(
+                    cfg.vocab, cfg.hidden, spec, dtype=dtype, name="wte",
+                    init_std=cfg.init_std).weight
+                h["tied_name"] = tied_w.name
+                logits = ht.linear(xr, tied_w)
+            else:
+                lm_head = ColumnParallelLinear(
+                    cfg.hidden, cfg.vocab, spec, bias=False, dtype=dtype,
+                    name="lm_head", init_std=cfg.init_std)
+                logits = lm_head(xr)
             per_tok = vocab_parallel_cross_entropy(logits, labels, cfg.vocab)
             loss = ht.reduce_mean(per_tok)
             h["labels"] = labels

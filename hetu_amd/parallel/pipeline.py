@@ -97,10 +97,12 @@ class PipelineRunner:
 
     def __init__(self, spec: PipelineSpec, stage: StageModule,
                  device: torch.device, ctx=None, scaler=None,
-                 recompute: bool = False, offload: bool = False):
+                 recompute: bool = False, offload: bool = False,
+                 schedule: str = "1f1b"):
         self.spec = spec
         self.stage = stage
         self.device = device
+        self.schedule = schedule      # "1f1b" (PipeDream-flush) | "gpipe"
         self.scaler = scaler          # engine.amp.GradScaler or None
         self.recompute = recompute    # rerun fwd in bwd instead of caching
         self.offloader = None
@@ -194,9 +196,11 @@ class PipelineRunner:
         _p2p(self.comm, [], [(t, self.prev)])
         return t
 
-    # ---- one training step (1F1B) ---------------------------------------
+    # ---- one training step (1F1B | GPipe) --------------------------------
     def step(self, micro_batches: List[Dict]):
         """Returns the mean micro-batch loss on the LAST stage, else None."""
+        if self.schedule == "gpipe":
+            return self._step_gpipe(micro_batches)
         M = len(micro_batches)
         h = self.stage.h
         for b in self.grad_bufs:
@@ -257,6 +261,53 @@ class PipelineRunner:
             dx = self._bwd(pending.pop(0), gin)
             if not self.is_first:
                 _p2p(self.comm, [(dx, self.prev)], [])
+
+        return self._finish(M, losses)
+
+    def _step_gpipe(self, micro_batches: List[Dict]):
+        """GPipe schedule (reference executable_graph.cc:803): every
+        forward first, then every backward in reverse order; peak memory
+        holds all M activation caches."""
+        M = len(micro_batches)
+        for b in self.grad_bufs:
+            b.zero_()
+        losses = []
+        pending: List = []
+        for i in range(M):
+            act = self._recv_act() if not self.is_first else None
+            out, saved = self._fwd(i, micro_batches, act)
+            pending.append(saved)
+            if self.is_last:
+                losses.append(out.clone())
+            else:
+                _p2p(self.comm, [(out, self.next)], [])
+        for i in reversed(range(M)):
+            gin = None
+            if not self.is_last:
+                gin = torch.empty(self._act_shape, dtype=self._act_dtype,
+                                  device=self.device)
+                _p2p(self.comm, [], [(gin, self.next)])
+            dx = self._bwd(pending.pop(), gin)
+            if not self.is_first:
+                _p2p(self.comm, [(dx, self.prev)], [])
+        return self._finish(M, losses)
+
+    def _finish(self, M: int, losses):
+        h = self.stage.h
+        # ---- shared-weight p2p (tied wte/lm_head): first and last stage
+        # exchange + sum the tied parameter's accumulated grads so both
+        # copies update identically (reference executable_graph.cc:929-933)
+        tied = self.stage.h.get("tied_name")
+        if tied is not None and self.spec.pp > 1 \
+                and (self.is_first or self.is_last):
+            idx = next(i for i, p in enumerate(self.params)
+                       if p.name == tied)
+            peer = self.spec.peer(self.rank,
+                                  self.spec.pp - 1 if self.is_first else 0)
+            buf = self.grad_bufs[idx]
+            other = torch.empty_like(buf)
+            _p2p(self.comm, [(buf, peer)], [(other, peer)])
+            buf += other
 
         # ---- update: grad allreduce over dp + optimizer, once ----
         # per-micro-batch losses are token means; the step optimizes their

@@ -35,7 +35,9 @@ stage = build_llama_pipeline_stage(cfg, pspec, micro_batch=1, seq_len=16,
                                    dtype=torch.float32, lr=1e-3)
 runner = PipelineRunner(pspec, stage, torch.device("cpu"),
                         offload=bool(int(os.environ.get(
-                            "HETU_TEST_OFFLOAD", "0"))))
+                            "HETU_TEST_OFFLOAD", "0"))),
+                        schedule=os.environ.get("HETU_TEST_SCHEDULE",
+                                                "1f1b"))
 gen = torch.Generator().manual_seed(99)
 ids = torch.randint(0, cfg.vocab, (M, 1, 16), generator=gen)
 labels = torch.randint(0, cfg.vocab, (M, 16), generator=gen)
@@ -162,3 +164,86 @@ def test_pp2_offload_matches_single(single_losses):
     assert losses is not None
     assert np.allclose(losses, single_losses, rtol=2e-4, atol=1e-4), \
         f"pp2+offload {losses} vs single {single_losses}"
+
+
+def test_pp2_gpipe_matches_single(single_losses):
+    """GPipe schedule (all-forward-then-all-backward) computes the same
+    step as 1F1B (reference executable_graph.cc:803)."""
+    gl = _launch(2, {"HETU_TEST_SCHEDULE": "gpipe"}, 29567)
+    assert np.allclose(gl, single_losses, rtol=2e-4, atol=1e-5), \
+        (gl, single_losses)
+
+
+TIED_WORKER = r"""
+import json, os, sys
+sys.path.insert(0, os.environ["HETU_REPO"])
+import torch
+from hetu_amd.models.gpt import GPTConfig, build_gpt_pipeline_stage
+from hetu_amd.parallel.pipeline import PipelineSpec, PipelineRunner
+from hetu_amd.parallel.comm import comm_backend
+
+ws = int(os.environ.get("WORLD_SIZE", "1"))
+pp = ws if ws > 1 else 1
+M = 2
+cfg = GPTConfig(n_layer=2, n_head=2, n_kv_head=2, hidden=32,
+                ffn_hidden=64, vocab=96, max_seq=8, tie_embeddings=True)
+pspec = PipelineSpec(pp=pp)
+comm = comm_backend(torch.device("cpu"))
+stage = build_gpt_pipeline_stage(cfg, pspec, micro_batch=1, seq_len=8,
+                                 dtype=torch.float32, lr=1e-2)
+runner = PipelineRunner(pspec, stage, torch.device("cpu"))
+gen = torch.Generator().manual_seed(7)
+ids = torch.randint(0, cfg.vocab, (M, 1, 8), generator=gen)
+labels = torch.randint(0, cfg.vocab, (M, 8), generator=gen)
+h = stage.h
+losses = []
+for step in range(4):
+    mbs = []
+    for m in range(M):
+        feed = {}
+        if "input_ids" in h:
+            feed[h["input_ids"]] = ids[m]
+        if "labels" in h:
+            feed[h["labels"]] = labels[m].reshape(-1)
+        mbs.append(feed)
+    loss = runner.step(mbs)
+    if loss is not None:
+        losses.append(float(loss))
+if losses:
+    print("LOSSES:" + json.dumps(losses))
+# the tied copies must remain bit-identical across stages: report a
+# checksum of the local wte copy
+for p in stage.graph.parameters:
+    if p.name.startswith("wte.weight"):
+        print(f"WSUM:{p.get_data().double().sum().item()!r}")
+"""
+
+
+def test_pp2_tied_embeddings_matches_single():
+    """Shared wte/lm_head across first/last stage: pp2 must match the
+    single-stage tied model, and both stages' copies stay identical
+    (shared-weight grad p2p, reference executable_graph.cc:929-933)."""
+    single = _launch(1, {}, 29571, script=TIED_WORKER)
+    env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29573"}
+    procs = []
+    for r in range(2):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="2", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, "-c", TIED_WORKER],
+                                      env=env, stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    losses = None
+    wsums = []
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=300)
+        assert p.returncode in (0, -6), f"rank {r}: {out}\n{err}"
+        for ln in out.splitlines():
+            if ln.startswith("LOSSES:"):
+                losses = json.loads(ln[len("LOSSES:"):])
+            elif ln.startswith("WSUM:"):
+                wsums.append(float(ln[len("WSUM:"):]))
+    assert losses is not None
+    assert np.allclose(losses, single, rtol=2e-4, atol=1e-5), \
+        (losses, single)
+    assert len(wsums) == 2
+    assert abs(wsums[0] - wsums[1]) < 1e-9, wsums
