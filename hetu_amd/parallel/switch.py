@@ -104,16 +104,71 @@ def switch_params(plan: List[Dict], comm: Optional[CommBackend] = None):
                         [s.stop - s.start for s in in_b],
                         dtype=ent["dst"].dtype, device=ent["dst"].device)
                     recvs.append((buf, srank, ent["dst"], in_b))
-    ops = []
+    # ---- ParamBuffer staging (reference switch_exec_graph.h:76-160):
+    # all fragments headed to the same peer coalesce into ONE flat buffer
+    # per direction, so the batched p2p moves few LARGE messages over the
+    # xGMI links instead of one message per overlap region.  Fragment
+    # order is the plan order, which is identical on both endpoints.
+    send_groups: Dict[int, List[torch.Tensor]] = {}
     for t, dst in sends:
-        ops.append(dist.P2POp(dist.isend, t, dst))
-    for buf, src, _, _ in recvs:
-        ops.append(dist.P2POp(dist.irecv, buf, src))
+        send_groups.setdefault(dst, []).append(t.reshape(-1))
+    recv_groups: Dict[int, List] = {}
+    for buf, src, dst_t, in_b in recvs:
+        recv_groups.setdefault(src, []).append((buf, dst_t, in_b))
+    ops = []
+    flat_sends = {d: (torch.cat(ts) if len(ts) > 1 else ts[0])
+                  for d, ts in send_groups.items()}
+    flat_recvs = {}
+    for s, items in recv_groups.items():
+        n = sum(b.numel() for b, _, _ in items)
+        flat_recvs[s] = torch.empty(
+            n, dtype=items[0][0].dtype, device=items[0][0].device)
+    for d in sorted(flat_sends):
+        ops.append(dist.P2POp(dist.isend, flat_sends[d], d))
+    for s in sorted(flat_recvs):
+        ops.append(dist.P2POp(dist.irecv, flat_recvs[s], s))
     if ops and dist.is_initialized():
         for r in dist.batch_isend_irecv(ops):
             r.wait()
-    for buf, _, dst, in_b in recvs:
-        dst[in_b].copy_(buf)
+    for s, items in recv_groups.items():
+        flat = flat_recvs[s]
+        off = 0
+        for buf, dst_t, in_b in items:
+            n = buf.numel()
+            dst_t[in_b].copy_(flat[off:off + n].view(buf.shape))
+            off += n
+
+
+def plan_entries(pa, pb, src_t, dst_t):
+    """Migration plan entries for one tensor pair — one entry, or one per
+    fused section: a sectioned shard is [s0_loc|s1_loc|...], so each
+    section is its own plainly-chunked dim-0 migration (reference qkv
+    reorder, ht_safetensors.py:113)."""
+    gshape = pb.ds.global_shape(tuple(pb.shape)) if pb.ds is not None \
+        else tuple(pb.shape)
+    secs = getattr(pb, "shard_sections", None) or \
+        (getattr(pa, "shard_sections", None) if pa is not None else None)
+    base = {
+        "src_ds": pa.ds if pa is not None else None,
+        "src_group": pa.device_group if pa is not None else None,
+        "dst_ds": pb.ds, "dst_group": pb.device_group,
+    }
+    if not secs:
+        return [dict(base, src=src_t, dst=dst_t, global_shape=gshape)]
+    tp_a = pa.ds.get_dim(0) if (pa is not None and pa.ds) else 1
+    tp_b = pb.ds.get_dim(0) if pb.ds is not None else 1
+    out = []
+    off_a = off_b = 0
+    for s in secs:
+        la, lb = s // tp_a, s // tp_b
+        out.append(dict(
+            base,
+            src=(src_t[off_a:off_a + la] if src_t is not None else None),
+            dst=dst_t[off_b:off_b + lb],
+            global_shape=(s,) + gshape[1:]))
+        off_a += la
+        off_b += lb
+    return out
 
 
 def switch_graph_params(graph_a, graph_b,
@@ -125,36 +180,7 @@ def switch_graph_params(graph_a, graph_b,
     comm = comm or comm_backend()
     by_name_a = {p.name.split(":")[0]: p for p in graph_a.parameters}
     plan = []
-
-    def entries(pa, pb, src_t, dst_t):
-        """One plan entry — or one per fused section: a sectioned shard is
-        [s0_loc|s1_loc|...], so each section is its own plainly-chunked
-        dim-0 migration (reference qkv reorder, ht_safetensors.py:113)."""
-        gshape = pb.ds.global_shape(tuple(pb.shape)) if pb.ds is not None \
-            else tuple(pb.shape)
-        secs = getattr(pb, "shard_sections", None) or \
-            (getattr(pa, "shard_sections", None) if pa is not None else None)
-        base = {
-            "src_ds": pa.ds if pa is not None else None,
-            "src_group": pa.device_group if pa is not None else None,
-            "dst_ds": pb.ds, "dst_group": pb.device_group,
-        }
-        if not secs:
-            return [dict(base, src=src_t, dst=dst_t, global_shape=gshape)]
-        tp_a = pa.ds.get_dim(0) if (pa is not None and pa.ds) else 1
-        tp_b = pb.ds.get_dim(0) if pb.ds is not None else 1
-        out = []
-        off_a = off_b = 0
-        for s in secs:
-            la, lb = s // tp_a, s // tp_b
-            out.append(dict(
-                base,
-                src=(src_t[off_a:off_a + la] if src_t is not None else None),
-                dst=dst_t[off_b:off_b + lb],
-                global_shape=(s,) + gshape[1:]))
-            off_a += la
-            off_b += lb
-        return out
+    entries = plan_entries
 
     for pb in graph_b.parameters:
         name = pb.name.split(":")[0]
