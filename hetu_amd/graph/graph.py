@@ -161,6 +161,7 @@ class Graph:
             need_ops.reverse()
             # clone in topo (= creation) order
             mapping: Dict[int, Tensor] = {}
+            need_ops_cloned: List[Op] = []
             for op in need_ops:
                 new_in = [mapping.get(t.id, t) for t in op.inputs]
                 attrs = {k: v for k, v in op.attrs.items()
@@ -171,12 +172,30 @@ class Graph:
                     new_t.ds = old_t.ds
                     new_t.device_group = old_t.device_group
                     mapping[old_t.id] = new_t
+                need_ops_cloned.append(new_op)
                 total += 1
             # rewire backward references to the clones
             for op in bwd_ops:
                 for i, t in enumerate(op.inputs):
                     if t.id in mapping:
                         op.inputs[i] = mapping[t.id]
+            # scheduling: the clones' inputs (block inputs) are ready right
+            # after forward, so Kahn's min-id order would run EVERY scope's
+            # clones at the fwd->bwd boundary and re-materialize all blocks
+            # at once (worse than no recompute).  Key each scope's clones
+            # just before their earliest backward consumer so they run
+            # lazily, one block at a time.
+            clone_out_ids = {t.id for op_ in need_ops_cloned
+                             for t in op_.outputs}
+            first_consumer = None
+            for op in bwd_ops:
+                if any(t.id in clone_out_ids for t in op.inputs):
+                    if first_consumer is None or op.id < first_consumer:
+                        first_consumer = op.id
+            if first_consumer is not None:
+                for j, op_ in enumerate(need_ops_cloned):
+                    op_.attrs["_sched_key"] = \
+                        first_consumer - 0.5 + j * 1e-6
         return total
 
     # ---- topology --------------------------------------------------------

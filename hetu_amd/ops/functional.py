@@ -264,13 +264,17 @@ def _rope_ref(x, cos, sin, backward: bool):
 # ---------------------------------------------------------------------------
 
 def softmax_fwd(x, dim=-1):
-    if _use_hip("softmax", x) and dim in (-1, x.ndim - 1):
+    # the HIP kernel vectorizes rows by 8; narrow rows (e.g. MoE gate
+    # logits over a handful of experts) take the torch path
+    if _use_hip("softmax", x) and dim in (-1, x.ndim - 1) \
+            and x.shape[-1] % 8 == 0:
         return ext().softmax_fwd(x.contiguous())
     return torch.softmax(x.float(), dim=dim).to(x.dtype)
 
 
 def softmax_bwd(dy, y, dim=-1):
-    if _use_hip("softmax", dy) and dim in (-1, dy.ndim - 1):
+    if _use_hip("softmax", dy) and dim in (-1, dy.ndim - 1) \
+            and dy.shape[-1] % 8 == 0:
         return ext().softmax_bwd(dy.contiguous(), y.contiguous())
     dyf, yf = dy.float(), y.float()
     dx = (dyf - (dyf * yf).sum(dim, keepdim=True)) * yf
@@ -720,45 +724,49 @@ def varlen_attention_bwd(dout, q, k, v, out, lse, cu_seqlens,
 # (DGELU_BGRAD).  CPU path = the exact reference composition.
 # ---------------------------------------------------------------------------
 
-_LT_OK = [None]     # None = untested; False = no algos on this build
-
-
-def _lt_available() -> bool:
-    return _LT_OK[0] is not False
+# per-direction capability: this hipBLASLt build (gfx950) has DGELU_BGRAD
+# solutions but NO GELU_AUX_BIAS solutions, so the forward composes while
+# the backward still fuses dgelu + bias-grad into the dgrad GEMM
+_LT_FWD = [None]
+_LT_BWD = [None]
 
 
 def linear_gelu_aux(x2d, w, b):
     """returns (gelu(x@w^T+b), pre-gelu aux)."""
-    if _use_hip("ltgemm", x2d) and _lt_available():
+    if _use_hip("ltgemm", x2d) and _LT_FWD[0] is not False:
         try:
             out = tuple(ext().lt_linear_gelu_aux(x2d.contiguous(),
                                                  w.contiguous(),
                                                  b.contiguous()))
-            _LT_OK[0] = True
+            _LT_FWD[0] = True
             return out
         except RuntimeError as e:
             if "no algo" not in str(e):
                 raise
-            # this hipBLASLt build has no GELU_AUX solutions: compose
-            # (bias still fused via F.linear; gelu = hand HIP kernel)
-            _LT_OK[0] = False
+            # no GELU_AUX solutions: compose (bias still fused via
+            # F.linear; gelu = hand HIP kernel; aux = pre-gelu kept)
+            _LT_FWD[0] = False
             print("[hetu_amd] hipBLASLt GELU_AUX epilogue unavailable, "
-                  "using composed MLP path")
+                  "composing the MLP forward")
     h = torch.nn.functional.linear(x2d, w.to(x2d.dtype), b.to(x2d.dtype))
     return gelu_fwd(h), h
 
 
 def dgelu_bgrad(dy2d, w, aux):
     """returns (dgelu(dy@w, aux), colsum(dgelu(...)))."""
-    if _use_hip("ltgemm", dy2d) and _LT_OK[0] is True:
+    if _use_hip("ltgemm", dy2d) and _LT_BWD[0] is not False:
         try:
-            return tuple(ext().lt_dgelu_bgrad(dy2d.contiguous(),
-                                              w.contiguous(),
-                                              aux.contiguous()))
+            out = tuple(ext().lt_dgelu_bgrad(dy2d.contiguous(),
+                                             w.contiguous(),
+                                             aux.contiguous()))
+            _LT_BWD[0] = True
+            return out
         except RuntimeError as e:
             if "no algo" not in str(e):
                 raise
-            _LT_OK[0] = False
+            _LT_BWD[0] = False
+            print("[hetu_amd] hipBLASLt DGELU_BGRAD epilogue unavailable, "
+                  "composing the MLP backward")
     da = torch.matmul(dy2d, w.to(dy2d.dtype))
     dh = gelu_bwd(da, aux)
     return dh, colsum(dh).to(dh.dtype)

@@ -411,14 +411,14 @@ def test_lt_epilogue_fused_mlp():
     b1 = torch.randn(Ff, dtype=torch.bfloat16, device=dev) * 0.1
     wproj = torch.randn(H, Ff, dtype=torch.bfloat16, device=dev) * 0.05
     a, aux = F.linear_gelu_aux(x, wfc, b1)
-    if F._LT_OK[0] is not True:
-        pytest.skip("hipBLASLt GELU_AUX epilogue unavailable on this build")
     href = torch.nn.functional.linear(x.float(), wfc.float(), b1.float())
     aref = torch.nn.functional.gelu(href, approximate="tanh")
     assert (aux.float() - href).abs().max().item() < 0.15
     assert (a.float() - aref).abs().max().item() < 0.15
     dy = torch.randn(M, H, dtype=torch.bfloat16, device=dev)
     dh, db = F.dgelu_bgrad(dy, wproj, aux)
+    if F._LT_BWD[0] is not True:
+        pytest.skip("hipBLASLt DGELU_BGRAD unavailable on this build")
     da = torch.matmul(dy.float(), wproj.float())
     dh_ref = F.gelu_bwd(da.bfloat16(), aux).float()
     assert (dh.float() - dh_ref).abs().max().item() < 0.2, \

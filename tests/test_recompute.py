@@ -98,6 +98,14 @@ def test_recompute_llama_block_scopes():
         vb, _ = gb.run([hb["loss"], hb["train_op"]],
                        {hb["input_ids"]: ids, hb["labels"]: lab}, ctx=ctxb)
         assert abs(float(va) - float(vb)) < 2e-5, (float(va), float(vb))
-    pa = _peak_live_bytes(ga, [ha["loss"], ha["train_op"]])
-    pb = _peak_live_bytes(gb, [hb["loss"], hb["train_op"]])
+    # at a tiny hidden size the peak is weight-grad-dominated; the memory
+    # win is asserted at an activation-dominated shape (batch 16)
+    torch.manual_seed(1)
+    ga2, ha2 = build_llama_train_graph(cfg, 16, 16, dtype=torch.float32,
+                                       lr=1e-3)
+    torch.manual_seed(1)
+    gb2, hb2 = build_llama_train_graph(cfg, 16, 16, dtype=torch.float32,
+                                       lr=1e-3, recompute=True)
+    pa = _peak_live_bytes(ga2, [ha2["loss"], ha2["train_op"]])
+    pb = _peak_live_bytes(gb2, [hb2["loss"], hb2["train_op"]])
     assert pb < pa, (pa, pb)
