@@ -20,7 +20,9 @@ import os
 
 # gradient-allreduce / backward overlap on the comm stream (disable with
 # HETU_AMD_COMM_OVERLAP=0 if a RCCL/capture combination misbehaves)
-_COMM_OVERLAP = os.environ.get("HETU_AMD_COMM_OVERLAP", "1") == "1" 
+_COMM_OVERLAP = os.environ.get("HETU_AMD_COMM_OVERLAP", "1") == "1"
+# per-op allocated-memory trace (debug): records the op at the peak
+_MEM_TRACE = os.environ.get("HETU_AMD_MEM_TRACE", "0") == "1"
 
 
 class ExecContext:
@@ -50,13 +52,17 @@ class ExecContext:
 
 
 class _Plan:
-    __slots__ = ("topo", "last_use", "fetch_ids")
+    __slots__ = ("topo", "last_use", "fetch_ids", "dead_outputs")
 
     def __init__(self, topo: List[Op], last_use: Dict[int, int],
-                 fetch_ids: List[int]):
+                 fetch_ids: List[int], dead_outputs: Dict[int, list]):
         self.topo = topo
         self.last_use = last_use   # tensor_id -> index of last consuming op
         self.fetch_ids = fetch_ids
+        # op index -> output tensor ids with NO consumer in this plan
+        # (e.g. recompute rewires backward-saved tensors to clones): freed
+        # right after the producing op instead of living the whole run
+        self.dead_outputs = dead_outputs
 
 
 class Executor:
@@ -65,6 +71,7 @@ class Executor:
         self._plan_pool: Dict = {}
         self.ctx: Optional[ExecContext] = None
         self._inflight: Dict[int, "torch.cuda.Event"] = {}
+        self._mem_peak = (0, "", "", -1)
 
     def bind_context(self, ctx: ExecContext):
         self.ctx = ctx
@@ -101,7 +108,14 @@ class Executor:
             # fetched tensors are never freed
             for tid in fetch_ids:
                 last_use.pop(tid, None)
-            plan = _Plan(topo, last_use, fetch_ids)
+            dead: Dict[int, list] = {}
+            fset = set(fetch_ids)
+            for i, op in enumerate(topo):
+                dd = [t.id for t in op.outputs
+                      if t.id not in last_use and t.id not in fset]
+                if dd:
+                    dead[i] = dd
+            plan = _Plan(topo, last_use, fetch_ids, dead)
             self._plan_pool[key] = plan
         return plan
 
@@ -190,6 +204,10 @@ class Executor:
                 prof.end(tok)
             else:
                 outs = op.interface.compute(op, ins, ctx)
+            if _MEM_TRACE and ctx.device.type == "cuda":
+                a = torch.cuda.memory_allocated(ctx.device)
+                if a > self._mem_peak[0]:
+                    self._mem_peak = (a, op.type, op.name, i)
             for t, v in zip(op.outputs, outs):
                 values[t.id] = v
             if keep_values is not None:
@@ -200,6 +218,8 @@ class Executor:
             for t in op.inputs:
                 if plan.last_use.get(t.id) == i and t.id in values:
                     del values[t.id]
+            for tid in plan.dead_outputs.get(i, ()):
+                values.pop(tid, None)
 
         # join any comm-stream work not consumed by an op (e.g. fetched
         # tensors) back into the compute stream
