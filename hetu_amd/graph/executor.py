@@ -207,7 +207,15 @@ class Executor:
             if _MEM_TRACE and ctx.device.type == "cuda":
                 a = torch.cuda.memory_allocated(ctx.device)
                 if a > self._mem_peak[0]:
-                    self._mem_peak = (a, op.type, op.name, i)
+                    census = {}
+                    id2n = {t.id: t.name for o2 in plan.topo
+                            for t in o2.outputs}
+                    for tid, v in values.items():
+                        if isinstance(v, torch.Tensor) and v.is_cuda:
+                            census[id2n.get(tid, tid)] =                                 v.numel() * v.element_size()
+                    top = sorted(census.items(), key=lambda kv: -kv[1])[:12]
+                    self._mem_peak = (a, op.type, op.name, i, top,
+                                      sum(census.values()))
             for t, v in zip(op.outputs, outs):
                 values[t.id] = v
             if keep_values is not None:

@@ -9,6 +9,7 @@ with bucketed flat-buffer allreduce overlapped with backward).
 """
 from __future__ import annotations
 
+import weakref
 from typing import Dict, List, Optional
 
 import torch
@@ -66,13 +67,17 @@ class AdamStepOp(OptimizerUpdateOp):
     copies put hundreds of tiny host-memcpy nodes in the hipGraph)."""
     type = "AdamStep"
 
-    _instances: List["AdamStepOp"] = []
+    # weak registry: a discarded graph (hot-switch pool eviction, elastic
+    # rebuild, tests) must release its op interfaces AND their fp32
+    # master/m/v states — a strong list here leaked ~12 bytes/param per
+    # dead graph build
+    _instances: "weakref.WeakSet[AdamStepOp]" = weakref.WeakSet()
     # (beta1, beta2, device) -> {"host": pinned[2], "dev": cuda[2], "step"}
     _shared_bc: Dict = {}
 
     def __init__(self):
         self.state: Dict = {}
-        AdamStepOp._instances.append(self)
+        AdamStepOp._instances.add(self)
 
     @classmethod
     def set_replay_step(cls, step: int):

@@ -21,8 +21,22 @@ for rc in (False, True):
                       h["labels"]: torch.randint(0, cfg.vocab, (B*S,), device=dev)})
     torch.cuda.synchronize()
     ex = g.executor()
+    mp = ex._mem_peak
     print(f"rc={rc}: cuda_peak={torch.cuda.max_memory_allocated()/1e9:.2f}GB "
-          f"executor_peak={ex._mem_peak[0]/1e9:.2f}GB at "
-          f"{ex._mem_peak[1]}:{ex._mem_peak[2]} (op {ex._mem_peak[3]})",
-          flush=True)
+          f"executor_peak={mp[0]/1e9:.2f}GB at {mp[1]}:{mp[2]} (op {mp[3]}) "
+          f"values_total={mp[5]/1e9:.2f}GB", flush=True)
+    for nm, nb in mp[4]:
+        print(f"    {nm}: {nb/1e6:.0f}MB", flush=True)
+    # allocator block histogram right now (post-step, synced)
+    import collections
+    hist = collections.Counter()
+    for seg in torch.cuda.memory_snapshot():
+        for blk in seg.get("blocks", []):
+            if blk.get("state") == "active_allocated":
+                hist[round(blk["size"] / 1e6)] += 1
+    big = sorted(((sz, n) for sz, n in hist.items() if sz >= 8),
+                 key=lambda x: -x[0] * x[1])[:12]
+    tot = sum(sz * n for sz, n in hist.items())
+    print(f"    post-step active blocks total {tot/1e3:.2f}GB; "
+          f"big: {big}", flush=True)
     del tr, g, h
