@@ -103,6 +103,16 @@ class PipelineRunner:
         self.stage = stage
         self.device = device
         self.schedule = schedule      # "1f1b" (PipeDream-flush) | "gpipe"
+        # hipGraph capture of the per-micro-batch fwd/bwd bodies (rotating
+        # pp+1 buffer slots); opt-in: HETU_AMD_PP_CAPTURE=1, GPU only
+        import os as _os2
+        self._capture = (device.type == "cuda" and not recompute
+                         and not offload
+                         and _os2.environ.get("HETU_AMD_PP_CAPTURE",
+                                              "0") == "1")
+        self._nslots = spec.pp + 1
+        self._slots = [None] * self._nslots
+        self._step_no = 0
         self.scaler = scaler          # engine.amp.GradScaler or None
         self.recompute = recompute    # rerun fwd in bwd instead of caching
         self.offloader = None
@@ -136,10 +146,85 @@ class PipelineRunner:
             from ..utils.profiler import MemorySnapshots
             self.mem_snapshots = MemorySnapshots()
 
+    # ---- hipGraph-captured per-slot stage bodies -------------------------
+    # (reference/VERDICT: each micro-batch re-ran the Python executor per
+    # stage — ~30 ms host overhead per micro-batch; capturing the fwd and
+    # bwd bodies into per-slot hipGraphs cuts that to a graph launch.  1F1B
+    # keeps up to pp in-flight micro-batches, so pp+1 rotating buffer sets.)
+    def _slot_capture(self, si, micro_batches, i):
+        h = self.stage.h
+        out_t = h["loss"] if self.is_last else h["act_out"]
+        slot = {"feeds": {}, "kept": {}}
+        for t, v in micro_batches[i].items():
+            slot["feeds"][t] = v.to(self.device).clone()
+        if not self.is_first:
+            slot["feeds"][h["act_in"]] = torch.empty(
+                self._act_shape, dtype=self._act_dtype, device=self.device)
+        if not self.is_last:
+            slot["gin"] = torch.empty(self._act_shape,
+                                      dtype=self._act_dtype,
+                                      device=self.device)
+        if "loss_seed" in h and self.is_last:
+            slot["seed"] = torch.ones((), dtype=torch.float32,
+                                      device=self.device)
+        torch.cuda.synchronize()
+        gf = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(gf):
+            slot["out"] = self.stage.graph.run(
+                [out_t], dict(slot["feeds"]), ctx=self.ctx,
+                keep_values=slot["kept"])[0]
+        bwd_feed = {}
+        if not self.is_last:
+            bwd_feed[h["grad_in"]] = slot["gin"]
+        elif "seed" in slot:
+            bwd_feed[h["loss_seed"]] = slot["seed"]
+        for t, v in slot["feeds"].items():
+            bwd_feed.setdefault(t, v)
+        fetches = ([] if self.is_first else [h["dx"]]) + h["param_grads"]
+        gb = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(gb):
+            slot["bwd_outs"] = self.stage.graph.run(
+                fetches, bwd_feed, ctx=self.ctx,
+                seed_values=dict(slot["kept"]))
+        slot["fwd_g"], slot["bwd_g"] = gf, gb
+        return slot
+
+    def _fwd_captured(self, i, micro_batches, act):
+        si = i % self._nslots
+        if self._slots[si] is None:
+            self._slots[si] = self._slot_capture(si, micro_batches, i)
+        slot = self._slots[si]
+        for t, v in micro_batches[i].items():
+            slot["feeds"][t].copy_(v, non_blocking=True)
+        if act is not None:
+            slot["feeds"][self.stage.h["act_in"]].copy_(act)
+        slot["fwd_g"].replay()
+        return slot["out"], ("slot", si)
+
+    def _bwd_captured(self, saved, gin):
+        _, si = saved
+        slot = self._slots[si]
+        if gin is not None and "gin" in slot:
+            slot["gin"].copy_(gin)
+        if "seed" in slot:
+            scale = self.scaler.scale if self.scaler is not None else 1.0
+            slot["seed"].fill_(float(scale))
+        slot["bwd_g"].replay()
+        res = slot["bwd_outs"]
+        dx = None
+        if not self.is_first:
+            dx, res = res[0], res[1:]
+        for buf, g in zip(self.grad_bufs, res):
+            if g is not None:
+                buf += g.float()
+        return dx
+
     # ---- fwd / bwd over the stage graph ---------------------------------
     def _fwd(self, i, micro_batches, act):
         if self.mem_snapshots is not None:
             self.mem_snapshots.mark(f"fwd_mb{i}", self.device)
+        if self._capture and self._step_no >= 1:
+            return self._fwd_captured(i, micro_batches, act)
         h = self.stage.h
         feed = dict(micro_batches[i])
         if not self.is_first:
@@ -162,6 +247,9 @@ class PipelineRunner:
     def _bwd(self, saved, gin):
         if self.mem_snapshots is not None:
             self.mem_snapshots.mark("bwd_mb", self.device)
+        if isinstance(saved, tuple) and len(saved) == 2 \
+                and saved[0] == "slot":
+            return self._bwd_captured(saved, gin)
         h = self.stage.h
         feed, cache = saved
         if self.offloader is not None and cache is not None \
@@ -199,6 +287,7 @@ class PipelineRunner:
     # ---- one training step (1F1B | GPipe) --------------------------------
     def step(self, micro_batches: List[Dict]):
         """Returns the mean micro-batch loss on the LAST stage, else None."""
+        self._step_no += 1
         if self.schedule == "gpipe":
             return self._step_gpipe(micro_batches)
         M = len(micro_batches)

@@ -94,3 +94,43 @@ def test_osdp_partial_sharding_plan_runs():
               for _ in range(3)]
     assert all(x == x for x in losses), losses
     assert losses[-1] < losses[0] + 0.5
+
+
+def test_pipeline_slot_capture_parity():
+    """HETU_AMD_PP_CAPTURE=1: the per-slot hipGraph-captured stage bodies
+    must reproduce the eager PipelineRunner's losses (pp=1 exercises the
+    rotating-slot capture machinery without p2p)."""
+    import os
+    from hetu_amd.models.gpt import GPTConfig, build_gpt_pipeline_stage
+    from hetu_amd.parallel.pipeline import PipelineRunner, PipelineSpec
+    dev = torch.device("cuda", 0)
+    cfg = GPTConfig(n_layer=4, n_head=4, n_kv_head=4, hidden=512,
+                    ffn_hidden=2048, vocab=50304, max_seq=512)
+    M = 4
+    results = {}
+    for cap in ("0", "1"):
+        os.environ["HETU_AMD_PP_CAPTURE"] = cap
+        torch.manual_seed(3)
+        pspec = PipelineSpec(pp=1)
+        stage = build_gpt_pipeline_stage(cfg, pspec, micro_batch=2,
+                                         seq_len=512,
+                                         dtype=torch.bfloat16, lr=1e-4)
+        runner = PipelineRunner(pspec, stage, dev)
+        h = stage.h
+        torch.manual_seed(11)
+        losses = []
+        for step in range(4):
+            mbs = [{h["input_ids"]: torch.randint(0, cfg.vocab, (2, 512),
+                                                  device=dev),
+                    h["labels"]: torch.randint(0, cfg.vocab, (1024,),
+                                               device=dev)}
+                   for _ in range(M)]
+            lv = runner.step(mbs)
+            torch.cuda.synchronize()
+            losses.append(float(lv.float()))
+        results[cap] = losses
+        if cap == "1":
+            assert runner._slots[0] is not None, "capture did not engage"
+    os.environ.pop("HETU_AMD_PP_CAPTURE", None)
+    for a, b in zip(results["0"], results["1"]):
+        assert abs(a - b) < 3e-2, results
