@@ -430,10 +430,20 @@ def flash_attn_bwd(dout, q, k, v, out, lse, causal: bool,
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _use_hip("fa", q):
-        return ext().flash_attn_bwd(dout.contiguous(), q.contiguous(),
-                                    k.contiguous(), v.contiguous(),
-                                    out.contiguous(), lse.contiguous(),
-                                    causal, scale)
+        e = ext()
+        if q.shape[1] == k.shape[1] and hasattr(e, "flash_attn_bwd_v3") \
+                and os.environ.get("HETU_AMD_FA_V3", "1") == "1":
+            # v3 bwd: 3-deep walking-pointer staging rings + counted
+            # vmcnt (bit-exact vs v2, ~6% faster on the causal bwd at
+            # S=2048 — gpurun_out/r2_fa_bench.log); non-GQA BHSD only
+            return e.flash_attn_bwd_v3(dout.contiguous(), q.contiguous(),
+                                       k.contiguous(), v.contiguous(),
+                                       out.contiguous(), lse.contiguous(),
+                                       causal, scale)
+        return e.flash_attn_bwd(dout.contiguous(), q.contiguous(),
+                                k.contiguous(), v.contiguous(),
+                                out.contiguous(), lse.contiguous(),
+                                causal, scale)
     return _attn_ref_bwd(dout, q, k, v, out, lse, causal, scale)
 
 
