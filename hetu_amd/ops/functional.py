@@ -431,7 +431,8 @@ def flash_attn_bwd(dout, q, k, v, out, lse, causal: bool,
         scale = 1.0 / math.sqrt(q.shape[-1])
     if _use_hip("fa", q):
         e = ext()
-        if q.shape[1] == k.shape[1] and hasattr(e, "flash_attn_bwd_v3") \
+        if q.shape[1] == k.shape[1] and q.shape[-1] == 128 \
+                and hasattr(e, "flash_attn_bwd_v3") \
                 and os.environ.get("HETU_AMD_FA_V3", "1") == "1":
             # v3 bwd: 3-deep walking-pointer staging rings + counted
             # vmcnt (bit-exact vs v2, ~6% faster on the causal bwd at
