@@ -682,14 +682,21 @@ def fused_qkv_attention_bwd(dout, qkv, out, lse, n_head: int,
 def varlen_attention_fwd(q, k, v, cu_seqlens, causal: bool = True,
                          scale: Optional[float] = None):
     """Packed-varlen attention (reference ParallelAttention.cc packed
-    path): q/k/v [T, H, D] with cu_seqlens [n+1] int delimiting the packed
-    sequences; each segment attends only within itself.  Runs the flash
-    kernel per segment (B=1 view) — segments in one packed bin are
-    length-bucketed by data/bucket.py so the launches stay few and fat.
+    path / FlashAttention.cu mha_varlen_fwd): q/k/v [T, H, D] with
+    cu_seqlens [n+1] delimiting the packed sequences; each segment attends
+    only within itself.  On GPU (D=128, no GQA mismatch constraint): ONE
+    kernel launch with device-side cu offsets — no host loop, no
+    `.tolist()` sync; blocks beyond a segment's tiles exit early.
     Returns (out [T, H, D], lse [H, T])."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     T, H, D = q.shape
+    if _use_hip("varlen", q) and D == 128:
+        cu32 = cu_seqlens.to(device=q.device, dtype=torch.int32)
+        o, lse = ext().flash_attn_varlen_fwd(
+            q.contiguous(), k.contiguous(), v.contiguous(), cu32, T,
+            causal, scale)
+        return o, lse
     out = torch.empty_like(q)
     lse = torch.empty(H, T, dtype=torch.float32, device=q.device)
     cu = cu_seqlens.tolist()

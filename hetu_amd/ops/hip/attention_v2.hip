@@ -50,7 +50,8 @@ __global__ __launch_bounds__(THREADS, 1) void fa2_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, bf16* __restrict__ O,
     float* __restrict__ LSE, int B, int H, int Hkv, int S, int Skv,
-    float scale, bool causal, FaStrides sq, FaStrides skv, FaStrides so) {
+    float scale, bool causal, FaStrides sq, FaStrides skv, FaStrides so,
+    const int* __restrict__ cu, int lse_T) {
   static_assert(D == 128, "fa2 fwd: D=128 only");
   constexpr int KBYTES = BN * D * 2;          // 16 KiB
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -65,6 +66,17 @@ __global__ __launch_bounds__(THREADS, 1) void fa2_fwd_kernel(
   const int b = bh / H;
   const int hkv = h / (H / Hkv);
   const int q0 = blockIdx.y * BM;
+  // packed-varlen mode (reference mha_varlen_fwd): blockIdx "batch" is a
+  // SEGMENT of the packed [T, H, D] layout; segment bounds come from the
+  // device-side cu_seqlens — one launch covers every ragged segment, no
+  // host loop, no host sync.
+  int row0 = 0;
+  if (cu != nullptr) {
+    row0 = cu[b];
+    S = cu[b + 1] - row0;
+    Skv = S;
+    if (q0 >= S) return;        // ragged tail tile of a shorter segment
+  }
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -72,9 +84,12 @@ __global__ __launch_bounds__(THREADS, 1) void fa2_fwd_kernel(
   const int iq = lane & 31;      // this lane's q row within the wave block
   const int hi = lane >> 5;
 
-  const bf16* Qb = Q + (int64_t)b * sq.bs + (int64_t)h * sq.hs;
-  const bf16* Kb = K + (int64_t)b * skv.bs + (int64_t)hkv * skv.hs;
-  const bf16* Vb = V + (int64_t)b * skv.bs + (int64_t)hkv * skv.hs;
+  const int64_t qb_off = cu ? (int64_t)row0 * sq.rs : (int64_t)b * sq.bs;
+  const int64_t kb_off = cu ? (int64_t)row0 * skv.rs
+                            : (int64_t)b * skv.bs;
+  const bf16* Qb = Q + qb_off + (int64_t)h * sq.hs;
+  const bf16* Kb = K + kb_off + (int64_t)hkv * skv.hs;
+  const bf16* Vb = V + kb_off + (int64_t)hkv * skv.hs;
 
   const int my_q = q0 + wid * 32 + iq;          // global q row (this lane)
   const int diag = Skv - S;                     // causal offset
@@ -300,8 +315,8 @@ __global__ __launch_bounds__(THREADS, 1) void fa2_fwd_kernel(
   // ---- epilogue -------------------------------------------------------
   if (my_q < S) {
     float inv = (l_i > 0.f) ? 1.f / l_i : 0.f;
-    bf16* orow = O + (int64_t)b * so.bs + (int64_t)h * so.hs
-                 + (int64_t)my_q * so.rs;
+    bf16* orow = O + (cu ? (int64_t)row0 * so.rs : (int64_t)b * so.bs)
+                 + (int64_t)h * so.hs + (int64_t)my_q * so.rs;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
 #pragma unroll
@@ -322,8 +337,9 @@ __global__ __launch_bounds__(THREADS, 1) void fa2_fwd_kernel(
       }
     }
     if (hi == 0) {
-      LSE[(int64_t)bh * S + my_q] =
-          (l_i > 0.f) ? m_i + __logf(l_i) : -INFINITY;
+      int64_t li = cu ? ((int64_t)h * lse_T + row0 + my_q)
+                      : ((int64_t)bh * S + my_q);
+      LSE[li] = (l_i > 0.f) ? m_i + __logf(l_i) : -INFINITY;
     }
   }
 }
@@ -341,7 +357,38 @@ void fa2_fwd_launch(const void* q, const void* k, const void* v, void* o,
   hipLaunchKernelGGL(fa2_fwd_kernel<128>, grid, dim3(THREADS), lds, stream,
                      (const bf16*)q, (const bf16*)k, (const bf16*)v,
                      (bf16*)o, lse, B, H, Hkv, S, Skv, scale, causal,
-                     sq, skv, so);
+                     sq, skv, so, (const int*)nullptr, 0);
+}
+
+// Packed-varlen entry (reference FlashAttention.cu mha_varlen_fwd): ONE
+// kernel launch over all ragged segments; q/k/v [T, H, D] packed, cu
+// int32 [nseg+1] on DEVICE.  Returns (o [T, H, D], lse [H, T] fp32).
+std::vector<torch::Tensor> flash_attn_varlen_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor cu,
+    int64_t max_seqlen, bool causal, double scale) {
+  TORCH_CHECK(q.dim() == 3 && q.is_contiguous() && k.is_contiguous() &&
+              v.is_contiguous(), "varlen: q/k/v [T, H, D] contiguous");
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "varlen: bf16 only");
+  TORCH_CHECK(cu.scalar_type() == at::kInt && cu.is_cuda() &&
+              cu.is_contiguous(), "varlen: cu int32 on device");
+  const int T = q.size(0), H = q.size(1), D = q.size(2);
+  const int Hkv = k.size(1);
+  TORCH_CHECK(D == 128, "fa2 varlen: D=128 only");
+  const int nseg = cu.numel() - 1;
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({H, (int64_t)T},
+                          q.options().dtype(at::kFloat));
+  FaStrides sq{0, (int64_t)D, (int64_t)H * D};
+  FaStrides skv{0, (int64_t)D, (int64_t)Hkv * D};
+  auto stream = hetu_current_stream();
+  dim3 grid(nseg * H, ((int)max_seqlen + BM - 1) / BM);
+  size_t lds = 2 * NBUF * (size_t)BN * 128 * 2;
+  hipLaunchKernelGGL(fa2_fwd_kernel<128>, grid, dim3(THREADS), lds, stream,
+                     (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+                     (const bf16*)v.data_ptr(), (bf16*)o.data_ptr(),
+                     lse.data_ptr<float>(), nseg, H, Hkv, 0, 0,
+                     (float)scale, causal, sq, skv, sq,
+                     cu.data_ptr<int>(), T);
 }
 
 // Fused-QKV entry: qkv [B, S, (H+2*Hkv)*D] straight from the column-

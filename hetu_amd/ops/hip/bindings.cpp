@@ -107,6 +107,9 @@ torch::Tensor flash_attn_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
 void rope_qk_inplace(torch::Tensor qkv, torch::Tensor cs, torch::Tensor sn,
                      int64_t n_rot, int64_t D, int64_t sign);
 // attention_v3.hip (experimental round-2 candidate)
+std::vector<torch::Tensor> flash_attn_varlen_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor cu,
+    int64_t max_seqlen, bool causal, double scale);
 std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q,
                                              torch::Tensor k,
                                              torch::Tensor v, bool causal,
@@ -163,6 +166,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("flash_attn_fwd_qkv", &flash_attn_fwd_qkv);
+  m.def("flash_attn_varlen_fwd", &flash_attn_varlen_fwd);
   m.def("flash_attn_bwd_qkv", &flash_attn_bwd_qkv);
   m.def("rope_qk_inplace", &rope_qk_inplace);
   m.def("flash_attn_fwd_v3", &flash_attn_fwd_v3);
