@@ -389,6 +389,7 @@ std::vector<torch::Tensor> flash_attn_varlen_fwd(
                      lse.data_ptr<float>(), nseg, H, Hkv, 0, 0,
                      (float)scale, causal, sq, skv, sq,
                      cu.data_ptr<int>(), T);
+  return {o, lse};
 }
 
 // Fused-QKV entry: qkv [B, S, (H+2*Hkv)*D] straight from the column-
