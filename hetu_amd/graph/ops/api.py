@@ -324,7 +324,7 @@ def check_finite(tensors) -> Tensor:
 
 
 def ring_attention(q, k, v, cp_ranks, causal=True, scale=None,
-                   split=None):
+                   split=None, seq_lens=None):
     """split: NORMAL (contiguous chunks) or SYM (zigzag [head|tail]
     halves, causal-load-balanced); default from
     HETU_AMD_ATTN_SPLIT (reference HETU_PARALLEL_ATTN_SPLIT_PATTERN)."""
@@ -335,6 +335,8 @@ def ring_attention(q, k, v, cp_ranks, causal=True, scale=None,
     return _cg().make_op(P.RingAttentionOp(), [q, k, v],
                          {"causal": causal, "scale": scale,
                           "split": split,
+                          "seq_lens": (list(seq_lens) if seq_lens
+                                       else None),
                           "cp_ranks": list(cp_ranks)}).output(0)
 
 

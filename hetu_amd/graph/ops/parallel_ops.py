@@ -173,7 +173,15 @@ class RingAttentionOp(OpInterface):
         from ...parallel.ring_attention import (ring_attn_fwd,
                                                 ring_attn_fwd_sym)
         q, k, v = inputs
-        if op.attrs.get("split", "NORMAL") == "SYM" \
+        if op.attrs.get("seq_lens") and ctx.comm is not None \
+                and len(op.attrs["cp_ranks"]) > 1:
+            from ...parallel.ring_attention import ring_attn_fwd_hetero
+            o, lse = ring_attn_fwd_hetero(q, k, v, ctx.comm,
+                                          op.attrs["cp_ranks"],
+                                          op.attrs["seq_lens"],
+                                          op.attrs.get("causal", True),
+                                          op.attrs.get("scale"))
+        elif op.attrs.get("split", "NORMAL") == "SYM" \
                 and op.attrs.get("causal", True) and ctx.comm is not None \
                 and len(op.attrs["cp_ranks"]) > 1:
             o, lse = ring_attn_fwd_sym(q, k, v, ctx.comm,
@@ -214,6 +222,13 @@ class RingAttentionGradOp(OpInterface):
         from ...parallel.ring_attention import (ring_attn_bwd,
                                                 ring_attn_bwd_sym)
         dout, q, k, v, o, lse = inputs
+        if op.attrs.get("seq_lens") and ctx.comm is not None \
+                and len(op.attrs["cp_ranks"]) > 1:
+            from ...parallel.ring_attention import ring_attn_bwd_hetero
+            return list(ring_attn_bwd_hetero(
+                dout, q, k, v, o, lse, ctx.comm, op.attrs["cp_ranks"],
+                op.attrs["seq_lens"], op.attrs.get("causal", True),
+                op.attrs.get("scale")))
         if op.attrs.get("split", "NORMAL") == "SYM" \
                 and op.attrs.get("causal", True) and ctx.comm is not None \
                 and len(op.attrs["cp_ranks"]) > 1:

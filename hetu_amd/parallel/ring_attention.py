@@ -88,6 +88,110 @@ def ring_attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return o, lse
 
 
+def ring_attn_fwd_hetero(q, k, v, comm, ranks, seq_lens, causal=True,
+                         scale=None):
+    """Heterogeneous context parallelism (reference hetero CP,
+    trainer.py:255-259 `[32k, 32k, 48k...]`): NORMAL split with UNEQUAL
+    per-rank seq shards — rank at ring position p owns global rows
+    [sum(seq_lens[:p]), +seq_lens[p]).  KV blocks of differing sizes
+    rotate the ring; recv shapes come from the static seq_lens."""
+    n = len(ranks)
+    if n <= 1 or comm is None:
+        return F.flash_attn_fwd(q, k, v, causal, scale)
+    my_pos = ranks.index(comm.rank)
+    offs = [0]
+    for s in seq_lens:
+        offs.append(offs[-1] + s)
+    my_off = offs[my_pos]
+    my_len = seq_lens[my_pos]
+    o, lse = None, None
+    kv_k, kv_v = k, v
+    src_pos = my_pos
+    B, H, _, D = q.shape
+    for step in range(n):
+        src_off = offs[src_pos]
+        src_len = seq_lens[src_pos]
+        skip = causal and src_off >= my_off + my_len
+        if not skip and src_len > 0 and my_len > 0:
+            blk_causal = causal and (src_pos == my_pos)
+            ob, lseb = F.flash_attn_fwd(q, kv_k, kv_v, blk_causal, scale)
+            if o is None:
+                o, lse = ob, lseb
+            else:
+                o, lse = _merge(o, lse, ob, lseb)
+        if step < n - 1:
+            nxt_src = (src_pos - 1) % n
+            shp = (B, H, seq_lens[nxt_src], D)
+            kv_k, kv_v = _ring_exchange_shaped(
+                comm, ranks, my_pos, [kv_k, kv_v], [shp, shp])
+            src_pos = nxt_src
+    if o is None:
+        o = torch.zeros_like(q)
+        lse = torch.full(q.shape[:-1], float("-inf"), dtype=torch.float32,
+                         device=q.device)
+    return o, lse
+
+
+def ring_attn_bwd_hetero(dout, q, k, v, o, lse, comm, ranks, seq_lens,
+                         causal=True, scale=None):
+    n = len(ranks)
+    if n <= 1 or comm is None:
+        return F.flash_attn_bwd(dout, q, k, v, o, lse, causal, scale)
+    my_pos = ranks.index(comm.rank)
+    offs = [0]
+    for s in seq_lens:
+        offs.append(offs[-1] + s)
+    my_off, my_len = offs[my_pos], seq_lens[my_pos]
+    dq = torch.zeros_like(q, dtype=torch.float32)
+    dk_acc = torch.zeros_like(k, dtype=torch.float32)
+    dv_acc = torch.zeros_like(v, dtype=torch.float32)
+    kv_k, kv_v = k, v
+    src_pos = my_pos
+    B, H, _, D = q.shape
+    for step in range(n):
+        src_off, src_len = offs[src_pos], seq_lens[src_pos]
+        skip = causal and src_off >= my_off + my_len
+        if not skip and src_len > 0 and my_len > 0:
+            blk_causal = causal and (src_pos == my_pos)
+            dqb, dkb, dvb = F.flash_attn_bwd(dout, q, kv_k, kv_v, o, lse,
+                                             blk_causal, scale)
+            dq += dqb.float()
+            dk_acc += dkb.float()
+            dv_acc += dvb.float()
+        if step < n - 1:
+            nxt_src = (src_pos - 1) % n
+            shp = (B, H, seq_lens[nxt_src], D)
+            kv_k, kv_v, dk_acc, dv_acc = _ring_exchange_shaped(
+                comm, ranks, my_pos, [kv_k, kv_v, dk_acc, dv_acc],
+                [shp, shp, shp, shp])
+            src_pos = nxt_src
+    # after n-1 hops the block here is owned by my_pos+1; one more
+    # exchange delivers the my_pos-owned accumulator home
+    home = (B, H, my_len, D)
+    dk_acc, dv_acc = _ring_exchange_shaped(
+        comm, ranks, my_pos, [dk_acc, dv_acc], [home, home])
+    return dq.to(q.dtype), dk_acc.to(k.dtype), dv_acc.to(v.dtype)
+
+
+def _ring_exchange_shaped(comm, ranks, my_pos, tensors, recv_shapes):
+    """Ring exchange where the incoming block's shape differs from the
+    outgoing one (hetero CP)."""
+    n = len(ranks)
+    nxt = ranks[(my_pos + 1) % n]
+    prv = ranks[(my_pos - 1) % n]
+    import torch.distributed as dist
+    recvs = [torch.empty(s, dtype=t.dtype, device=t.device)
+             for t, s in zip(tensors, recv_shapes)]
+    ops = []
+    for t in tensors:
+        ops.append(dist.P2POp(dist.isend, t.contiguous(), nxt))
+    for r in recvs:
+        ops.append(dist.P2POp(dist.irecv, r, prv))
+    for r in dist.batch_isend_irecv(ops):
+        r.wait()
+    return recvs
+
+
 def _sym_subblocks(my_pos: int, src_pos: int, n: int):
     """Sub-block schedule for the SYM (zigzag) split: rank p owns global
     chunks (p, 2n-1-p) as its [head | tail] halves.  Yields
