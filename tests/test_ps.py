@@ -154,3 +154,39 @@ def test_ps_two_ranks():
         out, err = p.communicate(timeout=300)
         ok = (p.returncode == 0 or p.returncode == -6) and "PSOK:" in out
         assert ok, f"rank {r} failed rc={p.returncode}:\n{out}\n{err}"
+
+
+def test_ps_server_transport_sparse_training():
+    """ps-lite-style server transport: 2 server shards over TCP, a worker
+    pushes sparse grads and the server-side optimizer converges an
+    embedding toward targets (reference kv_app.h Push/Pull +
+    PSFhandle_embedding server-side update)."""
+    import torch
+    from hetu_amd.ps.server import PSClient, PSServer
+    s1 = PSServer().start()
+    s2 = PSServer().start()
+    cli = PSClient([("127.0.0.1", s1.port), ("127.0.0.1", s2.port)])
+    try:
+        N, D = 64, 8
+        cli.register("emb", N, D, optimizer="sgd", lr=0.5, init_std=0.01)
+        torch.manual_seed(0)
+        target = torch.randn(N, D)
+        for step in range(200):
+            ids = torch.randint(0, N, (32,))
+            rows = cli.pull("emb", ids)
+            grads = rows - target[ids]          # d/drow of 0.5||row-t||^2
+            cli.push("emb", ids, grads)
+        rows = cli.pull("emb", torch.arange(N))
+        err = (rows - target).norm() / target.norm()
+        assert err < 0.05, float(err)
+        # adagrad path
+        cli.register("emb2", N, D, optimizer="adagrad", lr=1.0)
+        ids = torch.arange(N)
+        for step in range(100):
+            rows = cli.pull("emb2", ids)
+            cli.push("emb2", ids, rows - target)
+        err2 = (cli.pull("emb2", ids) - target).norm() / target.norm()
+        assert err2 < 0.2, float(err2)
+    finally:
+        cli.stop_servers()
+        cli.close()
