@@ -130,9 +130,12 @@ class GPTMLP(Module):
 
     def forward(self, x):
         if self.wfc.spec.tp == 1 and os.environ.get(
-                "HETU_AMD_FUSED_MLP", "1") == "1":
-            # hipBLASLt epilogue fusion (single-device/dp hot path):
-            # gelu rides the fc GEMM, dgelu+b1-grad ride the dgrad GEMM
+                "HETU_AMD_FUSED_MLP", "0") == "1":
+            # hipBLASLt epilogue fusion (GELU_AUX_BIAS fwd + DGELU_BGRAD
+            # bwd).  Opt-in: the ROCm 7.2 hipBLASLt Tensile catalog has no
+            # epilogue solutions at transformer shapes on gfx950 (probed,
+            # scripts/lt_probe2.py), and the composed-graph fallback
+            # measured ~1.3% faster than FusedMLPOp's fallback composition.
             return ht.fused_mlp(x, self.wfc.weight, self.wfc.bias,
                                 self.wproj.weight, self.wproj.bias)
         return self.wproj(ht.gelu(self.wfc(x)))
