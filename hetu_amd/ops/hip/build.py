@@ -70,6 +70,10 @@ def build(verbose: bool = True) -> str:
         "-fno-gpu-rdc",
         "-Wno-deprecated-declarations",
         "-Wno-unused-result",
+        # a non-void function falling off its end produced a silent
+        # GPU-side hang (see profiles/r02_capture_replay_bug.md era):
+        # make it a hard error
+        "-Werror=return-type",
     ] + [f"-I{p}" for p in inc] + [f"-I{py_inc}", f"-I{HERE}"]
     for src in srcs:
         base = os.path.basename(src).rsplit(".", 1)[0]
