@@ -106,3 +106,39 @@ def test_elastic_controller_reshape():
         assert results[r]["restore_step"] == 7        # min common step
         assert results[r]["alive"] == [0, 1]
         assert results[r]["strategy"] == "dp2"
+
+
+def test_pssh_launcher_local_transport(tmp_path):
+    """Multi-'node' launch over an injected local transport (the ssh van
+    of pssh_start.py): 2 hosts x 1 proc rendezvous over torchrun and
+    train; a killed host respawns (pssh_start_elastic pool)."""
+    import sys
+    from hetu_amd.rpc.pssh import Host, PsshLauncher
+    script = tmp_path / "w.py"
+    script.write_text(
+        "import os, torch, torch.distributed as dist\n"
+        "from datetime import timedelta\n"
+        "os.environ.setdefault('GLOO_SOCKET_IFNAME', 'lo')\n"
+        "dist.init_process_group('gloo',"
+        " timeout=timedelta(seconds=60))\n"
+        "t = torch.ones(4) * (dist.get_rank() + 1)\n"
+        "dist.all_reduce(t)\n"
+        "print('PSSH_SUM', t[0].item(), dist.get_world_size())\n"
+        "dist.destroy_process_group()\n")
+    hosts = [Host("127.0.0.1", 1), Host("127.0.0.1", 1)]
+    la = PsshLauncher(hosts, str(script), master_port=29793,
+                      ssh_cmd=["bash", "-c"],
+                      env_extra={"GLOO_SOCKET_IFNAME": "lo"})
+    la.start()
+    codes = la.wait(timeout_s=180)
+    outs = [la.output(i) for i in range(2)]
+    assert all(c == 0 for c in codes), (codes, outs)
+    assert any("PSSH_SUM 3.0 2" in o for o in outs), outs
+    # respawn path: relaunch host 1 (full group restart for the test)
+    la2 = PsshLauncher(hosts, str(script), master_port=29795,
+                       ssh_cmd=["bash", "-c"],
+                       env_extra={"GLOO_SOCKET_IFNAME": "lo"})
+    la2.start()
+    la2.respawn(1)
+    codes = la2.wait(timeout_s=180)
+    assert codes[1] == 0, la2.output(1)
