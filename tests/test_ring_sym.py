@@ -64,12 +64,13 @@ print("SYMOK")
 """
 
 
-def test_sym_ring_two_ranks_parity():
+def _run_sym(ws, port):
     env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
-            "MASTER_PORT": "29707", "GLOO_SOCKET_IFNAME": "lo"}
+            "MASTER_PORT": str(port), "GLOO_SOCKET_IFNAME": "lo"}
     procs = []
-    for r in range(2):
-        env = dict(env0, RANK=str(r), WORLD_SIZE="2", LOCAL_RANK=str(r))
+    for r in range(ws):
+        env = dict(env0, RANK=str(r), WORLD_SIZE=str(ws),
+                   LOCAL_RANK=str(r))
         procs.append(subprocess.Popen([sys.executable, "-c", SYM_WORKER],
                                       env=env, stdout=subprocess.PIPE,
                                       stderr=subprocess.PIPE, text=True))
@@ -77,6 +78,14 @@ def test_sym_ring_two_ranks_parity():
         out, err = p.communicate(timeout=300)
         ok = (p.returncode in (0, -6)) and "SYMOK" in out
         assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
+
+
+def test_sym_ring_two_ranks_parity():
+    _run_sym(2, 29707)
+
+
+def test_sym_ring_four_ranks_parity():
+    _run_sym(4, 29727)
 
 
 HET_WORKER = r"""
