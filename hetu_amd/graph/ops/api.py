@@ -290,6 +290,13 @@ def fused_qkv_attention(qkv, n_head, n_kv_head, head_dim, cos=None,
                           "scale": scale}).output()
 
 
+def fused_add_ln(x, r, w, b, eps=1e-5):
+    """Fused residual-add + LayerNorm (see nnops.FusedAddLNOp):
+    returns (ln_out, sum)."""
+    op = _cg().make_op(N.FusedAddLNOp(), [x, r, w, b], {"eps": eps})
+    return op.output(0), op.output(1)
+
+
 def fused_mlp(x, wfc, b1, wproj, b2=None):
     """Epilogue-fused transformer MLP (see nnops.FusedMLPOp):
     y = gelu(x @ wfc^T + b1) @ wproj^T (+ b2)."""

@@ -774,3 +774,86 @@ class FusedMLPGradOp(OpInterface):
         if op.attrs.get("with_b2"):
             outs.append(F.colsum(dy2).to(dy.dtype))
         return outs
+
+
+class FusedAddLNOp(OpInterface):
+    """Fused residual-add + LayerNorm (transformer pre-norm chain):
+    s = x + r; y = LN(s).  One kernel reads x and r once and writes both
+    s (the next residual) and y — no standalone elementwise add; the
+    backward folds the residual-grad accumulation into the LN-dx kernel
+    (reference keeps Add + FusedLayerNorm separate around Arithmetics.cu).
+    inputs: x, r, w, b; outputs: y, s, mean, rstd."""
+    type = "FusedAddLN"
+
+    def infer_meta(self, attrs, inputs):
+        x = inputs[0]
+        rows = list(x.shape[:-1])
+        return [TensorMeta(x.shape, x.dtype),
+                TensorMeta(x.shape, x.dtype),
+                TensorMeta(rows, torch.float32),
+                TensorMeta(rows, torch.float32)]
+
+    def deduce_states(self, op):
+        x = op.inputs[0]
+        for out in op.outputs:
+            out.ds = x.ds
+            out.device_group = x.device_group
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        x, r, w, b = inputs
+        y, s, mean, rstd = F.layernorm_fwd_res(x, r, w, b,
+                                               op.attrs["eps"])
+        return [y, s, mean, rstd]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        dy, ds_ext = g[0], g[1]
+        ins = [dy, op.outputs[1], op.inputs[2], op.outputs[2],
+               op.outputs[3]]
+        attrs = {"has_ext": ds_ext is not None}
+        if ds_ext is not None:
+            ins.append(ds_ext)
+        bwd = _make(gr, FusedAddLNGradOp(), ins, attrs,
+                    name="fused_addln_grad")
+        dsum = bwd.output(0)
+        return [dsum, dsum, bwd.output(1), bwd.output(2)]
+
+
+class FusedAddLNGradOp(OpInterface):
+    """inputs: dy, s, w, mean, rstd[, ds_ext];
+    outputs: dsum, dw, db."""
+    type = "FusedAddLNGrad"
+
+    def infer_meta(self, attrs, inputs):
+        s, w = inputs[1], inputs[2]
+        return [TensorMeta(s.shape, s.dtype),
+                TensorMeta(w.shape, w.dtype),
+                TensorMeta(w.shape, w.dtype)]
+
+    def deduce_states(self, op):
+        s = op.inputs[1]
+        op.outputs[0].ds = s.ds
+        if s.ds is not None:
+            n = s.ds.device_num
+            npart = s.ds.partial
+            for d in s.ds.split_dims():
+                if d != s.ndim - 1:
+                    npart *= s.ds.get_dim(d)
+            states = {}
+            if npart > 1:
+                states[-2] = npart
+            if n // max(npart, 1) > 1:
+                states[-1] = n // max(npart, 1)
+            wds = DistributedStates(n, states)
+            op.outputs[1].ds = wds
+            op.outputs[2].ds = wds
+        for out in op.outputs:
+            out.device_group = s.device_group
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        dy, s, w, mean, rstd = inputs[:5]
+        ds_ext = inputs[5] if op.attrs.get("has_ext") else None
+        dsum, dw, db = F.layernorm_bwd_res(dy, s, w, mean, rstd, ds_ext)
+        return [dsum, dw.to(w.dtype), db.to(w.dtype)]

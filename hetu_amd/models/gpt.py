@@ -183,6 +183,20 @@ class GPTBlock(Module):
         x = ht.add(x, self.mlp(self.ln2(x)))
         return x
 
+    def forward_chain(self, x, delta, B, S):
+        """Pre-norm chain form: (residual sum, pending delta) in/out, with
+        the residual add fused into each LayerNorm (FusedAddLNOp) — no
+        standalone elementwise adds in the block."""
+        if delta is None:
+            y1, s1 = self.ln1(x), x
+        else:
+            y1, s1 = ht.fused_add_ln(x, delta, self.ln1.weight,
+                                     self.ln1.bias, self.ln1.eps)
+        a = self.attn(y1, B, S)
+        y2, s2 = ht.fused_add_ln(s1, a, self.ln2.weight, self.ln2.bias,
+                                 self.ln2.eps)
+        return s2, self.mlp(y2)
+
 
 class GPTEmbedding(Module):
     """Token + learned position embeddings (positions offset per cp rank)."""
@@ -244,12 +258,21 @@ class GPTLMHeadModel(Module):
         B, S, cfg, spec = self.B, self.S, self.cfg, self.spec
         x = self.embed(input_ids)
         g = x.graph
+        fused_ln = os.environ.get("HETU_AMD_FUSED_ADDLN", "1") == "1"
+        delta = None
         for i, blk in enumerate(self.layers):
             cm = g.recompute_scope(i) if self.recompute \
                 else contextlib.nullcontext()
             with cm:
-                x = blk(x, B, S)
-        x = self.lnf(x)
+                if fused_ln:
+                    x, delta = blk.forward_chain(x, delta, B, S)
+                else:
+                    x = blk(x, B, S)
+        if fused_ln and delta is not None:
+            x, _ = ht.fused_add_ln(x, delta, self.lnf.weight,
+                                   self.lnf.bias, self.lnf.eps)
+        else:
+            x = self.lnf(x)
         xr = ht.reshape(x, (B * S, cfg.hidden), ds=spec.ds_tokens(0))
         if self.lm_head is None:
             logits = ht.linear(xr, self.embed.wte.weight)

@@ -159,6 +159,33 @@ def layernorm_bwd(dy, x, w, mean, rstd):
 # SwiGLU: y = silu(x1) * x2 over last-dim halves (reference SwiGLU.cu:14,30)
 # ---------------------------------------------------------------------------
 
+def layernorm_fwd_res(x, resid, w, b, eps):
+    """Fused residual-add + LayerNorm: s = x + resid; y = LN(s).
+    Returns (y, s, mean, rstd)."""
+    if _use_hip("ln", x):
+        return tuple(ext().layernorm_fwd_res(
+            x.contiguous(), resid.contiguous(),
+            w.contiguous().to(x.dtype), b.contiguous().to(x.dtype), eps))
+    s = x + resid
+    y, mean, rstd = layernorm_fwd(s, w, b, eps)
+    return y, s, mean, rstd
+
+
+def layernorm_bwd_res(dy, s, w, mean, rstd, ds_ext=None):
+    """returns (dsum = dLN/ds (+ ds_ext), dw, db); dsum is the grad of
+    both add inputs."""
+    if _use_hip("ln", s):
+        empty = torch.empty(0, dtype=s.dtype, device=s.device)
+        return tuple(ext().layernorm_bwd2_res(
+            dy.contiguous(), s.contiguous(), w.contiguous().to(s.dtype),
+            mean.contiguous(), rstd.contiguous(),
+            ds_ext.contiguous() if ds_ext is not None else empty))
+    dx, dw, db = layernorm_bwd(dy, s, w, mean, rstd)
+    if ds_ext is not None:
+        dx = dx + ds_ext
+    return dx, dw, db
+
+
 def swiglu_fwd(x):
     if _use_hip("swiglu", x):
         return ext().swiglu_fwd(x.contiguous())

@@ -108,11 +108,16 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ dy,
 
 // ---------------- LayerNorm ----------------
 
-template <typename T>
+// RES: fuse the transformer residual add into the norm — loads x+resid,
+// writes the sum (feeds the next residual) alongside y, killing the
+// standalone elementwise add kernel (the at::native tail).
+template <typename T, bool RES>
 __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ resid,
                                      const T* __restrict__ w,
                                      const T* __restrict__ b,
                                      T* __restrict__ y,
+                                     T* __restrict__ sum_out,
                                      float* __restrict__ mean_out,
                                      float* __restrict__ rstd_out,
                                      int64_t rows, int D, float eps) {
@@ -120,11 +125,20 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
   __shared__ float smem[16];
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + row * D;
+    const T* rr = RES ? resid + row * D : nullptr;
     T* yr = y + row * D;
+    T* sr = RES ? sum_out + row * D : nullptr;
     float s = 0.f, ss = 0.f;
     for (int i = threadIdx.x * V; i < D; i += BLOCK * V) {
       float v[VecIO<T>::VEC];
       VecIO<T>::load(xr + i, v);
+      if (RES) {
+        float rv[VecIO<T>::VEC];
+        VecIO<T>::load(rr + i, rv);
+#pragma unroll
+        for (int j = 0; j < V; ++j) v[j] += rv[j];
+        VecIO<T>::store(sr + i, v);
+      }
 #pragma unroll
       for (int j = 0; j < V; ++j) { s += v[j]; ss += v[j] * v[j]; }
     }
@@ -134,9 +148,10 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
     float var = ss / D - mu * mu;
     float rstd = rsqrtf(var + eps);
     if (threadIdx.x == 0) { mean_out[row] = mu; rstd_out[row] = rstd; }
+    const T* sum_r = RES ? sr : xr;
     for (int i = threadIdx.x * V; i < D; i += BLOCK * V) {
       float v[VecIO<T>::VEC], wv[VecIO<T>::VEC], bv[VecIO<T>::VEC];
-      VecIO<T>::load(xr + i, v);
+      VecIO<T>::load(sum_r + i, v);
       VecIO<T>::load(w + i, wv);
       VecIO<T>::load(b + i, bv);
 #pragma unroll
@@ -290,12 +305,14 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
   auto rstd = torch::empty({rows}, x.options().dtype(at::kFloat));
   auto stream = hetu_current_stream();
   DISPATCH_FLOAT(x, "layernorm_fwd", [&] {
-    hipLaunchKernelGGL(layernorm_fwd_kernel<scalar_t>, dim3(row_grid(rows)),
-                       dim3(BLOCK), 0, stream,
+    hipLaunchKernelGGL((layernorm_fwd_kernel<scalar_t, false>),
+                       dim3(row_grid(rows)), dim3(BLOCK), 0, stream,
                        (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)nullptr,
                        (const scalar_t*)w.data_ptr(),
                        (const scalar_t*)b.data_ptr(),
-                       (scalar_t*)y.data_ptr(), mean.data_ptr<float>(),
+                       (scalar_t*)y.data_ptr(), (scalar_t*)nullptr,
+                       mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), rows, D, (float)eps);
   });
   auto row_sizes = at::IntArrayRef(x.sizes().begin(), x.sizes().end() - 1);
@@ -352,6 +369,7 @@ __global__ void norm_bwd_dx_kernel(const T* __restrict__ dy,
                                    const T* __restrict__ w,
                                    const float* __restrict__ mean,
                                    const float* __restrict__ rstd,
+                                   const T* __restrict__ dresid,
                                    T* __restrict__ dx,
                                    int64_t rows, int D) {
   constexpr int V = VecIO<T>::VEC;
@@ -391,6 +409,14 @@ __global__ void norm_bwd_dx_kernel(const T* __restrict__ dy,
         float xhat = (xv[j] - mu) * r;
         float wdy = dv[j] * wv[j];
         o[j] = LN ? (wdy - c1 - xhat * c2) * r : (wdy - xhat * c2) * r;
+      }
+      if (dresid != nullptr) {
+        // fused residual-grad accumulation (the adjoint of add+LN):
+        // dx = dLN/dsum + dresid, no standalone add kernel in backward
+        float rv[VecIO<T>::VEC];
+        VecIO<T>::load(dresid + row * D + i, rv);
+#pragma unroll
+        for (int j = 0; j < V; ++j) o[j] += rv[j];
       }
       VecIO<T>::store(dxr + i, o);
     }
@@ -449,10 +475,10 @@ template <typename T, bool LN>
 void norm_bwd_v2_launch(const T* dy, const T* x, const T* w,
                         const float* mean, const float* rstd, T* dx,
                         float* dw, float* db, int64_t rows, int D,
-                        hipStream_t stream) {
+                        hipStream_t stream, const T* dresid = nullptr) {
   hipLaunchKernelGGL((norm_bwd_dx_kernel<T, LN>), dim3(v2_grid(rows)),
-                     dim3(256), 0, stream, dy, x, w, mean, rstd, dx, rows,
-                     D);
+                     dim3(256), 0, stream, dy, x, w, mean, rstd,
+                     dresid, dx, rows, D);
   // pick rows_chunk so the grid lands around ~2048 blocks
   int col_blocks = (D + 511) / 512;
   int target = (2048 + col_blocks - 1) / col_blocks;
@@ -509,4 +535,71 @@ std::vector<torch::Tensor> rmsnorm_bwd2(torch::Tensor dy, torch::Tensor x,
         D, stream);
   });
   return {dx, dw32.to(w.scalar_type())};
+}
+
+
+// ---- fused residual-add + LayerNorm (transformer pre-norm hot path) ----
+// fwd: s = x + resid; y = LN(s).  Returns {y, s, mean, rstd}.
+std::vector<torch::Tensor> layernorm_fwd_res(torch::Tensor x,
+                                             torch::Tensor resid,
+                                             torch::Tensor w,
+                                             torch::Tensor b, double eps) {
+  const int D = x.size(-1);
+  const int64_t rows = x.numel() / D;
+  TORCH_CHECK(D % 8 == 0, "layernorm: D must be a multiple of 8");
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              b.scalar_type() == x.scalar_type() &&
+              resid.scalar_type() == x.scalar_type(),
+              "layernorm_fwd_res: dtype mismatch");
+  TORCH_CHECK(resid.numel() == x.numel());
+  auto y = torch::empty_like(x);
+  auto s_out = torch::empty_like(x);
+  auto mean = torch::empty({rows}, x.options().dtype(at::kFloat));
+  auto rstd = torch::empty({rows}, x.options().dtype(at::kFloat));
+  auto stream = hetu_current_stream();
+  DISPATCH_FLOAT(x, "layernorm_fwd_res", [&] {
+    hipLaunchKernelGGL((layernorm_fwd_kernel<scalar_t, true>),
+                       dim3(row_grid(rows)), dim3(BLOCK), 0, stream,
+                       (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)resid.data_ptr(),
+                       (const scalar_t*)w.data_ptr(),
+                       (const scalar_t*)b.data_ptr(),
+                       (scalar_t*)y.data_ptr(),
+                       (scalar_t*)s_out.data_ptr(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       rows, D, (float)eps);
+  });
+  auto row_sizes = at::IntArrayRef(x.sizes().begin(), x.sizes().end() - 1);
+  return {y, s_out, mean.view(row_sizes), rstd.view(row_sizes)};
+}
+
+// bwd: given dy (grad of y) and optional ds_ext (grad of s from its other
+// consumers), returns {dsum = dLN/ds + ds_ext, dw, db} — dsum is the grad
+// of BOTH x and resid.
+std::vector<torch::Tensor> layernorm_bwd2_res(torch::Tensor dy,
+                                              torch::Tensor s,
+                                              torch::Tensor w,
+                                              torch::Tensor mean,
+                                              torch::Tensor rstd,
+                                              torch::Tensor ds_ext) {
+  const int D = s.size(-1);
+  TORCH_CHECK(w.scalar_type() == s.scalar_type() &&
+              dy.scalar_type() == s.scalar_type(),
+              "layernorm_bwd2_res: dy/w dtype must match s");
+  const int64_t rows = s.numel() / D;
+  TORCH_CHECK(D % 2 == 0);
+  auto dx = torch::empty_like(s);
+  auto dw32 = torch::zeros({D}, s.options().dtype(at::kFloat));
+  auto db32 = torch::zeros({D}, s.options().dtype(at::kFloat));
+  auto stream = hetu_current_stream();
+  const bool has_ext = ds_ext.defined() && ds_ext.numel() > 0;
+  DISPATCH_FLOAT(s, "layernorm_bwd2_res", [&] {
+    norm_bwd_v2_launch<scalar_t, true>(
+        (const scalar_t*)dy.data_ptr(), (const scalar_t*)s.data_ptr(),
+        (const scalar_t*)w.data_ptr(), mean.data_ptr<float>(),
+        rstd.data_ptr<float>(), (scalar_t*)dx.data_ptr(),
+        dw32.data_ptr<float>(), db32.data_ptr<float>(), rows, D, stream,
+        has_ext ? (const scalar_t*)ds_ext.data_ptr() : nullptr);
+  });
+  return {dx, dw32.to(w.scalar_type()), db32.to(w.scalar_type())};
 }
