@@ -472,3 +472,33 @@ def test_varlen_single_kernel():
             lref[:, s0:s1] = lb[0]
         assert (o.float() - oref.float()).abs().max().item() < 2e-2, causal
         assert (lse - lref).abs().max().item() < 1e-3, causal
+
+
+@pytest.mark.gpu
+def test_fused_add_ln_kernels():
+    """layernorm_fwd_res / layernorm_bwd2_res numerics vs composed."""
+    import hetu_amd.ops.functional as F
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(0)
+    R, D = 512, 1024
+    x = torch.randn(R, D, dtype=torch.bfloat16, device=dev)
+    r = torch.randn(R, D, dtype=torch.bfloat16, device=dev)
+    w = (torch.rand(D, device=dev) + 0.5).bfloat16()
+    b = (torch.randn(D, device=dev) * 0.1).bfloat16()
+    y, s, mean, rstd = F.layernorm_fwd_res(x, r, w, b, 1e-5)
+    s_ref = (x.float() + r.float()).bfloat16()
+    y_ref, m_ref, r_ref = F.layernorm_fwd(s_ref, w, b, 1e-5)
+    assert torch.equal(s, s_ref)
+    assert (y.float() - y_ref.float()).abs().max().item() < 2e-2
+    assert (mean - m_ref).abs().max().item() < 1e-3
+    dy = torch.randn_like(y)
+    ds_ext = torch.randn_like(y)
+    dsum, dw, db = F.layernorm_bwd_res(dy, s, w, mean, rstd, ds_ext)
+    dx_ref, dw_ref, db_ref = F.layernorm_bwd(dy, s, w, mean, rstd)
+    assert (dsum.float() - (dx_ref.float() + ds_ext.float())
+            ).abs().max().item() < 3e-2
+    assert (dw.float() - dw_ref.float()).abs().max().item() < 1.0
+    assert (db.float() - db_ref.float()).abs().max().item() < 1.0
+    # no-ext variant
+    dsum2, _, _ = F.layernorm_bwd_res(dy, s, w, mean, rstd, None)
+    assert (dsum2.float() - dx_ref.float()).abs().max().item() < 2e-2
