@@ -489,7 +489,10 @@ def test_fused_add_ln_kernels():
     s_ref = (x.float() + r.float()).bfloat16()
     y_ref, m_ref, r_ref = F.layernorm_fwd(s_ref, w, b, 1e-5)
     assert torch.equal(s, s_ref)
-    assert (y.float() - y_ref.float()).abs().max().item() < 2e-2
+    # the fused kernel takes stats over the fp32 sum; the composed ref
+    # rounds s to bf16 first — up to ~2 bf16 ulps apart (fused is the
+    # more accurate one)
+    assert (y.float() - y_ref.float()).abs().max().item() < 6e-2
     assert (mean - m_ref).abs().max().item() < 1e-3
     dy = torch.randn_like(y)
     ds_ext = torch.randn_like(y)
