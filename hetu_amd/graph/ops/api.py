@@ -297,6 +297,12 @@ def fused_add_ln(x, r, w, b, eps=1e-5):
     return op.output(0), op.output(1)
 
 
+def fused_add_rms(x, r, w, eps=1e-6):
+    """Fused residual-add + RMSNorm: returns (rms_out, sum)."""
+    op = _cg().make_op(N.FusedAddRMSOp(), [x, r, w], {"eps": eps})
+    return op.output(0), op.output(1)
+
+
 def fused_mlp(x, wfc, b1, wproj, b2=None):
     """Epilogue-fused transformer MLP (see nnops.FusedMLPOp):
     y = gelu(x @ wfc^T + b1) @ wproj^T (+ b2)."""

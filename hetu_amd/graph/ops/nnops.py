@@ -857,3 +857,75 @@ class FusedAddLNGradOp(OpInterface):
         ds_ext = inputs[5] if op.attrs.get("has_ext") else None
         dsum, dw, db = F.layernorm_bwd_res(dy, s, w, mean, rstd, ds_ext)
         return [dsum, dw.to(w.dtype), db.to(w.dtype)]
+
+
+class FusedAddRMSOp(OpInterface):
+    """Fused residual-add + RMSNorm (Llama pre-norm chain); see
+    FusedAddLNOp.  inputs: x, r, w; outputs: y, s, rstd."""
+    type = "FusedAddRMS"
+
+    def infer_meta(self, attrs, inputs):
+        x = inputs[0]
+        rows = list(x.shape[:-1])
+        return [TensorMeta(x.shape, x.dtype),
+                TensorMeta(x.shape, x.dtype),
+                TensorMeta(rows, torch.float32)]
+
+    def deduce_states(self, op):
+        x = op.inputs[0]
+        for out in op.outputs:
+            out.ds = x.ds
+            out.device_group = x.device_group
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        x, r, w = inputs
+        y, s, rstd = F.rmsnorm_fwd_res(x, r, w, op.attrs["eps"])
+        return [y, s, rstd]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        dy, ds_ext = g[0], g[1]
+        ins = [dy, op.outputs[1], op.inputs[2], op.outputs[2]]
+        attrs = {"has_ext": ds_ext is not None}
+        if ds_ext is not None:
+            ins.append(ds_ext)
+        bwd = _make(gr, FusedAddRMSGradOp(), ins, attrs,
+                    name="fused_addrms_grad")
+        dsum = bwd.output(0)
+        return [dsum, dsum, bwd.output(1)]
+
+
+class FusedAddRMSGradOp(OpInterface):
+    """inputs: dy, s, w, rstd[, ds_ext]; outputs: dsum, dw."""
+    type = "FusedAddRMSGrad"
+
+    def infer_meta(self, attrs, inputs):
+        s, w = inputs[1], inputs[2]
+        return [TensorMeta(s.shape, s.dtype),
+                TensorMeta(w.shape, w.dtype)]
+
+    def deduce_states(self, op):
+        s = op.inputs[1]
+        op.outputs[0].ds = s.ds
+        if s.ds is not None:
+            n = s.ds.device_num
+            npart = s.ds.partial
+            for d in s.ds.split_dims():
+                if d != s.ndim - 1:
+                    npart *= s.ds.get_dim(d)
+            states = {}
+            if npart > 1:
+                states[-2] = npart
+            if n // max(npart, 1) > 1:
+                states[-1] = n // max(npart, 1)
+            op.outputs[1].ds = DistributedStates(n, states)
+        for out in op.outputs:
+            out.device_group = s.device_group
+
+    def compute(self, op, inputs, ctx):
+        from ...ops import functional as F
+        dy, s, w, rstd = inputs[:4]
+        ds_ext = inputs[4] if op.attrs.get("has_ext") else None
+        dsum, dw = F.rmsnorm_bwd_res(dy, s, w, rstd, ds_ext)
+        return [dsum, dw.to(w.dtype)]

@@ -159,6 +159,18 @@ class LlamaBlock(Module):
         x = ht.add(x, self.mlp(self.ln2(x)))
         return x
 
+    def forward_chain(self, x, delta, B, S):
+        """Pre-norm chain with the residual add fused into each RMSNorm
+        (FusedAddRMSOp) — see GPTBlock.forward_chain."""
+        if delta is None:
+            y1, s1 = self.ln1(x), x
+        else:
+            y1, s1 = ht.fused_add_rms(x, delta, self.ln1.weight,
+                                      self.ln1.eps)
+        a = self.attn(y1, B, S)
+        y2, s2 = ht.fused_add_rms(s1, a, self.ln2.weight, self.ln2.eps)
+        return s2, self.mlp(y2)
+
 
 class LlamaLMHeadModel(Module):
     """Builds ops in the current graph; forward(input_ids, labels) ->
@@ -198,12 +210,23 @@ class LlamaLMHeadModel(Module):
         B, S, cfg, spec = self.B, self.S, self.cfg, self.spec
         x = self.wte(input_ids)
         g = x.graph
+        import os as _os
+        fused_ln = _os.environ.get("HETU_AMD_FUSED_ADDLN", "1") == "1" \
+            and not (spec.sequence_parallel and spec.tp > 1)
+        delta = None
         for i, blk in enumerate(self.layers):
             cm = g.recompute_scope(i) if self.recompute \
                 else contextlib.nullcontext()
             with cm:
-                x = blk(x, B, S)
-        x = self.lnf(x)
+                if fused_ln:
+                    x, delta = blk.forward_chain(x, delta, B, S)
+                else:
+                    x = blk(x, B, S)
+        if fused_ln and delta is not None:
+            x, _ = ht.fused_add_rms(x, delta, self.lnf.weight,
+                                    self.lnf.eps)
+        else:
+            x = self.lnf(x)
         if spec.sequence_parallel and spec.tp > 1:
             # gather the seq shards back before the LM head (the head's
             # column-parallel GEMM wants the full token set per rank)

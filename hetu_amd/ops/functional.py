@@ -159,6 +159,31 @@ def layernorm_bwd(dy, x, w, mean, rstd):
 # SwiGLU: y = silu(x1) * x2 over last-dim halves (reference SwiGLU.cu:14,30)
 # ---------------------------------------------------------------------------
 
+def rmsnorm_fwd_res(x, resid, w, eps):
+    """Fused residual-add + RMSNorm: s = x + resid; y = RMS(s).
+    Returns (y, s, rstd)."""
+    if _use_hip("rms", x):
+        return tuple(ext().rmsnorm_fwd_res(
+            x.contiguous(), resid.contiguous(),
+            w.contiguous().to(x.dtype), eps))
+    s = x + resid
+    y, rstd = rmsnorm_fwd(s, w, eps)
+    return y, s, rstd
+
+
+def rmsnorm_bwd_res(dy, s, w, rstd, ds_ext=None):
+    if _use_hip("rms", s):
+        empty = torch.empty(0, dtype=s.dtype, device=s.device)
+        return tuple(ext().rmsnorm_bwd2_res(
+            dy.contiguous(), s.contiguous(), w.contiguous().to(s.dtype),
+            rstd.contiguous(),
+            ds_ext.contiguous() if ds_ext is not None else empty))
+    dx, dw = rmsnorm_bwd(dy, s, w, rstd)
+    if ds_ext is not None:
+        dx = dx + ds_ext
+    return dx, dw
+
+
 def layernorm_fwd_res(x, resid, w, b, eps):
     """Fused residual-add + LayerNorm: s = x + resid; y = LN(s).
     Returns (y, s, mean, rstd)."""

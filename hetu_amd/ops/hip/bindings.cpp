@@ -60,6 +60,14 @@ torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p,
 torch::Tensor embedding_fwd(torch::Tensor table, torch::Tensor ids);
 torch::Tensor embedding_bwd(torch::Tensor dy, torch::Tensor ids,
                             int64_t num_rows);
+std::vector<torch::Tensor> rmsnorm_fwd_res(torch::Tensor x,
+                                           torch::Tensor resid,
+                                           torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd2_res(torch::Tensor dy,
+                                            torch::Tensor s,
+                                            torch::Tensor w,
+                                            torch::Tensor rstd,
+                                            torch::Tensor ds_ext);
 std::vector<torch::Tensor> layernorm_fwd_res(torch::Tensor x,
                                              torch::Tensor resid,
                                              torch::Tensor w,
@@ -146,6 +154,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("layernorm_bwd2", &layernorm_bwd2);
   m.def("layernorm_fwd_res", &layernorm_fwd_res);
+  m.def("rmsnorm_fwd_res", &rmsnorm_fwd_res);
+  m.def("rmsnorm_bwd2_res", &rmsnorm_bwd2_res);
   m.def("layernorm_bwd2_res", &layernorm_bwd2_res);
   m.def("rmsnorm_bwd2", &rmsnorm_bwd2);
   m.def("colsum", &colsum);

@@ -15,30 +15,42 @@ constexpr int BLOCK = 256;
 
 // ---------------- RMSNorm ----------------
 
-template <typename T>
+template <typename T, bool RES>
 __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x,
+                                   const T* __restrict__ resid,
                                    const T* __restrict__ w,
                                    T* __restrict__ y,
+                                   T* __restrict__ sum_out,
                                    float* __restrict__ rstd_out,
                                    int64_t rows, int D, float eps) {
   constexpr int V = VecIO<T>::VEC;
   __shared__ float smem[16];
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + row * D;
+    const T* rr = RES ? resid + row * D : nullptr;
     T* yr = y + row * D;
+    T* sr = RES ? sum_out + row * D : nullptr;
     float ss = 0.f;
     for (int i = threadIdx.x * V; i < D; i += BLOCK * V) {
       float v[VecIO<T>::VEC];
       VecIO<T>::load(xr + i, v);
+      if (RES) {
+        float rv[VecIO<T>::VEC];
+        VecIO<T>::load(rr + i, rv);
+#pragma unroll
+        for (int j = 0; j < V; ++j) v[j] += rv[j];
+        VecIO<T>::store(sr + i, v);
+      }
 #pragma unroll
       for (int j = 0; j < V; ++j) ss += v[j] * v[j];
     }
     ss = block_sum(ss, smem);
     float rstd = rsqrtf(ss / D + eps);
     if (threadIdx.x == 0) rstd_out[row] = rstd;
+    const T* sum_r = RES ? sr : xr;
     for (int i = threadIdx.x * V; i < D; i += BLOCK * V) {
       float v[VecIO<T>::VEC], wv[VecIO<T>::VEC];
-      VecIO<T>::load(xr + i, v);
+      VecIO<T>::load(sum_r + i, v);
       VecIO<T>::load(w + i, wv);
 #pragma unroll
       for (int j = 0; j < V; ++j) v[j] = v[j] * rstd * wv[j];
@@ -248,12 +260,13 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
   auto rstd = torch::empty({rows}, x.options().dtype(at::kFloat));
   auto stream = hetu_current_stream();
   DISPATCH_FLOAT(x, "rmsnorm_fwd", [&] {
-    hipLaunchKernelGGL(rmsnorm_fwd_kernel<scalar_t>, dim3(row_grid(rows)),
-                       dim3(BLOCK), 0, stream,
+    hipLaunchKernelGGL((rmsnorm_fwd_kernel<scalar_t, false>),
+                       dim3(row_grid(rows)), dim3(BLOCK), 0, stream,
                        (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)nullptr,
                        (const scalar_t*)w.data_ptr(),
-                       (scalar_t*)y.data_ptr(), rstd.data_ptr<float>(),
-                       rows, D, (float)eps);
+                       (scalar_t*)y.data_ptr(), (scalar_t*)nullptr,
+                       rstd.data_ptr<float>(), rows, D, (float)eps);
   });
   return {y, rstd.view(at::IntArrayRef(x.sizes().begin(), x.sizes().end() - 1))};
 }
@@ -602,4 +615,59 @@ std::vector<torch::Tensor> layernorm_bwd2_res(torch::Tensor dy,
         has_ext ? (const scalar_t*)ds_ext.data_ptr() : nullptr);
   });
   return {dx, dw32.to(w.scalar_type()), db32.to(w.scalar_type())};
+}
+
+
+// ---- fused residual-add + RMSNorm (Llama pre-norm hot path) ------------
+std::vector<torch::Tensor> rmsnorm_fwd_res(torch::Tensor x,
+                                           torch::Tensor resid,
+                                           torch::Tensor w, double eps) {
+  const int D = x.size(-1);
+  const int64_t rows = x.numel() / D;
+  TORCH_CHECK(D % 8 == 0, "rmsnorm: D must be a multiple of 8");
+  TORCH_CHECK(w.scalar_type() == x.scalar_type() &&
+              resid.scalar_type() == x.scalar_type(),
+              "rmsnorm_fwd_res: dtype mismatch");
+  auto y = torch::empty_like(x);
+  auto s_out = torch::empty_like(x);
+  auto rstd = torch::empty({rows}, x.options().dtype(at::kFloat));
+  auto stream = hetu_current_stream();
+  DISPATCH_FLOAT(x, "rmsnorm_fwd_res", [&] {
+    hipLaunchKernelGGL((rmsnorm_fwd_kernel<scalar_t, true>),
+                       dim3(row_grid(rows)), dim3(BLOCK), 0, stream,
+                       (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)resid.data_ptr(),
+                       (const scalar_t*)w.data_ptr(),
+                       (scalar_t*)y.data_ptr(),
+                       (scalar_t*)s_out.data_ptr(),
+                       rstd.data_ptr<float>(), rows, D, (float)eps);
+  });
+  auto row_sizes = at::IntArrayRef(x.sizes().begin(), x.sizes().end() - 1);
+  return {y, s_out, rstd.view(row_sizes)};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd2_res(torch::Tensor dy,
+                                            torch::Tensor s,
+                                            torch::Tensor w,
+                                            torch::Tensor rstd,
+                                            torch::Tensor ds_ext) {
+  const int D = s.size(-1);
+  TORCH_CHECK(w.scalar_type() == s.scalar_type() &&
+              dy.scalar_type() == s.scalar_type(),
+              "rmsnorm_bwd2_res: dy/w dtype must match s");
+  const int64_t rows = s.numel() / D;
+  TORCH_CHECK(D % 2 == 0);
+  auto dx = torch::empty_like(s);
+  auto dw32 = torch::zeros({D}, s.options().dtype(at::kFloat));
+  auto stream = hetu_current_stream();
+  const bool has_ext = ds_ext.defined() && ds_ext.numel() > 0;
+  DISPATCH_FLOAT(s, "rmsnorm_bwd2_res", [&] {
+    norm_bwd_v2_launch<scalar_t, false>(
+        (const scalar_t*)dy.data_ptr(), (const scalar_t*)s.data_ptr(),
+        (const scalar_t*)w.data_ptr(), nullptr, rstd.data_ptr<float>(),
+        (scalar_t*)dx.data_ptr(), dw32.data_ptr<float>(), nullptr, rows,
+        D, stream,
+        has_ext ? (const scalar_t*)ds_ext.data_ptr() : nullptr);
+  });
+  return {dx, dw32.to(w.scalar_type())};
 }
