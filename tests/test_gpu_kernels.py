@@ -505,3 +505,26 @@ def test_fused_add_ln_kernels():
     # no-ext variant
     dsum2, _, _ = F.layernorm_bwd_res(dy, s, w, mean, rstd, None)
     assert (dsum2.float() - dx_ref.float()).abs().max().item() < 2e-2
+
+
+@pytest.mark.gpu
+def test_fused_add_rms_kernels():
+    import hetu_amd.ops.functional as F
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(1)
+    R, D = 256, 512
+    x = torch.randn(R, D, dtype=torch.bfloat16, device=dev)
+    r = torch.randn(R, D, dtype=torch.bfloat16, device=dev)
+    w = (torch.rand(D, device=dev) + 0.5).bfloat16()
+    y, s, rstd = F.rmsnorm_fwd_res(x, r, w, 1e-6)
+    s_ref = (x.float() + r.float()).bfloat16()
+    y_ref, r_ref = F.rmsnorm_fwd(s_ref, w, 1e-6)
+    assert torch.equal(s, s_ref)
+    assert (y.float() - y_ref.float()).abs().max().item() < 6e-2
+    dy = torch.randn_like(y)
+    ds_ext = torch.randn_like(y)
+    dsum, dw = F.rmsnorm_bwd_res(dy, s, w, rstd, ds_ext)
+    dx_ref, dw_ref = F.rmsnorm_bwd(dy, s, w, rstd)
+    assert (dsum.float() - (dx_ref.float() + ds_ext.float())
+            ).abs().max().item() < 3e-2
+    assert (dw.float() - dw_ref.float()).abs().max().item() < 1.0
