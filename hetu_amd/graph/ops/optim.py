@@ -20,6 +20,20 @@ from .basics import _make
 from .comm import make_comm
 
 
+_ZERO_TOKENS: Dict = {}
+
+
+def _zero_token(device) -> torch.Tensor:
+    """Shared dummy scalar used as the dependency token of update ops —
+    allocating a fresh zeros(()) per op per step put hundreds of tiny
+    fill kernels in every captured step."""
+    t = _ZERO_TOKENS.get(device)
+    if t is None:
+        t = torch.zeros((), device=device)
+        _ZERO_TOKENS[device] = t
+    return t
+
+
 class OptimizerUpdateOp(OpInterface):
     """Base: inputs [param, grad]; output: dummy scalar (dependency token)."""
 
@@ -51,7 +65,7 @@ class SGDStepOp(OptimizerUpdateOp):
         else:
             upd = grad.float()
         param -= (lr * upd).to(param.dtype)
-        return [torch.zeros((), device=param.device)]
+        return [_zero_token(param.device)]
 
 
 class AdamStepOp(OptimizerUpdateOp):
@@ -130,7 +144,7 @@ class AdamStepOp(OptimizerUpdateOp):
                     a.get("weight_decay", 0.0), st["step"], out16, bc_dev)
         if out16 is None:
             param.copy_(st["master"])
-        return [torch.zeros((), device=param.device)]
+        return [_zero_token(param.device)]
 
 
 class ZeroAdamStepOp(OptimizerUpdateOp):
@@ -197,7 +211,7 @@ class ZeroAdamStepOp(OptimizerUpdateOp):
         else:
             full = out16
         param.reshape(-1).copy_(full[:numel])
-        return [torch.zeros((), device=param.device)]
+        return [_zero_token(param.device)]
 
 
 class GradBucketOp(OpInterface):
@@ -314,8 +328,8 @@ class GroupOp(OpInterface):
         op.outputs[0].ds = None
 
     def compute(self, op, inputs, ctx):
-        dev = inputs[0].device if inputs else "cpu"
-        return [torch.zeros((), device=dev)]
+        dev = inputs[0].device if inputs else torch.device("cpu")
+        return [_zero_token(dev)]
 
 
 class Optimizer:
