@@ -57,6 +57,11 @@ def run_elastic_training(build_fn: Callable, feed_fn: Callable,
     deterministic so every incarnation replays the same data.
     die_at: fault injection — this rank exits hard before that step.
     Returns {"losses": {step: loss}, "epoch": n, "final_world": k}.
+
+    The elastic controller runs as a thread next to rank 0 (the reference
+    hosts it in an external gRPC server): recovery covers any non-zero
+    rank dying; host the KV store + controller out-of-process (as the
+    tests and examples/elastic do for the store) to also survive rank 0.
     """
     from ..engine.runner import prepare_run_context
     from ..utils.checkpoint import (collect_adam_states, load_adam_states,

@@ -106,8 +106,10 @@ class PipelineRunner:
         # hipGraph capture of the per-micro-batch fwd/bwd bodies (rotating
         # pp+1 buffer slots); opt-in: HETU_AMD_PP_CAPTURE=1, GPU only
         import os as _os2
+        # 1F1B only: GPipe keeps all M micro-batches in flight, which
+        # would reuse a slot before its backward replays
         self._capture = (device.type == "cuda" and not recompute
-                         and not offload
+                         and not offload and schedule == "1f1b"
                          and _os2.environ.get("HETU_AMD_PP_CAPTURE",
                                               "0") == "1")
         self._nslots = spec.pp + 1
