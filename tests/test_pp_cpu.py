@@ -247,3 +247,14 @@ def test_pp2_tied_embeddings_matches_single():
         (losses, single)
     assert len(wsums) == 2
     assert abs(wsums[0] - wsums[1]) < 1e-9, wsums
+
+
+def test_pp2_dp2_matches_single(single_losses):
+    """pp2 x dp2 (4 ranks): both dp replicas get IDENTICAL data, so the
+    global-token-denominator per-replica loss is exactly single/2 and the
+    dp-summed gradients reproduce the single-process trajectory — every
+    step must equal half the single loss to tight tolerance."""
+    losses = _launch(4, {"HETU_TEST_PP": "2"}, 29581)
+    assert losses is not None
+    assert np.allclose([2 * v for v in losses], single_losses,
+                       rtol=5e-4, atol=2e-4), (losses, single_losses)
