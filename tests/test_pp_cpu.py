@@ -258,3 +258,11 @@ def test_pp2_dp2_matches_single(single_losses):
     assert losses is not None
     assert np.allclose([2 * v for v in losses], single_losses,
                        rtol=5e-4, atol=2e-4), (losses, single_losses)
+
+
+def test_pp2_tp2_matches_single(single_losses):
+    """pp2 x tp2 (4 ranks): tensor-parallel stages inside a pipeline."""
+    losses = _launch(4, {"HETU_TEST_PP": "2", "HETU_TEST_TP": "2"}, 29585)
+    assert losses is not None
+    assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
+        (losses, single_losses)
