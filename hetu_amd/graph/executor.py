@@ -23,6 +23,38 @@ import os
 _COMM_OVERLAP = os.environ.get("HETU_AMD_COMM_OVERLAP", "1") == "1"
 # per-op allocated-memory trace (debug): records the op at the peak
 _MEM_TRACE = os.environ.get("HETU_AMD_MEM_TRACE", "0") == "1"
+# per-op NaN/Inf guard (reference CheckNumeric/CheckFinite kernels,
+# hetu/impl/kernel + gradscaler): raises at the FIRST op whose output goes
+# non-finite, naming the op — turns a step-level NaN into a one-op bisect.
+# Synchronizes per op, so debug-only; skipped inside hipGraph capture.
+_CHECK_NUMERIC = os.environ.get("HETU_AMD_CHECK_NUMERIC", "0") == "1"
+
+
+class NonFiniteError(RuntimeError):
+    """Raised by HETU_AMD_CHECK_NUMERIC=1 when an op output contains
+    NaN/Inf. .op_type/.op_name/.out_index locate the producing op."""
+
+    def __init__(self, op_type: str, op_name: str, out_index: int,
+                 n_nan: int, n_inf: int, shape):
+        self.op_type, self.op_name = op_type, op_name
+        self.out_index = out_index
+        super().__init__(
+            f"non-finite output {out_index} of {op_type} '{op_name}' "
+            f"shape={tuple(shape)}: {n_nan} NaN, {n_inf} Inf")
+
+
+def _check_numeric(op, outs):
+    for j, v in enumerate(outs):
+        if not isinstance(v, torch.Tensor) or not v.is_floating_point():
+            continue
+        if v.is_cuda and torch.cuda.is_current_stream_capturing():
+            return
+        fin = torch.isfinite(v)
+        if not bool(fin.all()):
+            bad = ~fin
+            n_nan = int(torch.isnan(v).sum())
+            n_inf = int(bad.sum()) - n_nan
+            raise NonFiniteError(op.type, op.name, j, n_nan, n_inf, v.shape)
 
 
 class ExecContext:
@@ -204,6 +236,8 @@ class Executor:
                 prof.end(tok)
             else:
                 outs = op.interface.compute(op, ins, ctx)
+            if _CHECK_NUMERIC:
+                _check_numeric(op, outs)
             if _MEM_TRACE and ctx.device.type == "cuda":
                 a = torch.cuda.memory_allocated(ctx.device)
                 if a > self._mem_peak[0]:

@@ -76,3 +76,25 @@ print("SNAP OK")
                        capture_output=True, text=True, timeout=300)
     assert p.returncode == 0 and "SNAP OK" in p.stdout, \
         f"{p.stdout}\n{p.stderr}"
+
+
+def test_check_numeric_guard(monkeypatch):
+    """HETU_AMD_CHECK_NUMERIC: the executor raises at the first op whose
+    output goes non-finite, naming that op (reference CheckNumeric)."""
+    import pytest
+    import hetu_amd as ht
+    from hetu_amd.graph import executor as ex
+
+    monkeypatch.setattr(ex, "_CHECK_NUMERIC", True)
+    with ht.graph("define_and_run") as g:
+        x = ht.placeholder((4,), name="x")
+        y = ht.log(x)                     # log(-1) -> NaN
+        y.producer.name = "bad_log"
+        z = ht.add(y, y)
+        with pytest.raises(ex.NonFiniteError) as ei:
+            g.run([z], {x: torch.tensor([1.0, -1.0, 2.0, 3.0])})
+        assert ei.value.op_name == "bad_log"
+        assert ei.value.out_index == 0
+        # clean inputs pass untouched
+        (out,) = g.run([z], {x: torch.tensor([1.0, 1.0, 2.0, 3.0])})
+        assert torch.isfinite(out).all()
