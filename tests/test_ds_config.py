@@ -1,0 +1,106 @@
+"""JSON ds_parallel_config generate/read roundtrip (reference
+utils/parallel/generate_ds.py + read_ds.py config2ds)."""
+import json
+
+from hetu_amd.parallel.dstates import NULL_HETERO_DIM
+from hetu_amd.parallel.hetero import HeteroSpec
+from hetu_amd.parallel.pipeline import PipelineSpec
+from hetu_amd.utils.ds_config import (config2ds, convert_strategy,
+                                      generate_ds_parallel_config,
+                                      read_ds_parallel_config,
+                                      strategy_from_config,
+                                      write_ds_parallel_config)
+
+
+def test_convert_strategy_keeps_tp_in_node():
+    ltg, gpu_pos = convert_strategy([(8, 1), (4, 2)], 16, 4,
+                                    gpus_per_node=8)
+    for layer in ltg:
+        assert len(layer) == 2
+        for grp in layer:
+            nodes = {r // 8 for r in grp}
+            assert len(nodes) == 1 or len(grp) % 8 == 0
+    assert len(gpu_pos) == 16
+    # pipeline 1 (tp4 pp2) has two stages
+    stages1 = {gpu_pos[r][1] for r in gpu_pos if gpu_pos[r][0] == 1}
+    assert stages1 == {0, 1}
+
+
+def test_homogeneous_roundtrip(tmp_path):
+    cfg = generate_ds_parallel_config([(2, 2)], num_layers=4)
+    p = str(tmp_path / "ds.json")
+    write_ds_parallel_config(cfg, p)
+    entries = read_ds_parallel_config(p)
+    assert "input" in entries and "gpt.wte" in entries
+    assert "gpt.blocks.blocks0.attn.qkv" in entries
+    u, dgu = entries["gpt.blocks.blocks0.attn.qkv"]
+    assert u.hetero_dim == NULL_HETERO_DIM and u.size() == 1
+    ds = u.get(0)
+    assert ds.get_dim(1) == 2 and ds.device_num == 2
+    spec, stages = strategy_from_config(p)
+    assert isinstance(spec, PipelineSpec)
+    assert spec.pp == 2 and spec.tp == 2
+    assert stages[0] == [[0, 1], [2, 3]]
+    # dp==1 forces zero off
+    assert cfg["zero"] is False
+
+
+def test_hetero_roundtrip():
+    cfg = generate_ds_parallel_config([(4, 1), (2, 1), (2, 1)],
+                                      num_layers=2)
+    entries = read_ds_parallel_config(cfg)
+    u, dgu = entries["gpt.wte"]
+    assert u.is_hetero() and u.hetero_dim == -1 and u.size() == 3
+    # pipeline 0 is tp4: local layout (union stripped) splits dim 0 by 4
+    loc = u.get_local(0)
+    assert loc.get_dim(0) == 4 and loc.device_num == 4
+    loc2 = u.get_local(1)
+    assert loc2.get_dim(0) == 2 and loc2.device_num == 2
+    # placeholder input is hetero along the batch dim
+    ui, _ = entries["input"]
+    assert ui.hetero_dim == 0
+    spec, stages = strategy_from_config(cfg)
+    assert isinstance(spec, HeteroSpec)
+    assert sorted(p.tp for p in spec.pipelines) == [2, 2, 4]
+    # zero spread into variable entries (dp>1 keeps it on)
+    assert cfg["zero"] is True
+    assert u.get(0).zero
+
+
+def test_config2ds_orders():
+    e = {"split": {"0": [2]}, "dup": [3], "device_group_union": [[0, 1, 2,
+         3, 4, 5]], "type": "placeholder"}
+    u, _ = config2ds(e)
+    assert u.get(0).order == [0, -1]
+    e["type"] = "variable"
+    u, _ = config2ds(e)
+    assert u.get(0).order == [-1, 0]
+
+
+def test_interop_with_reference_style_json(tmp_path):
+    """A hand-written reference-schema file (llama key, rmsnorm names)
+    parses: the reader walks any tree shape."""
+    cfg = {
+        "zero": True, "devices": [0, 1],
+        "input": {"split": {"0": [2, 2]}, "dup": [1, 1],
+                  "device_group_union": [[0], [1]], "type": "placeholder"},
+        "llama": {"wte": {"split": {"0": [1, 1]}, "dup": [2, 2],
+                          "device_group_union": [[0], [1]],
+                          "type": "variable"},
+                  "blocks": {"blocks0": {
+                      "range": [0], "recompute": [False, False],
+                      "attn": {"qkv": {"split": {"1": [1, 1]},
+                                       "dup": [2, 2],
+                                       "device_group_union": [[0], [1]],
+                                       "type": "variable"}}}}},
+        "label": {"split": {"0": [2, 2]}, "dup": [1, 1],
+                  "device_group_union": [[0], [1]], "type": "placeholder"},
+    }
+    p = str(tmp_path / "ref.json")
+    with open(p, "w") as f:
+        json.dump(cfg, f)
+    entries = read_ds_parallel_config(p)
+    u, _ = entries["llama.wte"]
+    assert u.is_hetero() and u.size() == 2 and u.get(0).zero
+    spec, stages = strategy_from_config(p, model_key="llama")
+    assert isinstance(spec, HeteroSpec) and len(spec.pipelines) == 2
