@@ -35,6 +35,22 @@ class Trainer:
         self._static_feeds: Dict = {}
         self._loss_out = None
         self._step = 0
+        # host-side tracing (reference engine/trainer.py:22 wires
+        # torch.profiler): HETU_AMD_TORCH_PROFILE=<dir> records steps
+        # 2-4 (skip 0, warm 1) and writes a chrome trace there
+        self._prof = None
+        prof_dir = os.environ.get("HETU_AMD_TORCH_PROFILE", "")
+        if prof_dir:
+            acts = [torch.profiler.ProfilerActivity.CPU]
+            if device.type == "cuda":
+                acts.append(torch.profiler.ProfilerActivity.CUDA)
+            self._prof = torch.profiler.profile(
+                activities=acts,
+                schedule=torch.profiler.schedule(wait=1, warmup=1,
+                                                 active=3, repeat=1),
+                on_trace_ready=torch.profiler.tensorboard_trace_handler(
+                    prof_dir))
+            self._prof.start()
 
     # ---- plain step ------------------------------------------------------
     def run_step(self, feed: Dict):
@@ -70,8 +86,21 @@ class Trainer:
         self._cuda_graph.replay()
         return self._loss_out
 
+    def _prof_tick(self):
+        if self._prof is not None:
+            self._prof.step()
+            if self._step >= 5:            # schedule exhausted
+                self._prof.stop()
+                self._prof = None
+
     def step(self, feed: Dict):
         """Run one training step, transparently using capture when armed."""
+        try:
+            return self._step_impl(feed)
+        finally:
+            self._prof_tick()
+
+    def _step_impl(self, feed: Dict):
         if self._cuda_graph is None:
             if self.want_capture and self._step >= 1:
                 try:

@@ -98,3 +98,26 @@ def test_check_numeric_guard(monkeypatch):
         # clean inputs pass untouched
         (out,) = g.run([z], {x: torch.tensor([1.0, 1.0, 2.0, 3.0])})
         assert torch.isfinite(out).all()
+
+
+def test_torch_profiler_trace(tmp_path, monkeypatch):
+    """HETU_AMD_TORCH_PROFILE writes a chrome trace of steps 2-4
+    (reference trainer wires torch.profiler for host tracing)."""
+    import glob
+    import hetu_amd as ht
+    from hetu_amd.engine.trainer import Trainer
+
+    monkeypatch.setenv("HETU_AMD_TORCH_PROFILE", str(tmp_path))
+    with ht.graph("define_and_run") as g:
+        x = ht.placeholder((4, 8), name="x")
+        w = ht.variable(torch.randn(8, 8), name="w")
+        loss = ht.reduce_mean(ht.matmul(x, w))
+        opt = ht.Adam(lr=1e-3)
+        train_op = opt.minimize(loss)
+        tr = Trainer(g, {"loss": loss, "train_op": train_op},
+                     torch.device("cpu"))
+        for _ in range(6):
+            tr.step({x: torch.randn(4, 8)})
+    traces = glob.glob(str(tmp_path / "*.json")) + \
+        glob.glob(str(tmp_path / "**" / "*.json"), recursive=True)
+    assert traces, "no chrome trace written"
