@@ -176,6 +176,51 @@ def reduce_max(a: Tensor, dim=None, keepdim=False) -> Tensor:
                                               "keepdim": keepdim}).output()
 
 
+def reduce_min(a: Tensor, dim=None, keepdim=False) -> Tensor:
+    return _cg().make_op(B.ReduceOp_(), [a], {"mode": "min", "dim": dim,
+                                              "keepdim": keepdim}).output()
+
+
+def reduce_prod(a: Tensor, dim=None, keepdim=False) -> Tensor:
+    return _cg().make_op(B.ReduceOp_(), [a], {"mode": "prod", "dim": dim,
+                                              "keepdim": keepdim}).output()
+
+
+def norm(a: Tensor, p=2, dim=None, keepdim=False) -> Tensor:
+    from . import extra as E
+    return _cg().make_op(E.NormOp(), [a],
+                         {"p": p, "dim": dim,
+                          "keepdim": keepdim}).output()
+
+
+def softmax_cross_entropy(logits: Tensor, labels: Tensor,
+                          reduction="mean") -> Tensor:
+    """Dense soft-label CE; for integer labels use
+    softmax_cross_entropy_sparse."""
+    from . import extra as E
+    return _cg().make_op(E.SoftmaxCrossEntropyOp(), [logits, labels],
+                         {"reduction": reduction}).output()
+
+
+def broadcast_to(a: Tensor, shape) -> Tensor:
+    return _cg().make_op(B.BroadcastToOp(), [a],
+                         {"shape": tuple(shape)}).output()
+
+
+def group(*tensors) -> Tensor:
+    """Control-dependency join: fetch the returned scalar to force every
+    input to execute (reference group.cc)."""
+    return _cg().make_op(B.GroupOp(), list(tensors), {}).output()
+
+
+def ones_like(a: Tensor) -> Tensor:
+    return _cg().make_op(B.OnesLikeOp(), [a], {}).output()
+
+
+def zeros_like(a: Tensor) -> Tensor:
+    return _cg().make_op(B.ZerosLikeOp(), [a], {}).output()
+
+
 # ---- GEMM ------------------------------------------------------------------
 
 # ---- autocast (reference graph/autocast/autocast.cc:39-92: dtype

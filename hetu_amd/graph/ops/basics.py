@@ -109,6 +109,45 @@ def make_ones_like(graph, t: Tensor) -> Tensor:
     return _make(graph, OnesLikeOp(), [t], name=f"ones_like({t.name})").output()
 
 
+class BroadcastToOp(OpInterface):
+    """Expand to a target shape (reference Broadcast.cc); the gradient
+    sum-reduces back over the broadcast dims (ReduceToShapeOp)."""
+    type = "BroadcastTo"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(tuple(attrs["shape"]), inputs[0].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        return [inputs[0].expand(*op.attrs["shape"]).contiguous()]
+
+    def gradient(self, op, g):
+        gr = _g(op.outputs[0])
+        return [_make(gr, ReduceToShapeOp(), [g[0], op.inputs[0]],
+                      name="bcast_grad").output()]
+
+
+class GroupOp(OpInterface):
+    """Control-dependency join (reference group.cc): a zero scalar that
+    depends on every input — fetch it to force their execution in one
+    graph.run without materializing each value at the caller."""
+    type = "Group"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta([], torch.float32)]
+
+    def deduce_states(self, op):
+        op.outputs[0].ds = None
+        op.outputs[0].device_group = (op.inputs[0].device_group
+                                      if op.inputs else None)
+
+    def compute(self, op, inputs, ctx):
+        dev = inputs[0].device if inputs else "cpu"
+        return [torch.zeros((), device=dev)]
+
+    def gradient(self, op, g):
+        return [None] * len(op.inputs)
+
+
 # ---------------------------------------------------------------------------
 # Elementwise arithmetic (broadcasting; grads reduce back to input shape)
 # ---------------------------------------------------------------------------
@@ -647,7 +686,9 @@ class ReduceOp_(OpInterface):
                     if op.attrs["mode"] in ("sum", "mean"):
                         part *= n     # reduced over a split dim -> partial
                     else:
-                        raise ValueError("max/min reduce over split dim")
+                        raise ValueError(
+                            "max/min/prod reduce over a split dim has no "
+                            "partial-sum representation")
                 elif d >= 0:
                     nd_ = d - sum(1 for r in dims if r < d) if not keepdim else d
                     states[nd_] = n
@@ -701,16 +742,29 @@ class ReduceOp_(OpInterface):
             if dim is None:
                 return [x.max()]
             return [x.max(dim, keepdim=keepdim).values]
+        if mode == "min":
+            if dim is None:
+                return [x.min()]
+            return [x.min(dim, keepdim=keepdim).values]
+        if mode == "prod":
+            if dim is None:
+                return [x.prod()]
+            assert isinstance(dim, int), "prod reduces one dim at a time"
+            return [x.prod(dim, keepdim=keepdim)]
         raise ValueError(mode)
 
     def gradient(self, op, g):
         gr = _g(op.outputs[0])
         mode = op.attrs["mode"]
-        if mode not in ("sum", "mean"):
-            # max/min: subgradient routed to the extremal positions
+        if mode in ("max", "min"):
+            # subgradient routed to the extremal positions
             # (split across ties, matching the mask/count convention)
             return [_make(gr, ReduceExtremumGradOp(),
                           [g[0], op.inputs[0], op.outputs[0]],
+                          dict(op.attrs)).output()]
+        if mode == "prod":
+            return [_make(gr, ReduceProdGradOp(),
+                          [g[0], op.inputs[0]],
                           dict(op.attrs)).output()]
         return [_make(gr, ReduceGradOp(), [g[0], op.inputs[0]],
                       dict(op.attrs)).output()]
@@ -738,6 +792,26 @@ class ReduceExtremumGradOp(OpInterface):
         mask = (xf == ye).to(xf.dtype)
         return [(mask / mask.sum(dim, keepdim=True).clamp(min=1)
                  * ge).to(x.dtype)]
+
+
+class ReduceProdGradOp(OpInterface):
+    """dx for prod reduce.  Exact including zeros: replays the reduction
+    under torch.autograd on a detached leaf (cold path; reference
+    Reduce.cc prod grad)."""
+    type = "ReduceProdGrad"
+
+    def infer_meta(self, attrs, inputs):
+        return [TensorMeta(inputs[1].shape, inputs[1].dtype)]
+
+    def compute(self, op, inputs, ctx):
+        gy, x = inputs
+        dim = op.attrs.get("dim")
+        keepdim = op.attrs.get("keepdim", False)
+        with torch.enable_grad():
+            xl = x.detach().float().requires_grad_(True)
+            y = xl.prod() if dim is None else xl.prod(dim, keepdim=keepdim)
+            (dx,) = torch.autograd.grad(y, xl, gy.float())
+        return [dx.to(x.dtype)]
 
 
 class CheckFiniteOp(OpInterface):

@@ -219,6 +219,39 @@ class KLDivOp(_AutogradOp):
 
 
 @_register
+class NormOp(_AutogradOp):
+    """p-norm reduction (reference Norm.cc: vector p-norm over dims)."""
+    type = "Norm"
+
+    def fn(self, a):
+        return lambda x: torch.linalg.vector_norm(
+            x, ord=a.get("p", 2), dim=a.get("dim"),
+            keepdim=a.get("keepdim", False))
+
+
+@_register
+class SoftmaxCrossEntropyOp(_AutogradOp):
+    """Dense soft-label cross entropy (reference SoftmaxCrossEntropy.cc):
+    -sum(labels * log_softmax(logits), -1), reduced.  The sparse
+    integer-label variant is SoftmaxCrossEntropySparseOp (nnops.py)."""
+    type = "SoftmaxCrossEntropy"
+    grad_mask = [True, False]
+
+    def fn(self, a):
+        red = a.get("reduction", "mean")
+
+        def f(x, t):
+            ce = -(t.float()
+                   * torch.log_softmax(x.float(), -1)).sum(-1)
+            if red == "mean":
+                return ce.mean()
+            if red == "sum":
+                return ce.sum()
+            return ce
+        return f
+
+
+@_register
 class NLLOp(_AutogradOp):
     type = "NLLLoss"
     grad_mask = [True, False]

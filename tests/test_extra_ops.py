@@ -418,3 +418,76 @@ def test_fused_mlp_parity():
         outs[mode] = vals
     for i, (a, b) in enumerate(zip(outs["fused"], outs["composed"])):
         assert torch.allclose(a, b, atol=1e-4), (i, (a - b).abs().max())
+
+
+def test_min_prod_norm_reduce_grads():
+    x = torch.randn(3, 5, requires_grad=True)
+
+    def build_min():
+        p = ht.placeholder((3, 5), name="x")
+        return [p], ht.reduce_sum(ht.reduce_min(p, dim=1))
+    out, gx = _run(build_min, [x.detach()], wrt=[0])
+    ref = x.min(1).values.sum()
+    ref.backward()
+    assert torch.allclose(out, ref.detach())
+    assert torch.allclose(gx, x.grad, atol=1e-6)
+
+    x2 = torch.randn(3, 4, requires_grad=True)
+    x2d = x2.detach().clone()
+    x2d[1, 2] = 0.0                          # zero inside the product
+
+    def build_prod():
+        p = ht.placeholder((3, 4), name="x")
+        return [p], ht.reduce_sum(ht.reduce_prod(p, dim=1))
+    out, gx = _run(build_prod, [x2d], wrt=[0])
+    xr = x2d.clone().requires_grad_(True)
+    ref = xr.prod(1).sum()
+    ref.backward()
+    assert torch.allclose(out, ref.detach(), atol=1e-6)
+    assert torch.allclose(gx, xr.grad, atol=1e-5)
+
+    x3 = torch.randn(4, 6, requires_grad=True)
+
+    def build_norm():
+        p = ht.placeholder((4, 6), name="x")
+        return [p], ht.reduce_sum(ht.norm(p, p=2, dim=1))
+    out, gx = _run(build_norm, [x3.detach()], wrt=[0])
+    ref = torch.linalg.vector_norm(x3, 2, dim=1).sum()
+    ref.backward()
+    assert torch.allclose(out, ref.detach(), atol=1e-5)
+    assert torch.allclose(gx, x3.grad, atol=1e-5)
+
+
+def test_dense_softmax_ce_broadcast_group():
+    logits = torch.randn(5, 7, requires_grad=True)
+    labels = torch.softmax(torch.randn(5, 7), -1)
+
+    def build():
+        lg = ht.placeholder((5, 7), name="lg")
+        lb = ht.placeholder((5, 7), name="lb")
+        return [lg, lb], ht.softmax_cross_entropy(lg, lb)
+    out, g = _run(build, [logits.detach(), labels], wrt=[0])
+    ref = -(labels * torch.log_softmax(logits, -1)).sum(-1).mean()
+    ref.backward()
+    assert torch.allclose(out, ref.detach(), atol=1e-6)
+    assert torch.allclose(g, logits.grad, atol=1e-6)
+
+    # broadcast_to: grad sum-reduces back over the expanded dims
+    b = torch.randn(1, 4, requires_grad=True)
+
+    def build_b():
+        p = ht.placeholder((1, 4), name="b")
+        return [p], ht.reduce_sum(ht.mul(ht.broadcast_to(p, (3, 4)), 2.0))
+    out, g = _run(build_b, [b.detach()], wrt=[0])
+    ref = (b.expand(3, 4) * 2.0).sum()
+    ref.backward()
+    assert torch.allclose(out, ref.detach())
+    assert torch.allclose(g, b.grad)
+
+    # group: one scalar forcing several subgraphs, like/zeros helpers
+    def build_g():
+        p = ht.placeholder((2, 2), name="p")
+        tok = ht.group(ht.ones_like(p), ht.zeros_like(p), ht.exp(p))
+        return [p], tok
+    (out,) = _run(build_g, [torch.randn(2, 2)])
+    assert out.shape == () and float(out) == 0.0
