@@ -23,6 +23,11 @@ from typing import Dict, List, Optional, Tuple
 import torch
 import torch.distributed as dist
 
+# upcast bf16/fp16 reduction collectives to fp32 (reference runtime
+# fp32_comm_reduce flag, Communication.cc SplitAllReduce notes): trades 2x
+# collective bytes for bit-deterministic-across-bucketing grad sums
+_FP32_COMM = os.environ.get("HETU_AMD_FP32_COMM", "0") == "1"
+
 
 class CommBackend:
     """Wraps torch.distributed; caches subgroup ProcessGroups; provides
@@ -109,10 +114,57 @@ class CommBackend:
         red = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
                "min": dist.ReduceOp.MIN, "avg": dist.ReduceOp.SUM}[op]
         t = t.contiguous()
+        if _FP32_COMM and op in ("sum", "avg") \
+                and t.dtype in (torch.bfloat16, torch.float16):
+            t32 = t.float()
+            dist.all_reduce(t32, op=red, group=self.group(ranks))
+            if op == "avg":
+                t32 = t32 / len(ranks)
+            t.copy_(t32)
+            return t
         dist.all_reduce(t, op=red, group=self.group(ranks))
         if op == "avg":
             t = t / len(ranks)
         return t
+
+    def reduce(self, t: torch.Tensor, ranks: List[int], root: int,
+               op: str = "sum") -> torch.Tensor:
+        """Reduce-to-root (reference ncclReduce, nccl_comm_group.cu:373);
+        only `root` holds the reduced value afterwards."""
+        if len(ranks) <= 1 or not dist.is_initialized():
+            return t
+        red = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX,
+               "min": dist.ReduceOp.MIN}[op]
+        t = t.contiguous()
+        dist.reduce(t, dst=root, op=red, group=self.group(ranks))
+        return t
+
+    def gather(self, t: torch.Tensor, ranks: List[int], root: int
+               ) -> Optional[List[torch.Tensor]]:
+        """Gather tensors to `root` (reference nccl Gather :496); returns
+        the list on root, None elsewhere."""
+        if len(ranks) <= 1 or not dist.is_initialized():
+            return [t]
+        g = self.group(ranks)
+        t = t.contiguous()
+        outs = ([torch.empty_like(t) for _ in ranks]
+                if self.rank == root else None)
+        dist.gather(t, outs, dst=root, group=g)
+        return outs
+
+    def scatter(self, tensors: Optional[List[torch.Tensor]],
+                ranks: List[int], root: int, out: torch.Tensor
+                ) -> torch.Tensor:
+        """Scatter `tensors` (significant on root) into `out` on each
+        member (reference nccl Scatter :547)."""
+        if len(ranks) <= 1 or not dist.is_initialized():
+            out.copy_(tensors[0])
+            return out
+        g = self.group(ranks)
+        src = ([x.contiguous() for x in tensors]
+               if self.rank == root else None)
+        dist.scatter(out, src, src=root, group=g)
+        return out
 
     def allgather(self, t: torch.Tensor, ranks: List[int], dim: int = 0
                   ) -> torch.Tensor:
@@ -130,6 +182,9 @@ class CommBackend:
         dim. Falls back to allreduce+slice on gloo (no reduce_scatter)."""
         if len(ranks) <= 1 or not dist.is_initialized():
             return t
+        if _FP32_COMM and t.dtype in (torch.bfloat16, torch.float16):
+            return self.reducescatter(t.float(), ranks, dim,
+                                      my_index).to(t.dtype)
         g = self.group(ranks)
         n = len(ranks)
         if my_index is None:

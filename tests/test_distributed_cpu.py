@@ -221,3 +221,57 @@ def test_generic_resharding_partial_destination():
         out, err = p.communicate(timeout=300)
         ok = (p.returncode in (0, -6)) and "GENPOK" in out
         assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
+
+
+ROOTED_COLL_WORKER = r"""
+import os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+from hetu_amd.parallel.comm import comm_backend
+comm = comm_backend()
+rank, ws = comm.rank, comm.world_size
+ranks = list(range(ws))
+# reduce-to-root
+t = torch.full((3,), float(rank + 1))
+comm.reduce(t, ranks, root=0)
+if rank == 0:
+    assert torch.equal(t, torch.full((3,), 3.0)), t
+# gather
+outs = comm.gather(torch.full((2,), float(rank)), ranks, root=1)
+if rank == 1:
+    assert [float(o[0]) for o in outs] == [0.0, 1.0], outs
+else:
+    assert outs is None
+# scatter
+out = torch.empty(2)
+src = [torch.full((2,), 10.0), torch.full((2,), 20.0)] if rank == 0 else None
+comm.scatter(src, ranks, root=0, out=out)
+assert float(out[0]) == (10.0 if rank == 0 else 20.0), out
+# fp32-upcast allreduce path (flag forced on in-process)
+import hetu_amd.parallel.comm as cm
+cm._FP32_COMM = True
+b = torch.full((4,), 0.1, dtype=torch.bfloat16)
+comm.allreduce(b, ranks)
+assert b.dtype == torch.bfloat16
+assert abs(float(b[0]) - 0.2) < 2e-3, b
+print("ROOTOK")
+"""
+
+
+def test_rooted_collectives_two_ranks():
+    """reduce/gather/scatter to a root + HETU_AMD_FP32_COMM upcast
+    (reference ncclReduce/Gather/Scatter + fp32_comm_reduce)."""
+    import subprocess
+    import sys
+    procs = []
+    env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29637", "GLOO_SOCKET_IFNAME": "lo"}
+    for r in range(2):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="2", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, "-c",
+                                       ROOTED_COLL_WORKER], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=300)
+        ok = (p.returncode in (0, -6)) and "ROOTOK" in out
+        assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
