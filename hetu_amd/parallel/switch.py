@@ -109,29 +109,33 @@ def switch_params(plan: List[Dict], comm: Optional[CommBackend] = None):
     # per direction, so the batched p2p moves few LARGE messages over the
     # xGMI links instead of one message per overlap region.  Fragment
     # order is the plan order, which is identical on both endpoints.
-    send_groups: Dict[int, List[torch.Tensor]] = {}
+    # key by (peer, dtype): a mixed-precision plan (e.g. bf16 params +
+    # fp32 master copies in one call) cannot share a flat buffer
+    send_groups: Dict[Tuple[int, torch.dtype], List[torch.Tensor]] = {}
     for t, dst in sends:
-        send_groups.setdefault(dst, []).append(t.reshape(-1))
-    recv_groups: Dict[int, List] = {}
+        send_groups.setdefault((dst, t.dtype), []).append(t.reshape(-1))
+    recv_groups: Dict[Tuple[int, torch.dtype], List] = {}
     for buf, src, dst_t, in_b in recvs:
-        recv_groups.setdefault(src, []).append((buf, dst_t, in_b))
+        recv_groups.setdefault((src, buf.dtype), []).append(
+            (buf, dst_t, in_b))
     ops = []
-    flat_sends = {d: (torch.cat(ts) if len(ts) > 1 else ts[0])
-                  for d, ts in send_groups.items()}
+    flat_sends = {k: (torch.cat(ts) if len(ts) > 1 else ts[0])
+                  for k, ts in send_groups.items()}
     flat_recvs = {}
-    for s, items in recv_groups.items():
+    for k, items in recv_groups.items():
         n = sum(b.numel() for b, _, _ in items)
-        flat_recvs[s] = torch.empty(
+        flat_recvs[k] = torch.empty(
             n, dtype=items[0][0].dtype, device=items[0][0].device)
-    for d in sorted(flat_sends):
-        ops.append(dist.P2POp(dist.isend, flat_sends[d], d))
-    for s in sorted(flat_recvs):
-        ops.append(dist.P2POp(dist.irecv, flat_recvs[s], s))
+    for d, _ in sorted(flat_sends, key=lambda k: (k[0], str(k[1]))):
+        ops.append(dist.P2POp(dist.isend,
+                              flat_sends[(d, _)], d))
+    for s, _ in sorted(flat_recvs, key=lambda k: (k[0], str(k[1]))):
+        ops.append(dist.P2POp(dist.irecv, flat_recvs[(s, _)], s))
     if ops and dist.is_initialized():
         for r in dist.batch_isend_irecv(ops):
             r.wait()
-    for s, items in recv_groups.items():
-        flat = flat_recvs[s]
+    for k, items in recv_groups.items():
+        flat = flat_recvs[k]
         off = 0
         for buf, dst_t, in_b in items:
             n = buf.numel()
