@@ -271,6 +271,11 @@ def strategy_from_config(cfg: Union[str, Dict],
         pp = len(stages[0])
         tp = len(stages[0][0])
         return PipelineSpec(pp=pp, dp=1, tp=tp), stages
+    shapes = {(len(s), len(s[0])) for s in stages}
+    if len(shapes) == 1:
+        # identical pipelines: a plain homogeneous pp x dp x tp world
+        (pp, tp), = shapes
+        return PipelineSpec(pp=pp, dp=K, tp=tp), stages
     pipes = []
     for k in range(K):
         assert len(stages[k]) == 1, \

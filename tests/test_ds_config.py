@@ -103,4 +103,13 @@ def test_interop_with_reference_style_json(tmp_path):
     u, _ = entries["llama.wte"]
     assert u.is_hetero() and u.size() == 2 and u.get(0).zero
     spec, stages = strategy_from_config(p, model_key="llama")
-    assert isinstance(spec, HeteroSpec) and len(spec.pipelines) == 2
+    # two IDENTICAL tp1 pipelines collapse to homogeneous dp2
+    assert isinstance(spec, PipelineSpec)
+    assert (spec.pp, spec.dp, spec.tp) == (1, 2, 1)
+
+
+def test_homogeneous_dp_pipelines_collapse_to_pipeline_spec():
+    cfg = generate_ds_parallel_config([(2, 2), (2, 2)], num_layers=4)
+    spec, stages = strategy_from_config(cfg)
+    assert isinstance(spec, PipelineSpec)
+    assert (spec.pp, spec.dp, spec.tp) == (2, 2, 2)
