@@ -102,3 +102,13 @@ def test_elastic_driver_kill_one():
     assert p.returncode == 0, p.stderr
     assert "rank 2 exited rc=17" in p.stdout
     assert '"final_world": 2' in p.stdout
+
+
+def test_lobra_multi_tenant_lora():
+    p = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples", "lobra",
+                                      "train_multi_lora.py"),
+         "--steps", "40"],
+        capture_output=True, text=True, timeout=600)
+    assert p.returncode == 0, f"{p.stdout}\n{p.stderr}"
+    assert "LOBRA_OK" in p.stdout
