@@ -16,5 +16,6 @@ from .graph.ops.optim import SGD, Adam, Optimizer
 from .parallel.dstates import (DistributedStates, ds_dup, ds_partial,
                                ds_split)
 from .parallel.comm import CommBackend, comm_backend
+from .utils.profiler import MemorySnapshots, OpProfiler
 
 __version__ = "0.1.0"

@@ -229,6 +229,24 @@ def zeros_like(a: Tensor) -> Tensor:
 _AUTOCAST_STACK: List[torch.dtype] = []
 
 
+class recompute:
+    """with ht.recompute(): ... — ops built inside form one recompute
+    scope (reference context.py:223 hetu.recompute): their internal
+    activations are dropped after forward and recomputed in backward
+    (Graph.apply_recompute).  Scope indices auto-increment per graph."""
+
+    def __enter__(self):
+        g = _cg()
+        idx = getattr(g, "_rc_auto_idx", 0)
+        g._rc_auto_idx = idx + 1
+        self._scope = g.recompute_scope(idx)
+        self._scope.__enter__()
+        return self
+
+    def __exit__(self, *a):
+        self._scope.__exit__(*a)
+
+
 class autocast:
     """with ht.autocast(torch.bfloat16): matmul/linear/bmm/attention
     inputs are cast to the target dtype (fp32 params keep a cast edge, so

@@ -109,3 +109,32 @@ def test_recompute_llama_block_scopes():
     pa = _peak_live_bytes(ga2, [ha2["loss"], ha2["train_op"]])
     pb = _peak_live_bytes(gb2, [hb2["loss"], hb2["train_op"]])
     assert pb < pa, (pa, pb)
+
+
+def test_recompute_context_manager():
+    """ht.recompute() top-level context (reference hetu.recompute,
+    context.py:223) auto-indexes scopes and matches explicit scopes."""
+    import hetu_amd as H
+    torch.manual_seed(0)
+    g = DefineAndRunGraph("rc_ctx")
+    push_graph(g)
+    try:
+        x = ht.placeholder((16, 8), name="x")
+        tgt = ht.placeholder((16, 8), name="t")
+        cur = x
+        for i in range(2):
+            with H.recompute():
+                w = ht.variable(torch.randn(8, 8) * 0.3, name=f"w{i}")
+                cur = ht.tanh(ht.matmul(cur, w))
+        loss = ht.mse_loss(cur, tgt)
+        train = Adam(lr=1e-2).minimize(loss)   # applies recompute
+    finally:
+        pop_graph()
+    # scopes were tagged and backward uses clones
+    assert any("_rc_scope" in op.attrs for op in g.ops)
+    assert any(op.name.endswith("_rc") for op in g.ops)
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+    xd, td = torch.randn(16, 8), torch.randn(16, 8)
+    l0 = float(g.run([loss, train], {x: xd, tgt: td}, ctx=ctx)[0])
+    l1 = float(g.run([loss, train], {x: xd, tgt: td}, ctx=ctx)[0])
+    assert l1 < l0
