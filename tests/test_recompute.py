@@ -125,7 +125,9 @@ def test_recompute_context_manager():
         for i in range(2):
             with H.recompute():
                 w = ht.variable(torch.randn(8, 8) * 0.3, name=f"w{i}")
-                cur = ht.tanh(ht.matmul(cur, w))
+                # inner tanh output is scope-internal AND needed by the
+                # outer tanh's backward -> forces a recompute clone
+                cur = ht.tanh(ht.tanh(ht.matmul(cur, w)))
         loss = ht.mse_loss(cur, tgt)
         train = Adam(lr=1e-2).minimize(loss)   # applies recompute
     finally:
