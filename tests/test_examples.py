@@ -112,3 +112,12 @@ def test_lobra_multi_tenant_lora():
         capture_output=True, text=True, timeout=600)
     assert p.returncode == 0, f"{p.stdout}\n{p.stderr}"
     assert "LOBRA_OK" in p.stdout
+
+
+def test_efficiency_profile_attn_cpu():
+    p = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples", "efficiency",
+                                      "profile_attn.py"), "--allow-cpu"],
+        capture_output=True, text=True, timeout=300)
+    assert p.returncode == 0, f"{p.stdout}\n{p.stderr}"
+    assert "PROFILE_ATTN_OK" in p.stdout
