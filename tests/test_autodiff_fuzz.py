@@ -33,6 +33,15 @@ OPS = [
     ("triu", lambda t: ht.triu(t), torch.triu),
     ("roll", lambda t: ht.roll(t, 1, 0),
      lambda t: torch.roll(t, 1, 0)),
+    ("sub_min", lambda t: ht.sub(t, ht.reduce_min(t, dim=1, keepdim=True)),
+     lambda t: t - t.min(1, keepdim=True).values),
+    ("div_norm", lambda t: ht.div(t, ht.add(
+        ht.norm(t, p=2, dim=1, keepdim=True), 1.0)),
+     lambda t: t / (torch.linalg.vector_norm(t, 2, dim=1,
+                                             keepdim=True) + 1.0)),
+    ("bcast_mean", lambda t: ht.add(t, ht.broadcast_to(
+        ht.reduce_mean(t, dim=1, keepdim=True), (6, 6))),
+     lambda t: t + t.mean(1, keepdim=True).expand(6, 6)),
 ]
 
 
