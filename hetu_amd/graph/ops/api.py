@@ -210,7 +210,8 @@ def broadcast_to(a: Tensor, shape) -> Tensor:
 def group(*tensors) -> Tensor:
     """Control-dependency join: fetch the returned scalar to force every
     input to execute (reference group.cc)."""
-    return _cg().make_op(B.GroupOp(), list(tensors), {}).output()
+    from .optim import GroupOp
+    return _cg().make_op(GroupOp(), list(tensors), {}).output()
 
 
 def ones_like(a: Tensor) -> Tensor:

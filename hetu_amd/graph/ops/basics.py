@@ -126,28 +126,6 @@ class BroadcastToOp(OpInterface):
                       name="bcast_grad").output()]
 
 
-class GroupOp(OpInterface):
-    """Control-dependency join (reference group.cc): a zero scalar that
-    depends on every input — fetch it to force their execution in one
-    graph.run without materializing each value at the caller."""
-    type = "Group"
-
-    def infer_meta(self, attrs, inputs):
-        return [TensorMeta([], torch.float32)]
-
-    def deduce_states(self, op):
-        op.outputs[0].ds = None
-        op.outputs[0].device_group = (op.inputs[0].device_group
-                                      if op.inputs else None)
-
-    def compute(self, op, inputs, ctx):
-        dev = inputs[0].device if inputs else "cpu"
-        return [torch.zeros((), device=dev)]
-
-    def gradient(self, op, g):
-        return [None] * len(op.inputs)
-
-
 # ---------------------------------------------------------------------------
 # Elementwise arithmetic (broadcasting; grads reduce back to input shape)
 # ---------------------------------------------------------------------------
