@@ -243,13 +243,15 @@ class LlamaLMHeadModel(Module):
 def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
                                seq_len: int, dtype=torch.bfloat16,
                                lr: float = 1e-4,
-                               stage_layers=None):
+                               stage_layers=None, zero: bool = False):
     """Build THIS rank's pipeline-stage subgraph (see parallel.pipeline).
 
     Returns a StageModule whose graph exposes fwd (act_out|loss), bwd
     (dx + param grads via grad_in) and update (train_op fed by grad
     placeholders; dp-allreduce of accumulated grads happens here, once per
-    step) fetch sets."""
+    step) fetch sets.  zero=True shards the optimizer states over the
+    stage's dp group (ZeroAdamStepOp reduce-scatters the still-partial
+    accumulated grads itself — no bucket allreduce)."""
     from ..parallel.pipeline import StageModule
     B, S = micro_batch, seq_len
     sid = pspec.my_stage()
@@ -334,13 +336,14 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
         if not is_first:
             h["dx"] = grads[len(params)]
         # ---- update graph: grad placeholders -> (dp allreduce) -> Adam ---
-        from ..graph.ops.optim import (AdamStepOp, GroupOp,
+        from ..graph.ops.optim import (AdamStepOp, GroupOp, ZeroAdamStepOp,
                                        make_grad_buckets)
         from ..graph.ops.basics import _make
         from ..graph.ops.comm import make_comm
         grad_phs, updates = [], []
         opt_attrs = {"lr": lr, "beta1": 0.9, "beta2": 0.999, "eps": 1e-8,
                      "weight_decay": 0.0}
+        use_zero = zero and spec.dp > 1
         phs, pend = [], []
         for p, pg in zip(params, h["param_grads"]):
             gds = pg.ds if pg is not None else None
@@ -349,7 +352,7 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
                                 device_group=spec.device_group)
             grad_phs.append(ph)
             phs.append((p, ph))
-            if gds is not None and p.ds is not None \
+            if not use_zero and gds is not None and p.ds is not None \
                     and not gds.check_equal(p.ds) \
                     and gds.check_allreduce(p.ds):
                 pend.append((p, ph))
@@ -357,6 +360,13 @@ def build_llama_pipeline_stage(cfg: LlamaConfig, pspec, micro_batch: int,
             if pend else {}
         for p, ph in phs:
             gt = reduced.get(p.id, ph)
+            if use_zero:
+                # partial grads flow straight in: ZeroAdamStep does the
+                # reduce-scatter + local Adam + allgather itself
+                updates.append(_make(g, ZeroAdamStepOp(), [p, gt],
+                                     dict(opt_attrs),
+                                     name=f"zadam_{p.name}").output())
+                continue
             if gt is ph and ph.ds is not None and p.ds is not None \
                     and not ph.ds.check_equal(p.ds):
                 gt = make_comm(g, ph, p.ds, name=f"gred_{p.name}")

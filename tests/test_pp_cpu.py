@@ -32,7 +32,9 @@ cfg = LlamaConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
 pspec = PipelineSpec(pp=pp, dp=dp, tp=tp)
 comm = comm_backend(torch.device("cpu"))
 stage = build_llama_pipeline_stage(cfg, pspec, micro_batch=1, seq_len=16,
-                                   dtype=torch.float32, lr=1e-3)
+                                   dtype=torch.float32, lr=1e-3,
+                                   zero=bool(int(os.environ.get(
+                                       "HETU_TEST_ZERO", "0"))))
 runner = PipelineRunner(pspec, stage, torch.device("cpu"),
                         offload=bool(int(os.environ.get(
                             "HETU_TEST_OFFLOAD", "0"))),
@@ -266,3 +268,14 @@ def test_pp2_tp2_matches_single(single_losses):
     assert losses is not None
     assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
         (losses, single_losses)
+
+
+def test_pp2_dp2_zero_matches_single(single_losses):
+    """pp2 x dp2 with ZeRO optimizer-state sharding inside each stage's
+    dp group: trajectory must still reproduce the single run (losses are
+    half, as in test_pp2_dp2_matches_single)."""
+    losses = _launch(4, {"HETU_TEST_PP": "2", "HETU_TEST_ZERO": "1"},
+                     29593)
+    assert losses is not None
+    assert np.allclose([2 * v for v in losses], single_losses,
+                       rtol=5e-4, atol=2e-4), (losses, single_losses)
