@@ -131,3 +131,18 @@ def test_cp2_ring_attention_matches_single(single_losses):
     cp_losses = _launch(2, {"HETU_TEST_DP": "1", "HETU_TEST_CP": "2"}, 29536)
     assert np.allclose(cp_losses, single_losses, rtol=5e-4, atol=2e-4), \
         f"cp2 {cp_losses} vs single {single_losses}"
+
+
+def test_cp2_tp2_matches_single(single_losses):
+    """cp2 x tp2 (4 ranks): ring attention across cp groups whose members
+    are tensor-parallel (heads sharded) — both layouts must compose."""
+    losses = _launch(4, {"HETU_TEST_DP": "1", "HETU_TEST_CP": "2"}, 29538)
+    assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
+        f"cp2xtp2 {losses} vs single {single_losses}"
+
+
+def test_dp2_cp2_matches_single(single_losses):
+    """dp2 x cp2 (4 ranks): data-parallel replicas each running a cp ring."""
+    losses = _launch(4, {"HETU_TEST_DP": "2", "HETU_TEST_CP": "2"}, 29539)
+    assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
+        f"dp2xcp2 {losses} vs single {single_losses}"
