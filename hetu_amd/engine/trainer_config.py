@@ -30,13 +30,36 @@ class TrainingConfig:
     save_every: int = 0
     save_path: Optional[str] = None
     seed: int = 1234
+    # path to a layered JSON ds_parallel_config (reference generate_ds.py
+    # format): when set, dp/tp/pp/zero are read from the file instead
+    ds_parallel_config: Optional[str] = None
 
     @classmethod
     def from_yaml(cls, path: str) -> "TrainingConfig":
         with open(path) as f:
             raw = yaml.safe_load(f) or {}
         known = {f.name for f in dataclasses.fields(cls)}
-        return cls(**{k: v for k, v in raw.items() if k in known})
+        tc = cls(**{k: v for k, v in raw.items() if k in known})
+        if tc.ds_parallel_config:
+            tc.apply_ds_config(tc.ds_parallel_config)
+        return tc
+
+    def apply_ds_config(self, path: str) -> "TrainingConfig":
+        """Override the parallel strategy from a ds_parallel_config JSON
+        (homogeneous configs only; hetero worlds use HeteroSpec directly)."""
+        import json
+
+        from ..parallel.pipeline import PipelineSpec
+        from ..utils.ds_config import strategy_from_config
+        spec, _ = strategy_from_config(path)
+        if not isinstance(spec, PipelineSpec):
+            raise ValueError(
+                "ds_parallel_config is heterogeneous: drive it through "
+                "parallel.hetero.HeteroSpec (examples/malleus)")
+        self.pp, self.dp, self.tp = spec.pp, spec.dp, spec.tp
+        with open(path) as f:
+            self.zero = bool(json.load(f).get("zero", False))
+        return self
 
     def dtype(self):
         import torch

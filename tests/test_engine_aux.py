@@ -121,3 +121,19 @@ def test_torch_profiler_trace(tmp_path, monkeypatch):
     traces = glob.glob(str(tmp_path / "*.json")) + \
         glob.glob(str(tmp_path / "**" / "*.json"), recursive=True)
     assert traces, "no chrome trace written"
+
+
+def test_training_config_from_ds_parallel_json(tmp_path):
+    """A TrainingConfig yaml can point at a ds_parallel_config JSON; the
+    strategy (pp/dp/tp/zero) is read from the file."""
+    from hetu_amd.engine.trainer_config import TrainingConfig
+    from hetu_amd.utils.ds_config import (generate_ds_parallel_config,
+                                          write_ds_parallel_config)
+    dsp = str(tmp_path / "ds.json")
+    write_ds_parallel_config(
+        generate_ds_parallel_config([(2, 2), (2, 2)], num_layers=4), dsp)
+    yml = tmp_path / "run.yaml"
+    yml.write_text(f"model: gpt-tiny\nds_parallel_config: {dsp}\n")
+    tc = TrainingConfig.from_yaml(str(yml))
+    assert (tc.pp, tc.dp, tc.tp) == (2, 2, 2)
+    assert tc.zero is True
