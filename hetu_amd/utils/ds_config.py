@@ -235,6 +235,27 @@ def read_ds_parallel_config(src: Union[str, Dict]
     return out
 
 
+def recompute_layers_from_config(cfg: Union[str, Dict],
+                                 model_key: Optional[str] = None,
+                                 pipeline: int = 0) -> List[int]:
+    """Layer indices flagged for activation recompute in `pipeline`'s
+    column of the per-block recompute lists (reference generate_ds.py
+    per-block `recompute: [bool]*dp`)."""
+    if isinstance(cfg, str):
+        with open(cfg) as f:
+            cfg = json.load(f)
+    if model_key is None:
+        skip = {"input", "lm_head", "label", "zero", "devices"}
+        model_key = next(k for k, v in cfg.items()
+                         if k not in skip and isinstance(v, dict))
+    out = []
+    for blk in cfg[model_key]["blocks"].values():
+        flags = blk.get("recompute", [])
+        if pipeline < len(flags) and flags[pipeline]:
+            out.extend(blk["range"])
+    return sorted(out)
+
+
 def strategy_from_config(cfg: Union[str, Dict],
                          model_key: Optional[str] = None):
     """Recover the engine-level strategy from a config tree.

@@ -113,3 +113,10 @@ def test_homogeneous_dp_pipelines_collapse_to_pipeline_spec():
     spec, stages = strategy_from_config(cfg)
     assert isinstance(spec, PipelineSpec)
     assert (spec.pp, spec.dp, spec.tp) == (2, 2, 2)
+
+
+def test_recompute_layers_roundtrip():
+    from hetu_amd.utils.ds_config import recompute_layers_from_config
+    cfg = generate_ds_parallel_config([(2, 1)], num_layers=4,
+                                      recompute_layers=[1, 3])
+    assert recompute_layers_from_config(cfg) == [1, 3]
