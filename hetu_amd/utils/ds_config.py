@@ -235,6 +235,20 @@ def read_ds_parallel_config(src: Union[str, Dict]
     return out
 
 
+def strategy_to_ds_config(strategy, num_layers: int,
+                          model_key: str = "gpt",
+                          gpus_per_node: int = GPUS_PER_NODE) -> Dict:
+    """Expand a Galvatron search result (galvatron.cost_model.Strategy:
+    dp/tp/pp/zero/recompute_layers) into the layered JSON tree — the
+    auto-parallel output becomes a runnable, inspectable config file
+    (reference flow: search -> generate_ds -> train)."""
+    rc = list(range(int(getattr(strategy, "recompute_layers", 0) or 0)))
+    return generate_ds_parallel_config(
+        [(strategy.tp, strategy.pp)] * strategy.dp, num_layers,
+        zero=bool(getattr(strategy, "zero", False)), model_key=model_key,
+        recompute_layers=rc, gpus_per_node=gpus_per_node)
+
+
 def recompute_layers_from_config(cfg: Union[str, Dict],
                                  model_key: Optional[str] = None,
                                  pipeline: int = 0) -> List[int]:

@@ -120,3 +120,17 @@ def test_recompute_layers_roundtrip():
     cfg = generate_ds_parallel_config([(2, 1)], num_layers=4,
                                       recompute_layers=[1, 3])
     assert recompute_layers_from_config(cfg) == [1, 3]
+
+
+def test_strategy_export_roundtrip():
+    """Galvatron Strategy -> ds_parallel_config -> strategy recovery."""
+    from hetu_amd.galvatron.cost_model import Strategy
+    from hetu_amd.utils.ds_config import strategy_to_ds_config
+    s = Strategy(dp=2, tp=2, pp=2, zero=True, recompute_layers=2)
+    cfg = strategy_to_ds_config(s, num_layers=8)
+    spec, _ = strategy_from_config(cfg)
+    assert isinstance(spec, PipelineSpec)
+    assert (spec.pp, spec.dp, spec.tp) == (2, 2, 2)
+    assert cfg["zero"] is True
+    from hetu_amd.utils.ds_config import recompute_layers_from_config
+    assert recompute_layers_from_config(cfg) == [0, 1]
