@@ -9,6 +9,7 @@ strategies with one batched p2p when the bucket changes
 Run: python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
        --master-addr 127.0.0.1 examples/hotspa/hot_switch_train.py
 """
+import argparse
 import os
 import sys
 
@@ -28,9 +29,20 @@ STRATEGIES = {"short": dict(dp=2, tp=1),         # short seqs: data parallel
 
 
 def main():
+    ap = argparse.ArgumentParser()
+    # BASELINE config 5 is llama-13b on 8 GPUs; llama-tiny is the CPU/CI
+    # default so the same driver runs everywhere
+    ap.add_argument("--model", default="llama-tiny",
+                    choices=list(LLAMA_CONFIGS))
+    ap.add_argument("--dtype", default=None,
+                    help="default: bf16 on GPU, fp32 on CPU")
+    args = ap.parse_args()
     comm = comm_backend()
     device = comm.device
-    mcfg = LLAMA_CONFIGS["llama-tiny"]
+    mcfg = LLAMA_CONFIGS[args.model]
+    dt = {"bf16": torch.bfloat16, "fp32": torch.float32,
+          None: torch.bfloat16 if device.type == "cuda"
+          else torch.float32}[args.dtype]
 
     shapes = {}
 
@@ -39,11 +51,15 @@ def main():
         S = BUCKETS[key]
         B = 8 // spec.dp
         shapes[key] = (B, S)
-        g, h = build_llama_train_graph(mcfg, B, S, dtype=torch.float32,
+        g, h = build_llama_train_graph(mcfg, B, S, dtype=dt,
                                        lr=1e-4, spec=spec)
         return g, h
 
     trainer = HotSwitchTrainer(build, device)
+    ws = comm.world_size
+    for k in STRATEGIES:                      # scale strategies to world
+        STRATEGIES[k] = {kk: (ws if v == 2 and ws > 2 else v)
+                         for kk, v in STRATEGIES[k].items()}
     torch.manual_seed(7 + comm.rank)
     seq_lens = [400, 380, 1800, 2000, 300, 1900]   # mixed workload
     for sl in seq_lens:

@@ -138,7 +138,14 @@ class HotSwitchTrainer:
 
 
 def bucket_for_seq_len(seq_len: int, buckets) -> str:
-    """Smallest bucket holding seq_len (hotspa seq-len bucket dispatch)."""
+    """Smallest bucket holding seq_len (hotspa seq-len bucket dispatch).
+    `buckets` is either a list of int ceilings (key = str(ceiling)) or a
+    {name: ceiling} dict (key = name)."""
+    if isinstance(buckets, dict):
+        for k, b in sorted(buckets.items(), key=lambda kv: kv[1]):
+            if seq_len <= b:
+                return k
+        return max(buckets, key=buckets.get)
     for b in sorted(buckets):
         if seq_len <= b:
             return str(b)
