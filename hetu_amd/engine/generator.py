@@ -22,6 +22,31 @@ from ..models.llama import LlamaConfig, rope_tables
 from ..ops import functional as F
 
 
+def sample_next(logits: torch.Tensor, temperature: float = 0.0,
+                top_k: int = 0, top_p: float = 0.0,
+                gen: Optional[torch.Generator] = None) -> torch.Tensor:
+    """Next-token sampling: greedy (temperature<=0), temperature,
+    top-k, and nucleus (top-p) filters compose.  CPU multinomial with an
+    explicit generator keeps runs reproducible across devices."""
+    if temperature <= 0:
+        return logits.argmax(-1)
+    lg = logits.float() / temperature
+    if top_k:
+        kth = lg.topk(top_k, dim=-1).values[:, -1:]
+        lg = lg.masked_fill(lg < kth, float("-inf"))
+    if top_p and 0.0 < top_p < 1.0:
+        srt, idx = lg.sort(dim=-1, descending=True)
+        cum = torch.softmax(srt, dim=-1).cumsum(dim=-1)
+        # keep the smallest prefix with cumulative prob >= top_p (the
+        # first token always survives)
+        drop_sorted = cum - torch.softmax(srt, dim=-1) >= top_p
+        drop = drop_sorted.scatter(-1, idx, drop_sorted)
+        lg = lg.masked_fill(drop, float("-inf"))
+    probs = torch.softmax(lg, dim=-1).cpu()
+    return torch.multinomial(probs, 1, generator=gen).squeeze(-1) \
+        .to(logits.device)
+
+
 class LlamaKVCache:
     def __init__(self, cfg: LlamaConfig, batch: int, max_len: int,
                  device, dtype):
@@ -134,7 +159,7 @@ class LlamaGenerator:
     @torch.no_grad()
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
                  temperature: float = 0.0, top_k: int = 0,
-                 eos_id: Optional[int] = None,
+                 top_p: float = 0.0, eos_id: Optional[int] = None,
                  seed: int = 0) -> torch.Tensor:
         """input_ids [B, S] -> [B, S + new]; temperature 0 = greedy."""
         ids = input_ids.to(self.device)
@@ -147,16 +172,7 @@ class LlamaGenerator:
         out = [ids]
         alive = torch.ones(B, dtype=torch.bool)
         for _ in range(max_new_tokens):
-            if temperature <= 0:
-                nxt = logits.argmax(-1)
-            else:
-                lg = logits.float() / temperature
-                if top_k:
-                    kth = lg.topk(top_k, dim=-1).values[:, -1:]
-                    lg = lg.masked_fill(lg < kth, float("-inf"))
-                probs = torch.softmax(lg, dim=-1).cpu()
-                nxt = torch.multinomial(probs, 1, generator=gen) \
-                    .squeeze(-1).to(self.device)
+            nxt = sample_next(logits, temperature, top_k, top_p, gen)
             out.append(nxt.unsqueeze(1))
             if eos_id is not None:
                 alive &= (nxt.cpu() != eos_id)
@@ -234,7 +250,7 @@ class GPTGenerator:
     @torch.no_grad()
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
                  temperature: float = 0.0, top_k: int = 0,
-                 seed: int = 0) -> torch.Tensor:
+                 top_p: float = 0.0, seed: int = 0) -> torch.Tensor:
         cfg = self.cfg
         ids = input_ids.to(self.device)
         B, S = ids.shape
@@ -248,16 +264,7 @@ class GPTGenerator:
         out = [ids]
         pos = S
         for _ in range(max_new_tokens):
-            if temperature <= 0:
-                nxt = logits.argmax(-1)
-            else:
-                lg = logits.float() / temperature
-                if top_k:
-                    kth = lg.topk(top_k, dim=-1).values[:, -1:]
-                    lg = lg.masked_fill(lg < kth, float("-inf"))
-                nxt = torch.multinomial(torch.softmax(lg, -1).cpu(), 1,
-                                        generator=gen).squeeze(-1) \
-                    .to(self.device)
+            nxt = sample_next(logits, temperature, top_k, top_p, gen)
             out.append(nxt.unsqueeze(1))
             logits = self._forward(nxt.unsqueeze(1), kc, vc, pos)
             pos += 1

@@ -92,3 +92,27 @@ def test_gpt_incremental_matches_full_forward(tmp_path):
     vc = T.zeros_like(kc)
     logits_full = gen._forward(out[:, :15], kc, vc, 0)
     assert T.equal(logits_full.argmax(-1), out[:, 15])
+
+
+def test_sample_next_filters():
+    """sample_next: greedy at temperature 0; a tight nucleus keeps only
+    the argmax; top-k keeps the k best; sampling is seed-reproducible."""
+    import torch
+    from hetu_amd.engine.generator import sample_next
+    logits = torch.tensor([[2.0, 1.0, 0.5, -1.0],
+                           [0.0, 3.0, 2.9, -2.0]])
+    assert sample_next(logits).tolist() == [0, 1]
+    # top_p tiny -> nucleus is just the argmax even at high temperature
+    g = torch.Generator().manual_seed(0)
+    out = sample_next(logits, temperature=5.0, top_p=1e-6, gen=g)
+    assert out.tolist() == [0, 1]
+    # top_k=1 behaves identically
+    g = torch.Generator().manual_seed(0)
+    assert sample_next(logits, temperature=5.0, top_k=1,
+                       gen=g).tolist() == [0, 1]
+    # reproducible with the same seed
+    a = sample_next(logits, 1.0, 0, 0.9,
+                    torch.Generator().manual_seed(7))
+    b = sample_next(logits, 1.0, 0, 0.9,
+                    torch.Generator().manual_seed(7))
+    assert torch.equal(a, b)
