@@ -74,7 +74,8 @@ def _route(probs: torch.Tensor, E: int, C: int, K: int):
         pos_fl.scatter_(0, flat, toks)
         slot_of[:, k] = torch.where(keep, flat,
                                     torch.full_like(flat, -1))
-        base = base + onehot.sum(0).clamp(max=C)             if False else (pos_fl[:E * C].reshape(E, C) >= 0).sum(-1)
+        # recount filled slots per expert (robust to same-slot rewrites)
+        base = (pos_fl[:E * C].reshape(E, C) >= 0).sum(-1)
     pos = pos_fl[:E * C].reshape(E, C)
     wk = torch.where(slot_of >= 0, w, torch.zeros_like(w))
     return pos, slot_of, idx, wk
