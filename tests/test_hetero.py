@@ -74,9 +74,19 @@ from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
 from hetu_amd.engine.runner import prepare_run_context
 comm = comm_backend()
 rank = comm.rank
-specA = ParallelSpec(dp=1, tp=2, device_group=[0, 1])
-specB = ParallelSpec(dp=1, tp=1, device_group=[2])
-hs = HeteroSpec(pipelines=[specA, specB], weights=[2/3, 1/3])
+if os.environ.get("HETU_HETERO_FROM_CONFIG", "0") == "1":
+    # build the hetero world from a generated ds_parallel_config file
+    # (reference flow: strategy -> generate_ds -> config2ds -> run)
+    from hetu_amd.utils.ds_config import (generate_ds_parallel_config,
+                                          strategy_from_config)
+    cfg_js = generate_ds_parallel_config([(2, 1), (1, 1)], num_layers=2)
+    hs, _ = strategy_from_config(cfg_js)
+    assert isinstance(hs, HeteroSpec)
+    hs.weights = [2/3, 1/3]            # Malleus planner output
+else:
+    specA = ParallelSpec(dp=1, tp=2, device_group=[0, 1])
+    specB = ParallelSpec(dp=1, tp=1, device_group=[2])
+    hs = HeteroSpec(pipelines=[specA, specB], weights=[2/3, 1/3])
 pi = hs.my_pipeline()
 my_spec = hs.pipelines[pi]
 mb = hs.micro_batches(3)[pi]          # A: 2 rows, B: 1 row
@@ -183,3 +193,35 @@ def test_hetero_tp2_tp1_exact_parity():
     # the per-step loss parity above already pins the update semantics)
     for k, v in spar.items():
         assert abs(hpar[k] - v) < 5e-4, (k, hpar[k], v)
+
+
+def test_hetero_from_ds_config_parity():
+    """Same 3-rank hetero parity, but the HeteroSpec comes from a
+    generated ds_parallel_config JSON (config -> spec -> training)."""
+    env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29689", "GLOO_SOCKET_IFNAME": "lo",
+            "HETU_HETERO_FROM_CONFIG": "1"}
+    procs = []
+    for r in range(3):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="3", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, "-c", HETERO_WORKER],
+                                      env=env, stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    single = subprocess.run([sys.executable, "-c", SINGLE_WORKER],
+                            env={**os.environ, "HETU_REPO": REPO},
+                            capture_output=True, text=True, timeout=300)
+    assert single.returncode == 0, single.stderr
+    sl = json.loads(single.stdout.split("SLOSS:")[1].splitlines()[0])
+    hlosses = {}
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=300)
+        assert p.returncode in (0, -6), f"rank {r}: {out}\n{err}"
+        for line in out.splitlines():
+            if line.startswith("HLOSS:"):
+                _, rk, ls = line.split(":", 2)
+                hlosses[int(rk)] = json.loads(ls)
+    import numpy as np
+    # pipeline losses are weighted local means; weighted sum == single
+    comb = [2/3 * a + 1/3 * b
+            for a, b in zip(hlosses[0], hlosses[2])]
+    assert np.allclose(comb, sl, rtol=5e-4, atol=5e-4), (comb, sl)
