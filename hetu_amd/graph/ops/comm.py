@@ -26,28 +26,86 @@ from ..tensor import Tensor, TensorMeta
 from .basics import _g, _make
 
 
+def _stable_dims_match(src: DistributedStates,
+                       dst: DistributedStates) -> bool:
+    """Per-device shard indices must agree on every tensor dim whose split
+    count is unchanged.  The count-based predicates (check_*) ignore
+    `order`, but order encodes the device->shard mapping: e.g.
+    {0:2,1:2}[0,1] -> {0:2,1:2}[1,0] has equal counts yet every device
+    owns a different shard — that transition needs the generic path."""
+    dims = [d for d in set(list(src.states) + list(dst.states))
+            if d >= 0 and src.get_dim(d) == dst.get_dim(d)
+            and src.get_dim(d) > 1]
+    if not dims:
+        return True
+    for i in range(src.device_num):
+        si = src.map_device_to_state_index(i)
+        di = dst.map_device_to_state_index(i)
+        if any(si.get(d, 0) != di.get(d, 0) for d in dims):
+            return False
+    return True
+
+
+def _slice_contained(src: DistributedStates,
+                     dst: DistributedStates) -> bool:
+    """For the no-comm slice kind: each device's dst shard must lie inside
+    its src shard on every refined dim."""
+    for i in range(src.device_num):
+        si = src.map_device_to_state_index(i)
+        di = dst.map_device_to_state_index(i)
+        for d in dst.split_dims():
+            ns, nd = src.get_dim(d), dst.get_dim(d)
+            if nd > ns:
+                r = nd // ns
+                if not (si.get(d, 0) * r <= di.get(d, 0)
+                        < (si.get(d, 0) + 1) * r):
+                    return False
+    return True
+
+
+def _rs_positions_ok(src: DistributedStates, dst: DistributedStates,
+                     d: int) -> bool:
+    """reduce_scatter_tensor hands chunk i to group rank i: each device's
+    dst split index must equal its position in its partial group."""
+    for grp in src.group_devices_along(-2):
+        for pos, i in enumerate(grp):
+            if dst.map_device_to_state_index(i).get(d, 0) != pos:
+                return False
+    return True
+
+
 def deduce_comm_kind(src: DistributedStates, dst: DistributedStates):
     """Returns (kind, info) where kind in {identity, slice, allreduce,
-    allgather, reducescatter, zeropad}."""
+    allgather, reducescatter, zeropad, generic}.  Every fast kind also
+    verifies the per-device placement it assumes (order-aware); when the
+    counts fit but the device->shard mapping does not, the transition
+    falls through to the generic gather+reslice path, which is correct
+    for arbitrary layout pairs."""
     if src.check_equal(dst):
         return "identity", None
-    if src.check_allreduce(dst):
+    stable = _stable_dims_match(src, dst)
+    if src.check_allreduce(dst) and stable:
         return "allreduce", None
     # allgather: some split dim in src becomes dup in dst
     for d in src.split_dims():
-        if src.get_dim(d) > dst.get_dim(d) and src.check_allgather(dst, d):
+        if src.get_dim(d) > dst.get_dim(d) and src.check_allgather(dst, d) \
+                and stable:
             return "allgather", d
     # reduce-scatter: partial becomes a split dim
     for d in dst.split_dims():
-        if dst.get_dim(d) > src.get_dim(d) and src.check_reducescatter(dst, d):
+        if dst.get_dim(d) > src.get_dim(d) \
+                and src.check_reducescatter(dst, d) and stable \
+                and _rs_positions_ok(src, dst, d):
             return "reducescatter", d
     # scatter / split: dup becomes split — keep local shard, no comm
-    if src.check_split(dst) or any(
-            src.check_scatter(dst, d) for d in dst.split_dims()):
+    if (src.check_split(dst) or any(
+            src.check_scatter(dst, d) for d in dst.split_dims())) \
+            and stable and _slice_contained(src, dst):
         return "slice", None
     # adjoint of slice: split dim in src becomes partial in dst
     for d in src.split_dims():
-        if src.get_dim(d) > dst.get_dim(d) and src._same_but(dst, d, -2):
+        if src.get_dim(d) > dst.get_dim(d) and src._same_but(dst, d, -2) \
+                and stable:
             return "zeropad", d
     return "generic", None
 

@@ -275,3 +275,76 @@ def test_rooted_collectives_two_ranks():
         out, err = p.communicate(timeout=300)
         ok = (p.returncode in (0, -6)) and "ROOTOK" in out
         assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
+
+
+RESHARD_FUZZ_WORKER = r"""
+import itertools, os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+from hetu_amd.parallel.comm import comm_backend
+from hetu_amd.parallel.dstates import DistributedStates
+from hetu_amd.graph.graph import DefineAndRunGraph, push_graph, pop_graph
+from hetu_amd.graph.ops import api as ht
+from hetu_amd.engine.runner import prepare_run_context
+comm = comm_backend()
+rank, ws = comm.rank, comm.world_size
+assert ws == 4
+dg = tuple(range(4))
+D = DistributedStates
+LAYOUTS = {
+    "dup":  D(4, {-1: 4}, [-1]),
+    "s0":   D(4, {0: 4}, [0]),
+    "s1":   D(4, {1: 4}, [1]),
+    "s0d":  D(4, {0: 2, -1: 2}, [0, -1]),
+    "ds0":  D(4, {0: 2, -1: 2}, [-1, 0]),
+    "s01":  D(4, {0: 2, 1: 2}, [0, 1]),
+    "s10":  D(4, {0: 2, 1: 2}, [1, 0]),
+    "part": D(4, {-2: 4}, [-2]),
+    "p2s0": D(4, {-2: 2, 0: 2}, [-2, 0]),
+    "s0p2": D(4, {-2: 2, 0: 2}, [0, -2]),
+}
+DUP = LAYOUTS["dup"]
+glob = torch.arange(64.0).reshape(8, 8)
+my_index = rank
+bad = []
+for (sn, src), (dn, dst) in itertools.product(LAYOUTS.items(),
+                                              LAYOUTS.items()):
+    if dn in ("part", "p2s0", "s0p2"):
+        continue            # partial DST is adjoint-only; covered via src
+    g = DefineAndRunGraph(f"rf_{sn}_{dn}"); push_graph(g)
+    try:
+        x = ht.placeholder(tuple(src.local_shape((8, 8))), name="x",
+                           ds=src, device_group=dg)
+        y = ht.comm(x, dst)
+        z = y if dst.check_equal(DUP) else ht.comm(y, DUP)
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"))
+    feed = glob[src.local_slice((8, 8), my_index)].clone()
+    feed = feed / src.partial          # partial shares sum to the value
+    (zl,) = g.run([z], {x: feed}, ctx=ctx)
+    if not torch.allclose(zl, glob, atol=1e-5):
+        bad.append((sn, dn, float((zl - glob).abs().max())))
+print("RESHARD_BAD:" + repr(bad))
+"""
+
+
+def test_reshard_fuzz_four_ranks():
+    """Every (src, dst) DS-transition pair over 10 layouts on 4 ranks:
+    comm to dst then back to dup must reconstruct the global tensor
+    (exercises allreduce/allgather/RS/slice/zeropad AND every generic
+    fallback the 8-GPU strategies could hit)."""
+    import subprocess
+    import sys
+    procs = []
+    env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29641", "GLOO_SOCKET_IFNAME": "lo"}
+    for r in range(4):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="4", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, "-c",
+                                       RESHARD_FUZZ_WORKER], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=600)
+        ok = (p.returncode in (0, -6)) and "RESHARD_BAD:[]" in out
+        assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"

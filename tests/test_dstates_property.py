@@ -87,7 +87,10 @@ def test_group_devices_along_partitions_devices(ds):
 def test_comm_kind_deduction_consistency(ds, dim, move):
     """deduce_comm_kind must recognize the canonical transitions built
     from any source layout (reference Communication.h DoDeduceStates)."""
-    from hetu_amd.graph.ops.comm import deduce_comm_kind
+    from hetu_amd.graph.ops.comm import (_rs_positions_ok,
+                                         _slice_contained,
+                                         _stable_dims_match,
+                                         deduce_comm_kind)
     n = ds.device_num
     states = dict(ds.states)
     order = list(ds.order)
@@ -115,7 +118,12 @@ def test_comm_kind_deduction_consistency(ds, dim, move):
                 dedup.append(d)
         dst = mk(states, dedup)
         kind, d = deduce_comm_kind(ds, dst)
-        assert kind == "allgather" and d == dim, (ds, dst, kind)
+        if _stable_dims_match(ds, dst):
+            assert kind == "allgather" and d == dim, (ds, dst, kind)
+        else:
+            # order change moved another dim's device mapping: the fast
+            # kind would place shards wrongly — generic is required
+            assert kind == "generic", (ds, dst, kind)
     elif move == "reducescatter":
         if -2 in states or ds.dup <= 1 or dim in states:
             return
@@ -128,7 +136,10 @@ def test_comm_kind_deduction_consistency(ds, dim, move):
         dst_states[dim] = k
         dst = mk(dst_states, [d for d in order if d != -1] + [dim])
         kind, d = deduce_comm_kind(src, dst)
-        assert kind == "reducescatter" and d == dim, (src, dst, kind)
+        if _stable_dims_match(src, dst) and _rs_positions_ok(src, dst, dim):
+            assert kind == "reducescatter" and d == dim, (src, dst, kind)
+        else:
+            assert kind == "generic", (src, dst, kind)
     else:  # slice_dup: dup splits into a new dim
         if ds.dup <= 1 or dim in states:
             return
@@ -137,7 +148,10 @@ def test_comm_kind_deduction_consistency(ds, dim, move):
         dst_states[dim] = k
         dst = mk(dst_states, [d for d in order if d != -1] + [dim])
         kind, _ = deduce_comm_kind(ds, dst)
-        assert kind == "slice", (ds, dst, kind)
+        if _stable_dims_match(ds, dst) and _slice_contained(ds, dst):
+            assert kind == "slice", (ds, dst, kind)
+        else:
+            assert kind == "generic", (ds, dst, kind)
 
 
 @settings(max_examples=100, deadline=None)
