@@ -134,3 +134,17 @@ def test_strategy_export_roundtrip():
     assert cfg["zero"] is True
     from hetu_amd.utils.ds_config import recompute_layers_from_config
     assert recompute_layers_from_config(cfg) == [0, 1]
+
+
+def test_convert_strategy_tp_spanning_nodes():
+    """tp=16 spans two whole nodes (allowed when tp % gpus_per_node == 0);
+    remaining pipelines pack into what's left."""
+    ltg, gpu_pos = convert_strategy([(16, 1), (8, 1)], 24, 2,
+                                    gpus_per_node=8)
+    g16 = [g for g in ltg[0] if len(g) == 16][0]
+    assert {r // 8 for r in g16} == {0, 1} or \
+        {r // 8 for r in g16} == {1, 2} or \
+        {r // 8 for r in g16} == {0, 2}
+    g8 = [g for g in ltg[0] if len(g) == 8][0]
+    assert len({r // 8 for r in g8}) == 1
+    assert len(gpu_pos) == 24
