@@ -32,7 +32,9 @@ GPUS_PER_NODE = 8
 # strategy placement: tp_pp_list -> per-layer tp groups
 # --------------------------------------------------------------------------
 def convert_strategy(tp_pp_list: Sequence[Tuple[int, int]], ngpus: int,
-                     layers: int, gpus_per_node: int = GPUS_PER_NODE
+                     layers: int, gpus_per_node: int = GPUS_PER_NODE,
+                     hetero_layers: Optional[Sequence[Sequence[int]]] = None,
+                     rank_map: Optional[Dict[int, int]] = None
                      ) -> Tuple[List[List[List[int]]], Dict[int, Tuple[int, int]]]:
     """Place K=(len list) pipelines onto `ngpus` devices.
 
@@ -43,6 +45,13 @@ def convert_strategy(tp_pp_list: Sequence[Tuple[int, int]], ngpus: int,
     Placement rule: each stage's tp group must sit inside one node
     (tp<=gpus_per_node, or tp a multiple of whole nodes); stages are
     packed best-fit-decreasing so big tp groups claim fresh nodes first.
+
+    hetero_layers (reference generate_gpt_3d_config hetero_layers):
+    pipeline k's per-stage layer COUNTS (sum == layers) for non-uniform
+    splits — the Malleus hetero-pp partitioner's output.  rank_map
+    (reference rank_to_device_mapping/unused_rank): logical->physical
+    rank relabeling applied to every group (physical ranks absent from
+    the map's values are simply unused).
     """
     assert ngpus >= sum(tp * pp for tp, pp in tp_pp_list), \
         f"need {sum(tp * pp for tp, pp in tp_pp_list)} gpus, have {ngpus}"
@@ -79,19 +88,34 @@ def convert_strategy(tp_pp_list: Sequence[Tuple[int, int]], ngpus: int,
                 grp, free[n] = free[n][:tp], free[n][tp:]
             groups.append(grp)
         stage_groups[k] = groups
+    if hetero_layers is not None:
+        assert len(hetero_layers) == len(tp_pp_list)
+        for k, (tp, pp) in enumerate(tp_pp_list):
+            assert len(hetero_layers[k]) == pp \
+                and sum(hetero_layers[k]) == layers, \
+                f"pipeline {k}: stage layer counts must sum to {layers}"
     for k, (tp, pp) in enumerate(tp_pp_list):
         groups = stage_groups[k]
         for s, grp in enumerate(groups):
             for r in grp:
                 gpu_pos[r] = (k, s)
-        per = layers // pp
-        rem = layers % pp
+        if hetero_layers is not None:
+            counts = list(hetero_layers[k])
+        else:
+            per = layers // pp
+            rem = layers % pp
+            counts = [per + (1 if s < rem else 0) for s in range(pp)]
         lo = 0
         for s in range(pp):
-            hi = lo + per + (1 if s < rem else 0)
+            hi = lo + counts[s]
             for l in range(lo, hi):
                 layers_tp_groups[l].append(groups[s])
             lo = hi
+    if rank_map is not None:
+        layers_tp_groups = [[[rank_map.get(r, r) for r in grp]
+                             for grp in layer]
+                            for layer in layers_tp_groups]
+        gpu_pos = {rank_map.get(r, r): pos for r, pos in gpu_pos.items()}
     return layers_tp_groups, gpu_pos
 
 
@@ -109,21 +133,28 @@ def generate_ds_parallel_config(
         tp_pp_list: Sequence[Tuple[int, int]], num_layers: int,
         ngpus: Optional[int] = None, zero: bool = True,
         model_key: str = "gpt", recompute_layers: Sequence[int] = (),
-        gpus_per_node: int = GPUS_PER_NODE) -> Dict:
+        gpus_per_node: int = GPUS_PER_NODE,
+        hetero_layers: Optional[Sequence[Sequence[int]]] = None,
+        rank_map: Optional[Dict[int, int]] = None) -> Dict:
     """Expand a (possibly heterogeneous) tp_pp_list into the layered JSON
     tree (reference generate_ds.py:253 layout: input / <model>{wte, wpe,
-    blocks{...}, norm_final} / lm_head / label)."""
+    blocks{...}, norm_final} / lm_head / label).  hetero_layers/rank_map:
+    non-uniform per-pipeline stage splits and rank relabeling (reference
+    parallel_config.py generate_gpt_3d_config)."""
     if ngpus is None:
         ngpus = sum(tp * pp for tp, pp in tp_pp_list)
-    ltg, _ = convert_strategy(tp_pp_list, ngpus, num_layers, gpus_per_node)
+    ltg, _ = convert_strategy(tp_pp_list, ngpus, num_layers, gpus_per_node,
+                              hetero_layers=hetero_layers,
+                              rank_map=rank_map)
     dp = len(tp_pp_list)
     if dp == 1:
         zero = False
     dp_union = [dp] * dp
     tp_u = [[len(g) for g in layer] for layer in ltg]    # [layer][pipe]
+    used = sorted({r for layer in ltg for grp in layer for r in grp})
     cfg: Dict = {
         "zero": zero,
-        "devices": list(range(ngpus)),
+        "devices": used,
         "input": _entry({0: dp_union}, tp_u[0], ltg[0], "placeholder"),
         model_key: {
             "wte": _entry({0: tp_u[0]}, dp_union, ltg[0], "variable"),

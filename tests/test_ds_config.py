@@ -148,3 +148,30 @@ def test_convert_strategy_tp_spanning_nodes():
     g8 = [g for g in ltg[0] if len(g) == 8][0]
     assert len({r // 8 for r in g8}) == 1
     assert len(gpu_pos) == 24
+
+
+def test_hetero_layers_and_rank_map():
+    """Non-uniform per-pipeline stage splits + rank relabeling
+    (reference parallel_config.py generate_gpt_3d_config hetero_layers /
+    rank_to_device_mapping)."""
+    ltg, gpu_pos = convert_strategy(
+        [(2, 2), (1, 2)], 6, 4,
+        hetero_layers=[[3, 1], [2, 2]],
+        rank_map={0: 10, 1: 11})
+    # pipeline 0 (tp2): stage 0 owns layers 0-2, stage 1 owns layer 3
+    p0_stage_of_layer = [ltg[l][0] for l in range(4)]
+    assert p0_stage_of_layer[0] == p0_stage_of_layer[2]
+    assert p0_stage_of_layer[3] != p0_stage_of_layer[0]
+    # pipeline 1 (tp1): 2+2 split
+    p1 = [ltg[l][1] for l in range(4)]
+    assert p1[0] == p1[1] and p1[2] == p1[3] and p1[1] != p1[2]
+    # ranks 0/1 relabeled to 10/11 everywhere
+    all_ranks = {r for layer in ltg for grp in layer for r in grp}
+    assert 0 not in all_ranks and 1 not in all_ranks
+    assert {10, 11} <= all_ranks
+    assert gpu_pos[10][0] in (0, 1)
+    # generated config carries only the used (relabeled) devices
+    cfg = generate_ds_parallel_config([(2, 2), (1, 2)], num_layers=4,
+                                      hetero_layers=[[3, 1], [2, 2]],
+                                      rank_map={0: 10, 1: 11})
+    assert 0 not in cfg["devices"] and 10 in cfg["devices"]
