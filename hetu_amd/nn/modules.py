@@ -105,3 +105,133 @@ class Tanh(Module):
 class Sigmoid(Module):
     def forward(self, x):
         return ht.sigmoid(x)
+
+
+class Conv2d(Module):
+    """(reference nn/modules/conv.py)"""
+
+    def __init__(self, in_channels: int, out_channels: int, kernel_size,
+                 stride=1, padding=0, dilation=1, groups: int = 1,
+                 bias: bool = True, dtype=torch.float32,
+                 name: str = "conv"):
+        super().__init__()
+        ks = (kernel_size, kernel_size) if isinstance(kernel_size, int) \
+            else tuple(kernel_size)
+        w = init.he_normal((out_channels, in_channels // groups, *ks),
+                           dtype=dtype, name=f"{name}.weight")
+        self.weight = ht.variable(w, name=f"{name}.weight")
+        if bias:
+            self.bias = ht.variable(init.zeros((out_channels,), dtype),
+                                    name=f"{name}.bias")
+        else:
+            self.register_parameter("bias", None)
+        self.stride, self.padding = stride, padding
+        self.dilation, self.groups = dilation, groups
+
+    def forward(self, x):
+        return ht.conv2d(x, self.weight, self.bias, self.stride,
+                         self.padding, self.dilation, self.groups)
+
+
+class MaxPool2d(Module):
+    def __init__(self, kernel_size, stride=None, padding=0):
+        super().__init__()
+        self.k, self.s, self.p = kernel_size, stride, padding
+
+    def forward(self, x):
+        return ht.max_pool2d(x, self.k, self.s, self.p)
+
+
+class AvgPool2d(Module):
+    def __init__(self, kernel_size, stride=None, padding=0):
+        super().__init__()
+        self.k, self.s, self.p = kernel_size, stride, padding
+
+    def forward(self, x):
+        return ht.avg_pool2d(x, self.k, self.s, self.p)
+
+
+class BatchNorm2d(Module):
+    """(reference nn/modules/batchnorm.py)"""
+
+    def __init__(self, num_features: int, eps: float = 1e-5,
+                 dtype=torch.float32, name: str = "bn"):
+        super().__init__()
+        self.eps = eps
+        self.weight = ht.variable(init.ones((num_features,), dtype),
+                                  name=f"{name}.weight")
+        self.bias = ht.variable(init.zeros((num_features,), dtype),
+                                name=f"{name}.bias")
+
+    def forward(self, x):
+        return ht.batch_norm(x, self.weight, self.bias, self.eps)
+
+
+class InstanceNorm2d(Module):
+    def __init__(self, eps: float = 1e-5):
+        super().__init__()
+        self.eps = eps
+
+    def forward(self, x):
+        return ht.instance_norm(x, self.eps)
+
+
+class ZeroPad2d(Module):
+    """(reference nn/modules/padding.py ZeroPad2d)"""
+
+    def __init__(self, padding):
+        super().__init__()
+        p = (padding,) * 4 if isinstance(padding, int) else tuple(padding)
+        self.pad = p          # (left, right, top, bottom)
+
+    def forward(self, x):
+        l, r, t, b = self.pad
+        # torch pad convention: last dim first -> (left, right, top, bottom)
+        return ht.pad(x, [l, r, t, b], 0.0)
+
+
+class MSELoss(Module):
+    def forward(self, x, target):
+        return ht.mse_loss(x, target)
+
+
+class NLLLoss(Module):
+    def __init__(self, reduction="mean", ignore_index=-100):
+        super().__init__()
+        self.reduction, self.ignore_index = reduction, ignore_index
+
+    def forward(self, x, target):
+        return ht.nll_loss(x, target, self.reduction, self.ignore_index)
+
+
+class BCELoss(Module):
+    def __init__(self, reduction="mean"):
+        super().__init__()
+        self.reduction = reduction
+
+    def forward(self, x, target):
+        return ht.binary_cross_entropy(x, target, self.reduction)
+
+
+class KLDivLoss(Module):
+    def __init__(self, reduction="batchmean"):
+        super().__init__()
+        self.reduction = reduction
+
+    def forward(self, x, target):
+        return ht.kl_div(x, target, self.reduction)
+
+
+class CrossEntropyLoss(Module):
+    """Dense soft-label or sparse integer-label CE by target dtype."""
+
+    def __init__(self, reduction="mean"):
+        super().__init__()
+        self.reduction = reduction
+
+    def forward(self, x, target):
+        if target.dtype in (torch.int64, torch.int32):
+            per_tok = ht.softmax_cross_entropy_sparse(x, target)
+            return ht.reduce_mean(per_tok) if self.reduction == "mean" \
+                else ht.reduce_sum(per_tok)
+        return ht.softmax_cross_entropy(x, target, self.reduction)

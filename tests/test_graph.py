@@ -300,3 +300,30 @@ def test_grad_bucket_construction_and_placement():
         # dependency chains complete: within a small window of the last grad
         assert bp > last_in
         assert bp - last_in <= 3, (b.name, bp, last_in)
+
+
+def test_nn_module_layer_wrappers():
+    """Module-class wrappers over the vision/loss op families (reference
+    nn/modules/{conv,pooling,batchnorm,padding,loss}.py)."""
+    from hetu_amd import nn
+    torch.manual_seed(0)
+    with ht.graph("define_and_run") as g:
+        x = ht.placeholder((2, 3, 8, 8), name="x")
+        conv = nn.Conv2d(3, 4, 3, padding=1, name="c1")
+        bn = nn.BatchNorm2d(4)
+        net_out = nn.MaxPool2d(2)(bn(conv(x)))
+        pad_out = nn.ZeroPad2d(1)(x)
+        tgt = ht.placeholder((2, 4, 4, 4), name="tgt")
+        loss = nn.MSELoss()(net_out, tgt)
+        lg = ht.placeholder((6, 10), name="lg")
+        lab = ht.placeholder((6,), dtype=torch.int64, name="lab")
+        ce = nn.CrossEntropyLoss()(lg, lab)
+        res = g.run([net_out, pad_out, loss, ce],
+                    {x: torch.randn(2, 3, 8, 8),
+                     tgt: torch.randn(2, 4, 4, 4),
+                     lg: torch.randn(6, 10),
+                     lab: torch.randint(0, 10, (6,))})
+    out, padded, lv, cev = res
+    assert tuple(out.shape) == (2, 4, 4, 4)
+    assert tuple(padded.shape) == (2, 3, 10, 10)
+    assert float(lv) > 0 and float(cev) > 0
