@@ -22,10 +22,15 @@ from .runner import prepare_run_context
 
 class Trainer:
     def __init__(self, graph, handles: Dict, device: torch.device,
-                 capture: Optional[bool] = None):
+                 capture: Optional[bool] = None,
+                 lr_schedule=None):
+        """lr_schedule: callable step -> multiplier on the optimizer's
+        base lr (engine.lr_schedule); works under capture — the
+        multiplier rides the pinned bias-correction buffer."""
         self.graph = graph
         self.h = handles
         self.device = device
+        self.lr_schedule = lr_schedule
         self.ctx = prepare_run_context(graph, device)
         env = os.environ.get("HETU_AMD_CAPTURE", "auto")
         if capture is None:
@@ -54,6 +59,8 @@ class Trainer:
 
     # ---- plain step ------------------------------------------------------
     def run_step(self, feed: Dict):
+        if self.lr_schedule is not None:
+            AdamStepOp.set_lr_scale(self.lr_schedule(self._step))
         loss, _ = self.graph.run([self.h["loss"], self.h["train_op"]],
                                  feed, ctx=self.ctx)
         self._step += 1
@@ -81,8 +88,10 @@ class Trainer:
         # immediately after
 
     def replay(self):
+        scale = self.lr_schedule(self._step) \
+            if self.lr_schedule is not None else None
         self._step += 1
-        AdamStepOp.set_replay_step(self._step)
+        AdamStepOp.set_replay_step(self._step, lr_scale=scale)
         self._cuda_graph.replay()
         return self._loss_out
 

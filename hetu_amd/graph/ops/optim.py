@@ -93,12 +93,25 @@ class AdamStepOp(OptimizerUpdateOp):
         self.state: Dict = {}
         AdamStepOp._instances.add(self)
 
+    # global LR multiplier (schedules under hipGraph capture): the update
+    # is lr * (m/bc1) / (sqrt(v/bc2)+eps), so scaling bc1 by 1/s scales
+    # the WHOLE update by s exactly — the kernel's baked lr never changes,
+    # the scale rides the same pinned bc buffer the replay re-reads.
+    _lr_scale: float = 1.0
+
     @classmethod
-    def set_replay_step(cls, step: int):
+    def set_lr_scale(cls, scale: float):
+        cls._lr_scale = float(scale)
+
+    @classmethod
+    def set_replay_step(cls, step: int, lr_scale: Optional[float] = None):
         """Update the shared pinned bias-correction buffers for a graph
-        replay at optimizer step `step` (1-based)."""
+        replay at optimizer step `step` (1-based); lr_scale (if given)
+        also updates the global LR multiplier."""
+        if lr_scale is not None:
+            cls._lr_scale = float(lr_scale)
         for (b1, b2, _dev), sh in cls._shared_bc.items():
-            sh["host"][0] = 1.0 - b1 ** step
+            sh["host"][0] = (1.0 - b1 ** step) / cls._lr_scale
             sh["host"][1] = 1.0 - b2 ** step
             sh["step"] = step
         for inst in cls._instances:
@@ -132,15 +145,19 @@ class AdamStepOp(OptimizerUpdateOp):
         if "bc" in st:
             sh = st["bc"]
             if sh["step"] < st["step"]:
-                # first Adam op of this step: write + upload once
-                sh["host"][0] = 1.0 - a["beta1"] ** st["step"]
+                # first Adam op of this step: write + upload once (the
+                # LR multiplier folds into bc1, see _lr_scale)
+                sh["host"][0] = (1.0 - a["beta1"] ** st["step"]) \
+                    / AdamStepOp._lr_scale
                 sh["host"][1] = 1.0 - a["beta2"] ** st["step"]
                 sh["dev"].copy_(sh["host"], non_blocking=True)
                 sh["step"] = st["step"]
             bc_dev = sh["dev"]
         out16 = param if param.dtype != torch.float32 else None
+        lr = a["lr"] if bc_dev is not None \
+            else a["lr"] * AdamStepOp._lr_scale
         F.adam_step(st["master"], grad, st["m"], st["v"],
-                    a["lr"], a["beta1"], a["beta2"], a["eps"],
+                    lr, a["beta1"], a["beta2"], a["eps"],
                     a.get("weight_decay", 0.0), st["step"], out16, bc_dev)
         if out16 is None:
             param.copy_(st["master"])
@@ -204,7 +221,8 @@ class ZeroAdamStepOp(OptimizerUpdateOp):
         out16 = torch.empty(shard_elems, dtype=param.dtype,
                             device=param.device)
         F.adam_step(st["master"], gshard, st["m"], st["v"],
-                    a["lr"], a["beta1"], a["beta2"], a["eps"],
+                    a["lr"] * AdamStepOp._lr_scale,
+                    a["beta1"], a["beta2"], a["eps"],
                     a.get("weight_decay", 0.0), st["step"], out16, None)
         if n > 1:
             full = ctx.comm.allgather(out16, ranks, dim=0)

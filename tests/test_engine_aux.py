@@ -137,3 +137,43 @@ def test_training_config_from_ds_parallel_json(tmp_path):
     tc = TrainingConfig.from_yaml(str(yml))
     assert (tc.pp, tc.dp, tc.tp) == (2, 2, 2)
     assert tc.zero is True
+
+
+def test_lr_schedule_exact_scaling():
+    """A schedule multiplier of 0.5 must produce EXACTLY the trajectory
+    of half the base lr (the multiplier folds into the Adam update), and
+    schedules compose with the Trainer step loop."""
+    import hetu_amd as ht
+    from hetu_amd.engine.lr_schedule import (cosine_with_warmup,
+                                             linear_warmup)
+    from hetu_amd.engine.trainer import Trainer
+    from hetu_amd.graph.ops.optim import AdamStepOp
+
+    def build(lr):
+        torch.manual_seed(3)
+        with ht.graph("define_and_run") as g:
+            x = ht.placeholder((4, 8), name="x")
+            w = ht.variable(torch.randn(8, 8), name="w")
+            loss = ht.reduce_mean(ht.pow(ht.matmul(x, w), 2))
+            train = ht.Adam(lr=lr).minimize(loss)
+        return g, {"loss": loss, "train_op": train}, x, w
+
+    try:
+        xd = torch.randn(4, 8)
+        g1, h1, x1, w1 = build(0.05)
+        t1 = Trainer(g1, h1, torch.device("cpu"))
+        for _ in range(4):
+            t1.step({x1: xd})
+        g2, h2, x2, w2 = build(0.10)
+        t2 = Trainer(g2, h2, torch.device("cpu"),
+                     lr_schedule=lambda s: 0.5)
+        for _ in range(4):
+            t2.step({x2: xd})
+        assert torch.allclose(w1.get_data(), w2.get_data(), atol=1e-6)
+        # warmup actually ramps: step-0 update is smaller than lr
+        m = linear_warmup(4)
+        assert m(0) == 0.25 and m(3) == 1.0 and m(10) == 1.0
+        c = cosine_with_warmup(2, 10, min_ratio=0.1)
+        assert c(0) == 0.5 and abs(c(10) - 0.1) < 1e-6
+    finally:
+        AdamStepOp.set_lr_scale(1.0)
