@@ -48,7 +48,8 @@ def main():
         g, h = build_gpt_train_graph(mcfg, micro_batch=B, seq_len=S,
                                      dtype=tc.dtype(), lr=tc.lr, spec=spec,
                                      zero=tc.zero)
-    trainer = Trainer(g, h, device)
+    trainer = Trainer(g, h, device,
+                      lr_schedule=tc.lr_schedule())
     t0 = time.time()
     for step in range(tc.steps):
         ids = torch.randint(0, mcfg.vocab, (B, S), device=device)

@@ -25,6 +25,9 @@ class TrainingConfig:
     recompute: bool = False
     precision: str = "bf16"                   # bf16 | fp32 | fp16(+scaler)
     lr: float = 1e-4
+    lr_warmup_steps: int = 0
+    lr_decay: str = "none"            # none | cosine | inv_sqrt
+    lr_min_ratio: float = 0.1
     weight_decay: float = 0.0
     steps: int = 100
     save_every: int = 0
@@ -60,6 +63,19 @@ class TrainingConfig:
         with open(path) as f:
             self.zero = bool(json.load(f).get("zero", False))
         return self
+
+    def lr_schedule(self):
+        """Multiplier schedule for the Trainer (None when static)."""
+        from .lr_schedule import (cosine_with_warmup, inverse_sqrt,
+                                  linear_warmup)
+        if self.lr_decay == "cosine":
+            return cosine_with_warmup(self.lr_warmup_steps, self.steps,
+                                      self.lr_min_ratio)
+        if self.lr_decay == "inv_sqrt":
+            return inverse_sqrt(max(1, self.lr_warmup_steps))
+        if self.lr_warmup_steps > 0:
+            return linear_warmup(self.lr_warmup_steps)
+        return None
 
     def dtype(self):
         import torch

@@ -177,3 +177,14 @@ def test_lr_schedule_exact_scaling():
         assert c(0) == 0.5 and abs(c(10) - 0.1) < 1e-6
     finally:
         AdamStepOp.set_lr_scale(1.0)
+
+
+def test_training_config_lr_schedule(tmp_path):
+    from hetu_amd.engine.trainer_config import TrainingConfig
+    yml = tmp_path / "r.yaml"
+    yml.write_text("lr: 0.001\nlr_warmup_steps: 10\nlr_decay: cosine\n"
+                   "steps: 100\n")
+    tc = TrainingConfig.from_yaml(str(yml))
+    f = tc.lr_schedule()
+    assert f(0) < f(9) == 1.0 and f(99) <= 0.11
+    assert TrainingConfig().lr_schedule() is None
