@@ -178,6 +178,7 @@ class ZeroAdamStepOp(OptimizerUpdateOp):
 
     def __init__(self):
         self.state: Dict = {}
+        AdamStepOp._instances.add(self)     # set_replay_step syncs "step"
 
     def _dp_ranks(self, op, ctx):
         p = op.inputs[0]
@@ -209,7 +210,31 @@ class ZeroAdamStepOp(OptimizerUpdateOp):
             st["v"] = torch.zeros_like(st["master"])
             st["step"] = 0
             st["pad"] = pad
+            if param.is_cuda:
+                # same pinned bias-correction buffer as AdamStepOp: a
+                # hipGraph-captured ZeRO step re-reads corrections (and
+                # the LR multiplier) each replay instead of baking them
+                key = (a["beta1"], a["beta2"], param.device.index)
+                sh = AdamStepOp._shared_bc.get(key)
+                if sh is None:
+                    sh = {"host": torch.empty(2, dtype=torch.float32,
+                                              pin_memory=True),
+                          "dev": torch.empty(2, dtype=torch.float32,
+                                             device=param.device),
+                          "step": 0}
+                    AdamStepOp._shared_bc[key] = sh
+                st["bc"] = sh
         st["step"] += 1
+        bc_dev = None
+        if "bc" in st:
+            sh = st["bc"]
+            if sh["step"] < st["step"]:
+                sh["host"][0] = (1.0 - a["beta1"] ** st["step"]) \
+                    / AdamStepOp._lr_scale
+                sh["host"][1] = 1.0 - a["beta2"] ** st["step"]
+                sh["dev"].copy_(sh["host"], non_blocking=True)
+                sh["step"] = st["step"]
+            bc_dev = sh["dev"]
         gflat = grad.reshape(-1)
         if pad:
             gflat = torch.cat([gflat, gflat.new_zeros(pad)])
@@ -220,10 +245,11 @@ class ZeroAdamStepOp(OptimizerUpdateOp):
             gshard = gflat
         out16 = torch.empty(shard_elems, dtype=param.dtype,
                             device=param.device)
+        lr = a["lr"] if bc_dev is not None \
+            else a["lr"] * AdamStepOp._lr_scale
         F.adam_step(st["master"], gshard, st["m"], st["v"],
-                    a["lr"] * AdamStepOp._lr_scale,
-                    a["beta1"], a["beta2"], a["eps"],
-                    a.get("weight_decay", 0.0), st["step"], out16, None)
+                    lr, a["beta1"], a["beta2"], a["eps"],
+                    a.get("weight_decay", 0.0), st["step"], out16, bc_dev)
         if n > 1:
             full = ctx.comm.allgather(out16, ranks, dim=0)
         else:
