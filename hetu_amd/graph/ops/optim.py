@@ -53,7 +53,10 @@ class SGDStepOp(OptimizerUpdateOp):
 
     def compute(self, op, inputs, ctx):
         param, grad = inputs
-        lr = op.attrs["lr"]
+        # eager path honors the LR schedule; capture-safe schedules are
+        # an Adam feature (bc-buffer folding) — SGD under capture keeps
+        # the baked lr
+        lr = op.attrs["lr"] * AdamStepOp._lr_scale
         momentum = op.attrs.get("momentum", 0.0)
         if momentum > 0.0:
             buf = self.state.get("momentum_buffer")
