@@ -45,6 +45,21 @@ class OptimizerUpdateOp(OpInterface):
         op.outputs[0].device_group = op.inputs[0].device_group
 
 
+# device-resident LR multipliers (one per GPU): pinned host source +
+# device scalar; captured update ops read the device scalar, so
+# set_replay_step's host write + one H2D reaches every replay
+_LR_SCALE_BUFS: Dict[int, Dict] = {}
+
+
+def _lr_scale_dev(device) -> torch.Tensor:
+    sh = _LR_SCALE_BUFS.get(device.index)
+    if sh is None:
+        sh = {"host": torch.ones(1, dtype=torch.float32, pin_memory=True),
+              "dev": torch.ones(1, dtype=torch.float32, device=device)}
+        _LR_SCALE_BUFS[device.index] = sh
+    return sh["dev"]
+
+
 class SGDStepOp(OptimizerUpdateOp):
     type = "SGDStep"
 
@@ -53,10 +68,6 @@ class SGDStepOp(OptimizerUpdateOp):
 
     def compute(self, op, inputs, ctx):
         param, grad = inputs
-        # eager path honors the LR schedule; capture-safe schedules are
-        # an Adam feature (bc-buffer folding) — SGD under capture keeps
-        # the baked lr
-        lr = op.attrs["lr"] * AdamStepOp._lr_scale
         momentum = op.attrs.get("momentum", 0.0)
         if momentum > 0.0:
             buf = self.state.get("momentum_buffer")
@@ -67,7 +78,14 @@ class SGDStepOp(OptimizerUpdateOp):
             upd = buf
         else:
             upd = grad.float()
-        param -= (lr * upd).to(param.dtype)
+        if param.is_cuda:
+            # device-scalar multiplier: a captured step re-reads it each
+            # replay (set_replay_step updates the pinned source)
+            scale = _lr_scale_dev(param.device)
+            param -= (op.attrs["lr"] * upd * scale).to(param.dtype)
+        else:
+            lr = op.attrs["lr"] * AdamStepOp._lr_scale
+            param -= (lr * upd).to(param.dtype)
         return [_zero_token(param.device)]
 
 
@@ -105,6 +123,9 @@ class AdamStepOp(OptimizerUpdateOp):
     @classmethod
     def set_lr_scale(cls, scale: float):
         cls._lr_scale = float(scale)
+        for sh in _LR_SCALE_BUFS.values():
+            sh["host"][0] = cls._lr_scale
+            sh["dev"].copy_(sh["host"], non_blocking=True)
 
     @classmethod
     def set_replay_step(cls, step: int, lr_scale: Optional[float] = None):
@@ -112,7 +133,7 @@ class AdamStepOp(OptimizerUpdateOp):
         replay at optimizer step `step` (1-based); lr_scale (if given)
         also updates the global LR multiplier."""
         if lr_scale is not None:
-            cls._lr_scale = float(lr_scale)
+            cls.set_lr_scale(lr_scale)
         for (b1, b2, _dev), sh in cls._shared_bc.items():
             sh["host"][0] = (1.0 - b1 ** step) / cls._lr_scale
             sh["host"][1] = 1.0 - b2 ** step
