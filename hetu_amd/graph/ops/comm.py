@@ -12,7 +12,9 @@ Deduction order mirrors the reference: equal -> identity; check_scatter ->
 local slice (keep my shard); check_split -> local slice; check_allreduce ->
 ALL_REDUCE; check_allgather -> ALL_GATHER; check_reducescatter ->
 REDUCE_SCATTER; split->partial -> local zero-pad (adjoint of slice);
-otherwise a generic all-to-all style redistribution (not yet implemented).
+otherwise the generic gather+reslice redistribution (reduce partials,
+allgather every split dim, slice to the destination layout — correct for
+arbitrary pairs; exercised by tests/test_distributed_cpu.py reshard fuzz).
 """
 from __future__ import annotations
 
