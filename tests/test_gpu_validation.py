@@ -136,3 +136,42 @@ def test_pipeline_slot_capture_parity():
     os.environ.pop("HETU_AMD_PP_CAPTURE", None)
     for a, b in zip(results["0"], results["1"]):
         assert abs(a - b) < 3e-2, results
+
+
+def test_lr_schedule_under_capture():
+    """LR schedules under hipGraph capture: the multiplier folds into the
+    pinned Adam bias-correction buffer, so the CAPTURED trajectory must
+    match the EAGER trajectory with the same schedule exactly (and both
+    must differ from the unscheduled run)."""
+    from hetu_amd.engine.lr_schedule import cosine_with_warmup
+    from hetu_amd.engine.trainer import Trainer
+    from hetu_amd.graph.ops.optim import AdamStepOp
+    from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
+    dev = torch.device("cuda", 0)
+    cfg = GPTConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=256,
+                    ffn_hidden=512, vocab=2048, max_seq=128)
+    B, S = 4, 128
+    sched = cosine_with_warmup(3, 8, min_ratio=0.2)
+
+    def run(capture, schedule):
+        torch.manual_seed(11)
+        g, h = build_gpt_train_graph(cfg, micro_batch=B, seq_len=S,
+                                     dtype=torch.bfloat16, lr=3e-3)
+        tr = Trainer(g, h, dev, capture=capture, lr_schedule=schedule)
+        torch.manual_seed(13)
+        out = [float(_step(tr, h, cfg, B, S, dev).float())
+               for _ in range(8)]
+        torch.cuda.synchronize()
+        return out
+
+    try:
+        cap = run(True, sched)
+        eag = run(False, sched)
+        none = run(False, None)
+    finally:
+        AdamStepOp.set_lr_scale(1.0)
+    assert all(abs(a - b) < 2e-2 for a, b in zip(cap, eag)), (cap, eag)
+    # the schedule actually changed the trajectory
+    assert any(abs(a - b) > 1e-3 for a, b in zip(eag[3:], none[3:])), \
+        (eag, none)
+    assert all(v == v and v < 20 for v in cap)       # finite
