@@ -45,6 +45,26 @@ class OptimizerUpdateOp(OpInterface):
         op.outputs[0].device_group = op.inputs[0].device_group
 
 
+def _bc_refresh(st: Dict, a: Dict):
+    """Write + upload the shared pinned bias-correction buffer when this
+    step's (step, lr_scale) differs from what the buffer holds.  Tagged
+    by VALUE, not a monotonic step: the buffer is shared process-wide, so
+    a second graph built later starts at step 1 again and must overwrite
+    a stale higher-step value (the old `sh[step] < step` check silently
+    reused it)."""
+    if "bc" not in st:
+        return None
+    sh = st["bc"]
+    tag = (st["step"], AdamStepOp._lr_scale)
+    if sh.get("written") != tag:
+        sh["host"][0] = (1.0 - a["beta1"] ** st["step"]) \
+            / AdamStepOp._lr_scale
+        sh["host"][1] = 1.0 - a["beta2"] ** st["step"]
+        sh["dev"].copy_(sh["host"], non_blocking=True)
+        sh["written"] = tag
+    return sh["dev"]
+
+
 # device-resident LR multipliers (one per GPU): pinned host source +
 # device scalar; captured update ops read the device scalar, so
 # set_replay_step's host write + one H2D reaches every replay
@@ -137,7 +157,7 @@ class AdamStepOp(OptimizerUpdateOp):
         for (b1, b2, _dev), sh in cls._shared_bc.items():
             sh["host"][0] = (1.0 - b1 ** step) / cls._lr_scale
             sh["host"][1] = 1.0 - b2 ** step
-            sh["step"] = step
+            sh["written"] = (step, cls._lr_scale)
         for inst in cls._instances:
             if inst.state:
                 inst.state["step"] = step
@@ -161,22 +181,11 @@ class AdamStepOp(OptimizerUpdateOp):
                                               pin_memory=True),
                           "dev": torch.empty(2, dtype=torch.float32,
                                              device=param.device),
-                          "step": 0}
+                          "written": None}
                     AdamStepOp._shared_bc[key] = sh
                 st["bc"] = sh
         st["step"] += 1
-        bc_dev = None
-        if "bc" in st:
-            sh = st["bc"]
-            if sh["step"] < st["step"]:
-                # first Adam op of this step: write + upload once (the
-                # LR multiplier folds into bc1, see _lr_scale)
-                sh["host"][0] = (1.0 - a["beta1"] ** st["step"]) \
-                    / AdamStepOp._lr_scale
-                sh["host"][1] = 1.0 - a["beta2"] ** st["step"]
-                sh["dev"].copy_(sh["host"], non_blocking=True)
-                sh["step"] = st["step"]
-            bc_dev = sh["dev"]
+        bc_dev = _bc_refresh(st, a)
         out16 = param if param.dtype != torch.float32 else None
         lr = a["lr"] if bc_dev is not None \
             else a["lr"] * AdamStepOp._lr_scale
@@ -245,20 +254,11 @@ class ZeroAdamStepOp(OptimizerUpdateOp):
                                               pin_memory=True),
                           "dev": torch.empty(2, dtype=torch.float32,
                                              device=param.device),
-                          "step": 0}
+                          "written": None}
                     AdamStepOp._shared_bc[key] = sh
                 st["bc"] = sh
         st["step"] += 1
-        bc_dev = None
-        if "bc" in st:
-            sh = st["bc"]
-            if sh["step"] < st["step"]:
-                sh["host"][0] = (1.0 - a["beta1"] ** st["step"]) \
-                    / AdamStepOp._lr_scale
-                sh["host"][1] = 1.0 - a["beta2"] ** st["step"]
-                sh["dev"].copy_(sh["host"], non_blocking=True)
-                sh["step"] = st["step"]
-            bc_dev = sh["dev"]
+        bc_dev = _bc_refresh(st, a)
         gflat = grad.reshape(-1)
         if pad:
             gflat = torch.cat([gflat, gflat.new_zeros(pad)])

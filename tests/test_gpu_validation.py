@@ -166,7 +166,9 @@ def test_lr_schedule_under_capture():
 
     try:
         cap = run(True, sched)
+        AdamStepOp.set_lr_scale(1.0)
         eag = run(False, sched)
+        AdamStepOp.set_lr_scale(1.0)
         none = run(False, None)
     finally:
         AdamStepOp.set_lr_scale(1.0)
