@@ -286,6 +286,23 @@ def bmm(a: Tensor, b: Tensor) -> Tensor:
     return _cg().make_op(B.BatchMatMulOp(), [_ac(a), _ac(b)], {}).output()
 
 
+def baddbmm(inp: Tensor, a: Tensor, b: Tensor, beta: float = 1.0,
+            alpha: float = 1.0) -> Tensor:
+    """beta*inp + alpha*(a @ b), batched (reference Baddbmm.cu) — composed
+    over the bmm/add ops so the grads come for free."""
+    y = bmm(a, b)
+    if alpha != 1.0:
+        y = mul(y, alpha)
+    return add(mul(inp, beta) if beta != 1.0 else inp, y)
+
+
+def matvec(a: Tensor, v: Tensor) -> Tensor:
+    """[M,K] @ [K] -> [M] (reference MatVecMul.cu): routed through the
+    GEMM path on a [K,1] view."""
+    y = matmul(a, reshape(v, (int(v.shape[0]), 1)))
+    return reshape(y, (int(a.shape[0]),))
+
+
 # ---- nn --------------------------------------------------------------------
 
 def relu(a):

@@ -491,3 +491,38 @@ def test_dense_softmax_ce_broadcast_group():
         return [p], tok
     (out,) = _run(build_g, [torch.randn(2, 2)])
     assert out.shape == () and float(out) == 0.0
+
+
+def test_baddbmm_matvec():
+    a = torch.randn(3, 4, 5, requires_grad=True)
+    b = torch.randn(3, 5, 6, requires_grad=True)
+    inp = torch.randn(3, 4, 6, requires_grad=True)
+
+    def build():
+        pa = ht.placeholder((3, 4, 5), name="a")
+        pb = ht.placeholder((3, 5, 6), name="b")
+        pi = ht.placeholder((3, 4, 6), name="i")
+        return [pa, pb, pi], ht.reduce_sum(
+            ht.baddbmm(pi, pa, pb, beta=0.5, alpha=2.0))
+    out, ga, gb, gi = _run(build, [a.detach(), b.detach(), inp.detach()],
+                           wrt=[0, 1, 2])
+    ref = torch.baddbmm(inp, a, b, beta=0.5, alpha=2.0).sum()
+    ref.backward()
+    assert torch.allclose(out, ref.detach(), rtol=1e-4)
+    assert torch.allclose(ga, a.grad, rtol=1e-4, atol=1e-5)
+    assert torch.allclose(gb, b.grad, rtol=1e-4, atol=1e-5)
+    assert torch.allclose(gi, inp.grad)
+
+    m = torch.randn(4, 6, requires_grad=True)
+    v = torch.randn(6, requires_grad=True)
+
+    def build_mv():
+        pm = ht.placeholder((4, 6), name="m")
+        pv = ht.placeholder((6,), name="v")
+        return [pm, pv], ht.reduce_sum(ht.matvec(pm, pv))
+    out, gm, gv = _run(build_mv, [m.detach(), v.detach()], wrt=[0, 1])
+    ref = (m @ v).sum()
+    ref.backward()
+    assert torch.allclose(out, ref.detach(), rtol=1e-5)
+    assert torch.allclose(gm, m.grad, atol=1e-6)
+    assert torch.allclose(gv, v.grad, atol=1e-5)
