@@ -47,3 +47,30 @@ def test_bench_ws2_pp2():
     assert out is not None
     assert out["config"]["parallelism"] == "dp1_pp2"
     assert out["value"] > 0
+
+
+def test_bench_ws8_auto():
+    """8-rank gloo smoke of the exact round-end N=8 invocation shape:
+    auto search at ws=8, strategy broadcast to all ranks, whole-job
+    aggregation."""
+    procs = []
+    env0 = {**os.environ, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29671", "GLOO_SOCKET_IFNAME": "lo"}
+    for r in range(8):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="8", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen(
+            [sys.executable, BENCH, "--allow-cpu", "--model", "gpt-tiny",
+             "--gpus", "8", "--steps", "2", "--warmup", "1",
+             "--seq-len", "32", "--global-batch", "16"],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True))
+    line = None
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=900)
+        ok = p.returncode in (0, -6)
+        assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
+        for ln in out.splitlines():
+            if ln.startswith("{"):
+                line = json.loads(ln)
+    assert line is not None and line["n_gpus"] == 8
+    assert line["value"] > 0
