@@ -195,3 +195,39 @@ def test_gpt_moe_model_trains():
                       {h["input_ids"]: ids, h["labels"]: lab}, ctx=ctx)
         losses.append(float(lv))
     assert losses[-1] < losses[0] - 0.5, losses
+
+
+HIER_WORKER = r"""
+import os, sys, torch
+sys.path.insert(0, os.environ["HETU_REPO"])
+from hetu_amd.parallel.comm import comm_backend
+from hetu_amd.parallel.moe import alltoall, hierarchical_alltoall
+comm = comm_backend()
+rank, ws = comm.rank, comm.world_size
+assert ws == 4
+ranks = list(range(4))
+torch.manual_seed(17 + rank)
+x = torch.randn(4 * 3, 5)                 # P chunks of 3 rows
+flat = alltoall(comm, ranks, x.clone())
+hier = hierarchical_alltoall(comm, ranks, x.clone(), node_size=2)
+assert torch.allclose(flat, hier, atol=1e-6), \
+    (rank, (flat - hier).abs().max())
+print("HIEROK")
+"""
+
+
+def test_hierarchical_a2a_matches_flat():
+    """4 ranks as 2 nodes x 2 gpus: the 3-phase hierarchical all-to-all
+    must be elementwise identical to the flat all-to-all."""
+    env0 = {**os.environ, "HETU_REPO": REPO, "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": "29586", "GLOO_SOCKET_IFNAME": "lo"}
+    procs = []
+    for r in range(4):
+        env = dict(env0, RANK=str(r), WORLD_SIZE="4", LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen([sys.executable, "-c", HIER_WORKER],
+                                      env=env, stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=300)
+        ok = (p.returncode in (0, -6)) and "HIEROK" in out
+        assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
