@@ -39,7 +39,14 @@ class Trainer:
         self._cuda_graph = None
         self._static_feeds: Dict = {}
         self._loss_out = None
-        self._step = 0
+        # resume-aware: checkpoint load restores each Adam op's step
+        # counter BEFORE the Trainer is built; starting _step at 0 would
+        # make the first captured replay's set_replay_step clobber them
+        # (and restart any LR schedule)
+        self._step = max((op.interface.state.get("step", 0)
+                          for op in graph.ops
+                          if op.type in ("AdamStep", "ZeroAdamStep")
+                          and op.interface.state), default=0)
         # host-side tracing (reference engine/trainer.py:22 wires
         # torch.profiler): HETU_AMD_TORCH_PROFILE=<dir> records steps
         # 2-4 (skip 0, warm 1) and writes a chrome trace there

@@ -230,3 +230,41 @@ def test_tp2_save_tp4_load_resharding(tmp_path):
     assert len(losses) == 4
     for lv in losses:
         assert abs(lv - float(ref_loss)) < 1e-5, (lv, float(ref_loss))
+
+
+def test_trainer_resumes_step_counter(tmp_path):
+    """A Trainer built after restoring Adam states must continue the
+    optimizer step count (bias corrections + LR schedule), not restart
+    at 0."""
+    import hetu_amd as ht
+    from hetu_amd.engine.trainer import Trainer
+    from hetu_amd.utils.checkpoint import (collect_adam_states,
+                                           load_adam_states, load_model,
+                                           save_model)
+
+    def build():
+        torch.manual_seed(21)
+        with ht.graph("define_and_run") as g:
+            x = ht.placeholder((4, 8), name="x")
+            w = ht.variable(torch.randn(8, 8), name="w")
+            loss = ht.reduce_mean(ht.pow(ht.matmul(x, w), 2))
+            train = ht.Adam(lr=1e-3).minimize(loss)
+        return g, {"loss": loss, "train_op": train}, x
+
+    g, h, x = build()
+    tr = Trainer(g, h, torch.device("cpu"))
+    xd = torch.randn(4, 8)
+    for _ in range(3):
+        tr.step({x: xd})
+    sd = str(tmp_path / "ck")
+    save_model(g.parameters, sd, optimizer_states=collect_adam_states(g))
+    g2, h2, x2 = build()
+    load_model(g2.parameters, sd)
+    load_adam_states(g2, sd)
+    tr2 = Trainer(g2, h2, torch.device("cpu"))
+    assert tr2._step == 3
+    tr2.step({x2: xd})
+    steps = {op.interface.state.get("step")
+             for op in g2.ops if op.type == "AdamStep"
+             and op.interface.state}
+    assert steps == {4}, steps
