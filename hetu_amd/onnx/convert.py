@@ -119,6 +119,17 @@ def export_onnx(graph, fetches: Sequence, path: str = None) -> bytes:
             nodes.append(P.node_proto(
                 "Cast", ins, [o], name=op.name,
                 attrs={"to": _TORCH2DT[op.attrs["dtype"]]}))
+        elif t == "Reduce":
+            onnx_op = {"sum": "ReduceSum", "mean": "ReduceMean",
+                       "max": "ReduceMax", "min": "ReduceMin",
+                       "prod": "ReduceProd"}[op.attrs["mode"]]
+            attrs = {"keepdims": 1 if op.attrs.get("keepdim") else 0}
+            dim = op.attrs.get("dim")
+            if dim is not None:
+                attrs["axes"] = ([dim] if isinstance(dim, int)
+                                 else list(dim))
+            nodes.append(P.node_proto(onnx_op, ins, [o], name=op.name,
+                                      attrs=attrs))
         else:
             raise NotImplementedError(f"ONNX export: op {t}")
 
@@ -265,6 +276,18 @@ def import_onnx(blob) -> Tuple[object, Dict[str, object], List[object]]:
                 y = ht.embedding(get(ins[0]), get(ins[1]))
             elif op == "Cast":
                 y = ht.cast(get(ins[0]), _DT2TORCH[attrs["to"]])
+            elif op in ("ReduceSum", "ReduceMean", "ReduceMax",
+                        "ReduceMin", "ReduceProd"):
+                f = {"ReduceSum": ht.reduce_sum,
+                     "ReduceMean": ht.reduce_mean,
+                     "ReduceMax": ht.reduce_max,
+                     "ReduceMin": ht.reduce_min,
+                     "ReduceProd": ht.reduce_prod}[op]
+                axes = attrs.get("axes")
+                dim = (None if axes is None
+                       else axes[0] if len(axes) == 1 else list(axes))
+                y = f(get(ins[0]), dim=dim,
+                      keepdim=bool(attrs.get("keepdims", 0)))
             else:
                 raise NotImplementedError(f"ONNX import: op {op}")
             env[outs[0]] = y

