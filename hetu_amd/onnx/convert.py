@@ -87,8 +87,25 @@ def export_onnx(graph, fetches: Sequence, path: str = None) -> bytes:
                 b = bt
             nodes.append(P.node_proto("MatMul", [a, b], [o], name=op.name))
         elif t in ("Add", "Sub", "Mul", "Div", "Relu", "Gelu", "Sigmoid",
-                   "Tanh"):
+                   "Tanh", "Exp", "Log", "Sqrt", "Abs", "Neg"):
             nodes.append(P.node_proto(t, ins, [o], name=op.name))
+        elif t == "Concat":
+            nodes.append(P.node_proto(
+                "Concat", ins, [o], name=op.name,
+                attrs={"axis": int(op.attrs.get("dim", 0))}))
+        elif t in ("AddScalar", "SubScalar", "MulScalar", "DivScalar",
+                   "PowScalar"):
+            # scalar operand becomes a rank-0 initializer
+            sc = o + "/scalar"
+            val = torch.tensor(float(op.attrs["value"]),
+                               dtype=op.outputs[0].dtype)
+            inits.append(P.tensor_proto(sc, [], _TORCH2DT[val.dtype],
+                                        _tensor_raw(val)))
+            onnx_op = {"AddScalar": "Add", "SubScalar": "Sub",
+                       "MulScalar": "Mul", "DivScalar": "Div",
+                       "PowScalar": "Pow"}[t]
+            nodes.append(P.node_proto(onnx_op, [ins[0], sc], [o],
+                                      name=op.name))
         elif t == "Softmax":
             nodes.append(P.node_proto("Softmax", ins, [o], name=op.name,
                                       attrs={"axis": op.attrs.get("dim",
@@ -252,8 +269,17 @@ def import_onnx(blob) -> Tuple[object, Dict[str, object], List[object]]:
                 f = {"Add": ht.add, "Sub": ht.sub, "Mul": ht.mul,
                      "Div": ht.div}[op]
                 y = f(get(ins[0]), get(ins[1]))
-            elif op in ("Relu", "Gelu", "Sigmoid", "Tanh"):
+            elif op in ("Relu", "Gelu", "Sigmoid", "Tanh", "Exp", "Log",
+                        "Sqrt", "Neg"):
                 y = getattr(ht, op.lower())(get(ins[0]))
+            elif op == "Abs":
+                y = ht.abs_(get(ins[0]))
+            elif op == "Pow" and ins[1] in const \
+                    and const[ins[1]].numel() == 1:
+                y = ht.pow(get(ins[0]), float(const[ins[1]]))
+            elif op == "Concat":
+                y = ht.concat([get(i) for i in ins],
+                              dim=int(attrs.get("axis", 0)))
             elif op == "Softmax":
                 y = ht.softmax(get(ins[0]), dim=attrs.get("axis", -1))
             elif op == "Reshape":

@@ -132,3 +132,25 @@ def test_reduce_roundtrip():
     (out,) = g2.run(outputs, {inputs["x"]: xd}, ctx=ctx)
     ref = xd.mean(1, keepdim=True) + xd.max(1, keepdim=True).values
     assert torch.allclose(out, ref, atol=1e-6)
+
+
+def test_unary_concat_roundtrip():
+    import torch
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import DefineAndRunGraph, pop_graph, push_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.onnx.convert import export_onnx, import_onnx
+    g = DefineAndRunGraph("u")
+    push_graph(g)
+    try:
+        x = ht.placeholder((2, 3), name="x")
+        y = ht.concat([ht.exp(x), ht.abs_(ht.neg(x)),
+                       ht.log(ht.add(ht.abs_(x), 1.0))], dim=1)
+    finally:
+        pop_graph()
+    g2, inputs, outputs = import_onnx(export_onnx(g, [y]))
+    xd = torch.randn(2, 3)
+    ctx = prepare_run_context(g2, torch.device("cpu"), use_comm=False)
+    (out,) = g2.run(outputs, {inputs["x"]: xd}, ctx=ctx)
+    ref = torch.cat([xd.exp(), xd.neg().abs(), (xd.abs() + 1).log()], 1)
+    assert torch.allclose(out, ref, atol=1e-6)
