@@ -1,8 +1,10 @@
 """ONNX graph export/import over the minimal proto layer.
 
 Supported op subset (both directions): Linear/Gemm, MatMul, Add, Sub,
-Mul, Div, Relu, Gelu, Sigmoid, Tanh, Softmax, Reshape, Transpose,
-LayerNormalization, Gather (embedding), Cast.  Reference parity:
+Mul, Div (+Scalar variants as rank-0 initializers), Relu, Gelu, Sigmoid,
+Tanh, Exp, Log, Sqrt, Abs, Neg, Pow, Softmax, Reshape, Transpose, Concat,
+Slice, Where, Bool, Reduce{Sum,Mean,Max,Min,Prod}, LayerNormalization,
+Gather (embedding), Cast.  Reference parity:
 hetu/v1/python/hetu/onnx/{onnx2hetu,hetu2onnx}.
 """
 from __future__ import annotations
@@ -17,7 +19,7 @@ from . import proto as P
 
 
 _TORCH2DT = {torch.float32: P.DT_FLOAT, torch.int64: P.DT_INT64,
-             torch.int32: P.DT_INT32}
+             torch.int32: P.DT_INT32, torch.bool: P.DT_BOOL}
 _DT2TORCH = {v: k for k, v in _TORCH2DT.items()}
 
 
@@ -93,6 +95,27 @@ def export_onnx(graph, fetches: Sequence, path: str = None) -> bytes:
             nodes.append(P.node_proto(
                 "Concat", ins, [o], name=op.name,
                 attrs={"axis": int(op.attrs.get("dim", 0))}))
+        elif t == "Slice":
+            # opset-10+ Slice takes starts/ends/axes as inputs
+            start = int(op.attrs["start"])
+            end = start + int(op.attrs["length"])
+            axis = int(op.attrs["dim"])
+            extra_ins = []
+            for suffix, vals in (("/starts", [start]), ("/ends", [end]),
+                                 ("/axes", [axis])):
+                nmx = o + suffix
+                inits.append(P.tensor_proto(
+                    nmx, [1], P.DT_INT64,
+                    np.asarray(vals, dtype=np.int64).tobytes()))
+                extra_ins.append(nmx)
+            nodes.append(P.node_proto("Slice", ins + extra_ins, [o],
+                                      name=op.name))
+        elif t == "Where":
+            nodes.append(P.node_proto("Where", ins, [o], name=op.name))
+        elif t == "Bool":
+            nodes.append(P.node_proto(
+                "Cast", ins, [o], name=op.name,
+                attrs={"to": P.DT_BOOL}))
         elif t in ("AddScalar", "SubScalar", "MulScalar", "DivScalar",
                    "PowScalar"):
             # scalar operand becomes a rank-0 initializer
@@ -277,6 +300,16 @@ def import_onnx(blob) -> Tuple[object, Dict[str, object], List[object]]:
             elif op == "Pow" and ins[1] in const \
                     and const[ins[1]].numel() == 1:
                 y = ht.pow(get(ins[0]), float(const[ins[1]]))
+            elif op == "Slice":
+                starts = const[ins[1]].tolist()
+                ends = const[ins[2]].tolist()
+                axes = (const[ins[3]].tolist() if len(ins) > 3
+                        else list(range(len(starts))))
+                y = get(ins[0])
+                for s, e, ax in zip(starts, ends, axes):
+                    y = ht.slice_(y, int(ax), int(s), int(e) - int(s))
+            elif op == "Where":
+                y = ht.where(get(ins[0]), get(ins[1]), get(ins[2]))
             elif op == "Concat":
                 y = ht.concat([get(i) for i in ins],
                               dim=int(attrs.get("axis", 0)))

@@ -86,7 +86,7 @@ def parse_message(buf: bytes) -> Dict[int, List]:
 # ---------------------------------------------------------------------------
 # ONNX message builders
 # ---------------------------------------------------------------------------
-DT_FLOAT, DT_INT64, DT_INT32 = 1, 7, 6
+DT_FLOAT, DT_INT64, DT_INT32, DT_BOOL = 1, 7, 6, 9
 ATTR_FLOAT, ATTR_INT, ATTR_STRING, ATTR_TENSOR = 1, 2, 3, 4
 ATTR_FLOATS, ATTR_INTS = 6, 7
 

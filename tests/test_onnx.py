@@ -154,3 +154,26 @@ def test_unary_concat_roundtrip():
     (out,) = g2.run(outputs, {inputs["x"]: xd}, ctx=ctx)
     ref = torch.cat([xd.exp(), xd.neg().abs(), (xd.abs() + 1).log()], 1)
     assert torch.allclose(out, ref, atol=1e-6)
+
+
+def test_slice_where_roundtrip():
+    import torch
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.graph import DefineAndRunGraph, pop_graph, push_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.onnx.convert import export_onnx, import_onnx
+    g = DefineAndRunGraph("sw")
+    push_graph(g)
+    try:
+        x = ht.placeholder((4, 6), name="x")
+        sl = ht.slice_(x, 1, 2, 3)
+        y = ht.where(ht.bool_(sl), sl, ht.neg(sl))
+    finally:
+        pop_graph()
+    g2, inputs, outputs = import_onnx(export_onnx(g, [y]))
+    xd = torch.randn(4, 6)
+    ctx = prepare_run_context(g2, torch.device("cpu"), use_comm=False)
+    (out,) = g2.run(outputs, {inputs["x"]: xd}, ctx=ctx)
+    sref = xd[:, 2:5]
+    ref = torch.where(sref.bool(), sref, -sref)
+    assert torch.allclose(out, ref, atol=1e-6)
