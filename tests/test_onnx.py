@@ -71,7 +71,8 @@ def test_onnx_roundtrip_fuzz():
     from hypothesis import given, settings
     from hypothesis import strategies as st
 
-    ACTS = ["gelu", "relu", "sigmoid", "tanh"]
+    ACTS = ["gelu", "relu", "sigmoid", "tanh", "exp_c",
+            "abs_n", "red_mean"]
 
     @settings(max_examples=40, deadline=None)
     @given(st.lists(st.tuples(st.integers(2, 24),
@@ -91,7 +92,17 @@ def test_onnx_roundtrip_fuzz():
                                 name=f"w{li}")
                 bvar = ht.variable(torch.randn(w_out) * 0.1,
                                    name=f"b{li}")
-                cur = getattr(ht, ACTS[ai])(ht.linear(cur, w, bvar))
+                lin = ht.linear(cur, w, bvar)
+                a = ACTS[ai]
+                if a == "exp_c":          # bounded exp + scalar op
+                    cur = ht.exp(ht.mul(ht.tanh(lin), 0.5))
+                elif a == "abs_n":
+                    cur = ht.abs_(ht.neg(lin))
+                elif a == "red_mean":     # reduce + broadcast back
+                    cur = ht.sub(lin, ht.reduce_mean(lin, dim=1,
+                                                     keepdim=True))
+                else:
+                    cur = getattr(ht, a)(lin)
                 d = w_out
             y = ht.softmax(cur, dim=-1)
         finally:
