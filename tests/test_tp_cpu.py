@@ -146,3 +146,12 @@ def test_dp2_cp2_matches_single(single_losses):
     losses = _launch(4, {"HETU_TEST_DP": "2", "HETU_TEST_CP": "2"}, 29539)
     assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
         f"dp2xcp2 {losses} vs single {single_losses}"
+
+
+def test_dp2_tp2_zero_matches_single(single_losses):
+    """dp2 x tp2 with ZeRO (4 ranks): optimizer-state sharding over the
+    dp groups composes with tensor parallelism."""
+    losses = _launch(4, {"HETU_TEST_DP": "2", "HETU_TEST_ZERO": "1"},
+                     29540)
+    assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
+        f"dp2xtp2+zero {losses} vs single {single_losses}"
