@@ -155,3 +155,10 @@ def test_dp2_tp2_zero_matches_single(single_losses):
                      29540)
     assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
         f"dp2xtp2+zero {losses} vs single {single_losses}"
+
+
+def test_dp2_tp2_sp_matches_single(single_losses):
+    """dp2 x tp2 with sequence parallelism (4 ranks)."""
+    losses = _launch(4, {"HETU_TEST_DP": "2", "HETU_TEST_SP": "1"}, 29545)
+    assert np.allclose(losses, single_losses, rtol=5e-4, atol=2e-4), \
+        f"dp2xtp2+sp {losses} vs single {single_losses}"
