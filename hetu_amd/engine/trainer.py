@@ -126,8 +126,9 @@ class Trainer:
                     # silently skip one update and return stale loss)
                     return self.replay()
                 except Exception as e:  # noqa: BLE001
-                    print(f"[hetu_amd] hipGraph capture failed, running "
-                          f"eager: {e}")
+                    from ..utils.logging import get_logger
+                    get_logger().warning(
+                        "hipGraph capture failed, running eager: %s", e)
                     self.want_capture = False
             for t, v in feed.items():
                 feed[t] = v.to(self.device)

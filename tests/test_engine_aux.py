@@ -188,3 +188,14 @@ def test_training_config_lr_schedule(tmp_path):
     f = tc.lr_schedule()
     assert f(0) < f(9) == 1.0 and f(99) <= 0.11
     assert TrainingConfig().lr_schedule() is None
+
+
+def test_leveled_logger(monkeypatch, capsys):
+    import importlib
+    monkeypatch.setenv("HETU_AMD_LOG_LEVEL", "DEBUG")
+    from hetu_amd.utils import logging as hl
+    lg = hl.get_logger("hetu_amd.test")
+    lg.debug("dbg %d", 1)
+    lg.warning("warn")
+    err = capsys.readouterr().err
+    assert "dbg 1" in err and "warn" in err and "r0" in err
