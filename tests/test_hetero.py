@@ -225,3 +225,40 @@ def test_hetero_from_ds_config_parity():
     comb = [2/3 * a + 1/3 * b
             for a, b in zip(hlosses[0], hlosses[2])]
     assert np.allclose(comb, sl, rtol=5e-4, atol=5e-4), (comb, sl)
+
+
+def test_hetero_atom_partition_property():
+    """Property: for any pipeline tp-mix and fused-section layout, the
+    split-allreduce plan's atoms exactly partition the global extent and
+    every atom names exactly one shard per pipeline."""
+    import itertools
+    from hetu_amd.parallel.hetero import _shard_regions
+
+    for tps in itertools.product((1, 2, 4), repeat=3):
+        for secs in (None, [8, 8, 8], [16, 8]):
+            glen = 24 if secs is None else sum(secs)
+            if any(glen % t for t in tps) or \
+                    (secs and any(s % t for s in secs for t in tps)):
+                continue
+            cuts = {0, glen}
+            per_pipe = []
+            for tp in tps:
+                regs = [_shard_regions(glen, tp, t, secs)
+                        for t in range(tp)]
+                per_pipe.append(regs)
+                for shard in regs:
+                    for a, b in shard:
+                        cuts.update((a, b))
+            cuts = sorted(cuts)
+            covered = 0
+            for a, b in zip(cuts[:-1], cuts[1:]):
+                owners_per_pipe = []
+                for k, tp in enumerate(tps):
+                    owners = [t for t in range(tp)
+                              if any(x <= a and b <= y
+                                     for x, y in per_pipe[k][t])]
+                    owners_per_pipe.append(owners)
+                assert all(len(o) == 1 for o in owners_per_pipe), \
+                    (tps, secs, a, b, owners_per_pipe)
+                covered += b - a
+            assert covered == glen, (tps, secs, covered)
