@@ -175,3 +175,25 @@ def test_hetero_layers_and_rank_map():
                                       hetero_layers=[[3, 1], [2, 2]],
                                       rank_map={0: 10, 1: 11})
     assert 0 not in cfg["devices"] and 10 in cfg["devices"]
+
+
+def test_strategy_roundtrip_property():
+    """generate -> strategy_from_config recovers the (tp, pp) multiset
+    for a sweep of homogeneous and heterogeneous strategy lists."""
+    import itertools
+    cases = []
+    for tp, pp, dp in itertools.product((1, 2, 4), (1, 2, 4), (1, 2)):
+        if tp * pp * dp <= 16:
+            cases.append([(tp, pp)] * dp)
+    cases += [[(4, 1), (2, 1), (2, 1)], [(2, 1), (1, 1), (1, 1)],
+              [(4, 1), (4, 1)]]
+    for tp_pp in cases:
+        layers = 4
+        if any(layers % pp for _, pp in tp_pp):
+            continue
+        cfg = generate_ds_parallel_config(tp_pp, num_layers=layers)
+        spec, stages = strategy_from_config(cfg)
+        got = sorted((len(s[0]), len(s)) for s in stages)
+        assert got == sorted(tp_pp), (tp_pp, got)
+        if isinstance(spec, PipelineSpec):
+            assert [(spec.tp, spec.pp)] * spec.dp == list(tp_pp)
