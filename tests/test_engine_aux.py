@@ -199,3 +199,23 @@ def test_leveled_logger(monkeypatch, capsys):
     lg.warning("warn")
     err = capsys.readouterr().err
     assert "dbg 1" in err and "warn" in err and "r0" in err
+
+
+def test_package_import_hygiene():
+    """Importing the package surface must not initialize distributed
+    state or CUDA (workers decide that), and the public namespaces
+    resolve."""
+    import subprocess
+    import sys
+    code = (
+        "import hetu_amd as ht\n"
+        "from hetu_amd import nn, optim, models, data, parallel, utils\n"
+        "from hetu_amd.engine import Trainer, TrainingConfig\n"
+        "import torch, torch.distributed as dist\n"
+        "assert not dist.is_initialized()\n"
+        "assert not torch.cuda.is_initialized()\n"
+        "print('HYGIENE_OK')\n")
+    p = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert p.returncode == 0 and "HYGIENE_OK" in p.stdout, \
+        f"{p.stdout}\n{p.stderr}"
