@@ -349,3 +349,39 @@ def test_reshard_fuzz_four_ranks():
         out, err = p.communicate(timeout=600)
         ok = (p.returncode in (0, -6)) and "RESHARD_BAD:[]" in out
         assert ok, f"rank {r}: rc={p.returncode}\n{out}\n{err}"
+
+
+def test_comm_kind_census_regression():
+    """Every real strategy graph must keep its FAST comm kinds: a change
+    that silently reroutes a transition to the generic gather+reslice
+    path passes parity tests but costs a full materialization per step —
+    pin the expected kind census per strategy."""
+    from collections import Counter
+
+    import torch
+
+    from hetu_amd.models.gpt import GPTConfig, build_gpt_train_graph
+    from hetu_amd.nn.parallel import ParallelSpec
+
+    cfg = GPTConfig(n_layer=2, n_head=4, n_kv_head=4, hidden=64,
+                    ffn_hidden=128, vocab=128, max_seq=16)
+    expected = {
+        "dp2": {"allreduce": 2},
+        "tp2": {"allreduce": 10, "identity": 5},
+        "tp2sp": {"reducescatter": 5, "allgather": 9, "allreduce": 5,
+                  "slice": 4},
+        "dp2tp2": {"allreduce": 12, "identity": 5},
+        "cp2": {"allreduce": 1},
+    }
+    specs = {"dp2": ParallelSpec(dp=2), "tp2": ParallelSpec(tp=2),
+             "tp2sp": ParallelSpec(tp=2, sequence_parallel=True),
+             "dp2tp2": ParallelSpec(dp=2, tp=2),
+             "cp2": ParallelSpec(cp=2)}
+    for name, spec in specs.items():
+        g, h = build_gpt_train_graph(cfg, micro_batch=2, seq_len=16,
+                                     dtype=torch.float32, lr=1e-3,
+                                     spec=spec)
+        kinds = Counter(op.attrs["kind"][0] for op in g.ops
+                        if op.type == "Comm")
+        assert "generic" not in kinds, (name, kinds)
+        assert dict(kinds) == expected[name], (name, dict(kinds))
