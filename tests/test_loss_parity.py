@@ -107,3 +107,51 @@ def test_gpt_loss_curve_matches_pytorch():
         theirs.append(float(loss.detach()))
     for a, b in zip(ours, theirs):
         assert math.isfinite(a) and abs(a - b) < 5e-3, (ours, theirs)
+
+
+def test_lr_schedule_matches_torch_lambda_lr():
+    """Our multiplier schedules must track torch.optim LambdaLR exactly:
+    same tiny model, same cosine-with-warmup, identical loss curves."""
+    from hetu_amd.engine.lr_schedule import cosine_with_warmup
+    from hetu_amd.engine.runner import prepare_run_context
+    from hetu_amd.graph.ops.optim import AdamStepOp
+    from hetu_amd.graph.graph import DefineAndRunGraph, pop_graph, push_graph
+    from hetu_amd.graph.ops import api as ht
+    from hetu_amd.graph.ops.optim import Adam
+
+    torch.manual_seed(4)
+    d = 16
+    w0 = torch.randn(d, d) * 0.3
+    sched = cosine_with_warmup(2, 8, min_ratio=0.1)
+    lr = 5e-3
+
+    g = DefineAndRunGraph("lrp")
+    push_graph(g)
+    try:
+        x = ht.placeholder((4, d), name="x")
+        t = ht.placeholder((4, d), name="t")
+        w = ht.variable(w0.clone(), name="w")
+        loss = ht.mse_loss(ht.matmul(x, w), t)
+        train = Adam(lr=lr).minimize(loss)
+    finally:
+        pop_graph()
+    ctx = prepare_run_context(g, torch.device("cpu"), use_comm=False)
+
+    wr = w0.clone().requires_grad_(True)
+    opt = torch.optim.Adam([wr], lr=lr, betas=(0.9, 0.999), eps=1e-8)
+    lsched = torch.optim.lr_scheduler.LambdaLR(opt, sched)
+
+    gen = torch.Generator().manual_seed(9)
+    try:
+        for step in range(8):
+            xd = torch.randn(4, d, generator=gen)
+            td = torch.randn(4, d, generator=gen)
+            AdamStepOp.set_lr_scale(sched(step))
+            lv, _ = g.run([loss, train], {x: xd, t: td}, ctx=ctx)
+            rl = torch.nn.functional.mse_loss(xd @ wr, td)
+            opt.zero_grad(); rl.backward(); opt.step(); lsched.step()
+            assert abs(float(lv) - float(rl)) < 1e-5, (step, float(lv),
+                                                       float(rl))
+        assert torch.allclose(w.get_data(), wr.detach(), atol=1e-5)
+    finally:
+        AdamStepOp.set_lr_scale(1.0)
